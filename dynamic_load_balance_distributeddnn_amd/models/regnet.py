@@ -1,0 +1,113 @@
+"""CIFAR-shape RegNet X/Y family (grouped conv, optional SE, GroupNorm(32)).
+
+Architecture parity with reference Net/RegNet.py: stem conv3x3(64),
+four stages of bottleneck-ratio-1 blocks with grouped 3x3 convs, SE block
+for the Y variants, adaptive avg pool + linear head.
+"""
+
+from __future__ import annotations
+
+import torch.nn as nn
+import torch.nn.functional as F
+
+from ..ops.layers import Conv2d, GroupNormAct, Linear
+
+_GN = 32
+
+
+def _gn_groups(channels: int) -> int:
+    """32 where it divides (reference layout); otherwise the largest of
+    16/8/4/2/1 that does.  The reference's RegNetX-200MF crashes as
+    shipped — GroupNorm(32, 24) at Net/RegNet.py:32 with width 24 — and
+    BASELINE config #4 requires that model to run, so this is a
+    deliberate fix (documented in SURVEY parity notes)."""
+    for g in (32, 16, 8, 4, 2):
+        if channels % g == 0:
+            return g
+    return 1
+
+
+class _SqueezeExcite(nn.Module):
+    """adaptive_avg_pool(1) -> 1x1 reduce -> ReLU -> 1x1 expand -> sigmoid-mul."""
+
+    def __init__(self, channels, se_channels):
+        super().__init__()
+        self.reduce = Conv2d(channels, se_channels, 1, bias=True)
+        self.expand = Conv2d(se_channels, channels, 1, bias=True)
+
+    def forward(self, x):
+        s = F.adaptive_avg_pool2d(x, 1)
+        s = self.expand(F.relu(self.reduce(s)))
+        return x * s.sigmoid()
+
+
+class _Block(nn.Module):
+    def __init__(self, w_in, w_out, stride, group_width, bottleneck_ratio, se_ratio):
+        super().__init__()
+        w_b = int(round(w_out * bottleneck_ratio))
+        self.a = nn.Sequential(
+            Conv2d(w_in, w_b, 1),
+            GroupNormAct(_gn_groups(w_b), w_b, relu=True),
+            Conv2d(w_b, w_b, 3, stride=stride, padding=1,
+                   groups=w_b // group_width),
+            GroupNormAct(_gn_groups(w_b), w_b, relu=True),
+        )
+        self.se = (_SqueezeExcite(w_b, int(round(w_in * se_ratio)))
+                   if se_ratio > 0 else None)
+        self.b = nn.Sequential(
+            Conv2d(w_b, w_out, 1),
+            GroupNormAct(_gn_groups(w_out), w_out),
+        )
+        self.proj = None
+        if stride != 1 or w_in != w_out:
+            self.proj = nn.Sequential(
+                Conv2d(w_in, w_out, 1, stride=stride),
+                GroupNormAct(_gn_groups(w_out), w_out),
+            )
+
+    def forward(self, x):
+        out = self.a(x)
+        if self.se is not None:
+            out = self.se(out)
+        out = self.b(out)
+        out = out + (self.proj(x) if self.proj is not None else x)
+        return F.relu(out, inplace=True)
+
+
+class RegNet(nn.Module):
+    def __init__(self, depths, widths, strides, group_width,
+                 bottleneck_ratio=1, se_ratio=0.0, num_classes=10):
+        super().__init__()
+        self.stem = nn.Sequential(
+            Conv2d(3, 64, 3, padding=1),
+            GroupNormAct(_GN, 64, relu=True),
+        )
+        blocks, cin = [], 64
+        for depth, width, stage_stride in zip(depths, widths, strides):
+            for i in range(depth):
+                s = stage_stride if i == 0 else 1
+                blocks.append(_Block(cin, width, s, group_width,
+                                     bottleneck_ratio, se_ratio))
+                cin = width
+        self.body = nn.Sequential(*blocks)
+        self.head = Linear(widths[-1], num_classes)
+
+    def forward(self, x):
+        out = self.body(self.stem(x))
+        out = F.adaptive_avg_pool2d(out, 1).flatten(1)
+        return self.head(out)
+
+
+def RegNetX_200MF(num_classes=10):
+    return RegNet((1, 1, 4, 7), (24, 56, 152, 368), (1, 1, 2, 2),
+                  group_width=8, num_classes=num_classes)
+
+
+def RegNetX_400MF(num_classes=10):
+    return RegNet((1, 2, 7, 12), (32, 64, 160, 384), (1, 1, 2, 2),
+                  group_width=16, num_classes=num_classes)
+
+
+def RegNetY_400MF(num_classes=10):
+    return RegNet((1, 2, 7, 12), (32, 64, 160, 384), (1, 1, 2, 2),
+                  group_width=16, se_ratio=0.25, num_classes=num_classes)
