@@ -1,0 +1,183 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: DenseNet-121, global batch 512, bf16, synthetic
+CIFAR-10-shape data — images/sec over the whole node (BASELINE.json
+headline metric).
+
+Single process:   python bench.py --steps 30 --warmup 10
+Multi-GPU (driver): python -m torch.distributed.run --nnodes=1
+    --nproc-per-node N --master-addr 127.0.0.1 bench.py --gpus N ...
+
+The global batch stays fixed at 512 as GPUs scale (strong scaling, the
+reference's own experiment shape: DBS splits ONE global batch across
+workers).  Every timed step runs the full training iteration: zero, bf16
+forward, loss, backward, weighted bucketed RCCL all-reduce, fused SGD
+step.  Rank 0 prints one JSON line.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import numpy as np
+import torch
+import torch.distributed as dist
+import torch.nn.functional as F
+
+
+def get_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=None,
+                   help="world size (informational; env WORLD_SIZE wins)")
+    p.add_argument("--steps", type=int, default=30)
+    p.add_argument("--warmup", type=int, default=10)
+    p.add_argument("--model", default="densenet",
+                   choices=["densenet", "resnet50", "resnet", "regnetx200",
+                            "regnet", "googlenet", "transformer", "mnistnet"])
+    p.add_argument("--global-batch", type=int, default=512)
+    p.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"])
+    p.add_argument("--device", default=None, help="override (e.g. cpu for debug)")
+    return p.parse_args()
+
+
+def build(model_name, num_classes=10):
+    from dynamic_load_balance_distributeddnn_amd import models as M
+
+    table = {
+        "densenet": lambda: M.DenseNet121(num_classes),
+        "resnet50": lambda: M.ResNet50(num_classes),
+        "resnet": lambda: M.ResNet101(num_classes),
+        "regnetx200": lambda: M.RegNetX_200MF(num_classes),
+        "regnet": lambda: M.RegNetY_400MF(num_classes),
+        "googlenet": lambda: M.GoogLeNet(num_classes),
+        "mnistnet": lambda: M.MnistNet(),
+        "transformer": lambda: M.build_model("transformer"),
+    }
+    return table[model_name]()
+
+
+def main():
+    args = get_args()
+    rank = int(os.environ.get("RANK", 0))
+    world = int(os.environ.get("WORLD_SIZE", args.gpus or 1))
+    local_rank = int(os.environ.get("LOCAL_RANK", rank))
+
+    if args.device:
+        device = torch.device(args.device)
+    elif torch.cuda.is_available():
+        device = torch.device(f"cuda:{local_rank % max(1, torch.cuda.device_count())}")
+        torch.cuda.set_device(device)
+    else:
+        device = torch.device("cpu")
+
+    if world > 1:
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29531")
+        backend = "nccl" if device.type == "cuda" else "gloo"
+        dist.init_process_group(backend, rank=rank, world_size=world)
+
+    from dynamic_load_balance_distributeddnn_amd.models import LM_CONFIG
+    from dynamic_load_balance_distributeddnn_amd.parallel import \
+        GradientSynchronizer
+    from dynamic_load_balance_distributeddnn_amd.parallel.optim import FlatSGD
+    from dynamic_load_balance_distributeddnn_amd.scheduler import DBSScheduler
+
+    torch.manual_seed(1234)
+    is_lm = args.model == "transformer"
+    model = build(args.model).to(device)
+    sched = DBSScheduler(world, args.global_batch)
+    my_batch = int(sched.batches[rank])
+
+    sync = GradientSynchronizer(model, defer=is_lm)
+    sync.set_weight(float(sched.weights[rank]))
+    opt = FlatSGD(sync, lr=0.01, momentum=0.9)
+
+    # synthetic per-rank batch, pre-staged on device
+    g = torch.Generator().manual_seed(1234 + rank)
+    if is_lm:
+        bptt = LM_CONFIG["bptt"]
+        ntokens = LM_CONFIG["ntokens"]
+        x = torch.randint(0, ntokens, (bptt, my_batch), generator=g).to(device)
+        y = torch.randint(0, ntokens, (bptt * my_batch,), generator=g).to(device)
+        items_per_step = args.global_batch * bptt  # tokens
+        metric, unit = "tokens_per_sec", "tokens/s"
+    else:
+        shape = (1, 28, 28) if args.model == "mnistnet" else (3, 32, 32)
+        x = torch.randn(my_batch, *shape, generator=g).to(device)
+        y = torch.randint(0, 10, (my_batch,), generator=g).to(device)
+        items_per_step = args.global_batch
+        metric, unit = "images_per_sec", "images/s"
+
+    amp = (torch.autocast("cuda", dtype=torch.bfloat16)
+           if args.dtype == "bf16" and device.type == "cuda"
+           else torch.autocast("cpu", enabled=False))
+    criterion = F.nll_loss if is_lm else F.cross_entropy
+
+    def step():
+        sync.zero()
+        with amp:
+            out = model(x)
+            if is_lm:
+                out = out.reshape(-1, ntokens)
+            loss = criterion(out, y)
+        loss.backward()
+        if is_lm:
+            torch.nn.utils.clip_grad_norm_(model.parameters(), 0.25)
+        sync.finish()
+        opt.step()
+        return loss
+
+    model.train()
+    for _ in range(args.warmup):
+        step()
+
+    if world > 1:
+        dist.barrier()
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step()
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    if world > 1:
+        dist.barrier()
+    elapsed = time.perf_counter() - t0
+
+    # MAX over ranks (slowest rank defines the job)
+    if world > 1:
+        t = torch.tensor([elapsed], device=device if device.type == "cuda" else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    if rank == 0:
+        value = items_per_step * args.steps / elapsed
+        print(json.dumps({
+            "metric": metric,
+            "value": round(value, 2),
+            "unit": unit,
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1e3, 3),
+            "higher_is_better": True,
+            "scaling": "strong",
+            "vs_baseline": None,
+            "dtype": args.dtype,
+            "data": "synthetic",
+            "config": {
+                "model": "DenseNet-121" if args.model == "densenet" else args.model,
+                "global_batch": args.global_batch,
+                "seq_len": LM_CONFIG["bptt"] if is_lm else None,
+                "parallelism": f"dbs-dp{world}",
+            },
+        }))
+
+    if world > 1:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -28,6 +28,7 @@ import torch.nn.functional as F
 from . import data as D
 from .models import LM_CONFIG, build_model
 from .parallel import GradientSynchronizer, StepTimer
+from .parallel.optim import FlatSGD
 from .scheduler import DBSScheduler, exchange_times
 from .utils import FaultInjector, StatsRecorder
 from .utils.lr_policy import apply_lr, one_cycle_lr
@@ -54,8 +55,6 @@ class Trainer:
         self.model.to(device)
         self._sync_initial_weights()
 
-        self.optimizer = torch.optim.SGD(self.model.parameters(),
-                                         lr=args.learning_rate, momentum=0.9)
         # MnistNet quirk preserved: cross_entropy over a log_softmax output
         # (reference dbs.py:371-374 + Net/MnistNet.py:27).
         self.criterion = F.nll_loss if self.is_lm else F.cross_entropy
@@ -63,6 +62,10 @@ class Trainer:
         # LM path clips gradients between backward and the reduce
         # (dbs.py:274), so its buckets launch deferred at finish().
         self.sync = GradientSynchronizer(self.model, defer=self.is_lm)
+        # Flat fused SGD-momentum over the grad arena (reference SGD
+        # semantics, dbs.py:369; one HIP kernel per step on GPU).
+        self.optimizer = FlatSGD(self.sync, lr=args.learning_rate,
+                                 momentum=0.9)
         self.timer = StepTimer(device)
         self.sched = DBSScheduler(world_size, args.batch_size,
                                   enabled=args.dynamic_batch_size)

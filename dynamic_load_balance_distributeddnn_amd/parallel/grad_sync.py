@@ -65,12 +65,14 @@ class GradientSynchronizer:
         ordered = list(reversed(self.params))
         self.buckets: list[_Bucket] = []
         self._param_bucket: dict[int, _Bucket] = {}
+        self.offsets: dict[int, tuple[int, int]] = {}  # id(p) -> (off, numel)
         offset = 0
         cur_params: list = []
         cur_start = 0
         elem = self.arena.element_size()
         for p in ordered:
             n = p.numel()
+            self.offsets[id(p)] = (offset, n)
             p.grad = self.arena.narrow(0, offset, n).view_as(p)
             cur_params.append(p)
             offset += n

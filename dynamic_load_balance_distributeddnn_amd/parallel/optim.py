@@ -1,0 +1,55 @@
+"""Flat fused SGD-with-momentum.
+
+The reference's optimizer is torch.optim.SGD(lr, momentum=0.9)
+(dbs.py:369), stepped per-tensor — up to 362 tensors for DenseNet-121.
+Here parameters are re-homed into ONE flat fp32 arena laid out identically
+to GradientSynchronizer's gradient arena, and momentum is a third flat
+buffer, so the whole update is a single HBM-bound HIP kernel
+(ops/csrc/sgd.hip) per step: v = mu*v + g ; p -= lr*v.
+
+Math matches torch SGD with dampening=0, nesterov=False exactly.
+CPU (debug mode) runs the same flat update with three torch ops.
+"""
+
+from __future__ import annotations
+
+import torch
+
+from .grad_sync import GradientSynchronizer
+
+__all__ = ["FlatSGD"]
+
+
+class FlatSGD:
+    def __init__(self, sync: GradientSynchronizer, lr: float,
+                 momentum: float = 0.9):
+        self.sync = sync
+        self.momentum = momentum
+        self.param_groups = [{"lr": lr}]  # LR-policy-compatible surface
+
+        total = sync.arena.numel()
+        device = sync.arena.device
+        self.param_arena = torch.empty(total, dtype=torch.float32, device=device)
+        self.momentum_buf = torch.zeros(total, dtype=torch.float32, device=device)
+        with torch.no_grad():
+            for p in sync.params:
+                off, n = sync.offsets[id(p)]
+                view = self.param_arena.narrow(0, off, n).view_as(p)
+                view.copy_(p.data)
+                p.data = view  # re-home the parameter into the arena
+
+    @torch.no_grad()
+    def step(self) -> None:
+        lr = float(self.param_groups[0]["lr"])
+        g = self.sync.arena
+        if g.is_cuda:
+            from ..ops import ext
+
+            ext().sgd_momentum(self.param_arena, g, self.momentum_buf,
+                               lr, self.momentum)
+        else:
+            self.momentum_buf.mul_(self.momentum).add_(g)
+            self.param_arena.add_(self.momentum_buf, alpha=-lr)
+
+    def zero_grad(self, set_to_none: bool = False) -> None:
+        self.sync.zero()
