@@ -39,6 +39,9 @@ def get_args():
     p.add_argument("--global-batch", type=int, default=512)
     p.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"])
     p.add_argument("--device", default=None, help="override (e.g. cpu for debug)")
+    p.add_argument("--no-channels-last", action="store_true",
+                   help="disable NHWC layout (NHWC is the GPU default; the "
+                        "gfx950 kernels are NHWC-native)")
     return p.parse_args()
 
 
@@ -87,6 +90,10 @@ def main():
     torch.manual_seed(1234)
     is_lm = args.model == "transformer"
     model = build(args.model).to(device)
+    channels_last = (device.type == "cuda" and not is_lm
+                     and not args.no_channels_last)
+    if channels_last:
+        model = model.to(memory_format=torch.channels_last)
     sched = DBSScheduler(world, args.global_batch)
     my_batch = int(sched.batches[rank])
 
@@ -107,6 +114,8 @@ def main():
         shape = (1, 28, 28) if args.model == "mnistnet" else (3, 32, 32)
         x = torch.randn(my_batch, *shape, generator=g).to(device)
         y = torch.randint(0, 10, (my_batch,), generator=g).to(device)
+        if channels_last:
+            x = x.to(memory_format=torch.channels_last)
         items_per_step = args.global_batch
         metric, unit = "images_per_sec", "images/s"
 

@@ -53,6 +53,13 @@ class Trainer:
         torch.manual_seed(seed)
         self.model = build_model(args.model, _num_classes(args.dataset))
         self.model.to(device)
+        if device.type == "cuda":
+            # GPU path: bf16 compute (fp32 master weights/grads) and NHWC
+            # activations — the layout the gfx950 kernels are written for.
+            if self.amp_dtype is None:
+                self.amp_dtype = torch.bfloat16
+            if not self.is_lm:
+                self.model = self.model.to(memory_format=torch.channels_last)
         self._sync_initial_weights()
 
         # MnistNet quirk preserved: cross_entropy over a log_softmax output
@@ -165,6 +172,8 @@ class Trainer:
         else:
             for inputs, target in source:
                 inputs = inputs.to(self.device, non_blocking=True)
+                if self.device.type == "cuda":
+                    inputs = inputs.contiguous(memory_format=torch.channels_last)
                 target = target.to(self.device, non_blocking=True)
                 loss = self._step(inputs, target, epoch, steps)
                 epoch_loss += loss.detach()

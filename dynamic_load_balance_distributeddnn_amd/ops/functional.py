@@ -20,7 +20,7 @@ import torch.nn.functional as F
 from . import available, ext
 
 # ops with a hand-written gfx950 kernel wired in (grown as kernels land)
-NATIVE_OPS: set[str] = set()
+NATIVE_OPS: set[str] = {"group_norm_act"}
 
 
 def _use_native(name: str, x: torch.Tensor) -> bool:
@@ -42,7 +42,8 @@ def group_norm_act(x, num_groups, weight, bias, eps=1e-5, relu=False):
     ReLU directly after nearly every GroupNorm — SURVEY.md K4/K5)."""
     if _use_native("group_norm_act", x):
         from . import native
-        return native.group_norm_act(x, num_groups, weight, bias, eps, relu)
+        if native.gn_native_ok(x, num_groups, weight):
+            return native.group_norm_act(x, num_groups, weight, bias, eps, relu)
     out = F.group_norm(x, num_groups, weight, bias, eps)
     return F.relu(out, inplace=True) if relu else out
 

@@ -29,6 +29,13 @@ import torch.distributed as dist
 __all__ = ["GradientSynchronizer"]
 
 
+def _strided_view(arena: torch.Tensor, offset: int, p: torch.Tensor):
+    """A view into the flat arena shaped like ``p`` INCLUDING its stride
+    layout (e.g. channels_last conv weights), so re-homed tensors keep the
+    memory format the kernels expect.  ``p`` must be dense."""
+    return arena.as_strided(p.shape, p.stride(), offset)
+
+
 class _Bucket:
     __slots__ = ("start", "end", "params", "pending", "work", "launched")
 
@@ -73,7 +80,7 @@ class GradientSynchronizer:
         for p in ordered:
             n = p.numel()
             self.offsets[id(p)] = (offset, n)
-            p.grad = self.arena.narrow(0, offset, n).view_as(p)
+            p.grad = _strided_view(self.arena, offset, p)
             cur_params.append(p)
             offset += n
             if (offset - cur_start) * elem >= bucket_bytes:

@@ -31,10 +31,12 @@ class FlatSGD:
         device = sync.arena.device
         self.param_arena = torch.empty(total, dtype=torch.float32, device=device)
         self.momentum_buf = torch.zeros(total, dtype=torch.float32, device=device)
+        from .grad_sync import _strided_view
+
         with torch.no_grad():
             for p in sync.params:
                 off, n = sync.offsets[id(p)]
-                view = self.param_arena.narrow(0, off, n).view_as(p)
+                view = _strided_view(self.param_arena, off, p)
                 view.copy_(p.data)
                 p.data = view  # re-home the parameter into the arena
 

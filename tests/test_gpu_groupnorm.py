@@ -1,0 +1,93 @@
+"""Fused GroupNorm(+ReLU) kernel vs plain fp32 PyTorch reference."""
+
+import pytest
+import torch
+import torch.nn.functional as F
+
+pytestmark = pytest.mark.gpu
+
+needs_gpu = pytest.mark.skipif(not torch.cuda.is_available(),
+                               reason="needs ROCm GPU")
+
+SHAPES = [
+    # (N, C, H, W, G, relu) — spans the zoo: Cg=1, Cg=3, big-C, GoogLeNet G=8
+    (4, 64, 32, 32, 32, True),
+    (4, 32, 32, 32, 32, False),   # Cg = 1
+    (2, 24, 16, 16, 8, True),     # Cg = 3 (RegNetX-200MF stage 1)
+    (3, 2208, 4, 4, 32, True),    # C > 8*block (DenseNet-161 tail)
+    (2, 192, 32, 32, 8, False),   # GoogLeNet stem
+    (5, 1024, 8, 8, 32, True),
+]
+
+
+def _native_gn(xb, G, gamma, beta, relu):
+    from dynamic_load_balance_distributeddnn_amd.ops import native
+
+    return native.group_norm_act(xb, G, gamma, beta, 1e-5, relu)
+
+
+@needs_gpu
+@pytest.mark.parametrize("shape", SHAPES)
+def test_gn_forward_matches_fp32(shape):
+    N, C, H, W, G, relu = shape
+    torch.manual_seed(0)
+    x = torch.randn(N, C, H, W, device="cuda")
+    gamma = torch.randn(C, device="cuda") * 0.5 + 1.0
+    beta = torch.randn(C, device="cuda") * 0.1
+    xb = x.bfloat16()
+
+    ref = F.group_norm(xb.float(), G, gamma, beta, 1e-5)
+    if relu:
+        ref = F.relu(ref)
+
+    xcl = xb.float().bfloat16().to(memory_format=torch.channels_last)
+    out = _native_gn(xcl, G, gamma, beta, relu)
+    assert out.is_contiguous(memory_format=torch.channels_last)
+    torch.testing.assert_close(out.float(), ref, rtol=2e-2, atol=3e-2)
+
+
+@needs_gpu
+@pytest.mark.parametrize("shape", SHAPES)
+def test_gn_backward_matches_fp32(shape):
+    N, C, H, W, G, relu = shape
+    torch.manual_seed(1)
+    x = torch.randn(N, C, H, W, device="cuda").bfloat16()
+    gamma = (torch.randn(C, device="cuda") * 0.5 + 1.0)
+    beta = torch.randn(C, device="cuda") * 0.1
+    dz = torch.randn(N, C, H, W, device="cuda").bfloat16()
+
+    # fp32 reference of the same composed op on the same quantized inputs
+    x32 = x.float().requires_grad_()
+    g32 = gamma.clone().requires_grad_()
+    b32 = beta.clone().requires_grad_()
+    ref = F.group_norm(x32, G, g32, b32, 1e-5)
+    if relu:
+        ref = F.relu(ref)
+    ref.backward(dz.float())
+
+    xcl = x.to(memory_format=torch.channels_last).requires_grad_()
+    gk = gamma.clone().requires_grad_()
+    bk = beta.clone().requires_grad_()
+    out = _native_gn(xcl, G, gk, bk, relu)
+    out.backward(dz.to(memory_format=torch.channels_last))
+
+    torch.testing.assert_close(xcl.grad.float(), x32.grad,
+                               rtol=5e-2, atol=5e-2)
+    torch.testing.assert_close(gk.grad, g32.grad, rtol=2e-2, atol=2e-1)
+    torch.testing.assert_close(bk.grad, b32.grad, rtol=2e-2, atol=2e-1)
+
+
+@needs_gpu
+def test_gn_autograd_gradcheck_small():
+    """End-to-end through a tiny GroupNormAct module in bf16 training mode."""
+    from dynamic_load_balance_distributeddnn_amd.ops.layers import \
+        GroupNormAct
+
+    torch.manual_seed(2)
+    m = GroupNormAct(4, 16, relu=True).cuda()
+    x = torch.randn(2, 16, 8, 8, device="cuda").bfloat16() \
+        .to(memory_format=torch.channels_last).requires_grad_()
+    y = m(x)
+    y.sum().backward()
+    assert torch.isfinite(x.grad.float()).all()
+    assert m.weight.grad is not None and m.weight.grad.dtype == torch.float32
