@@ -71,7 +71,84 @@ static std::vector<torch::Tensor> gn_bwd(torch::Tensor x, torch::Tensor dz,
   return {dx, dgamma, dbeta};
 }
 
+extern "C" void dlb_conv_fwd(const void* x, const void* w, void* y,
+                             const float* bias, int N, int IH, int IW, int Ci,
+                             int OH, int OW, int Co, int R, int S, int stride,
+                             int pad, hipStream_t stream);
+extern "C" void dlb_conv_bwd_data(const void* dy, const void* wt, void* dx,
+                                  int N, int IH, int IW, int Ci, int OH,
+                                  int OW, int Co, int R, int S, int stride,
+                                  int pad, hipStream_t stream);
+extern "C" void dlb_conv_wrw(const void* x, const void* dy, float* dw, int N,
+                             int IH, int IW, int Ci, int OH, int OW, int Co,
+                             int R, int S, int stride, int pad,
+                             hipStream_t stream);
+
+static inline bool is_cl(const torch::Tensor& t) {
+  return t.is_contiguous(torch::MemoryFormat::ChannelsLast);
+}
+
+// x [N,Ci,H,W] channels_last bf16; w [Co,Ci,R,S] channels_last bf16;
+// bias fp32 [Co] or empty.  Returns y [N,Co,OH,OW] channels_last bf16.
+static torch::Tensor conv_fwd(torch::Tensor x, torch::Tensor w,
+                              c10::optional<torch::Tensor> bias,
+                              int64_t stride, int64_t pad) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(is_cl(x) && is_cl(w), "conv expects channels_last tensors");
+  const int N = x.size(0), Ci = x.size(1), IH = x.size(2), IW = x.size(3);
+  const int Co = w.size(0), R = w.size(2), S = w.size(3);
+  TORCH_CHECK(w.size(1) == Ci);
+  const int OH = (IH + 2 * pad - R) / stride + 1;
+  const int OW = (IW + 2 * pad - S) / stride + 1;
+  auto y = torch::empty({N, Co, OH, OW},
+                        x.options().memory_format(torch::MemoryFormat::ChannelsLast));
+  const float* bp = nullptr;
+  if (bias.has_value() && bias->defined()) {
+    TORCH_CHECK(bias->scalar_type() == torch::kFloat32);
+    bp = bias->data_ptr<float>();
+  }
+  auto stream = at::hip::getCurrentHIPStream();
+  dlb_conv_fwd(x.data_ptr(), w.data_ptr(), y.data_ptr(), bp, N, IH, IW, Ci,
+               OH, OW, Co, R, S, (int)stride, (int)pad, stream.stream());
+  return y;
+}
+
+// dy [N,Co,OH,OW] channels_last bf16; wt [Ci, R*S*Co] bf16 contiguous
+// (flipped kernel, built host-side).  Returns dx channels_last bf16.
+static torch::Tensor conv_bwd_data(torch::Tensor dy, torch::Tensor wt,
+                                   int64_t IH, int64_t IW, int64_t Ci,
+                                   int64_t R, int64_t S, int64_t stride,
+                                   int64_t pad) {
+  TORCH_CHECK(dy.is_cuda() && is_cl(dy) && wt.is_contiguous());
+  const int N = dy.size(0), Co = dy.size(1), OH = dy.size(2), OW = dy.size(3);
+  auto dx = torch::empty({N, Ci, IH, IW},
+                         dy.options().memory_format(torch::MemoryFormat::ChannelsLast));
+  auto stream = at::hip::getCurrentHIPStream();
+  dlb_conv_bwd_data(dy.data_ptr(), wt.data_ptr(), dx.data_ptr(), N, (int)IH,
+                    (int)IW, (int)Ci, OH, OW, Co, (int)R, (int)S, (int)stride,
+                    (int)pad, stream.stream());
+  return dx;
+}
+
+// Returns dw fp32 [Co, R*S*Ci] (the [Co][R][S][Ci] channels_last image).
+static torch::Tensor conv_wrw(torch::Tensor x, torch::Tensor dy, int64_t R,
+                              int64_t S, int64_t stride, int64_t pad) {
+  TORCH_CHECK(x.is_cuda() && is_cl(x) && is_cl(dy));
+  const int N = x.size(0), Ci = x.size(1), IH = x.size(2), IW = x.size(3);
+  const int Co = dy.size(1), OH = dy.size(2), OW = dy.size(3);
+  auto dw = torch::zeros({Co, R * S * Ci},
+                         x.options().dtype(torch::kFloat32));
+  auto stream = at::hip::getCurrentHIPStream();
+  dlb_conv_wrw(x.data_ptr(), dy.data_ptr(), dw.data_ptr<float>(), N, IH, IW,
+               Ci, OH, OW, Co, (int)R, (int)S, (int)stride, (int)pad,
+               stream.stream());
+  return dw;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("conv_fwd", &conv_fwd, "Implicit-GEMM NHWC bf16 conv forward");
+  m.def("conv_bwd_data", &conv_bwd_data, "NHWC bf16 conv input-grad");
+  m.def("conv_wrw", &conv_wrw, "NHWC bf16 conv weight-grad (fp32 out)");
   m.def("sgd_momentum", &sgd_momentum,
         "Fused SGD momentum step over flat arenas (gfx950)");
   m.def("gn_fwd", &gn_fwd, "Fused GroupNorm(+ReLU) forward, NHWC bf16");

@@ -1,0 +1,517 @@
+// Implicit-GEMM NHWC bf16 convolution on MFMA (gfx950) — fwd, bwd-data, wrw.
+//
+// Replaces the MIOpen/CK conv path for the zoo's conv shapes (SURVEY.md
+// K1/K2): 1x1, 3x3 and 5x5 convs (stride 1/2, square pad) over
+// CIFAR-scale activations at large batch.  GEMM view (forward):
+//     C[M][Co] = A[M][K] x B[K][Co]
+//     M = N*OH*OW (one row per output pixel), K = R*S*Ci
+// A is materialized on the fly from the NHWC input (im2col addressing
+// with zero-fill at borders); B is the channels_last weight, whose
+// natural memory layout [Co][R][S][Ci] is exactly the [n][k] LDS image
+// the B-fragment reads want — no weight reshape on the host for fwd.
+//
+// MFMA: v_mfma_f32_16x16x32_bf16.  The k-reduction is invariant under
+// any permutation applied identically to A and B (verified on hardware
+// by tools/mfma_probe), so fragments use the CONTIGUOUS k-map
+// k = (lane/16)*8 + j — one 16-byte LDS read per fragment.  C/D map:
+// col = lane&15, row = (lane>>4)*4 + reg (cdna_hip_programming.md §3).
+//
+// Tiles: template <BM, BN, WM, WN>; 4 waves (256 threads); BK = 32.
+// Loaders vectorize (bf16x8) when the channel count is a multiple of 8
+// and fall back to per-element gathers otherwise (stem Ci=3, MnistNet).
+
+#include "common.h"
+
+typedef __hip_bfloat16 bf16;
+typedef __attribute__((ext_vector_type(8))) short bf16x8_t;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+#define CONV_BLOCK 256
+#define BK 32
+
+struct ConvParams {
+  const bf16* x;   // [N, IH, IW, Ci]
+  const bf16* w;   // [Co, R, S, Ci]  (channels_last natural layout)
+  bf16* y;         // [N, OH, OW, Co]
+  const float* bias;  // [Co] or nullptr
+  int N, IH, IW, Ci, OH, OW, Co, R, S, stride, pad;
+  int M, K;        // M = N*OH*OW, K = R*S*Ci
+};
+
+// im2col 8-element load for GEMM row m (output pixel), k-chunk k..k+7.
+__device__ inline bf16x8_t im2col_load8(const bf16* __restrict__ x, int m,
+                                        int k, int IH, int IW, int Ci, int OH,
+                                        int OW, int S, int stride, int pad,
+                                        int K, bool vec) {
+  bf16x8_t v = {0, 0, 0, 0, 0, 0, 0, 0};
+  const int n = m / (OH * OW);
+  const int rem = m % (OH * OW);
+  const int oh = rem / OW, ow = rem % OW;
+  if (vec) {
+    const int rs = k / Ci, ci = k % Ci;
+    const int r = rs / S, s = rs % S;
+    const int ih = oh * stride - pad + r;
+    const int iw = ow * stride - pad + s;
+    if (ih >= 0 && ih < IH && iw >= 0 && iw < IW)
+      v = *reinterpret_cast<const bf16x8_t*>(
+          x + (((long)n * IH + ih) * IW + iw) * Ci + ci);
+  } else {
+    bf16* vv = reinterpret_cast<bf16*>(&v);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int kk = k + j;
+      if (kk < K) {
+        const int rs = kk / Ci, ci = kk % Ci;
+        const int r = rs / S, s = rs % S;
+        const int ih = oh * stride - pad + r;
+        const int iw = ow * stride - pad + s;
+        if (ih >= 0 && ih < IH && iw >= 0 && iw < IW)
+          vv[j] = x[(((long)n * IH + ih) * IW + iw) * Ci + ci];
+      }
+    }
+  }
+  return v;
+}
+
+// ---------------------------------------------------------------- forward
+template <int BM, int BN, int WM, int WN>
+__global__ void __launch_bounds__(CONV_BLOCK)
+conv_fwd_kernel(const ConvParams p) {
+  constexpr int WTM = BM / WM;        // wave tile rows
+  constexpr int WTN = BN / WN;        // wave tile cols
+  constexpr int FA = WTM / 16;
+  constexpr int FB = WTN / 16;
+  constexpr int LDA = BK + 8;         // padded LDS row (elements)
+  constexpr int LDB = BK + 8;
+
+  __shared__ bf16 a_lds[BM * LDA];
+  __shared__ bf16 b_lds[BN * LDB];    // [n][k] image
+
+  const int t = threadIdx.x;
+  const int lane = t & 63;
+  const int wave = t >> 6;
+  const int wr = wave / WN, wc = wave % WN;
+  const int m0 = blockIdx.x * BM;
+  const int n0 = blockIdx.y * BN;
+  const bool vec = (p.Ci & 7) == 0;
+
+  f32x4 acc[FA][FB];
+#pragma unroll
+  for (int i = 0; i < FA; ++i)
+#pragma unroll
+    for (int j = 0; j < FB; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  constexpr int ACH = BM * (BK / 8);
+  constexpr int BCH = BN * (BK / 8);
+
+  for (int kt = 0; kt < p.K; kt += BK) {
+    for (int c = t; c < ACH; c += CONV_BLOCK) {
+      const int row = c / (BK / 8);
+      const int k8 = (c % (BK / 8)) * 8;
+      const int m = m0 + row;
+      const int k = kt + k8;
+      bf16x8_t v = {0, 0, 0, 0, 0, 0, 0, 0};
+      if (m < p.M && k < p.K)
+        v = im2col_load8(p.x, m, k, p.IH, p.IW, p.Ci, p.OH, p.OW, p.S,
+                         p.stride, p.pad, p.K, vec);
+      *reinterpret_cast<bf16x8_t*>(&a_lds[row * LDA + k8]) = v;
+    }
+    for (int c = t; c < BCH; c += CONV_BLOCK) {
+      const int nrow = c / (BK / 8);
+      const int k8 = (c % (BK / 8)) * 8;
+      const int n = n0 + nrow;
+      const int k = kt + k8;
+      bf16x8_t v = {0, 0, 0, 0, 0, 0, 0, 0};
+      if (n < p.Co && k + 7 < p.K)
+        v = *reinterpret_cast<const bf16x8_t*>(p.w + (long)n * p.K + k);
+      else if (n < p.Co) {
+        bf16* vv = reinterpret_cast<bf16*>(&v);
+        for (int j = 0; j < 8 && k + j < p.K; ++j)
+          vv[j] = p.w[(long)n * p.K + k + j];
+      }
+      *reinterpret_cast<bf16x8_t*>(&b_lds[nrow * LDB + k8]) = v;
+    }
+    __syncthreads();
+
+    bf16x8_t afrag[FA], bfrag[FB];
+#pragma unroll
+    for (int i = 0; i < FA; ++i) {
+      const int row = wr * WTM + i * 16 + (lane & 15);
+      afrag[i] = *reinterpret_cast<const bf16x8_t*>(
+          &a_lds[row * LDA + (lane >> 4) * 8]);
+    }
+#pragma unroll
+    for (int j = 0; j < FB; ++j) {
+      const int col = wc * WTN + j * 16 + (lane & 15);
+      bfrag[j] = *reinterpret_cast<const bf16x8_t*>(
+          &b_lds[col * LDB + (lane >> 4) * 8]);
+    }
+#pragma unroll
+    for (int i = 0; i < FA; ++i)
+#pragma unroll
+      for (int j = 0; j < FB; ++j)
+        acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int i = 0; i < FA; ++i) {
+#pragma unroll
+    for (int j = 0; j < FB; ++j) {
+      const int col = n0 + wc * WTN + j * 16 + (lane & 15);
+      if (col >= p.Co) continue;
+      const float b = p.bias ? p.bias[col] : 0.f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int m = m0 + wr * WTM + i * 16 + (lane >> 4) * 4 + r;
+        if (m < p.M)
+          p.y[(long)m * p.Co + col] = __float2bfloat16(acc[i][j][r] + b);
+      }
+    }
+  }
+}
+
+// ------------------------------------------------------------- bwd data
+// dx[n,ih,iw,ci] = sum_{r,s,co} dy[n,oh,ow,co] * w[co,r,s,ci]
+//   oh = (ih + pad - r)/stride  (valid when divisible & in range)
+// GEMM rows = input pixels, k = (r*S+s)*Co + co, cols = Ci.
+// B image [ci][k] is precomputed host-side (wt: [Ci][R*S*Co], kernel
+// flipped), so this reuses the forward's structure with a different
+// A-loader.
+struct ConvBwdParams {
+  const bf16* dy;  // [N, OH, OW, Co]
+  const bf16* wt;  // [Ci][R*S*Co]
+  bf16* dx;        // [N, IH, IW, Ci]
+  int N, IH, IW, Ci, OH, OW, Co, R, S, stride, pad;
+  int M, K;  // M = N*IH*IW, K = R*S*Co
+};
+
+__device__ inline bf16x8_t dcol_load8(const ConvBwdParams& p, int m, int k,
+                                      bool vec) {
+  bf16x8_t v = {0, 0, 0, 0, 0, 0, 0, 0};
+  const int n = m / (p.IH * p.IW);
+  const int rem = m % (p.IH * p.IW);
+  const int ih = rem / p.IW, iw = rem % p.IW;
+  if (vec) {
+    const int rs = k / p.Co, co = k % p.Co;
+    const int r = rs / p.S, s = rs % p.S;
+    const int ohn = ih + p.pad - r;
+    const int own = iw + p.pad - s;
+    if (ohn >= 0 && own >= 0 && ohn % p.stride == 0 && own % p.stride == 0) {
+      const int oh = ohn / p.stride, ow = own / p.stride;
+      if (oh < p.OH && ow < p.OW)
+        v = *reinterpret_cast<const bf16x8_t*>(
+            p.dy + (((long)n * p.OH + oh) * p.OW + ow) * p.Co + co);
+    }
+  } else {
+    bf16* vv = reinterpret_cast<bf16*>(&v);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int kk = k + j;
+      if (kk < p.K) {
+        const int rs = kk / p.Co, co = kk % p.Co;
+        const int r = rs / p.S, s = rs % p.S;
+        const int ohn = ih + p.pad - r;
+        const int own = iw + p.pad - s;
+        if (ohn >= 0 && own >= 0 && ohn % p.stride == 0 &&
+            own % p.stride == 0) {
+          const int oh = ohn / p.stride, ow = own / p.stride;
+          if (oh < p.OH && ow < p.OW)
+            vv[j] = p.dy[(((long)n * p.OH + oh) * p.OW + ow) * p.Co + co];
+        }
+      }
+    }
+  }
+  return v;
+}
+
+template <int BM, int BN, int WM, int WN>
+__global__ void __launch_bounds__(CONV_BLOCK)
+conv_bwd_data_kernel(const ConvBwdParams p) {
+  constexpr int WTM = BM / WM;
+  constexpr int WTN = BN / WN;
+  constexpr int FA = WTM / 16;
+  constexpr int FB = WTN / 16;
+  constexpr int LDA = BK + 8;
+  constexpr int LDB = BK + 8;
+
+  __shared__ bf16 a_lds[BM * LDA];
+  __shared__ bf16 b_lds[BN * LDB];
+
+  const int t = threadIdx.x;
+  const int lane = t & 63;
+  const int wave = t >> 6;
+  const int wr = wave / WN, wc = wave % WN;
+  const int m0 = blockIdx.x * BM;
+  const int n0 = blockIdx.y * BN;
+  const bool vec = (p.Co & 7) == 0;
+
+  f32x4 acc[FA][FB];
+#pragma unroll
+  for (int i = 0; i < FA; ++i)
+#pragma unroll
+    for (int j = 0; j < FB; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  constexpr int ACH = BM * (BK / 8);
+  constexpr int BCH = BN * (BK / 8);
+
+  for (int kt = 0; kt < p.K; kt += BK) {
+    for (int c = t; c < ACH; c += CONV_BLOCK) {
+      const int row = c / (BK / 8);
+      const int k8 = (c % (BK / 8)) * 8;
+      const int m = m0 + row;
+      const int k = kt + k8;
+      bf16x8_t v = {0, 0, 0, 0, 0, 0, 0, 0};
+      if (m < p.M && k < p.K) v = dcol_load8(p, m, k, vec);
+      *reinterpret_cast<bf16x8_t*>(&a_lds[row * LDA + k8]) = v;
+    }
+    for (int c = t; c < BCH; c += CONV_BLOCK) {
+      const int nrow = c / (BK / 8);
+      const int k8 = (c % (BK / 8)) * 8;
+      const int n = n0 + nrow;
+      const int k = kt + k8;
+      bf16x8_t v = {0, 0, 0, 0, 0, 0, 0, 0};
+      if (n < p.Ci && k + 7 < p.K)
+        v = *reinterpret_cast<const bf16x8_t*>(p.wt + (long)n * p.K + k);
+      else if (n < p.Ci) {
+        bf16* vv = reinterpret_cast<bf16*>(&v);
+        for (int j = 0; j < 8 && k + j < p.K; ++j)
+          vv[j] = p.wt[(long)n * p.K + k + j];
+      }
+      *reinterpret_cast<bf16x8_t*>(&b_lds[nrow * LDB + k8]) = v;
+    }
+    __syncthreads();
+
+    bf16x8_t afrag[FA], bfrag[FB];
+#pragma unroll
+    for (int i = 0; i < FA; ++i) {
+      const int row = wr * WTM + i * 16 + (lane & 15);
+      afrag[i] = *reinterpret_cast<const bf16x8_t*>(
+          &a_lds[row * LDA + (lane >> 4) * 8]);
+    }
+#pragma unroll
+    for (int j = 0; j < FB; ++j) {
+      const int col = wc * WTN + j * 16 + (lane & 15);
+      bfrag[j] = *reinterpret_cast<const bf16x8_t*>(
+          &b_lds[col * LDB + (lane >> 4) * 8]);
+    }
+#pragma unroll
+    for (int i = 0; i < FA; ++i)
+#pragma unroll
+      for (int j = 0; j < FB; ++j)
+        acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int i = 0; i < FA; ++i) {
+#pragma unroll
+    for (int j = 0; j < FB; ++j) {
+      const int col = n0 + wc * WTN + j * 16 + (lane & 15);
+      if (col >= p.Ci) continue;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int m = m0 + wr * WTM + i * 16 + (lane >> 4) * 4 + r;
+        if (m < p.M)
+          p.dx[(long)m * p.Ci + col] = __float2bfloat16(acc[i][j][r]);
+      }
+    }
+  }
+}
+
+// ------------------------------------------------------------------ wrw
+// dW[co][k=(r,s,ci)] = sum_m dy[m][co] * im2col(x)[m][k]
+// Split-K over M with fp32 atomic accumulation into dw (caller zeros).
+// Fragments need transposed LDS reads (operands are [m][*] images);
+// v1 uses per-element ds reads there.
+struct WrwParams {
+  const bf16* x;   // [N, IH, IW, Ci]
+  const bf16* dy;  // [N, OH, OW, Co]
+  float* dw;       // [Co][R*S*Ci] fp32 (natural channels_last layout)
+  int N, IH, IW, Ci, OH, OW, Co, R, S, stride, pad;
+  int M, K;        // M = N*OH*OW (reduction dim), K = R*S*Ci (gemm cols)
+  int m_per_split;
+};
+
+template <int BCO, int BKN>
+__global__ void __launch_bounds__(CONV_BLOCK)
+conv_wrw_kernel(const WrwParams p) {
+  constexpr int WM = (BCO >= 32) ? 2 : 1;
+  constexpr int WN = 4 / WM;
+  constexpr int WTM = BCO / WM;
+  constexpr int WTN = BKN / WN;
+  constexpr int FA = WTM / 16;
+  constexpr int FB = WTN / 16;
+  constexpr int LDR = BCO + 8;   // dy chunk row: [mm][co]
+  constexpr int LDX = BKN + 8;   // x chunk row:  [mm][rsci]
+
+  __shared__ bf16 dy_lds[BK * LDR];
+  __shared__ bf16 x_lds[BK * LDX];
+
+  const int t = threadIdx.x;
+  const int lane = t & 63;
+  const int wave = t >> 6;
+  const int wr = wave / WN, wc = wave % WN;
+  const int co0 = blockIdx.x * BCO;
+  const int k0 = blockIdx.y * BKN;
+  const int mstart = blockIdx.z * p.m_per_split;
+  const int mend = min(p.M, mstart + p.m_per_split);
+  const bool xvec = (p.Ci & 7) == 0;
+  const bool dvec = (p.Co & 7) == 0;
+
+  f32x4 acc[FA][FB];
+#pragma unroll
+  for (int i = 0; i < FA; ++i)
+#pragma unroll
+    for (int j = 0; j < FB; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  constexpr int DCH = BK * (BCO / 8);
+  constexpr int XCH = BK * (BKN / 8);
+
+  for (int mt = mstart; mt < mend; mt += BK) {
+    for (int c = t; c < DCH; c += CONV_BLOCK) {
+      const int mm = c / (BCO / 8);
+      const int c8 = (c % (BCO / 8)) * 8;
+      const int m = mt + mm;
+      bf16x8_t v = {0, 0, 0, 0, 0, 0, 0, 0};
+      if (m < mend && co0 + c8 < p.Co) {
+        if (dvec && co0 + c8 + 7 < p.Co)
+          v = *reinterpret_cast<const bf16x8_t*>(p.dy + (long)m * p.Co + co0 +
+                                                 c8);
+        else {
+          bf16* vv = reinterpret_cast<bf16*>(&v);
+          for (int j = 0; j < 8 && co0 + c8 + j < p.Co; ++j)
+            vv[j] = p.dy[(long)m * p.Co + co0 + c8 + j];
+        }
+      }
+      *reinterpret_cast<bf16x8_t*>(&dy_lds[mm * LDR + c8]) = v;
+    }
+    for (int c = t; c < XCH; c += CONV_BLOCK) {
+      const int mm = c / (BKN / 8);
+      const int k8 = (c % (BKN / 8)) * 8;
+      const int m = mt + mm;
+      const int k = k0 + k8;
+      bf16x8_t v = {0, 0, 0, 0, 0, 0, 0, 0};
+      if (m < mend && k < p.K)
+        v = im2col_load8(p.x, m, k, p.IH, p.IW, p.Ci, p.OH, p.OW, p.S,
+                         p.stride, p.pad, p.K, xvec);
+      *reinterpret_cast<bf16x8_t*>(&x_lds[mm * LDX + k8]) = v;
+    }
+    __syncthreads();
+
+    bf16x8_t afrag[FA], bfrag[FB];
+#pragma unroll
+    for (int i = 0; i < FA; ++i) {
+      const int co = wr * WTM + i * 16 + (lane & 15);
+      bf16* dst = reinterpret_cast<bf16*>(&afrag[i]);
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        dst[j] = dy_lds[((lane >> 4) * 8 + j) * LDR + co];
+    }
+#pragma unroll
+    for (int j = 0; j < FB; ++j) {
+      const int kk = wc * WTN + j * 16 + (lane & 15);
+      bf16* dst = reinterpret_cast<bf16*>(&bfrag[j]);
+#pragma unroll
+      for (int q = 0; q < 8; ++q)
+        dst[q] = x_lds[((lane >> 4) * 8 + q) * LDX + kk];
+    }
+#pragma unroll
+    for (int i = 0; i < FA; ++i)
+#pragma unroll
+      for (int j = 0; j < FB; ++j)
+        acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int i = 0; i < FA; ++i) {
+#pragma unroll
+    for (int j = 0; j < FB; ++j) {
+      const int kk = k0 + wc * WTN + j * 16 + (lane & 15);
+      if (kk >= p.K) continue;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int co = co0 + wr * WTM + i * 16 + (lane >> 4) * 4 + r;
+        if (co < p.Co) atomicAdd(&p.dw[(long)co * p.K + kk], acc[i][j][r]);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------- launch
+static void pick_tile(int Co, int& bm, int& bn) {
+  if (Co >= 128) { bm = 128; bn = 128; }
+  else if (Co >= 64) { bm = 128; bn = 64; }
+  else if (Co >= 32) { bm = 128; bn = 32; }
+  else { bm = 256; bn = 16; }
+}
+
+extern "C" void dlb_conv_fwd(const void* x, const void* w, void* y,
+                             const float* bias, int N, int IH, int IW, int Ci,
+                             int OH, int OW, int Co, int R, int S, int stride,
+                             int pad, hipStream_t stream) {
+  ConvParams p{(const bf16*)x, (const bf16*)w, (bf16*)y, bias, N, IH, IW, Ci,
+               OH, OW, Co, R, S, stride, pad, N * OH * OW, R * S * Ci};
+  int bm, bn;
+  pick_tile(Co, bm, bn);
+  dim3 grid(cdiv(p.M, bm), cdiv(Co, bn));
+  if (bm == 128 && bn == 128)
+    hipLaunchKernelGGL((conv_fwd_kernel<128, 128, 2, 2>), grid,
+                       dim3(CONV_BLOCK), 0, stream, p);
+  else if (bm == 128 && bn == 64)
+    hipLaunchKernelGGL((conv_fwd_kernel<128, 64, 2, 2>), grid,
+                       dim3(CONV_BLOCK), 0, stream, p);
+  else if (bm == 128 && bn == 32)
+    hipLaunchKernelGGL((conv_fwd_kernel<128, 32, 4, 1>), grid,
+                       dim3(CONV_BLOCK), 0, stream, p);
+  else
+    hipLaunchKernelGGL((conv_fwd_kernel<256, 16, 4, 1>), grid,
+                       dim3(CONV_BLOCK), 0, stream, p);
+}
+
+extern "C" void dlb_conv_bwd_data(const void* dy, const void* wt, void* dx,
+                                  int N, int IH, int IW, int Ci, int OH,
+                                  int OW, int Co, int R, int S, int stride,
+                                  int pad, hipStream_t stream) {
+  ConvBwdParams p{(const bf16*)dy, (const bf16*)wt, (bf16*)dx, N, IH, IW, Ci,
+                  OH, OW, Co, R, S, stride, pad, N * IH * IW, R * S * Co};
+  int bm, bn;
+  pick_tile(Ci, bm, bn);
+  dim3 grid(cdiv(p.M, bm), cdiv(Ci, bn));
+  if (bm == 128 && bn == 128)
+    hipLaunchKernelGGL((conv_bwd_data_kernel<128, 128, 2, 2>), grid,
+                       dim3(CONV_BLOCK), 0, stream, p);
+  else if (bm == 128 && bn == 64)
+    hipLaunchKernelGGL((conv_bwd_data_kernel<128, 64, 2, 2>), grid,
+                       dim3(CONV_BLOCK), 0, stream, p);
+  else if (bm == 128 && bn == 32)
+    hipLaunchKernelGGL((conv_bwd_data_kernel<128, 32, 4, 1>), grid,
+                       dim3(CONV_BLOCK), 0, stream, p);
+  else
+    hipLaunchKernelGGL((conv_bwd_data_kernel<256, 16, 4, 1>), grid,
+                       dim3(CONV_BLOCK), 0, stream, p);
+}
+
+extern "C" void dlb_conv_wrw(const void* x, const void* dy, float* dw, int N,
+                             int IH, int IW, int Ci, int OH, int OW, int Co,
+                             int R, int S, int stride, int pad,
+                             hipStream_t stream) {
+  WrwParams p{(const bf16*)x, (const bf16*)dy, dw, N, IH, IW, Ci, OH, OW,
+              Co, R, S, stride, pad, N * OH * OW, R * S * Ci, 0};
+  int splits = 16;
+  while (splits > 1 && p.M / splits < 8 * BK) splits >>= 1;
+  p.m_per_split = cdiv(cdiv(p.M, splits), BK) * BK;
+  splits = cdiv(p.M, p.m_per_split);
+  const int BCO = (Co >= 32) ? 32 : 16;
+  dim3 grid(cdiv(Co, BCO), cdiv(p.K, 64), splits);
+  if (BCO == 32)
+    hipLaunchKernelGGL((conv_wrw_kernel<32, 64>), grid, dim3(CONV_BLOCK), 0,
+                       stream, p);
+  else
+    hipLaunchKernelGGL((conv_wrw_kernel<16, 64>), grid, dim3(CONV_BLOCK), 0,
+                       stream, p);
+}

@@ -20,7 +20,7 @@ import torch.nn.functional as F
 from . import available, ext
 
 # ops with a hand-written gfx950 kernel wired in (grown as kernels land)
-NATIVE_OPS: set[str] = {"group_norm_act"}
+NATIVE_OPS: set[str] = {"group_norm_act", "conv2d"}
 
 
 def _use_native(name: str, x: torch.Tensor) -> bool:
@@ -31,9 +31,16 @@ def _use_native(name: str, x: torch.Tensor) -> bool:
 
 
 def conv2d(x, weight, bias=None, stride=1, padding=0, groups=1):
+    if (x.is_cuda and x.dtype == torch.float32
+            and torch.is_autocast_enabled("cuda")):
+        # network input enters fp32; the conv path computes in bf16
+        x = x.to(torch.bfloat16)
     if _use_native("conv2d", x):
         from . import native
-        return native.conv2d(x, weight, bias, stride, padding, groups)
+        if native.conv_native_ok(x, weight, stride, padding, groups):
+            s = stride if isinstance(stride, int) else stride[0]
+            p = padding if isinstance(padding, int) else padding[0]
+            return native.conv2d(x, weight, bias, s, p, groups)
     return F.conv2d(x, weight, bias, stride=stride, padding=padding, groups=groups)
 
 

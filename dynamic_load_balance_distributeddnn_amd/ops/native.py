@@ -43,6 +43,75 @@ def group_norm_act(x, num_groups, weight, bias, eps=1e-5, relu=False):
     return _GroupNormAct.apply(x, num_groups, weight, bias, eps, relu)
 
 
+class _Conv2d(torch.autograd.Function):
+    """NHWC bf16 implicit-GEMM conv (gfx950 MFMA kernels).
+
+    Weight/bias parameters stay fp32 masters; the forward casts the
+    weight to a channels_last bf16 copy itself (no autocast wrapping),
+    and the backward returns an fp32 weight grad directly — matching the
+    bf16-compute / fp32-master-grad regime of the whole framework.
+    """
+
+    @staticmethod
+    def forward(ctx, x, weight, bias, stride, padding):
+        wcl = weight.detach().to(torch.bfloat16) \
+            .contiguous(memory_format=torch.channels_last)
+        b32 = bias.detach().float() if bias is not None else None
+        y = ext().conv_fwd(x, wcl, b32, stride, padding)
+        ctx.save_for_backward(x, wcl)
+        ctx.conv_args = (stride, padding, bias is not None)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, wcl = ctx.saved_tensors
+        stride, padding, has_bias = ctx.conv_args
+        dy = dy.contiguous(memory_format=torch.channels_last)
+        co, ci, r, s = wcl.shape
+
+        dx = None
+        if ctx.needs_input_grad[0]:
+            # B image for the data-grad GEMM: [Ci][R*S*Co], kernel flipped
+            w4 = wcl.permute(0, 2, 3, 1)          # [Co,R,S,Ci] view
+            wt = w4.flip(1, 2).permute(3, 1, 2, 0).reshape(ci, r * s * co) \
+                .contiguous()
+            dx = ext().conv_bwd_data(dy, wt, x.size(2), x.size(3), ci, r, s,
+                                     stride, padding)
+
+        dweight = None
+        if ctx.needs_input_grad[1]:
+            dwf = ext().conv_wrw(x, dy, r, s, stride, padding)  # [Co, RSCi]
+            # fp32 [Co,R,S,Ci] image -> [Co,Ci,R,S] channels_last-strided
+            dweight = dwf.view(co, r, s, ci).permute(0, 3, 1, 2)
+
+        dbias = None
+        if has_bias and ctx.needs_input_grad[2]:
+            dbias = dy.sum(dim=(0, 2, 3), dtype=torch.float32)
+
+        return dx, dweight, dbias, None, None
+
+
+def conv2d(x, weight, bias=None, stride=1, padding=0, groups=1):
+    assert groups == 1, "grouped conv uses its own kernel path"
+    return _Conv2d.apply(x, weight, bias, stride, padding)
+
+
+def conv_native_ok(x, weight, stride, padding, groups) -> bool:
+    """Envelope of the implicit-GEMM kernels: square stride/pad,
+    ungrouped, bf16 channels_last activations."""
+    if groups != 1 or x.dim() != 4 or x.dtype != torch.bfloat16:
+        return False
+    if not x.is_contiguous(memory_format=torch.channels_last):
+        return False
+    s = stride if isinstance(stride, int) else stride[0]
+    p = padding if isinstance(padding, int) else padding[0]
+    if not isinstance(stride, int) and stride[0] != stride[1]:
+        return False
+    if not isinstance(padding, int) and padding[0] != padding[1]:
+        return False
+    return s in (1, 2) and p >= 0
+
+
 def gn_native_ok(x, num_groups, weight) -> bool:
     """Shape/dtype envelope the fused GN kernel covers (everything the
     zoo produces on the GPU path)."""
