@@ -324,8 +324,13 @@ conv_bwd_data_kernel(const ConvBwdParams p) {
 // ------------------------------------------------------------------ wrw
 // dW[co][k=(r,s,ci)] = sum_m dy[m][co] * im2col(x)[m][k]
 // Split-K over M with fp32 atomic accumulation into dw (caller zeros).
-// Fragments need transposed LDS reads (operands are [m][*] images);
-// v1 uses per-element ds reads there.
+//
+// v2 design notes (v1 measured 267us/call = 46% of the DenseNet step):
+// - wide k-tile (BKN=256) so dy is re-read K/256 times instead of K/64;
+// - both operands staged TRANSPOSED into LDS ([co][m] / [k][m]) via
+//   per-element ds_writes once, so every MFMA fragment read is one
+//   16-byte ds_read (reduction dim m must be register-resident per lane);
+// - split count sized to fill 256 CUs.
 struct WrwParams {
   const bf16* x;   // [N, IH, IW, Ci]
   const bf16* dy;  // [N, OH, OW, Co]
@@ -338,22 +343,19 @@ struct WrwParams {
 template <int BCO, int BKN>
 __global__ void __launch_bounds__(CONV_BLOCK)
 conv_wrw_kernel(const WrwParams p) {
-  constexpr int WM = (BCO >= 32) ? 2 : 1;
-  constexpr int WN = 4 / WM;
-  constexpr int WTM = BCO / WM;
-  constexpr int WTN = BKN / WN;
-  constexpr int FA = WTM / 16;
+  // waves tile the [BCO][BKN] output: 1 x 4 (each wave BCO x BKN/4)
+  constexpr int WTN = BKN / 4;
+  constexpr int FA = BCO / 16;
   constexpr int FB = WTN / 16;
-  constexpr int LDR = BCO + 8;   // dy chunk row: [mm][co]
-  constexpr int LDX = BKN + 8;   // x chunk row:  [mm][rsci]
+  constexpr int LMD = BK + 2;    // [co][m] rows, padded
+  constexpr int LMX = BK + 2;    // [k][m] rows, padded
 
-  __shared__ bf16 dy_lds[BK * LDR];
-  __shared__ bf16 x_lds[BK * LDX];
+  __shared__ bf16 dy_t[BCO * LMD];   // [co][mm]
+  __shared__ bf16 x_t[BKN * LMX];    // [kk][mm]
 
   const int t = threadIdx.x;
   const int lane = t & 63;
   const int wave = t >> 6;
-  const int wr = wave / WN, wc = wave % WN;
   const int co0 = blockIdx.x * BCO;
   const int k0 = blockIdx.y * BKN;
   const int mstart = blockIdx.z * p.m_per_split;
@@ -371,6 +373,7 @@ conv_wrw_kernel(const WrwParams p) {
   constexpr int XCH = BK * (BKN / 8);
 
   for (int mt = mstart; mt < mend; mt += BK) {
+    // stage dy chunk transposed: load [mm][co] contiguously, write [co][mm]
     for (int c = t; c < DCH; c += CONV_BLOCK) {
       const int mm = c / (BCO / 8);
       const int c8 = (c % (BCO / 8)) * 8;
@@ -386,8 +389,11 @@ conv_wrw_kernel(const WrwParams p) {
             vv[j] = p.dy[(long)m * p.Co + co0 + c8 + j];
         }
       }
-      *reinterpret_cast<bf16x8_t*>(&dy_lds[mm * LDR + c8]) = v;
+      const bf16* vv = reinterpret_cast<const bf16*>(&v);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) dy_t[(c8 + j) * LMD + mm] = vv[j];
     }
+    // stage x im2col chunk transposed: [kk][mm]
     for (int c = t; c < XCH; c += CONV_BLOCK) {
       const int mm = c / (BKN / 8);
       const int k8 = (c % (BKN / 8)) * 8;
@@ -397,26 +403,25 @@ conv_wrw_kernel(const WrwParams p) {
       if (m < mend && k < p.K)
         v = im2col_load8(p.x, m, k, p.IH, p.IW, p.Ci, p.OH, p.OW, p.S,
                          p.stride, p.pad, p.K, xvec);
-      *reinterpret_cast<bf16x8_t*>(&x_lds[mm * LDX + k8]) = v;
+      const bf16* vv = reinterpret_cast<const bf16*>(&v);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) x_t[(k8 + j) * LMX + mm] = vv[j];
     }
     __syncthreads();
 
+    // fragments are now plain 16-byte reads along m
     bf16x8_t afrag[FA], bfrag[FB];
 #pragma unroll
     for (int i = 0; i < FA; ++i) {
-      const int co = wr * WTM + i * 16 + (lane & 15);
-      bf16* dst = reinterpret_cast<bf16*>(&afrag[i]);
-#pragma unroll
-      for (int j = 0; j < 8; ++j)
-        dst[j] = dy_lds[((lane >> 4) * 8 + j) * LDR + co];
+      const int co = i * 16 + (lane & 15);
+      afrag[i] = *reinterpret_cast<const bf16x8_t*>(
+          &dy_t[co * LMD + (lane >> 4) * 8]);
     }
 #pragma unroll
     for (int j = 0; j < FB; ++j) {
-      const int kk = wc * WTN + j * 16 + (lane & 15);
-      bf16* dst = reinterpret_cast<bf16*>(&bfrag[j]);
-#pragma unroll
-      for (int q = 0; q < 8; ++q)
-        dst[q] = x_lds[((lane >> 4) * 8 + q) * LDX + kk];
+      const int kk = wave * WTN + j * 16 + (lane & 15);
+      bfrag[j] = *reinterpret_cast<const bf16x8_t*>(
+          &x_t[kk * LMX + (lane >> 4) * 8]);
     }
 #pragma unroll
     for (int i = 0; i < FA; ++i)
@@ -431,11 +436,11 @@ conv_wrw_kernel(const WrwParams p) {
   for (int i = 0; i < FA; ++i) {
 #pragma unroll
     for (int j = 0; j < FB; ++j) {
-      const int kk = k0 + wc * WTN + j * 16 + (lane & 15);
+      const int kk = k0 + wave * WTN + j * 16 + (lane & 15);
       if (kk >= p.K) continue;
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        const int co = co0 + wr * WTM + i * 16 + (lane >> 4) * 4 + r;
+        const int co = co0 + i * 16 + (lane >> 4) * 4 + r;
         if (co < p.Co) atomicAdd(&p.dw[(long)co * p.K + kk], acc[i][j][r]);
       }
     }
@@ -502,16 +507,19 @@ extern "C" void dlb_conv_wrw(const void* x, const void* dy, float* dw, int N,
                              hipStream_t stream) {
   WrwParams p{(const bf16*)x, (const bf16*)dy, dw, N, IH, IW, Ci, OH, OW,
               Co, R, S, stride, pad, N * OH * OW, R * S * Ci, 0};
-  int splits = 16;
-  while (splits > 1 && p.M / splits < 8 * BK) splits >>= 1;
+  const int BCO = (Co >= 32) ? 32 : 16;
+  const int BKN = 256;
+  const long tiles = (long)cdiv(Co, BCO) * cdiv(p.K, BKN);
+  // pick splits so total blocks ~ 2x256 CUs, capped by reduction depth
+  int splits = (int)std::min<long>(std::max<long>(1, 512 / tiles),
+                                   std::max<long>(1, p.M / (8 * BK)));
   p.m_per_split = cdiv(cdiv(p.M, splits), BK) * BK;
   splits = cdiv(p.M, p.m_per_split);
-  const int BCO = (Co >= 32) ? 32 : 16;
-  dim3 grid(cdiv(Co, BCO), cdiv(p.K, 64), splits);
+  dim3 grid(cdiv(Co, BCO), cdiv(p.K, BKN), splits);
   if (BCO == 32)
-    hipLaunchKernelGGL((conv_wrw_kernel<32, 64>), grid, dim3(CONV_BLOCK), 0,
+    hipLaunchKernelGGL((conv_wrw_kernel<32, 256>), grid, dim3(CONV_BLOCK), 0,
                        stream, p);
   else
-    hipLaunchKernelGGL((conv_wrw_kernel<16, 64>), grid, dim3(CONV_BLOCK), 0,
+    hipLaunchKernelGGL((conv_wrw_kernel<16, 256>), grid, dim3(CONV_BLOCK), 0,
                        stream, p);
 }
