@@ -347,8 +347,8 @@ conv_wrw_kernel(const WrwParams p) {
   constexpr int WTN = BKN / 4;
   constexpr int FA = BCO / 16;
   constexpr int FB = WTN / 16;
-  constexpr int LMD = BK + 2;    // [co][m] rows, padded
-  constexpr int LMX = BK + 2;    // [k][m] rows, padded
+  constexpr int LMD = BK + 8;    // [co][m] rows, 16B-aligned for b128
+  constexpr int LMX = BK + 8;
 
   __shared__ bf16 dy_t[BCO * LMD];   // [co][mm]
   __shared__ bf16 x_t[BKN * LMX];    // [kk][mm]
@@ -455,10 +455,20 @@ static void pick_tile(int Co, int& bm, int& bn) {
   else { bm = 256; bn = 16; }
 }
 
+extern "C" bool dlb_conv3x3_fwd_halo(const void* x, const void* w, void* y,
+                                     const float* bias, int N, int H, int W,
+                                     int Ci, int Co, hipStream_t stream);
+extern "C" bool dlb_conv3x3_wrw_halo(const void* x, const void* dy, float* dw,
+                                     int N, int H, int W, int Ci, int Co,
+                                     hipStream_t stream);
+
 extern "C" void dlb_conv_fwd(const void* x, const void* w, void* y,
                              const float* bias, int N, int IH, int IW, int Ci,
                              int OH, int OW, int Co, int R, int S, int stride,
                              int pad, hipStream_t stream) {
+  if (R == 3 && S == 3 && stride == 1 && pad == 1 &&
+      dlb_conv3x3_fwd_halo(x, w, y, bias, N, IH, IW, Ci, Co, stream))
+    return;
   ConvParams p{(const bf16*)x, (const bf16*)w, (bf16*)y, bias, N, IH, IW, Ci,
                OH, OW, Co, R, S, stride, pad, N * OH * OW, R * S * Ci};
   int bm, bn;
@@ -505,6 +515,9 @@ extern "C" void dlb_conv_wrw(const void* x, const void* dy, float* dw, int N,
                              int IH, int IW, int Ci, int OH, int OW, int Co,
                              int R, int S, int stride, int pad,
                              hipStream_t stream) {
+  if (R == 3 && S == 3 && stride == 1 && pad == 1 &&
+      dlb_conv3x3_wrw_halo(x, dy, dw, N, IH, IW, Ci, Co, stream))
+    return;
   WrwParams p{(const bf16*)x, (const bf16*)dy, dw, N, IH, IW, Ci, OH, OW,
               Co, R, S, stride, pad, N * OH * OW, R * S * Ci, 0};
   const int BCO = (Co >= 32) ? 32 : 16;
