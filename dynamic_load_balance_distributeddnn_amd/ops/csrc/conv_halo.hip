@@ -142,29 +142,30 @@ conv3x3_fwd_halo(const bf16* __restrict__ x, const bf16* __restrict__ w,
 // ------------------------------------------------------------------ wrw
 // dW[co][tap*Ci+ci] += sum_m dy[m][co] * x[tap(m)][ci]
 // Block: [32 co] x [9 taps x 32 ci = 288 k-cols], m-split over chunks of
-// 32 output pixels.  Per chunk: dy staged transposed ([co][m]); the x
-// halo band is staged once and expanded into an im2col-transposed image
-// ([k][m]) inside LDS — global x traffic is 1x.
-// Requires OH*OW % 32 == 0 (chunk never crosses a sample boundary).
+// 128 output pixels (whole output rows).  Per chunk: dy staged
+// transposed ([co][m]) with vector fragment reads; the x halo band is
+// staged once (global traffic 1x) and B-fragments gather from it
+// directly (8 scalar LDS reads per fragment — exactly the consumption,
+// cheaper than materializing the im2col image).
+// Requires OH*OW % 128 == 0... actually % BM == 0 handled by loop guard;
+// W must be a power of two <= 32 (runtime shifts).
 __global__ void __launch_bounds__(HBLOCK)
 conv3x3_wrw_halo(const bf16* __restrict__ x, const bf16* __restrict__ dy,
                  float* __restrict__ dw, const int N, const int H,
                  const int W, const int Ci, const int Co,
-                 const int m_per_split) {
+                 const int m_per_split, const int wshift) {
   constexpr int BCO = 32;
-  constexpr int BKC = 9 * CI_CHUNK;   // 288 k-cols per block
-  constexpr int BM = 32;              // reduction chunk
-  constexpr int LMD = BM + 8;  // 16B-aligned rows for ds_read_b128
-  // wave k-fragment ownership: 18 fragments split 5/5/4/4
+  constexpr int BM = 128;             // reduction chunk (whole rows)
+  constexpr int LMD = BM + 8;         // 16B-aligned [co][m] rows
   const int t = threadIdx.x;
   const int lane = t & 63;
   const int wave = t >> 6;
+  // 18 k-fragments split 5/5/4/4 across the 4 waves
   const int fb_count = (wave < 2) ? 5 : 4;
   const int fb_base = (wave < 2) ? wave * 5 : 10 + (wave - 2) * 4;
 
-  __shared__ bf16 dy_t[BCO * LMD];           // [co][m]
-  __shared__ bf16 col_t[BKC * LMD];          // [k][m] im2col image
-  __shared__ bf16 halo[102 * CI_CHUNK];      // raw band (worst 3x34 pix)
+  __shared__ bf16 dy_t[BCO * LMD];        // [co][m]
+  __shared__ bf16 halo[204 * CI_CHUNK];   // band: (BM/W+2) x (W+2) pixels
 
   const int co0 = blockIdx.x * BCO;
   const int ci0 = blockIdx.y * CI_CHUNK;
@@ -172,10 +173,11 @@ conv3x3_wrw_halo(const bf16* __restrict__ x, const bf16* __restrict__ dy,
   const int M = N * H * W;
   const int mstart = blockIdx.z * m_per_split;
   const int mend = min(M, mstart + m_per_split);
+  const int wmask = W - 1;
 
-  const int rows_per_chunk = BM / W > 0 ? BM / W : 1;  // W<=32
+  const int rows_per_chunk = BM >> wshift;        // BM/W (W<=32 => >=4)
   const int hh_rows = rows_per_chunk + 2;
-  const int hw_cols = (W < BM ? W : BM) + 2;
+  const int hw_cols = W + 2;
 
   f32x4 acc[2][5];
 #pragma unroll
@@ -184,7 +186,7 @@ conv3x3_wrw_halo(const bf16* __restrict__ x, const bf16* __restrict__ dy,
     for (int j = 0; j < 5; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
   for (int mt = mstart; mt < mend; mt += BM) {
-    // ---- stage dy transposed
+    // ---- stage dy transposed [co][mm]
     constexpr int DCH = BM * (BCO / 8);
     for (int c = t; c < DCH; c += HBLOCK) {
       const int mm = c / (BCO / 8);
@@ -204,8 +206,7 @@ conv3x3_wrw_halo(const bf16* __restrict__ x, const bf16* __restrict__ dy,
     }
     // ---- stage x band: rows [py0-1 .. py0+rows], cols [-1..W]
     const int n = mt / (H * W);
-    const int rem = mt % (H * W);
-    const int py0 = rem / W;  // chunk starts at (py0, 0): mt % W == 0 since W|32... (W in {8,16,32}; for W<32, 32%W==0)
+    const int py0 = (mt % (H * W)) >> wshift;   // chunk starts at col 0
     const int band_ch = hh_rows * hw_cols * (CI_CHUNK / 8);
     for (int c = t; c < band_ch; c += HBLOCK) {
       const int c8 = (c % (CI_CHUNK / 8)) * 8;
@@ -219,46 +220,36 @@ conv3x3_wrw_halo(const bf16* __restrict__ x, const bf16* __restrict__ dy,
       *reinterpret_cast<bf16x8_t*>(&halo[(hh * hw_cols + ww) * CI_CHUNK + c8]) = v;
     }
     __syncthreads();
-    // ---- expand halo -> col_t[k][m]: k = tap*32 + ci_local
-    // chunk: (k, m8): thread gathers 8 m values for one k
-    constexpr int CCH = BKC * (BM / 8);
-    for (int c = t; c < CCH; c += HBLOCK) {
-      const int m8 = (c % (BM / 8)) * 8;
-      const int k = c / (BM / 8);
-      const int tap = k / CI_CHUNK, cil = k % CI_CHUNK;
-      const int r = tap / 3, s = tap % 3;
-      bf16 tmp[8];
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const int mm = m8 + j;
-        const int py = mm / W, px = mm % W;  // relative to chunk start
-        // halo row = (py + r - 1) - (py0-1-py0) ... halo covers py0-1+hh
-        const int hh = py + r;               // since band starts at py0-1
-        const int ww = px + s;               // band cols start at -1
-        tmp[j] = halo[(hh * hw_cols + ww) * CI_CHUNK + cil];
-      }
-#pragma unroll
-      for (int j = 0; j < 8; ++j) col_t[k * LMD + m8 + j] = tmp[j];
-    }
-    __syncthreads();
 
-    // ---- fragments: A = dy_t rows (co), B = col_t rows (k); kdim = m
-    bf16x8_t afrag[2];
+    // ---- 4 MFMA sub-steps of 32 m each
 #pragma unroll
-    for (int i = 0; i < 2; ++i) {
-      const int co = i * 16 + (lane & 15);
-      afrag[i] = *reinterpret_cast<const bf16x8_t*>(
-          &dy_t[co * LMD + (lane >> 4) * 8]);
-    }
-    for (int j = 0; j < 5; ++j) {
-      if (j >= fb_count) break;
-      const int kk = (fb_base + j) * 16 + (lane & 15);
-      bf16x8_t bfrag = *reinterpret_cast<const bf16x8_t*>(
-          &col_t[kk * LMD + (lane >> 4) * 8]);
+    for (int sub = 0; sub < 4; ++sub) {
+      const int msub = sub * 32;
+      bf16x8_t afrag[2];
 #pragma unroll
-      for (int i = 0; i < 2; ++i)
-        acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            afrag[i], bfrag, acc[i][j], 0, 0, 0);
+      for (int i = 0; i < 2; ++i) {
+        const int co = i * 16 + (lane & 15);
+        afrag[i] = *reinterpret_cast<const bf16x8_t*>(
+            &dy_t[co * LMD + msub + (lane >> 4) * 8]);
+      }
+      for (int j = 0; j < 5; ++j) {
+        if (j >= fb_count) break;
+        const int kk = (fb_base + j) * 16 + (lane & 15);
+        const int tap = kk / CI_CHUNK, cil = kk % CI_CHUNK;
+        const int r = tap / 3, sxx = tap % 3;
+        bf16x8_t bfrag;
+        bf16* bp = reinterpret_cast<bf16*>(&bfrag);
+#pragma unroll
+        for (int q = 0; q < 8; ++q) {
+          const int mm = msub + (lane >> 4) * 8 + q;
+          const int py = mm >> wshift, px = mm & wmask;
+          bp[q] = halo[((py + r) * hw_cols + px + sxx) * CI_CHUNK + cil];
+        }
+#pragma unroll
+        for (int i = 0; i < 2; ++i)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag[i], bfrag, acc[i][j], 0, 0, 0);
+      }
     }
     __syncthreads();
   }
@@ -305,16 +296,18 @@ extern "C" bool dlb_conv3x3_wrw_halo(const void* x, const void* dy, float* dw,
                                      int N, int H, int W, int Ci, int Co,
                                      hipStream_t stream) {
   if (Ci % CI_CHUNK != 0 || Co % 8 != 0) return false;
-  if ((H * W) % 32 != 0 || W > 32) return false;
+  if (W > 32 || (W & (W - 1)) != 0 || (H * W) % 128 != 0) return false;
   const int M = N * H * W;
+  int wshift = 0;
+  while ((1 << wshift) < W) ++wshift;
   const long tiles = (long)cdiv(Co, 32) * (Ci / CI_CHUNK);
-  int splits = (int)std::min<long>(std::max<long>(1, 512 / tiles),
-                                   std::max<long>(1, M / (8 * 32)));
-  int m_per_split = cdiv(cdiv(M, splits), 32) * 32;
+  int splits = (int)std::min<long>(std::max<long>(1, 1024 / tiles),
+                                   std::max<long>(1, M / (4 * 128)));
+  int m_per_split = cdiv(cdiv(M, splits), 128) * 128;
   splits = cdiv(M, m_per_split);
   dim3 grid(cdiv(Co, 32), Ci / CI_CHUNK, splits);
   hipLaunchKernelGGL(conv3x3_wrw_halo, grid, dim3(HBLOCK), 0, stream,
                      (const bf16*)x, (const bf16*)dy, dw, N, H, W, Ci, Co,
-                     m_per_split);
+                     m_per_split, wshift);
   return true;
 }
