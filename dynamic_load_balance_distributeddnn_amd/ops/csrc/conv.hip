@@ -524,8 +524,8 @@ extern "C" void dlb_conv_wrw(const void* x, const void* dy, float* dw, int N,
   const int BKN = 256;
   const long tiles = (long)cdiv(Co, BCO) * cdiv(p.K, BKN);
   // pick splits so total blocks ~ 2x256 CUs, capped by reduction depth
-  int splits = (int)std::min<long>(std::max<long>(1, 512 / tiles),
-                                   std::max<long>(1, p.M / (8 * BK)));
+  int splits = (int)std::min<long>(std::max<long>(1, 2048 / tiles),
+                                   std::max<long>(1, p.M / (4 * BK)));
   p.m_per_split = cdiv(cdiv(p.M, splits), BK) * BK;
   splits = cdiv(p.M, p.m_per_split);
   dim3 grid(cdiv(Co, BCO), cdiv(p.K, BKN), splits);

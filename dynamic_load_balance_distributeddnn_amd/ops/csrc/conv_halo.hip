@@ -232,8 +232,9 @@ conv3x3_wrw_halo(const bf16* __restrict__ x, const bf16* __restrict__ dy,
         afrag[i] = *reinterpret_cast<const bf16x8_t*>(
             &dy_t[co * LMD + msub + (lane >> 4) * 8]);
       }
+#pragma unroll
       for (int j = 0; j < 5; ++j) {
-        if (j >= fb_count) break;
+        if (j >= fb_count) continue;
         const int kk = (fb_base + j) * 16 + (lane & 15);
         const int tap = kk / CI_CHUNK, cil = kk % CI_CHUNK;
         const int r = tap / 3, sxx = tap % 3;
@@ -255,8 +256,9 @@ conv3x3_wrw_halo(const bf16* __restrict__ x, const bf16* __restrict__ dy,
   }
 
   // ---- publish: dw[co][tap*Ci + ci0 + cil]
+#pragma unroll
   for (int j = 0; j < 5; ++j) {
-    if (j >= fb_count) break;
+    if (j >= fb_count) continue;
     const int klocal = (fb_base + j) * 16 + (lane & 15);
     const int tap = klocal / CI_CHUNK, cil = klocal % CI_CHUNK;
     const int kg = tap * Ci + ci0 + cil;
@@ -301,8 +303,8 @@ extern "C" bool dlb_conv3x3_wrw_halo(const void* x, const void* dy, float* dw,
   int wshift = 0;
   while ((1 << wshift) < W) ++wshift;
   const long tiles = (long)cdiv(Co, 32) * (Ci / CI_CHUNK);
-  int splits = (int)std::min<long>(std::max<long>(1, 1024 / tiles),
-                                   std::max<long>(1, M / (4 * 128)));
+  int splits = (int)std::min<long>(std::max<long>(1, 2048 / tiles),
+                                   std::max<long>(1, M / (2 * 128)));
   int m_per_split = cdiv(cdiv(M, splits), 128) * 128;
   splits = cdiv(M, m_per_split);
   dim3 grid(cdiv(Co, 32), Ci / CI_CHUNK, splits);
