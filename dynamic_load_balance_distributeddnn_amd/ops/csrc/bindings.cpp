@@ -81,8 +81,10 @@ extern "C" void dlb_conv_bwd_data(const void* dy, const void* wt, void* dx,
                                   int pad, hipStream_t stream);
 extern "C" void dlb_conv_wrw(const void* x, const void* dy, float* dw, int N,
                              int IH, int IW, int Ci, int OH, int OW, int Co,
-                             int R, int S, int stride, int pad,
+                             int R, int S, int stride, int pad, int splits,
                              hipStream_t stream);
+extern "C" int dlb_conv_wrw_nsplits(int N, int OH, int OW, int Ci, int Co,
+                                    int R, int S);
 
 static inline bool is_cl(const torch::Tensor& t) {
   return t.is_contiguous(torch::MemoryFormat::ChannelsLast);
@@ -131,18 +133,21 @@ static torch::Tensor conv_bwd_data(torch::Tensor dy, torch::Tensor wt,
 }
 
 // Returns dw fp32 [Co, R*S*Ci] (the [Co][R][S][Ci] channels_last image).
+// Split-K partials go to per-split slabs (no atomic contention) and are
+// reduced with one sum kernel here.
 static torch::Tensor conv_wrw(torch::Tensor x, torch::Tensor dy, int64_t R,
                               int64_t S, int64_t stride, int64_t pad) {
   TORCH_CHECK(x.is_cuda() && is_cl(x) && is_cl(dy));
   const int N = x.size(0), Ci = x.size(1), IH = x.size(2), IW = x.size(3);
   const int Co = dy.size(1), OH = dy.size(2), OW = dy.size(3);
-  auto dw = torch::zeros({Co, R * S * Ci},
-                         x.options().dtype(torch::kFloat32));
+  const int splits = dlb_conv_wrw_nsplits(N, OH, OW, Ci, Co, (int)R, (int)S);
+  auto part = torch::zeros({splits, Co, R * S * Ci},
+                           x.options().dtype(torch::kFloat32));
   auto stream = at::hip::getCurrentHIPStream();
-  dlb_conv_wrw(x.data_ptr(), dy.data_ptr(), dw.data_ptr<float>(), N, IH, IW,
-               Ci, OH, OW, Co, (int)R, (int)S, (int)stride, (int)pad,
-               stream.stream());
-  return dw;
+  dlb_conv_wrw(x.data_ptr(), dy.data_ptr(), part.data_ptr<float>(), N, IH,
+               IW, Ci, OH, OW, Co, (int)R, (int)S, (int)stride, (int)pad,
+               splits, stream.stream());
+  return splits == 1 ? part.squeeze(0) : part.sum(0);
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {

@@ -432,6 +432,9 @@ conv_wrw_kernel(const WrwParams p) {
     __syncthreads();
   }
 
+  // plain stores into this split's slab (no atomics — contention-free;
+  // the host reduces over splits with one sum kernel)
+  float* slab = p.dw + (long)blockIdx.z * p.Co * p.K;
 #pragma unroll
   for (int i = 0; i < FA; ++i) {
 #pragma unroll
@@ -441,7 +444,7 @@ conv_wrw_kernel(const WrwParams p) {
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int co = co0 + i * 16 + (lane >> 4) * 4 + r;
-        if (co < p.Co) atomicAdd(&p.dw[(long)co * p.K + kk], acc[i][j][r]);
+        if (co < p.Co) slab[(long)co * p.K + kk] = acc[i][j][r];
       }
     }
   }
@@ -460,7 +463,7 @@ extern "C" bool dlb_conv3x3_fwd_halo(const void* x, const void* w, void* y,
                                      int Ci, int Co, hipStream_t stream);
 extern "C" bool dlb_conv3x3_wrw_halo(const void* x, const void* dy, float* dw,
                                      int N, int H, int W, int Ci, int Co,
-                                     hipStream_t stream);
+                                     int splits, hipStream_t stream);
 
 extern "C" void dlb_conv_fwd(const void* x, const void* w, void* y,
                              const float* bias, int N, int IH, int IW, int Ci,
@@ -511,24 +514,37 @@ extern "C" void dlb_conv_bwd_data(const void* dy, const void* wt, void* dx,
                        dim3(CONV_BLOCK), 0, stream, p);
 }
 
+// split-count query: how many per-split slabs the wrw launch will write.
+extern "C" int dlb_conv_wrw_nsplits(int N, int OH, int OW, int Ci, int Co,
+                                    int R, int S) {
+  const int M = N * OH * OW;
+  const int K = R * S * Ci;
+  long tiles;
+  if (R == 3 && S == 3 && Ci % 32 == 0 && Co % 8 == 0 && OW <= 32 &&
+      (OW & (OW - 1)) == 0 && (OH * OW) % 128 == 0)
+    tiles = (long)cdiv(Co, 32) * (Ci / 32);            // halo path
+  else
+    tiles = (long)cdiv(Co, (Co >= 32) ? 32 : 16) * cdiv(K, 256);
+  int splits = (int)std::min<long>(std::max<long>(1, 2048 / tiles),
+                                   std::max<long>(1, M / (4 * BK)));
+  // both paths round m_per_split to their chunk; recompute exact count
+  const int chunk = (R == 3 && S == 3 && Ci % 32 == 0 && Co % 8 == 0 &&
+                     OW <= 32 && (OW & (OW - 1)) == 0 &&
+                     (OH * OW) % 128 == 0) ? 128 : BK;
+  int mps = cdiv(cdiv(M, splits), chunk) * chunk;
+  return cdiv(M, mps);
+}
+
 extern "C" void dlb_conv_wrw(const void* x, const void* dy, float* dw, int N,
                              int IH, int IW, int Ci, int OH, int OW, int Co,
-                             int R, int S, int stride, int pad,
+                             int R, int S, int stride, int pad, int splits,
                              hipStream_t stream) {
-  if (R == 3 && S == 3 && stride == 1 && pad == 1 &&
-      dlb_conv3x3_wrw_halo(x, dy, dw, N, IH, IW, Ci, Co, stream))
-    return;
   WrwParams p{(const bf16*)x, (const bf16*)dy, dw, N, IH, IW, Ci, OH, OW,
               Co, R, S, stride, pad, N * OH * OW, R * S * Ci, 0};
   const int BCO = (Co >= 32) ? 32 : 16;
-  const int BKN = 256;
-  const long tiles = (long)cdiv(Co, BCO) * cdiv(p.K, BKN);
-  // pick splits so total blocks ~ 2x256 CUs, capped by reduction depth
-  int splits = (int)std::min<long>(std::max<long>(1, 2048 / tiles),
-                                   std::max<long>(1, p.M / (4 * BK)));
   p.m_per_split = cdiv(cdiv(p.M, splits), BK) * BK;
   splits = cdiv(p.M, p.m_per_split);
-  dim3 grid(cdiv(Co, BCO), cdiv(p.K, BKN), splits);
+  dim3 grid(cdiv(Co, BCO), cdiv(p.K, 256), splits);
   if (BCO == 32)
     hipLaunchKernelGGL((conv_wrw_kernel<32, 256>), grid, dim3(CONV_BLOCK), 0,
                        stream, p);

@@ -256,6 +256,7 @@ conv3x3_wrw_halo(const bf16* __restrict__ x, const bf16* __restrict__ dy,
   }
 
   // ---- publish: dw[co][tap*Ci + ci0 + cil]
+  float* slab = dw + (long)blockIdx.z * Co * K;
 #pragma unroll
   for (int j = 0; j < 5; ++j) {
     if (j >= fb_count) continue;
@@ -267,7 +268,7 @@ conv3x3_wrw_halo(const bf16* __restrict__ x, const bf16* __restrict__ dy,
 #pragma unroll
       for (int rr = 0; rr < 4; ++rr) {
         const int co = co0 + i * 16 + (lane >> 4) * 4 + rr;
-        if (co < Co) atomicAdd(&dw[(long)co * K + kg], acc[i][j][rr]);
+        if (co < Co) slab[(long)co * K + kg] = acc[i][j][rr];
       }
     }
   }
@@ -296,15 +297,12 @@ extern "C" bool dlb_conv3x3_fwd_halo(const void* x, const void* w, void* y,
 
 extern "C" bool dlb_conv3x3_wrw_halo(const void* x, const void* dy, float* dw,
                                      int N, int H, int W, int Ci, int Co,
-                                     hipStream_t stream) {
+                                     int splits, hipStream_t stream) {
   if (Ci % CI_CHUNK != 0 || Co % 8 != 0) return false;
   if (W > 32 || (W & (W - 1)) != 0 || (H * W) % 128 != 0) return false;
   const int M = N * H * W;
   int wshift = 0;
   while ((1 << wshift) < W) ++wshift;
-  const long tiles = (long)cdiv(Co, 32) * (Ci / CI_CHUNK);
-  int splits = (int)std::min<long>(std::max<long>(1, 2048 / tiles),
-                                   std::max<long>(1, M / (2 * 128)));
   int m_per_split = cdiv(cdiv(M, splits), 128) * 128;
   splits = cdiv(M, m_per_split);
   dim3 grid(cdiv(Co, 32), Ci / CI_CHUNK, splits);
