@@ -140,7 +140,11 @@ static torch::Tensor conv_wrw(torch::Tensor x, torch::Tensor dy, int64_t R,
   TORCH_CHECK(x.is_cuda() && is_cl(x) && is_cl(dy));
   const int N = x.size(0), Ci = x.size(1), IH = x.size(2), IW = x.size(3);
   const int Co = dy.size(1), OH = dy.size(2), OW = dy.size(3);
-  const int splits = dlb_conv_wrw_nsplits(N, OH, OW, Ci, Co, (int)R, (int)S);
+  int splits = dlb_conv_wrw_nsplits(N, OH, OW, Ci, Co, (int)R, (int)S);
+  if (const char* env = getenv("DLB_WRW_SPLITS")) {
+    int v = atoi(env);
+    if (v > 0 && v < splits) splits = v;
+  }
   auto part = torch::zeros({splits, Co, R * S * Ci},
                            x.options().dtype(torch::kFloat32));
   auto stream = at::hip::getCurrentHIPStream();

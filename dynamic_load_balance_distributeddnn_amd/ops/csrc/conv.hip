@@ -323,35 +323,38 @@ conv_bwd_data_kernel(const ConvBwdParams p) {
 
 // ------------------------------------------------------------------ wrw
 // dW[co][k=(r,s,ci)] = sum_m dy[m][co] * im2col(x)[m][k]
-// Split-K over M with fp32 atomic accumulation into dw (caller zeros).
 //
-// v2 design notes (v1 measured 267us/call = 46% of the DenseNet step):
-// - wide k-tile (BKN=256) so dy is re-read K/256 times instead of K/64;
-// - both operands staged TRANSPOSED into LDS ([co][m] / [k][m]) via
-//   per-element ds_writes once, so every MFMA fragment read is one
-//   16-byte ds_read (reduction dim m must be register-resident per lane);
-// - split count sized to fill 256 CUs.
+// v3: software-pipelined split-K GEMM.  The reduction dim (m = output
+// pixels) is huge and both operands are m-major in memory, so each
+// m-chunk is loaded to REGISTERS first, the previous chunk's LDS image
+// is consumed by MFMAs while those loads are in flight, then the chunk
+// is written (transposed) to LDS for the next round — global latency
+// hides under compute (the un-pipelined v1/v2 measured 4-16us PER CHUNK
+// of pure latency).  Partials go to per-split slabs; host reduces.
 struct WrwParams {
   const bf16* x;   // [N, IH, IW, Ci]
   const bf16* dy;  // [N, OH, OW, Co]
-  float* dw;       // [Co][R*S*Ci] fp32 (natural channels_last layout)
+  float* dw;       // [splits][Co][R*S*Ci] fp32 slabs
   int N, IH, IW, Ci, OH, OW, Co, R, S, stride, pad;
-  int M, K;        // M = N*OH*OW (reduction dim), K = R*S*Ci (gemm cols)
+  int M, K;
   int m_per_split;
 };
 
 template <int BCO, int BKN>
 __global__ void __launch_bounds__(CONV_BLOCK)
 conv_wrw_kernel(const WrwParams p) {
-  // waves tile the [BCO][BKN] output: 1 x 4 (each wave BCO x BKN/4)
-  constexpr int WTN = BKN / 4;
+  constexpr int BM = 64;              // m-chunk per pipeline stage
+  constexpr int WTN = BKN / 4;        // wave k-columns
   constexpr int FA = BCO / 16;
   constexpr int FB = WTN / 16;
-  constexpr int LMD = BK + 8;    // [co][m] rows, 16B-aligned for b128
-  constexpr int LMX = BK + 8;
+  constexpr int LMD = BM + 8;         // 16B-aligned [row][m] strides
+  constexpr int DCH = BM * (BCO / 8); // dy chunks per stage
+  constexpr int XCH = BM * (BKN / 8); // x chunks per stage
+  constexpr int DPT = (DCH + CONV_BLOCK - 1) / CONV_BLOCK;
+  constexpr int XPT = (XCH + CONV_BLOCK - 1) / CONV_BLOCK;
 
-  __shared__ bf16 dy_t[BCO * LMD];   // [co][mm]
-  __shared__ bf16 x_t[BKN * LMX];    // [kk][mm]
+  __shared__ bf16 dy_t[BCO * LMD];    // [co][m]
+  __shared__ bf16 x_t[BKN * LMD];     // [k][m]
 
   const int t = threadIdx.x;
   const int lane = t & 63;
@@ -369,71 +372,111 @@ conv_wrw_kernel(const WrwParams p) {
 #pragma unroll
     for (int j = 0; j < FB; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
-  constexpr int DCH = BK * (BCO / 8);
-  constexpr int XCH = BK * (BKN / 8);
+  bf16x8_t dreg[DPT], xreg[XPT];
 
-  for (int mt = mstart; mt < mend; mt += BK) {
-    // stage dy chunk transposed: load [mm][co] contiguously, write [co][mm]
-    for (int c = t; c < DCH; c += CONV_BLOCK) {
-      const int mm = c / (BCO / 8);
-      const int c8 = (c % (BCO / 8)) * 8;
-      const int m = mt + mm;
+  auto load_chunk = [&](int mt) {
+#pragma unroll
+    for (int u = 0; u < DPT; ++u) {
+      const int c = t + u * CONV_BLOCK;
       bf16x8_t v = {0, 0, 0, 0, 0, 0, 0, 0};
-      if (m < mend && co0 + c8 < p.Co) {
-        if (dvec && co0 + c8 + 7 < p.Co)
-          v = *reinterpret_cast<const bf16x8_t*>(p.dy + (long)m * p.Co + co0 +
-                                                 c8);
-        else {
-          bf16* vv = reinterpret_cast<bf16*>(&v);
-          for (int j = 0; j < 8 && co0 + c8 + j < p.Co; ++j)
-            vv[j] = p.dy[(long)m * p.Co + co0 + c8 + j];
+      if (c < DCH) {
+        const int mm = c / (BCO / 8);
+        const int c8 = (c % (BCO / 8)) * 8;
+        const int m = mt + mm;
+        if (m < mend && co0 + c8 < p.Co) {
+          if (dvec && co0 + c8 + 7 < p.Co)
+            v = *reinterpret_cast<const bf16x8_t*>(p.dy + (long)m * p.Co +
+                                                   co0 + c8);
+          else {
+            bf16* vv = reinterpret_cast<bf16*>(&v);
+            for (int j = 0; j < 8 && co0 + c8 + j < p.Co; ++j)
+              vv[j] = p.dy[(long)m * p.Co + co0 + c8 + j];
+          }
         }
       }
-      const bf16* vv = reinterpret_cast<const bf16*>(&v);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) dy_t[(c8 + j) * LMD + mm] = vv[j];
+      dreg[u] = v;
     }
-    // stage x im2col chunk transposed: [kk][mm]
-    for (int c = t; c < XCH; c += CONV_BLOCK) {
-      const int mm = c / (BKN / 8);
-      const int k8 = (c % (BKN / 8)) * 8;
-      const int m = mt + mm;
-      const int k = k0 + k8;
+#pragma unroll
+    for (int u = 0; u < XPT; ++u) {
+      const int c = t + u * CONV_BLOCK;
       bf16x8_t v = {0, 0, 0, 0, 0, 0, 0, 0};
-      if (m < mend && k < p.K)
-        v = im2col_load8(p.x, m, k, p.IH, p.IW, p.Ci, p.OH, p.OW, p.S,
-                         p.stride, p.pad, p.K, xvec);
-      const bf16* vv = reinterpret_cast<const bf16*>(&v);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) x_t[(k8 + j) * LMX + mm] = vv[j];
+      if (c < XCH) {
+        const int mm = c / (BKN / 8);
+        const int k8 = (c % (BKN / 8)) * 8;
+        const int m = mt + mm;
+        const int k = k0 + k8;
+        if (m < mend && k < p.K)
+          v = im2col_load8(p.x, m, k, p.IH, p.IW, p.Ci, p.OH, p.OW, p.S,
+                           p.stride, p.pad, p.K, xvec);
+      }
+      xreg[u] = v;
     }
-    __syncthreads();
+  };
 
-    // fragments are now plain 16-byte reads along m
-    bf16x8_t afrag[FA], bfrag[FB];
+  auto write_chunk = [&]() {
 #pragma unroll
-    for (int i = 0; i < FA; ++i) {
-      const int co = i * 16 + (lane & 15);
-      afrag[i] = *reinterpret_cast<const bf16x8_t*>(
-          &dy_t[co * LMD + (lane >> 4) * 8]);
+    for (int u = 0; u < DPT; ++u) {
+      const int c = t + u * CONV_BLOCK;
+      if (c < DCH) {
+        const int mm = c / (BCO / 8);
+        const int c8 = (c % (BCO / 8)) * 8;
+        const bf16* vv = reinterpret_cast<const bf16*>(&dreg[u]);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) dy_t[(c8 + j) * LMD + mm] = vv[j];
+      }
     }
 #pragma unroll
-    for (int j = 0; j < FB; ++j) {
-      const int kk = wave * WTN + j * 16 + (lane & 15);
-      bfrag[j] = *reinterpret_cast<const bf16x8_t*>(
-          &x_t[kk * LMX + (lane >> 4) * 8]);
+    for (int u = 0; u < XPT; ++u) {
+      const int c = t + u * CONV_BLOCK;
+      if (c < XCH) {
+        const int mm = c / (BKN / 8);
+        const int k8 = (c % (BKN / 8)) * 8;
+        const bf16* vv = reinterpret_cast<const bf16*>(&xreg[u]);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) x_t[(k8 + j) * LMD + mm] = vv[j];
+      }
     }
+  };
+
+  // prologue: load + write chunk 0
+  load_chunk(mstart);
+  write_chunk();
+  __syncthreads();
+
+  for (int mt = mstart; mt < mend; mt += BM) {
+    // issue next chunk's global loads (latency hides under the MFMAs)
+    if (mt + BM < mend) load_chunk(mt + BM);
+
+    // consume current LDS image: BM/32 sub-steps of k-depth 32
 #pragma unroll
-    for (int i = 0; i < FA; ++i)
+    for (int sub = 0; sub < BM / 32; ++sub) {
+      bf16x8_t afrag[FA], bfrag[FB];
 #pragma unroll
-      for (int j = 0; j < FB; ++j)
-        acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
+      for (int i = 0; i < FA; ++i) {
+        const int co = i * 16 + (lane & 15);
+        afrag[i] = *reinterpret_cast<const bf16x8_t*>(
+            &dy_t[co * LMD + sub * 32 + (lane >> 4) * 8]);
+      }
+#pragma unroll
+      for (int j = 0; j < FB; ++j) {
+        const int kk = wave * WTN + j * 16 + (lane & 15);
+        bfrag[j] = *reinterpret_cast<const bf16x8_t*>(
+            &x_t[kk * LMD + sub * 32 + (lane >> 4) * 8]);
+      }
+#pragma unroll
+      for (int i = 0; i < FA; ++i)
+#pragma unroll
+        for (int j = 0; j < FB; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
+    }
     __syncthreads();
+    if (mt + BM < mend) {
+      write_chunk();
+      __syncthreads();
+    }
   }
 
-  // plain stores into this split's slab (no atomics — contention-free;
-  // the host reduces over splits with one sum kernel)
   float* slab = p.dw + (long)blockIdx.z * p.Co * p.K;
 #pragma unroll
   for (int i = 0; i < FA; ++i) {
@@ -519,19 +562,11 @@ extern "C" int dlb_conv_wrw_nsplits(int N, int OH, int OW, int Ci, int Co,
                                     int R, int S) {
   const int M = N * OH * OW;
   const int K = R * S * Ci;
-  long tiles;
-  if (R == 3 && S == 3 && Ci % 32 == 0 && Co % 8 == 0 && OW <= 32 &&
-      (OW & (OW - 1)) == 0 && (OH * OW) % 128 == 0)
-    tiles = (long)cdiv(Co, 32) * (Ci / 32);            // halo path
-  else
-    tiles = (long)cdiv(Co, (Co >= 32) ? 32 : 16) * cdiv(K, 256);
-  int splits = (int)std::min<long>(std::max<long>(1, 2048 / tiles),
-                                   std::max<long>(1, M / (4 * BK)));
-  // both paths round m_per_split to their chunk; recompute exact count
-  const int chunk = (R == 3 && S == 3 && Ci % 32 == 0 && Co % 8 == 0 &&
-                     OW <= 32 && (OW & (OW - 1)) == 0 &&
-                     (OH * OW) % 128 == 0) ? 128 : BK;
-  int mps = cdiv(cdiv(M, splits), chunk) * chunk;
+  const long tiles = (long)cdiv(Co, (Co >= 64) ? 64 : ((Co >= 32) ? 32 : 16))
+                     * cdiv(K, 128);
+  int splits = (int)std::min<long>(std::max<long>(1, 1024 / tiles),
+                                   std::max<long>(1, M / (16 * 64)));
+  int mps = cdiv(cdiv(M, splits), 64) * 64;
   return cdiv(M, mps);
 }
 
@@ -541,14 +576,19 @@ extern "C" void dlb_conv_wrw(const void* x, const void* dy, float* dw, int N,
                              hipStream_t stream) {
   WrwParams p{(const bf16*)x, (const bf16*)dy, dw, N, IH, IW, Ci, OH, OW,
               Co, R, S, stride, pad, N * OH * OW, R * S * Ci, 0};
-  const int BCO = (Co >= 32) ? 32 : 16;
-  p.m_per_split = cdiv(cdiv(p.M, splits), BK) * BK;
+  p.m_per_split = cdiv(cdiv(p.M, splits), 64) * 64;
   splits = cdiv(p.M, p.m_per_split);
-  dim3 grid(cdiv(Co, BCO), cdiv(p.K, 256), splits);
-  if (BCO == 32)
-    hipLaunchKernelGGL((conv_wrw_kernel<32, 256>), grid, dim3(CONV_BLOCK), 0,
+  if (Co >= 64) {
+    dim3 grid(cdiv(Co, 64), cdiv(p.K, 128), splits);
+    hipLaunchKernelGGL((conv_wrw_kernel<64, 128>), grid, dim3(CONV_BLOCK), 0,
                        stream, p);
-  else
-    hipLaunchKernelGGL((conv_wrw_kernel<16, 256>), grid, dim3(CONV_BLOCK), 0,
+  } else if (Co >= 32) {
+    dim3 grid(cdiv(Co, 32), cdiv(p.K, 128), splits);
+    hipLaunchKernelGGL((conv_wrw_kernel<32, 128>), grid, dim3(CONV_BLOCK), 0,
                        stream, p);
+  } else {
+    dim3 grid(cdiv(Co, 16), cdiv(p.K, 128), splits);
+    hipLaunchKernelGGL((conv_wrw_kernel<16, 128>), grid, dim3(CONV_BLOCK), 0,
+                       stream, p);
+  }
 }
