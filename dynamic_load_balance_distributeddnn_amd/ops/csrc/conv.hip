@@ -29,6 +29,14 @@ typedef __attribute__((ext_vector_type(4))) float f32x4;
 #define CONV_BLOCK 256
 #define BK 32
 
+struct ConvGeom {
+  FastDiv fd_pix;   // divide m by OH*OW (or IH*IW for bwd)
+  FastDiv fd_w;     // divide rem by OW (or IW)
+  FastDiv fd_c;     // divide k by Ci (fwd/wrw) or Co (bwd)
+  FastDiv fd_s;     // divide rs by S
+  int flat;         // 1x1 / stride 1 / pad 0: im2col is the identity
+};
+
 struct ConvParams {
   const bf16* x;   // [N, IH, IW, Ci]
   const bf16* w;   // [Co, R, S, Ci]  (channels_last natural layout)
@@ -36,22 +44,29 @@ struct ConvParams {
   const float* bias;  // [Co] or nullptr
   int N, IH, IW, Ci, OH, OW, Co, R, S, stride, pad;
   int M, K;        // M = N*OH*OW, K = R*S*Ci
+  ConvGeom g;
 };
 
 // im2col 8-element load for GEMM row m (output pixel), k-chunk k..k+7.
+// Address math is branch-light: magic-multiply divisions (ConvGeom) and
+// a flat fast path for 1x1/s1/p0 where im2col is the identity.
 __device__ inline bf16x8_t im2col_load8(const bf16* __restrict__ x, int m,
                                         int k, int IH, int IW, int Ci, int OH,
                                         int OW, int S, int stride, int pad,
-                                        int K, bool vec) {
+                                        int K, bool vec, const ConvGeom& g) {
   bf16x8_t v = {0, 0, 0, 0, 0, 0, 0, 0};
-  const int n = m / (OH * OW);
-  const int rem = m % (OH * OW);
-  const int oh = rem / OW, ow = rem % OW;
+  if (g.flat)
+    return *reinterpret_cast<const bf16x8_t*>(x + (long)m * Ci + k);
+  unsigned n, rem, oh, ow;
+  g.fd_pix.divmod(m, n, rem);
+  g.fd_w.divmod(rem, oh, ow);
   if (vec) {
-    const int rs = k / Ci, ci = k % Ci;
-    const int r = rs / S, s = rs % S;
-    const int ih = oh * stride - pad + r;
-    const int iw = ow * stride - pad + s;
+    unsigned rs, ci;
+    g.fd_c.divmod(k, rs, ci);
+    unsigned r, sx;
+    g.fd_s.divmod(rs, r, sx);
+    const int ih = (int)oh * stride - pad + r;
+    const int iw = (int)ow * stride - pad + sx;
     if (ih >= 0 && ih < IH && iw >= 0 && iw < IW)
       v = *reinterpret_cast<const bf16x8_t*>(
           x + (((long)n * IH + ih) * IW + iw) * Ci + ci);
@@ -61,10 +76,11 @@ __device__ inline bf16x8_t im2col_load8(const bf16* __restrict__ x, int m,
     for (int j = 0; j < 8; ++j) {
       const int kk = k + j;
       if (kk < K) {
-        const int rs = kk / Ci, ci = kk % Ci;
-        const int r = rs / S, s = rs % S;
-        const int ih = oh * stride - pad + r;
-        const int iw = ow * stride - pad + s;
+        unsigned rs, ci, r, sx;
+        g.fd_c.divmod(kk, rs, ci);
+        g.fd_s.divmod(rs, r, sx);
+        const int ih = (int)oh * stride - pad + r;
+        const int iw = (int)ow * stride - pad + sx;
         if (ih >= 0 && ih < IH && iw >= 0 && iw < IW)
           vv[j] = x[(((long)n * IH + ih) * IW + iw) * Ci + ci];
       }
@@ -113,7 +129,7 @@ conv_fwd_kernel(const ConvParams p) {
       bf16x8_t v = {0, 0, 0, 0, 0, 0, 0, 0};
       if (m < p.M && k < p.K)
         v = im2col_load8(p.x, m, k, p.IH, p.IW, p.Ci, p.OH, p.OW, p.S,
-                         p.stride, p.pad, p.K, vec);
+                         p.stride, p.pad, p.K, vec, p.g);
       *reinterpret_cast<bf16x8_t*>(&a_lds[row * LDA + k8]) = v;
     }
     for (int c = t; c < BCH; c += CONV_BLOCK) {
@@ -185,19 +201,23 @@ struct ConvBwdParams {
   bf16* dx;        // [N, IH, IW, Ci]
   int N, IH, IW, Ci, OH, OW, Co, R, S, stride, pad;
   int M, K;  // M = N*IH*IW, K = R*S*Co
+  ConvGeom g;  // fd_pix: /(IH*IW), fd_w: /IW, fd_c: /Co, fd_s: /S
 };
 
 __device__ inline bf16x8_t dcol_load8(const ConvBwdParams& p, int m, int k,
                                       bool vec) {
   bf16x8_t v = {0, 0, 0, 0, 0, 0, 0, 0};
-  const int n = m / (p.IH * p.IW);
-  const int rem = m % (p.IH * p.IW);
-  const int ih = rem / p.IW, iw = rem % p.IW;
+  if (p.g.flat)
+    return *reinterpret_cast<const bf16x8_t*>(p.dy + (long)m * p.Co + k);
+  unsigned n, rem, ih, iw;
+  p.g.fd_pix.divmod(m, n, rem);
+  p.g.fd_w.divmod(rem, ih, iw);
   if (vec) {
-    const int rs = k / p.Co, co = k % p.Co;
-    const int r = rs / p.S, s = rs % p.S;
-    const int ohn = ih + p.pad - r;
-    const int own = iw + p.pad - s;
+    unsigned rs, co, r, s;
+    p.g.fd_c.divmod(k, rs, co);
+    p.g.fd_s.divmod(rs, r, s);
+    const int ohn = (int)ih + p.pad - (int)r;
+    const int own = (int)iw + p.pad - (int)s;
     if (ohn >= 0 && own >= 0 && ohn % p.stride == 0 && own % p.stride == 0) {
       const int oh = ohn / p.stride, ow = own / p.stride;
       if (oh < p.OH && ow < p.OW)
@@ -210,10 +230,11 @@ __device__ inline bf16x8_t dcol_load8(const ConvBwdParams& p, int m, int k,
     for (int j = 0; j < 8; ++j) {
       const int kk = k + j;
       if (kk < p.K) {
-        const int rs = kk / p.Co, co = kk % p.Co;
-        const int r = rs / p.S, s = rs % p.S;
-        const int ohn = ih + p.pad - r;
-        const int own = iw + p.pad - s;
+        unsigned rs, co, r, s;
+        p.g.fd_c.divmod(kk, rs, co);
+        p.g.fd_s.divmod(rs, r, s);
+        const int ohn = (int)ih + p.pad - (int)r;
+        const int own = (int)iw + p.pad - (int)s;
         if (ohn >= 0 && own >= 0 && ohn % p.stride == 0 &&
             own % p.stride == 0) {
           const int oh = ohn / p.stride, ow = own / p.stride;
@@ -338,6 +359,7 @@ struct WrwParams {
   int N, IH, IW, Ci, OH, OW, Co, R, S, stride, pad;
   int M, K;
   int m_per_split;
+  ConvGeom g;
 };
 
 template <int BCO, int BKN>
@@ -407,7 +429,7 @@ conv_wrw_kernel(const WrwParams p) {
         const int k = k0 + k8;
         if (m < mend && k < p.K)
           v = im2col_load8(p.x, m, k, p.IH, p.IW, p.Ci, p.OH, p.OW, p.S,
-                           p.stride, p.pad, p.K, xvec);
+                           p.stride, p.pad, p.K, xvec, p.g);
       }
       xreg[u] = v;
     }
@@ -516,7 +538,12 @@ extern "C" void dlb_conv_fwd(const void* x, const void* w, void* y,
       dlb_conv3x3_fwd_halo(x, w, y, bias, N, IH, IW, Ci, Co, stream))
     return;
   ConvParams p{(const bf16*)x, (const bf16*)w, (bf16*)y, bias, N, IH, IW, Ci,
-               OH, OW, Co, R, S, stride, pad, N * OH * OW, R * S * Ci};
+               OH, OW, Co, R, S, stride, pad, N * OH * OW, R * S * Ci, {}};
+  p.g.fd_pix.init(OH * OW);
+  p.g.fd_w.init(OW);
+  p.g.fd_c.init(Ci);
+  p.g.fd_s.init(S);
+  p.g.flat = (R == 1 && S == 1 && stride == 1 && pad == 0) ? 1 : 0;
   int bm, bn;
   pick_tile(Co, bm, bn);
   dim3 grid(cdiv(p.M, bm), cdiv(Co, bn));
@@ -539,7 +566,12 @@ extern "C" void dlb_conv_bwd_data(const void* dy, const void* wt, void* dx,
                                   int OW, int Co, int R, int S, int stride,
                                   int pad, hipStream_t stream) {
   ConvBwdParams p{(const bf16*)dy, (const bf16*)wt, (bf16*)dx, N, IH, IW, Ci,
-                  OH, OW, Co, R, S, stride, pad, N * IH * IW, R * S * Co};
+                  OH, OW, Co, R, S, stride, pad, N * IH * IW, R * S * Co, {}};
+  p.g.fd_pix.init(IH * IW);
+  p.g.fd_w.init(IW);
+  p.g.fd_c.init(Co);
+  p.g.fd_s.init(S);
+  p.g.flat = (R == 1 && S == 1 && stride == 1 && pad == 0) ? 1 : 0;
   int bm, bn;
   pick_tile(Ci, bm, bn);
   dim3 grid(cdiv(p.M, bm), cdiv(Ci, bn));
@@ -575,7 +607,12 @@ extern "C" void dlb_conv_wrw(const void* x, const void* dy, float* dw, int N,
                              int R, int S, int stride, int pad, int splits,
                              hipStream_t stream) {
   WrwParams p{(const bf16*)x, (const bf16*)dy, dw, N, IH, IW, Ci, OH, OW,
-              Co, R, S, stride, pad, N * OH * OW, R * S * Ci, 0};
+              Co, R, S, stride, pad, N * OH * OW, R * S * Ci, 0, {}};
+  p.g.fd_pix.init(OH * OW);
+  p.g.fd_w.init(OW);
+  p.g.fd_c.init(Ci);
+  p.g.fd_s.init(S);
+  p.g.flat = (R == 1 && S == 1 && stride == 1 && pad == 0) ? 1 : 0;
   p.m_per_split = cdiv(cdiv(p.M, splits), 64) * 64;
   splits = cdiv(p.M, p.m_per_split);
   if (Co >= 64) {
