@@ -370,13 +370,16 @@ conv_wrw_kernel(const WrwParams p) {
   constexpr int FA = BCO / 16;
   constexpr int FB = WTN / 16;
   constexpr int LMD = BM + 8;         // 16B-aligned [row][m] strides
-  constexpr int DCH = BM * (BCO / 8); // dy chunks per stage
-  constexpr int XCH = BM * (BKN / 8); // x chunks per stage
+  constexpr int DCH = BM * (BCO / 8);
+  constexpr int XCH = BM * (BKN / 8);
   constexpr int DPT = (DCH + CONV_BLOCK - 1) / CONV_BLOCK;
   constexpr int XPT = (XCH + CONV_BLOCK - 1) / CONV_BLOCK;
 
-  __shared__ bf16 dy_t[BCO * LMD];    // [co][m]
-  __shared__ bf16 x_t[BKN * LMD];     // [k][m]
+  // double-buffered LDS: write stage i+1 while MFMAs consume stage i;
+  // ONE barrier per iteration (the un-pipelined form measured idle on
+  // every PMC: latency-bound, not memory- or compute-bound).
+  __shared__ bf16 dy_t[2][BCO * LMD];
+  __shared__ bf16 x_t[2][BKN * LMD];
 
   const int t = threadIdx.x;
   const int lane = t & 63;
@@ -435,7 +438,7 @@ conv_wrw_kernel(const WrwParams p) {
     }
   };
 
-  auto write_chunk = [&]() {
+  auto write_chunk = [&](int buf) {
 #pragma unroll
     for (int u = 0; u < DPT; ++u) {
       const int c = t + u * CONV_BLOCK;
@@ -444,7 +447,7 @@ conv_wrw_kernel(const WrwParams p) {
         const int c8 = (c % (BCO / 8)) * 8;
         const bf16* vv = reinterpret_cast<const bf16*>(&dreg[u]);
 #pragma unroll
-        for (int j = 0; j < 8; ++j) dy_t[(c8 + j) * LMD + mm] = vv[j];
+        for (int j = 0; j < 8; ++j) dy_t[buf][(c8 + j) * LMD + mm] = vv[j];
       }
     }
 #pragma unroll
@@ -455,21 +458,20 @@ conv_wrw_kernel(const WrwParams p) {
         const int k8 = (c % (BKN / 8)) * 8;
         const bf16* vv = reinterpret_cast<const bf16*>(&xreg[u]);
 #pragma unroll
-        for (int j = 0; j < 8; ++j) x_t[(k8 + j) * LMD + mm] = vv[j];
+        for (int j = 0; j < 8; ++j) x_t[buf][(k8 + j) * LMD + mm] = vv[j];
       }
     }
   };
 
-  // prologue: load + write chunk 0
   load_chunk(mstart);
-  write_chunk();
+  write_chunk(0);
   __syncthreads();
 
+  int buf = 0;
   for (int mt = mstart; mt < mend; mt += BM) {
-    // issue next chunk's global loads (latency hides under the MFMAs)
-    if (mt + BM < mend) load_chunk(mt + BM);
+    const bool more = mt + BM < mend;
+    if (more) load_chunk(mt + BM);
 
-    // consume current LDS image: BM/32 sub-steps of k-depth 32
 #pragma unroll
     for (int sub = 0; sub < BM / 32; ++sub) {
       bf16x8_t afrag[FA], bfrag[FB];
@@ -477,13 +479,13 @@ conv_wrw_kernel(const WrwParams p) {
       for (int i = 0; i < FA; ++i) {
         const int co = i * 16 + (lane & 15);
         afrag[i] = *reinterpret_cast<const bf16x8_t*>(
-            &dy_t[co * LMD + sub * 32 + (lane >> 4) * 8]);
+            &dy_t[buf][co * LMD + sub * 32 + (lane >> 4) * 8]);
       }
 #pragma unroll
       for (int j = 0; j < FB; ++j) {
         const int kk = wave * WTN + j * 16 + (lane & 15);
         bfrag[j] = *reinterpret_cast<const bf16x8_t*>(
-            &x_t[kk * LMD + sub * 32 + (lane >> 4) * 8]);
+            &x_t[buf][kk * LMD + sub * 32 + (lane >> 4) * 8]);
       }
 #pragma unroll
       for (int i = 0; i < FA; ++i)
@@ -492,11 +494,9 @@ conv_wrw_kernel(const WrwParams p) {
           acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
     }
+    if (more) write_chunk(buf ^ 1);  // other buffer: no barrier needed first
     __syncthreads();
-    if (mt + BM < mend) {
-      write_chunk();
-      __syncthreads();
-    }
+    buf ^= 1;
   }
 
   float* slab = p.dw + (long)blockIdx.z * p.Co * p.K;
@@ -515,87 +515,12 @@ conv_wrw_kernel(const WrwParams p) {
   }
 }
 
-// ---------------------------------------------------------------- launch
-static void pick_tile(int Co, int& bm, int& bn) {
-  if (Co >= 128) { bm = 128; bn = 128; }
-  else if (Co >= 64) { bm = 128; bn = 64; }
-  else if (Co >= 32) { bm = 128; bn = 32; }
-  else { bm = 256; bn = 16; }
-}
-
-extern "C" bool dlb_conv3x3_fwd_halo(const void* x, const void* w, void* y,
-                                     const float* bias, int N, int H, int W,
-                                     int Ci, int Co, hipStream_t stream);
-extern "C" bool dlb_conv3x3_wrw_halo(const void* x, const void* dy, float* dw,
-                                     int N, int H, int W, int Ci, int Co,
-                                     int splits, hipStream_t stream);
-
-extern "C" void dlb_conv_fwd(const void* x, const void* w, void* y,
-                             const float* bias, int N, int IH, int IW, int Ci,
-                             int OH, int OW, int Co, int R, int S, int stride,
-                             int pad, hipStream_t stream) {
-  if (R == 3 && S == 3 && stride == 1 && pad == 1 &&
-      dlb_conv3x3_fwd_halo(x, w, y, bias, N, IH, IW, Ci, Co, stream))
-    return;
-  ConvParams p{(const bf16*)x, (const bf16*)w, (bf16*)y, bias, N, IH, IW, Ci,
-               OH, OW, Co, R, S, stride, pad, N * OH * OW, R * S * Ci, {}};
-  p.g.fd_pix.init(OH * OW);
-  p.g.fd_w.init(OW);
-  p.g.fd_c.init(Ci);
-  p.g.fd_s.init(S);
-  p.g.flat = (R == 1 && S == 1 && stride == 1 && pad == 0) ? 1 : 0;
-  int bm, bn;
-  pick_tile(Co, bm, bn);
-  dim3 grid(cdiv(p.M, bm), cdiv(Co, bn));
-  if (bm == 128 && bn == 128)
-    hipLaunchKernelGGL((conv_fwd_kernel<128, 128, 2, 2>), grid,
-                       dim3(CONV_BLOCK), 0, stream, p);
-  else if (bm == 128 && bn == 64)
-    hipLaunchKernelGGL((conv_fwd_kernel<128, 64, 2, 2>), grid,
-                       dim3(CONV_BLOCK), 0, stream, p);
-  else if (bm == 128 && bn == 32)
-    hipLaunchKernelGGL((conv_fwd_kernel<128, 32, 4, 1>), grid,
-                       dim3(CONV_BLOCK), 0, stream, p);
-  else
-    hipLaunchKernelGGL((conv_fwd_kernel<256, 16, 4, 1>), grid,
-                       dim3(CONV_BLOCK), 0, stream, p);
-}
-
-extern "C" void dlb_conv_bwd_data(const void* dy, const void* wt, void* dx,
-                                  int N, int IH, int IW, int Ci, int OH,
-                                  int OW, int Co, int R, int S, int stride,
-                                  int pad, hipStream_t stream) {
-  ConvBwdParams p{(const bf16*)dy, (const bf16*)wt, (bf16*)dx, N, IH, IW, Ci,
-                  OH, OW, Co, R, S, stride, pad, N * IH * IW, R * S * Co, {}};
-  p.g.fd_pix.init(IH * IW);
-  p.g.fd_w.init(IW);
-  p.g.fd_c.init(Co);
-  p.g.fd_s.init(S);
-  p.g.flat = (R == 1 && S == 1 && stride == 1 && pad == 0) ? 1 : 0;
-  int bm, bn;
-  pick_tile(Ci, bm, bn);
-  dim3 grid(cdiv(p.M, bm), cdiv(Ci, bn));
-  if (bm == 128 && bn == 128)
-    hipLaunchKernelGGL((conv_bwd_data_kernel<128, 128, 2, 2>), grid,
-                       dim3(CONV_BLOCK), 0, stream, p);
-  else if (bm == 128 && bn == 64)
-    hipLaunchKernelGGL((conv_bwd_data_kernel<128, 64, 2, 2>), grid,
-                       dim3(CONV_BLOCK), 0, stream, p);
-  else if (bm == 128 && bn == 32)
-    hipLaunchKernelGGL((conv_bwd_data_kernel<128, 32, 4, 1>), grid,
-                       dim3(CONV_BLOCK), 0, stream, p);
-  else
-    hipLaunchKernelGGL((conv_bwd_data_kernel<256, 16, 4, 1>), grid,
-                       dim3(CONV_BLOCK), 0, stream, p);
-}
-
 // split-count query: how many per-split slabs the wrw launch will write.
 extern "C" int dlb_conv_wrw_nsplits(int N, int OH, int OW, int Ci, int Co,
                                     int R, int S) {
   const int M = N * OH * OW;
   const int K = R * S * Ci;
-  const long tiles = (long)cdiv(Co, (Co >= 64) ? 64 : ((Co >= 32) ? 32 : 16))
-                     * cdiv(K, 128);
+  const long tiles = (long)cdiv(Co, (Co >= 32) ? 32 : 16) * cdiv(K, 128);
   int splits = (int)std::min<long>(std::max<long>(1, 1024 / tiles),
                                    std::max<long>(1, M / (16 * 64)));
   int mps = cdiv(cdiv(M, splits), 64) * 64;
@@ -615,11 +540,7 @@ extern "C" void dlb_conv_wrw(const void* x, const void* dy, float* dw, int N,
   p.g.flat = (R == 1 && S == 1 && stride == 1 && pad == 0) ? 1 : 0;
   p.m_per_split = cdiv(cdiv(p.M, splits), 64) * 64;
   splits = cdiv(p.M, p.m_per_split);
-  if (Co >= 64) {
-    dim3 grid(cdiv(Co, 64), cdiv(p.K, 128), splits);
-    hipLaunchKernelGGL((conv_wrw_kernel<64, 128>), grid, dim3(CONV_BLOCK), 0,
-                       stream, p);
-  } else if (Co >= 32) {
+  if (Co >= 32) {
     dim3 grid(cdiv(Co, 32), cdiv(p.K, 128), splits);
     hipLaunchKernelGGL((conv_wrw_kernel<32, 128>), grid, dim3(CONV_BLOCK), 0,
                        stream, p);
