@@ -515,6 +515,77 @@ conv_wrw_kernel(const WrwParams p) {
   }
 }
 
+// ---------------------------------------------------------------- launch
+static void pick_tile(int Co, int& bm, int& bn) {
+  if (Co >= 128) { bm = 128; bn = 128; }
+  else if (Co >= 64) { bm = 128; bn = 64; }
+  else if (Co >= 32) { bm = 128; bn = 32; }
+  else { bm = 256; bn = 16; }
+}
+
+extern "C" bool dlb_conv3x3_fwd_halo(const void* x, const void* w, void* y,
+                                     const float* bias, int N, int H, int W,
+                                     int Ci, int Co, hipStream_t stream);
+
+extern "C" void dlb_conv_fwd(const void* x, const void* w, void* y,
+                             const float* bias, int N, int IH, int IW, int Ci,
+                             int OH, int OW, int Co, int R, int S, int stride,
+                             int pad, hipStream_t stream) {
+  if (R == 3 && S == 3 && stride == 1 && pad == 1 &&
+      dlb_conv3x3_fwd_halo(x, w, y, bias, N, IH, IW, Ci, Co, stream))
+    return;
+  ConvParams p{(const bf16*)x, (const bf16*)w, (bf16*)y, bias, N, IH, IW, Ci,
+               OH, OW, Co, R, S, stride, pad, N * OH * OW, R * S * Ci, {}};
+  p.g.fd_pix.init(OH * OW);
+  p.g.fd_w.init(OW);
+  p.g.fd_c.init(Ci);
+  p.g.fd_s.init(S);
+  p.g.flat = (R == 1 && S == 1 && stride == 1 && pad == 0) ? 1 : 0;
+  int bm, bn;
+  pick_tile(Co, bm, bn);
+  dim3 grid(cdiv(p.M, bm), cdiv(Co, bn));
+  if (bm == 128 && bn == 128)
+    hipLaunchKernelGGL((conv_fwd_kernel<128, 128, 2, 2>), grid,
+                       dim3(CONV_BLOCK), 0, stream, p);
+  else if (bm == 128 && bn == 64)
+    hipLaunchKernelGGL((conv_fwd_kernel<128, 64, 2, 2>), grid,
+                       dim3(CONV_BLOCK), 0, stream, p);
+  else if (bm == 128 && bn == 32)
+    hipLaunchKernelGGL((conv_fwd_kernel<128, 32, 4, 1>), grid,
+                       dim3(CONV_BLOCK), 0, stream, p);
+  else
+    hipLaunchKernelGGL((conv_fwd_kernel<256, 16, 4, 1>), grid,
+                       dim3(CONV_BLOCK), 0, stream, p);
+}
+
+extern "C" void dlb_conv_bwd_data(const void* dy, const void* wt, void* dx,
+                                  int N, int IH, int IW, int Ci, int OH,
+                                  int OW, int Co, int R, int S, int stride,
+                                  int pad, hipStream_t stream) {
+  ConvBwdParams p{(const bf16*)dy, (const bf16*)wt, (bf16*)dx, N, IH, IW, Ci,
+                  OH, OW, Co, R, S, stride, pad, N * IH * IW, R * S * Co, {}};
+  p.g.fd_pix.init(IH * IW);
+  p.g.fd_w.init(IW);
+  p.g.fd_c.init(Co);
+  p.g.fd_s.init(S);
+  p.g.flat = (R == 1 && S == 1 && stride == 1 && pad == 0) ? 1 : 0;
+  int bm, bn;
+  pick_tile(Ci, bm, bn);
+  dim3 grid(cdiv(p.M, bm), cdiv(Ci, bn));
+  if (bm == 128 && bn == 128)
+    hipLaunchKernelGGL((conv_bwd_data_kernel<128, 128, 2, 2>), grid,
+                       dim3(CONV_BLOCK), 0, stream, p);
+  else if (bm == 128 && bn == 64)
+    hipLaunchKernelGGL((conv_bwd_data_kernel<128, 64, 2, 2>), grid,
+                       dim3(CONV_BLOCK), 0, stream, p);
+  else if (bm == 128 && bn == 32)
+    hipLaunchKernelGGL((conv_bwd_data_kernel<128, 32, 4, 1>), grid,
+                       dim3(CONV_BLOCK), 0, stream, p);
+  else
+    hipLaunchKernelGGL((conv_bwd_data_kernel<256, 16, 4, 1>), grid,
+                       dim3(CONV_BLOCK), 0, stream, p);
+}
+
 // split-count query: how many per-split slabs the wrw launch will write.
 extern "C" int dlb_conv_wrw_nsplits(int N, int OH, int OW, int Ci, int Co,
                                     int R, int S) {
