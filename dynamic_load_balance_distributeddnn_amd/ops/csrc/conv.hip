@@ -48,43 +48,67 @@ struct ConvParams {
 };
 
 // im2col 8-element load for GEMM row m (output pixel), k-chunk k..k+7.
-// Address math is branch-light: magic-multiply divisions (ConvGeom) and
-// a flat fast path for 1x1/s1/p0 where im2col is the identity.
+// BRANCHLESS: the load is always issued at a clamped in-bounds address
+// and invalid lanes are zeroed with VALU selects afterwards — a
+// conditional load compiles to an exec-branch with a dependent
+// vmcnt(0) per element (CDNA guide §5 trap 4c), which serialized the
+// whole staging pipeline in earlier versions.
+__device__ inline bf16x8_t zero8() {
+  bf16x8_t z = {0, 0, 0, 0, 0, 0, 0, 0};
+  return z;
+}
+
+__device__ inline bf16x8_t mask8(bf16x8_t v, bool ok) {
+  union { bf16x8_t h; int4 i; } u;
+  u.h = v;
+  u.i.x = ok ? u.i.x : 0;
+  u.i.y = ok ? u.i.y : 0;
+  u.i.z = ok ? u.i.z : 0;
+  u.i.w = ok ? u.i.w : 0;
+  return u.h;
+}
+
 __device__ inline bf16x8_t im2col_load8(const bf16* __restrict__ x, int m,
                                         int k, int IH, int IW, int Ci, int OH,
                                         int OW, int S, int stride, int pad,
-                                        int K, bool vec, const ConvGeom& g) {
-  bf16x8_t v = {0, 0, 0, 0, 0, 0, 0, 0};
-  if (g.flat)
-    return *reinterpret_cast<const bf16x8_t*>(x + (long)m * Ci + k);
+                                        int K, int Mmax, bool vec,
+                                        const ConvGeom& g) {
+  bool ok = (m < Mmax) & (k < K);
+  m = ok ? m : 0;
+  k = ok ? k : 0;
+  if (g.flat) {
+    bf16x8_t v = *reinterpret_cast<const bf16x8_t*>(x + (long)m * Ci + k);
+    return mask8(v, ok);
+  }
   unsigned n, rem, oh, ow;
   g.fd_pix.divmod(m, n, rem);
   g.fd_w.divmod(rem, oh, ow);
   if (vec) {
-    unsigned rs, ci;
+    unsigned rs, ci, r, sx;
     g.fd_c.divmod(k, rs, ci);
-    unsigned r, sx;
     g.fd_s.divmod(rs, r, sx);
-    const int ih = (int)oh * stride - pad + r;
-    const int iw = (int)ow * stride - pad + sx;
-    if (ih >= 0 && ih < IH && iw >= 0 && iw < IW)
-      v = *reinterpret_cast<const bf16x8_t*>(
-          x + (((long)n * IH + ih) * IW + iw) * Ci + ci);
-  } else {
-    bf16* vv = reinterpret_cast<bf16*>(&v);
+    const int ih = (int)oh * stride - pad + (int)r;
+    const int iw = (int)ow * stride - pad + (int)sx;
+    ok &= (ih >= 0) & (ih < IH) & (iw >= 0) & (iw < IW);
+    const long off = ok ? (((long)n * IH + ih) * IW + iw) * Ci + ci : 0;
+    bf16x8_t v = *reinterpret_cast<const bf16x8_t*>(x + off);
+    return mask8(v, ok);
+  }
+  bf16x8_t v;
+  bf16* vv = reinterpret_cast<bf16*>(&v);
 #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      const int kk = k + j;
-      if (kk < K) {
-        unsigned rs, ci, r, sx;
-        g.fd_c.divmod(kk, rs, ci);
-        g.fd_s.divmod(rs, r, sx);
-        const int ih = (int)oh * stride - pad + r;
-        const int iw = (int)ow * stride - pad + sx;
-        if (ih >= 0 && ih < IH && iw >= 0 && iw < IW)
-          vv[j] = x[(((long)n * IH + ih) * IW + iw) * Ci + ci];
-      }
-    }
+  for (int j = 0; j < 8; ++j) {
+    const int kk = k + j;
+    unsigned rs, ci, r, sx;
+    g.fd_c.divmod(kk < K ? kk : 0, rs, ci);
+    g.fd_s.divmod(rs, r, sx);
+    const int ih = (int)oh * stride - pad + (int)r;
+    const int iw = (int)ow * stride - pad + (int)sx;
+    const bool e = ok & (kk < K) & (ih >= 0) & (ih < IH) & (iw >= 0) &
+                   (iw < IW);
+    const long off = e ? (((long)n * IH + ih) * IW + iw) * Ci + ci : 0;
+    bf16 t = x[off];
+    vv[j] = e ? t : (bf16)__float2bfloat16(0.f);
   }
   return v;
 }
@@ -126,10 +150,8 @@ conv_fwd_kernel(const ConvParams p) {
       const int k8 = (c % (BK / 8)) * 8;
       const int m = m0 + row;
       const int k = kt + k8;
-      bf16x8_t v = {0, 0, 0, 0, 0, 0, 0, 0};
-      if (m < p.M && k < p.K)
-        v = im2col_load8(p.x, m, k, p.IH, p.IW, p.Ci, p.OH, p.OW, p.S,
-                         p.stride, p.pad, p.K, vec, p.g);
+      bf16x8_t v = im2col_load8(p.x, m, k, p.IH, p.IW, p.Ci, p.OH, p.OW,
+                                p.S, p.stride, p.pad, p.K, p.M, vec, p.g);
       *reinterpret_cast<bf16x8_t*>(&a_lds[row * LDA + k8]) = v;
     }
     for (int c = t; c < BCH; c += CONV_BLOCK) {
@@ -137,10 +159,12 @@ conv_fwd_kernel(const ConvParams p) {
       const int k8 = (c % (BK / 8)) * 8;
       const int n = n0 + nrow;
       const int k = kt + k8;
-      bf16x8_t v = {0, 0, 0, 0, 0, 0, 0, 0};
-      if (n < p.Co && k + 7 < p.K)
-        v = *reinterpret_cast<const bf16x8_t*>(p.w + (long)n * p.K + k);
-      else if (n < p.Co) {
+      // K % 8 == 0 whenever Ci % 8 == 0; clamp+mask keeps it branchless,
+      // the generic tail only triggers for the tiny non-vec stems.
+      const bool ok = (n < p.Co) & (k + 7 < p.K);
+      const long off = ok ? (long)n * p.K + k : 0;
+      bf16x8_t v = mask8(*reinterpret_cast<const bf16x8_t*>(p.w + off), ok);
+      if (!ok && n < p.Co) {
         bf16* vv = reinterpret_cast<bf16*>(&v);
         for (int j = 0; j < 8 && k + j < p.K; ++j)
           vv[j] = p.w[(long)n * p.K + k + j];
@@ -206,43 +230,46 @@ struct ConvBwdParams {
 
 __device__ inline bf16x8_t dcol_load8(const ConvBwdParams& p, int m, int k,
                                       bool vec) {
-  bf16x8_t v = {0, 0, 0, 0, 0, 0, 0, 0};
-  if (p.g.flat)
-    return *reinterpret_cast<const bf16x8_t*>(p.dy + (long)m * p.Co + k);
+  bool ok = (m < p.M) & (k < p.K);
+  m = ok ? m : 0;
+  k = ok ? k : 0;
+  if (p.g.flat) {
+    bf16x8_t v = *reinterpret_cast<const bf16x8_t*>(p.dy + (long)m * p.Co + k);
+    return mask8(v, ok);
+  }
   unsigned n, rem, ih, iw;
   p.g.fd_pix.divmod(m, n, rem);
   p.g.fd_w.divmod(rem, ih, iw);
   if (vec) {
-    unsigned rs, co, r, s;
+    unsigned rs, co, r, sx;
     p.g.fd_c.divmod(k, rs, co);
-    p.g.fd_s.divmod(rs, r, s);
+    p.g.fd_s.divmod(rs, r, sx);
     const int ohn = (int)ih + p.pad - (int)r;
-    const int own = (int)iw + p.pad - (int)s;
-    if (ohn >= 0 && own >= 0 && ohn % p.stride == 0 && own % p.stride == 0) {
-      const int oh = ohn / p.stride, ow = own / p.stride;
-      if (oh < p.OH && ow < p.OW)
-        v = *reinterpret_cast<const bf16x8_t*>(
-            p.dy + (((long)n * p.OH + oh) * p.OW + ow) * p.Co + co);
-    }
-  } else {
-    bf16* vv = reinterpret_cast<bf16*>(&v);
+    const int own = (int)iw + p.pad - (int)sx;
+    const int oh = ohn / p.stride, ow = own / p.stride;
+    ok &= (ohn >= 0) & (own >= 0) & (oh * p.stride == ohn) &
+          (ow * p.stride == own) & (oh < p.OH) & (ow < p.OW);
+    const long off = ok ? (((long)n * p.OH + oh) * p.OW + ow) * p.Co + co : 0;
+    bf16x8_t v = *reinterpret_cast<const bf16x8_t*>(p.dy + off);
+    return mask8(v, ok);
+  }
+  bf16x8_t v;
+  bf16* vv = reinterpret_cast<bf16*>(&v);
 #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      const int kk = k + j;
-      if (kk < p.K) {
-        unsigned rs, co, r, s;
-        p.g.fd_c.divmod(kk, rs, co);
-        p.g.fd_s.divmod(rs, r, s);
-        const int ohn = (int)ih + p.pad - (int)r;
-        const int own = (int)iw + p.pad - (int)s;
-        if (ohn >= 0 && own >= 0 && ohn % p.stride == 0 &&
-            own % p.stride == 0) {
-          const int oh = ohn / p.stride, ow = own / p.stride;
-          if (oh < p.OH && ow < p.OW)
-            vv[j] = p.dy[(((long)n * p.OH + oh) * p.OW + ow) * p.Co + co];
-        }
-      }
-    }
+  for (int j = 0; j < 8; ++j) {
+    const int kk = k + j;
+    unsigned rs, co, r, sx;
+    p.g.fd_c.divmod(kk < p.K ? kk : 0, rs, co);
+    p.g.fd_s.divmod(rs, r, sx);
+    const int ohn = (int)ih + p.pad - (int)r;
+    const int own = (int)iw + p.pad - (int)sx;
+    const int oh = ohn / p.stride, ow = own / p.stride;
+    const bool e = ok & (kk < p.K) & (ohn >= 0) & (own >= 0) &
+                   (oh * p.stride == ohn) & (ow * p.stride == own) &
+                   (oh < p.OH) & (ow < p.OW);
+    const long off = e ? (((long)n * p.OH + oh) * p.OW + ow) * p.Co + co : 0;
+    bf16 t = p.dy[off];
+    vv[j] = e ? t : (bf16)__float2bfloat16(0.f);
   }
   return v;
 }
@@ -283,8 +310,7 @@ conv_bwd_data_kernel(const ConvBwdParams p) {
       const int k8 = (c % (BK / 8)) * 8;
       const int m = m0 + row;
       const int k = kt + k8;
-      bf16x8_t v = {0, 0, 0, 0, 0, 0, 0, 0};
-      if (m < p.M && k < p.K) v = dcol_load8(p, m, k, vec);
+      bf16x8_t v = dcol_load8(p, m, k, vec);
       *reinterpret_cast<bf16x8_t*>(&a_lds[row * LDA + k8]) = v;
     }
     for (int c = t; c < BCH; c += CONV_BLOCK) {
@@ -292,10 +318,10 @@ conv_bwd_data_kernel(const ConvBwdParams p) {
       const int k8 = (c % (BK / 8)) * 8;
       const int n = n0 + nrow;
       const int k = kt + k8;
-      bf16x8_t v = {0, 0, 0, 0, 0, 0, 0, 0};
-      if (n < p.Ci && k + 7 < p.K)
-        v = *reinterpret_cast<const bf16x8_t*>(p.wt + (long)n * p.K + k);
-      else if (n < p.Ci) {
+      const bool ok = (n < p.Ci) & (k + 7 < p.K);
+      const long off = ok ? (long)n * p.K + k : 0;
+      bf16x8_t v = mask8(*reinterpret_cast<const bf16x8_t*>(p.wt + off), ok);
+      if (!ok && n < p.Ci) {
         bf16* vv = reinterpret_cast<bf16*>(&v);
         for (int j = 0; j < 8 && k + j < p.K; ++j)
           vv[j] = p.wt[(long)n * p.K + k + j];
@@ -403,38 +429,29 @@ conv_wrw_kernel(const WrwParams p) {
 #pragma unroll
     for (int u = 0; u < DPT; ++u) {
       const int c = t + u * CONV_BLOCK;
-      bf16x8_t v = {0, 0, 0, 0, 0, 0, 0, 0};
-      if (c < DCH) {
-        const int mm = c / (BCO / 8);
-        const int c8 = (c % (BCO / 8)) * 8;
-        const int m = mt + mm;
-        if (m < mend && co0 + c8 < p.Co) {
-          if (dvec && co0 + c8 + 7 < p.Co)
-            v = *reinterpret_cast<const bf16x8_t*>(p.dy + (long)m * p.Co +
-                                                   co0 + c8);
-          else {
-            bf16* vv = reinterpret_cast<bf16*>(&v);
-            for (int j = 0; j < 8 && co0 + c8 + j < p.Co; ++j)
-              vv[j] = p.dy[(long)m * p.Co + co0 + c8 + j];
-          }
-        }
+      const int mm = c / (BCO / 8);
+      const int c8 = (c % (BCO / 8)) * 8;
+      const int m = mt + mm;
+      const bool ok = (c < DCH) & (m < mend) & (co0 + c8 + 7 < p.Co);
+      const long off = ok ? (long)m * p.Co + co0 + c8 : 0;
+      bf16x8_t v = mask8(*reinterpret_cast<const bf16x8_t*>(p.dy + off), ok);
+      if (!ok && c < DCH && m < mend && co0 + c8 < p.Co) {
+        bf16* vv = reinterpret_cast<bf16*>(&v);
+        for (int j = 0; j < 8 && co0 + c8 + j < p.Co; ++j)
+          vv[j] = p.dy[(long)m * p.Co + co0 + c8 + j];
       }
       dreg[u] = v;
     }
 #pragma unroll
     for (int u = 0; u < XPT; ++u) {
       const int c = t + u * CONV_BLOCK;
-      bf16x8_t v = {0, 0, 0, 0, 0, 0, 0, 0};
-      if (c < XCH) {
-        const int mm = c / (BKN / 8);
-        const int k8 = (c % (BKN / 8)) * 8;
-        const int m = mt + mm;
-        const int k = k0 + k8;
-        if (m < mend && k < p.K)
-          v = im2col_load8(p.x, m, k, p.IH, p.IW, p.Ci, p.OH, p.OW, p.S,
-                           p.stride, p.pad, p.K, xvec, p.g);
-      }
-      xreg[u] = v;
+      const int mm = c / (BKN / 8);
+      const int k8 = (c % (BKN / 8)) * 8;
+      const int m = mt + mm;
+      const int k = k0 + k8;
+      xreg[u] = im2col_load8(p.x, m, k, p.IH, p.IW, p.Ci, p.OH, p.OW, p.S,
+                             p.stride, p.pad, p.K, mend, xvec, p.g);
+      if (c >= XCH) xreg[u] = zero8();
     }
   };
 

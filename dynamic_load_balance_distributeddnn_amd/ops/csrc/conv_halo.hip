@@ -71,11 +71,14 @@ conv3x3_fwd_halo(const bf16* __restrict__ x, const bf16* __restrict__ w,
       const int pix = c / (CI_CHUNK / 8);
       const int hh = pix / HW, ww = pix % HW;
       const int ih = oh0 - 1 + hh, iw = ow0 - 1 + ww;
-      bf16x8_t v = {0, 0, 0, 0, 0, 0, 0, 0};
-      if (ih >= 0 && ih < H && iw >= 0 && iw < W)
-        v = *reinterpret_cast<const bf16x8_t*>(
-            x + (((long)n * H + ih) * W + iw) * Ci + ci0 + c8);
-      *reinterpret_cast<bf16x8_t*>(&halo[(hh * HW + ww) * HPAD + c8]) = v;
+      // branchless: always load a clamped address, mask invalid lanes
+      const bool ok = (ih >= 0) & (ih < H) & (iw >= 0) & (iw < W);
+      const long off = ok ? (((long)n * H + ih) * W + iw) * Ci + ci0 + c8 : 0;
+      bf16x8_t v = *reinterpret_cast<const bf16x8_t*>(x + off);
+      union { bf16x8_t h; int4 q; } u2; u2.h = v;
+      u2.q.x = ok ? u2.q.x : 0; u2.q.y = ok ? u2.q.y : 0;
+      u2.q.z = ok ? u2.q.z : 0; u2.q.w = ok ? u2.q.w : 0;
+      *reinterpret_cast<bf16x8_t*>(&halo[(hh * HW + ww) * HPAD + c8]) = u2.h;
     }
     // ---- stage weights [co][tap*32+ci] from w[co][tap*Ci + ci]
     constexpr int WCH = BN * 9 * (CI_CHUNK / 8);
@@ -84,11 +87,13 @@ conv3x3_fwd_halo(const bf16* __restrict__ x, const bf16* __restrict__ w,
       const int rest = c / (CI_CHUNK / 8);
       const int tap = rest % 9;
       const int co = rest / 9;
-      bf16x8_t v = {0, 0, 0, 0, 0, 0, 0, 0};
-      if (co0 + co < Co)
-        v = *reinterpret_cast<const bf16x8_t*>(
-            w + (long)(co0 + co) * K + tap * Ci + ci0 + c8);
-      *reinterpret_cast<bf16x8_t*>(&wlds[co * WLD + tap * CI_CHUNK + c8]) = v;
+      const bool ok = co0 + co < Co;
+      const long off = ok ? (long)(co0 + co) * K + tap * Ci + ci0 + c8 : 0;
+      bf16x8_t v = *reinterpret_cast<const bf16x8_t*>(w + off);
+      union { bf16x8_t h; int4 q; } u2; u2.h = v;
+      u2.q.x = ok ? u2.q.x : 0; u2.q.y = ok ? u2.q.y : 0;
+      u2.q.z = ok ? u2.q.z : 0; u2.q.w = ok ? u2.q.w : 0;
+      *reinterpret_cast<bf16x8_t*>(&wlds[co * WLD + tap * CI_CHUNK + c8]) = u2.h;
     }
     __syncthreads();
 
