@@ -401,9 +401,6 @@ conv_wrw_kernel(const WrwParams p) {
   constexpr int DPT = (DCH + CONV_BLOCK - 1) / CONV_BLOCK;
   constexpr int XPT = (XCH + CONV_BLOCK - 1) / CONV_BLOCK;
 
-  // double-buffered LDS: write stage i+1 while MFMAs consume stage i;
-  // ONE barrier per iteration (the un-pipelined form measured idle on
-  // every PMC: latency-bound, not memory- or compute-bound).
   __shared__ bf16 dy_t[2][BCO * LMD];
   __shared__ bf16 x_t[2][BKN * LMD];
 
@@ -608,17 +605,25 @@ extern "C" int dlb_conv_wrw_nsplits(int N, int OH, int OW, int Ci, int Co,
                                     int R, int S) {
   const int M = N * OH * OW;
   const int K = R * S * Ci;
-  const long tiles = (long)cdiv(Co, (Co >= 32) ? 32 : 16) * cdiv(K, 128);
+  const long tiles = (long)cdiv(Co, (Co >= 64) ? 64 : ((Co >= 32) ? 32 : 16))
+                     * cdiv(K, 128);
   int splits = (int)std::min<long>(std::max<long>(1, 1024 / tiles),
                                    std::max<long>(1, M / (16 * 64)));
   int mps = cdiv(cdiv(M, splits), 64) * 64;
   return cdiv(M, mps);
 }
 
+extern "C" bool dlb_conv3x3_wrw_halo(const void* x, const void* dy, float* dw,
+                                     int N, int H, int W, int Ci, int Co,
+                                     int splits, hipStream_t stream);
+
 extern "C" void dlb_conv_wrw(const void* x, const void* dy, float* dw, int N,
                              int IH, int IW, int Ci, int OH, int OW, int Co,
                              int R, int S, int stride, int pad, int splits,
                              hipStream_t stream) {
+  if (R == 3 && S == 3 && stride == 1 && pad == 1 &&
+      dlb_conv3x3_wrw_halo(x, dy, dw, N, IH, IW, Ci, Co, splits, stream))
+    return;
   WrwParams p{(const bf16*)x, (const bf16*)dy, dw, N, IH, IW, Ci, OH, OW,
               Co, R, S, stride, pad, N * OH * OW, R * S * Ci, 0, {}};
   p.g.fd_pix.init(OH * OW);
@@ -628,7 +633,11 @@ extern "C" void dlb_conv_wrw(const void* x, const void* dy, float* dw, int N,
   p.g.flat = (R == 1 && S == 1 && stride == 1 && pad == 0) ? 1 : 0;
   p.m_per_split = cdiv(cdiv(p.M, splits), 64) * 64;
   splits = cdiv(p.M, p.m_per_split);
-  if (Co >= 32) {
+  if (Co >= 64) {
+    dim3 grid(cdiv(Co, 64), cdiv(p.K, 128), splits);
+    hipLaunchKernelGGL((conv_wrw_kernel<64, 128>), grid, dim3(CONV_BLOCK), 0,
+                       stream, p);
+  } else if (Co >= 32) {
     dim3 grid(cdiv(Co, 32), cdiv(p.K, 128), splits);
     hipLaunchKernelGGL((conv_wrw_kernel<32, 128>), grid, dim3(CONV_BLOCK), 0,
                        stream, p);
