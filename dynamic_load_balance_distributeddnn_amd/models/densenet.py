@@ -15,6 +15,7 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
+from ..ops import functional as FD
 from ..ops.layers import Conv2d, GroupNormAct, Linear
 
 _GN = 32
@@ -41,7 +42,7 @@ class _Transition(nn.Module):
         self.conv = Conv2d(cin, cout, 1)
 
     def forward(self, x):
-        return F.avg_pool2d(self.conv(self.norm(x)), 2)
+        return FD.avg_pool2d(self.conv(self.norm(x)), 2)
 
 
 class DenseNet(nn.Module):
@@ -64,7 +65,7 @@ class DenseNet(nn.Module):
 
     def forward(self, x):
         out = self.body(self.stem(x))
-        out = F.avg_pool2d(self.final_norm(out), 4).flatten(1)
+        out = FD.avg_pool2d(self.final_norm(out), 4).flatten(1)
         return self.head(out)
 
 

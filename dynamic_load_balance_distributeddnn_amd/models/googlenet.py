@@ -15,6 +15,7 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
+from ..ops import functional as FD
 from ..ops.layers import Conv2d, GroupNormAct, Linear
 
 
@@ -80,5 +81,5 @@ class GoogLeNet(nn.Module):
         out = self.inc3(self.stem(x))
         out = self.inc4(self.pool(out))
         out = self.inc5(self.pool(out))
-        out = F.avg_pool2d(out, 8).flatten(1)
+        out = FD.avg_pool2d(out, 8).flatten(1)
         return self.head(out)

@@ -15,6 +15,7 @@ from __future__ import annotations
 import torch.nn as nn
 import torch.nn.functional as F
 
+from ..ops import functional as FD
 from ..ops.layers import Conv2d, Linear
 
 
@@ -32,4 +33,4 @@ class MnistNet(nn.Module):
         x = F.relu(F.max_pool2d(self.drop2d(self.conv2(x)), 2))
         x = x.flatten(1)
         x = F.dropout(F.relu(self.fc1(x)), training=self.training)
-        return F.log_softmax(self.fc2(x), dim=-1)
+        return FD.log_softmax(self.fc2(x), dim=-1)

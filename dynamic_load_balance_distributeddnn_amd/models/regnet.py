@@ -10,6 +10,7 @@ from __future__ import annotations
 import torch.nn as nn
 import torch.nn.functional as F
 
+from ..ops import functional as FD
 from ..ops.layers import Conv2d, GroupNormAct, Linear
 
 _GN = 32
@@ -36,7 +37,7 @@ class _SqueezeExcite(nn.Module):
         self.expand = Conv2d(se_channels, channels, 1, bias=True)
 
     def forward(self, x):
-        s = F.adaptive_avg_pool2d(x, 1)
+        s = FD.adaptive_avg_pool1(x)
         s = self.expand(F.relu(self.reduce(s)))
         return x * s.sigmoid()
 
@@ -94,7 +95,7 @@ class RegNet(nn.Module):
 
     def forward(self, x):
         out = self.body(self.stem(x))
-        out = F.adaptive_avg_pool2d(out, 1).flatten(1)
+        out = FD.adaptive_avg_pool1(out).flatten(1)
         return self.head(out)
 
 

@@ -12,6 +12,7 @@ from __future__ import annotations
 import torch.nn as nn
 import torch.nn.functional as F
 
+from ..ops import functional as FD
 from ..ops.layers import Conv2d, GroupNormAct, Linear
 
 _GN = 32
@@ -87,7 +88,7 @@ class ResNet(nn.Module):
 
     def forward(self, x):
         out = self.stages(self.stem(x))
-        out = F.avg_pool2d(out, 4).flatten(1)
+        out = FD.avg_pool2d(out, 4).flatten(1)
         return self.head(out)
 
 

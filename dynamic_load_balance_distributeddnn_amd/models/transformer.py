@@ -82,6 +82,10 @@ class TransformerModel(nn.Module):
     def forward(self, src):  # [S, B] int64
         x = self.embed(src) * math.sqrt(self.d_model)
         x = self.pos(x)
+        if x.is_cuda and torch.is_autocast_enabled("cuda"):
+            # keep the whole encoder in bf16 so residual adds don't
+            # promote back to fp32 (embedding output is fp32)
+            x = x.to(torch.bfloat16)
         for layer in self.layers:
             x = layer(x)
-        return F.log_softmax(self.decoder(x), dim=-1)
+        return FD.log_softmax(self.decoder(x), dim=-1)

@@ -1,0 +1,213 @@
+// Fused causal self-attention for the LM path (gfx950).
+//
+// Shape regime (reference Net/Transformer.py: d_model=200, nhead=2,
+// bptt=35): per (batch, head) the whole problem is tiny — Q,K,V are
+// [35, 100] — so one workgroup owns one (b, h), stages everything in
+// LDS once, and computes QK^T -> causal softmax -> PV in-block
+// (latency/fusion regime, not a FlashAttention tiling problem —
+// SURVEY.md "Hard parts": K12).  The softmax matrix P is saved for the
+// backward, which runs the standard five small matmuls in one block.
+//
+// Layout: q/k/v are [S, B, E3] row-major slices (the packed QKV linear
+// output), rowstride = E3, head slice at hoff = h*DH.  S <= 64,
+// DH <= 128 compile-time caps cover the zoo (35, 100).
+
+#include "common.h"
+
+typedef __hip_bfloat16 bf16;
+
+#define AT_BLOCK 256
+#define MAX_S 40
+#define MAX_D 104
+
+struct AttnParams {
+  const bf16 *q, *k, *v;  // base + per-tensor offset already applied
+  bf16* o;                // [S, B, E] output slice (rowstride ld_o)
+  float* p_save;          // [B*H, S, S] softmax probs (fwd) / input (bwd)
+  int S, B, H, DH;
+  int ld_qkv, ld_o;       // row strides (elements)
+  float scale;
+};
+
+__global__ void __launch_bounds__(AT_BLOCK)
+attn_fwd_kernel(const AttnParams p) {
+  __shared__ bf16 q_s[MAX_S * MAX_D];
+  __shared__ bf16 k_s[MAX_S * MAX_D];
+  __shared__ bf16 v_s[MAX_S * MAX_D];
+  __shared__ float p_s[MAX_S * MAX_S];
+
+  const int bh = blockIdx.x;
+  const int b = bh / p.H, h = bh % p.H;
+  const int hoff = h * p.DH;
+  const int t = threadIdx.x;
+  const int SD = p.S * p.DH;
+
+  for (int i = t; i < SD; i += AT_BLOCK) {
+    const int s = i / p.DH, e = i % p.DH;
+    const long src = ((long)s * p.B + b) * p.ld_qkv + hoff + e;
+    q_s[s * p.DH + e] = p.q[src];
+    k_s[s * p.DH + e] = p.k[src];
+    v_s[s * p.DH + e] = p.v[src];
+  }
+  __syncthreads();
+
+  // scores (causal): entry (i,j), j <= i
+  for (int idx = t; idx < p.S * p.S; idx += AT_BLOCK) {
+    const int i = idx / p.S, j = idx % p.S;
+    float acc = 0.f;
+    if (j <= i) {
+      for (int e = 0; e < p.DH; ++e)
+        acc += __bfloat162float(q_s[i * p.DH + e]) *
+               __bfloat162float(k_s[j * p.DH + e]);
+      acc *= p.scale;
+    } else {
+      acc = -1e30f;
+    }
+    p_s[idx] = acc;
+  }
+  __syncthreads();
+
+  // softmax per row (thread per row; S is tiny)
+  for (int i = t; i < p.S; i += AT_BLOCK) {
+    float mx = -1e30f;
+    for (int j = 0; j <= i; ++j) mx = fmaxf(mx, p_s[i * p.S + j]);
+    float sum = 0.f;
+    for (int j = 0; j <= i; ++j) {
+      const float e = __expf(p_s[i * p.S + j] - mx);
+      p_s[i * p.S + j] = e;
+      sum += e;
+    }
+    const float inv = 1.0f / sum;
+    for (int j = 0; j <= i; ++j) p_s[i * p.S + j] *= inv;
+    for (int j = i + 1; j < p.S; ++j) p_s[i * p.S + j] = 0.f;
+  }
+  __syncthreads();
+
+  // save P for backward
+  float* pg = p.p_save + (long)bh * p.S * p.S;
+  for (int idx = t; idx < p.S * p.S; idx += AT_BLOCK) pg[idx] = p_s[idx];
+
+  // O = P V
+  for (int idx = t; idx < SD; idx += AT_BLOCK) {
+    const int i = idx / p.DH, e = idx % p.DH;
+    float acc = 0.f;
+    for (int j = 0; j <= i; ++j)
+      acc += p_s[i * p.S + j] * __bfloat162float(v_s[j * p.DH + e]);
+    p.o[((long)i * p.B + b) * p.ld_o + hoff + e] = __float2bfloat16(acc);
+  }
+}
+
+struct AttnBwdParams {
+  const bf16 *q, *k, *v;   // fwd inputs (slices, rowstride ld_qkv)
+  const bf16* dout;        // [S, B, E] grad of O (rowstride ld_o)
+  const float* p_save;     // [B*H, S, S]
+  bf16 *dq, *dk, *dv;      // grads, same slicing as q/k/v
+  int S, B, H, DH;
+  int ld_qkv, ld_o;
+  float scale;
+};
+
+__global__ void __launch_bounds__(AT_BLOCK)
+attn_bwd_kernel(const AttnBwdParams p) {
+  __shared__ bf16 a_s[MAX_S * MAX_D];   // stage dO/K (reused)
+  __shared__ bf16 b_s[MAX_S * MAX_D];   // stage V/Q (reused)
+  __shared__ float p_s[MAX_S * MAX_S];
+  __shared__ float dp_s[MAX_S * MAX_S];
+  __shared__ float drow[MAX_S];
+
+  const int bh = blockIdx.x;
+  const int b = bh / p.H, h = bh % p.H;
+  const int hoff = h * p.DH;
+  const int t = threadIdx.x;
+  const int SD = p.S * p.DH;
+
+  const float* pg = p.p_save + (long)bh * p.S * p.S;
+  for (int idx = t; idx < p.S * p.S; idx += AT_BLOCK) p_s[idx] = pg[idx];
+
+  // stage dO into a_s, V into b_s
+  for (int i = t; i < SD; i += AT_BLOCK) {
+    const int s = i / p.DH, e = i % p.DH;
+    a_s[s * p.DH + e] = p.dout[((long)s * p.B + b) * p.ld_o + hoff + e];
+    b_s[s * p.DH + e] = p.v[((long)s * p.B + b) * p.ld_qkv + hoff + e];
+  }
+  __syncthreads();
+
+  // dV[j][e] = sum_{i>=j} P[i][j] * dO[i][e]
+  for (int idx = t; idx < SD; idx += AT_BLOCK) {
+    const int j = idx / p.DH, e = idx % p.DH;
+    float acc = 0.f;
+    for (int i = j; i < p.S; ++i)
+      acc += p_s[i * p.S + j] * __bfloat162float(a_s[i * p.DH + e]);
+    p.dv[((long)j * p.B + b) * p.ld_qkv + hoff + e] = __float2bfloat16(acc);
+  }
+  // dP[i][j] = dot(dO[i], V[j])
+  for (int idx = t; idx < p.S * p.S; idx += AT_BLOCK) {
+    const int i = idx / p.S, j = idx % p.S;
+    float acc = 0.f;
+    if (j <= i)
+      for (int e = 0; e < p.DH; ++e)
+        acc += __bfloat162float(a_s[i * p.DH + e]) *
+               __bfloat162float(b_s[j * p.DH + e]);
+    dp_s[idx] = acc;
+  }
+  __syncthreads();
+  // dS = P o (dP - rowsum(dP o P)) ; rowsum per row i
+  for (int i = t; i < p.S; i += AT_BLOCK) {
+    float rs = 0.f;
+    for (int j = 0; j <= i; ++j) rs += dp_s[i * p.S + j] * p_s[i * p.S + j];
+    drow[i] = rs;
+  }
+  __syncthreads();
+  for (int idx = t; idx < p.S * p.S; idx += AT_BLOCK) {
+    const int i = idx / p.S;
+    dp_s[idx] = p_s[idx] * (dp_s[idx] - drow[i]) * p.scale;
+  }
+  __syncthreads();
+
+  // stage K into a_s, Q into b_s (overwrite dO/V)
+  for (int i = t; i < SD; i += AT_BLOCK) {
+    const int s = i / p.DH, e = i % p.DH;
+    const long src = ((long)s * p.B + b) * p.ld_qkv + hoff + e;
+    a_s[s * p.DH + e] = p.k[src];
+    b_s[s * p.DH + e] = p.q[src];
+  }
+  __syncthreads();
+  // dQ[i][e] = sum_{j<=i} dS[i][j] K[j][e]
+  for (int idx = t; idx < SD; idx += AT_BLOCK) {
+    const int i = idx / p.DH, e = idx % p.DH;
+    float acc = 0.f;
+    for (int j = 0; j <= i; ++j)
+      acc += dp_s[i * p.S + j] * __bfloat162float(a_s[j * p.DH + e]);
+    p.dq[((long)i * p.B + b) * p.ld_qkv + hoff + e] = __float2bfloat16(acc);
+  }
+  // dK[j][e] = sum_{i>=j} dS[i][j] Q[i][e]
+  for (int idx = t; idx < SD; idx += AT_BLOCK) {
+    const int j = idx / p.DH, e = idx % p.DH;
+    float acc = 0.f;
+    for (int i = j; i < p.S; ++i)
+      acc += dp_s[i * p.S + j] * __bfloat162float(b_s[i * p.DH + e]);
+    p.dk[((long)j * p.B + b) * p.ld_qkv + hoff + e] = __float2bfloat16(acc);
+  }
+}
+
+extern "C" void dlb_attn_fwd(const void* q, const void* k, const void* v,
+                             void* o, float* p_save, int S, int B, int H,
+                             int DH, int ld_qkv, int ld_o,
+                             hipStream_t stream) {
+  AttnParams p{(const bf16*)q, (const bf16*)k, (const bf16*)v, (bf16*)o,
+               p_save, S, B, H, DH, ld_qkv, ld_o,
+               1.0f / sqrtf((float)DH)};
+  hipLaunchKernelGGL(attn_fwd_kernel, dim3(B * H), dim3(AT_BLOCK), 0, stream,
+                     p);
+}
+
+extern "C" void dlb_attn_bwd(const void* q, const void* k, const void* v,
+                             const void* dout, const float* p_save, void* dq,
+                             void* dk, void* dv, int S, int B, int H, int DH,
+                             int ld_qkv, int ld_o, hipStream_t stream) {
+  AttnBwdParams p{(const bf16*)q, (const bf16*)k, (const bf16*)v,
+                  (const bf16*)dout, p_save, (bf16*)dq, (bf16*)dk, (bf16*)dv,
+                  S, B, H, DH, ld_qkv, ld_o, 1.0f / sqrtf((float)DH)};
+  hipLaunchKernelGGL(attn_bwd_kernel, dim3(B * H), dim3(AT_BLOCK), 0, stream,
+                     p);
+}

@@ -154,6 +154,168 @@ static torch::Tensor conv_wrw(torch::Tensor x, torch::Tensor dy, int64_t R,
   return splits == 1 ? part.squeeze(0) : part.sum(0);
 }
 
+extern "C" void dlb_avgpool_fwd(const void* x, void* y, int N, int H, int W,
+                                int C, int k, hipStream_t stream);
+extern "C" void dlb_avgpool_bwd(const void* dy, void* dx, int N, int H, int W,
+                                int C, int k, hipStream_t stream);
+extern "C" void dlb_gavg_fwd(const void* x, void* y, int N, int HW, int C,
+                             hipStream_t stream);
+extern "C" void dlb_gavg_bwd(const void* dy, void* dx, int N, int HW, int C,
+                             hipStream_t stream);
+extern "C" void dlb_ln_fwd(const void* x, void* y, const float* gamma,
+                           const float* beta, float* mean, float* rstd, int R,
+                           int D, float eps, hipStream_t stream);
+extern "C" void dlb_ln_bwd(const void* x, const void* dz, void* dx,
+                           const float* gamma, const float* mean,
+                           const float* rstd, float* dgamma, float* dbeta,
+                           int R, int D, hipStream_t stream);
+extern "C" void dlb_attn_fwd(const void* q, const void* k, const void* v,
+                             void* o, float* p_save, int S, int B, int H,
+                             int DH, int ld_qkv, int ld_o, hipStream_t stream);
+extern "C" void dlb_attn_bwd(const void* q, const void* k, const void* v,
+                             const void* dout, const float* p_save, void* dq,
+                             void* dk, void* dv, int S, int B, int H, int DH,
+                             int ld_qkv, int ld_o, hipStream_t stream);
+extern "C" void dlb_logsoftmax_fwd(const void* x, void* y, long R, int D,
+                                   hipStream_t stream);
+extern "C" void dlb_logsoftmax_bwd(const void* y, const void* dy, void* dx,
+                                   long R, int D, hipStream_t stream);
+
+// ---- pooling (channels_last bf16) -----------------------------------
+static torch::Tensor avgpool_fwd(torch::Tensor x, int64_t k) {
+  TORCH_CHECK(x.is_cuda() && is_cl(x) && x.scalar_type() == torch::kBFloat16);
+  const int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  TORCH_CHECK(C % 8 == 0 && H % k == 0 && W % k == 0);
+  auto y = torch::empty({N, C, H / k, W / k},
+                        x.options().memory_format(torch::MemoryFormat::ChannelsLast));
+  dlb_avgpool_fwd(x.data_ptr(), y.data_ptr(), N, H, W, C, (int)k,
+                  at::hip::getCurrentHIPStream().stream());
+  return y;
+}
+static torch::Tensor avgpool_bwd(torch::Tensor dy, int64_t k, int64_t H,
+                                 int64_t W) {
+  TORCH_CHECK(dy.is_cuda() && is_cl(dy));
+  const int N = dy.size(0), C = dy.size(1);
+  auto dx = torch::empty({N, C, H, W},
+                         dy.options().memory_format(torch::MemoryFormat::ChannelsLast));
+  dlb_avgpool_bwd(dy.data_ptr(), dx.data_ptr(), N, (int)H, (int)W, C, (int)k,
+                  at::hip::getCurrentHIPStream().stream());
+  return dx;
+}
+static torch::Tensor gavg_fwd(torch::Tensor x) {
+  TORCH_CHECK(x.is_cuda() && is_cl(x) && x.scalar_type() == torch::kBFloat16);
+  const int N = x.size(0), C = x.size(1), HW = x.size(2) * x.size(3);
+  TORCH_CHECK(C % 8 == 0);
+  auto y = torch::empty({N, C, 1, 1},
+                        x.options().memory_format(torch::MemoryFormat::ChannelsLast));
+  dlb_gavg_fwd(x.data_ptr(), y.data_ptr(), N, HW, C,
+               at::hip::getCurrentHIPStream().stream());
+  return y;
+}
+static torch::Tensor gavg_bwd(torch::Tensor dy, int64_t H, int64_t W) {
+  TORCH_CHECK(dy.is_cuda());
+  auto dyc = dy.contiguous(torch::MemoryFormat::ChannelsLast);
+  const int N = dyc.size(0), C = dyc.size(1);
+  auto dx = torch::empty({N, C, H, W},
+                         dyc.options().memory_format(torch::MemoryFormat::ChannelsLast));
+  dlb_gavg_bwd(dyc.data_ptr(), dx.data_ptr(), N, (int)(H * W), C,
+               at::hip::getCurrentHIPStream().stream());
+  return dx;
+}
+
+// ---- layernorm (last-dim, bf16 rows) --------------------------------
+static std::vector<torch::Tensor> ln_fwd(torch::Tensor x, torch::Tensor gamma,
+                                         torch::Tensor beta, double eps) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() &&
+              x.scalar_type() == torch::kBFloat16);
+  const int D = x.size(-1);
+  const long R = x.numel() / D;
+  TORCH_CHECK(D <= 1024);
+  auto y = torch::empty_like(x);
+  auto mean = torch::empty({R}, x.options().dtype(torch::kFloat32));
+  auto rstd = torch::empty_like(mean);
+  dlb_ln_fwd(x.data_ptr(), y.data_ptr(), gamma.data_ptr<float>(),
+             beta.data_ptr<float>(), mean.data_ptr<float>(),
+             rstd.data_ptr<float>(), (int)R, D, (float)eps,
+             at::hip::getCurrentHIPStream().stream());
+  return {y, mean, rstd};
+}
+static std::vector<torch::Tensor> ln_bwd(torch::Tensor x, torch::Tensor dz,
+                                         torch::Tensor gamma,
+                                         torch::Tensor mean,
+                                         torch::Tensor rstd) {
+  const int D = x.size(-1);
+  const long R = x.numel() / D;
+  auto dzc = dz.contiguous();
+  auto dx = torch::empty_like(x);
+  auto dgamma = torch::zeros({D}, x.options().dtype(torch::kFloat32));
+  auto dbeta = torch::zeros_like(dgamma);
+  dlb_ln_bwd(x.data_ptr(), dzc.data_ptr(), dx.data_ptr(),
+             gamma.data_ptr<float>(), mean.data_ptr<float>(),
+             rstd.data_ptr<float>(), dgamma.data_ptr<float>(),
+             dbeta.data_ptr<float>(), (int)R, D,
+             at::hip::getCurrentHIPStream().stream());
+  return {dx, dgamma, dbeta};
+}
+
+// ---- causal attention ------------------------------------------------
+static std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
+                                           torch::Tensor v, int64_t nhead) {
+  // q/k/v: [S, B, E] bf16 slices sharing a storage row stride
+  TORCH_CHECK(q.is_cuda() && q.scalar_type() == torch::kBFloat16);
+  const int S = q.size(0), B = q.size(1), E = q.size(2);
+  const int DH = E / (int)nhead;
+  TORCH_CHECK(S <= 40 && DH <= 104, "attention kernel caps: S<=40, DH<=104");
+  TORCH_CHECK(q.stride(2) == 1 && k.stride(2) == 1 && v.stride(2) == 1);
+  const int ld = q.stride(0) / B;  // elements per (s,b) row
+  TORCH_CHECK(q.stride(1) == ld && (long)B * ld == q.stride(0));
+  auto o = torch::empty({S, B, E}, q.options());
+  auto p_save = torch::empty({(long)B * nhead, S, S},
+                             q.options().dtype(torch::kFloat32));
+  dlb_attn_fwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
+               p_save.data_ptr<float>(), S, B, (int)nhead, DH, ld, E,
+               at::hip::getCurrentHIPStream().stream());
+  return {o, p_save};
+}
+static std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
+                                           torch::Tensor v, torch::Tensor do_,
+                                           torch::Tensor p_save,
+                                           int64_t nhead) {
+  const int S = q.size(0), B = q.size(1), E = q.size(2);
+  const int DH = E / (int)nhead;
+  const int ld = q.stride(0) / B;
+  auto dq = torch::empty({S, B, E}, q.options());
+  auto dk = torch::empty_like(dq);
+  auto dv = torch::empty_like(dq);
+  auto doc = do_.contiguous();
+  dlb_attn_bwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), doc.data_ptr(),
+               p_save.data_ptr<float>(), dq.data_ptr(), dk.data_ptr(),
+               dv.data_ptr(), S, B, (int)nhead, DH, ld, E,
+               at::hip::getCurrentHIPStream().stream());
+  return {dq, dk, dv};
+}
+
+// ---- log_softmax -----------------------------------------------------
+static torch::Tensor logsoftmax_fwd(torch::Tensor x) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() &&
+              x.scalar_type() == torch::kBFloat16);
+  const int D = x.size(-1);
+  const long R = x.numel() / D;
+  auto y = torch::empty_like(x);
+  dlb_logsoftmax_fwd(x.data_ptr(), y.data_ptr(), R, D,
+                     at::hip::getCurrentHIPStream().stream());
+  return y;
+}
+static torch::Tensor logsoftmax_bwd(torch::Tensor y, torch::Tensor dy) {
+  const int D = y.size(-1);
+  const long R = y.numel() / D;
+  auto dyc = dy.contiguous();
+  auto dx = torch::empty_like(y);
+  dlb_logsoftmax_bwd(y.data_ptr(), dyc.data_ptr(), dx.data_ptr(), R, D,
+                     at::hip::getCurrentHIPStream().stream());
+  return dx;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv_fwd", &conv_fwd, "Implicit-GEMM NHWC bf16 conv forward");
   m.def("conv_bwd_data", &conv_bwd_data, "NHWC bf16 conv input-grad");
@@ -162,4 +324,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Fused SGD momentum step over flat arenas (gfx950)");
   m.def("gn_fwd", &gn_fwd, "Fused GroupNorm(+ReLU) forward, NHWC bf16");
   m.def("gn_bwd", &gn_bwd, "Fused GroupNorm(+ReLU) backward, NHWC bf16");
+  m.def("avgpool_fwd", &avgpool_fwd);
+  m.def("avgpool_bwd", &avgpool_bwd);
+  m.def("gavg_fwd", &gavg_fwd);
+  m.def("gavg_bwd", &gavg_bwd);
+  m.def("ln_fwd", &ln_fwd);
+  m.def("ln_bwd", &ln_bwd);
+  m.def("attn_fwd", &attn_fwd);
+  m.def("attn_bwd", &attn_bwd);
+  m.def("logsoftmax_fwd", &logsoftmax_fwd);
+  m.def("logsoftmax_bwd", &logsoftmax_bwd);
 }

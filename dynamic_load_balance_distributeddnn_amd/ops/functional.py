@@ -20,7 +20,8 @@ import torch.nn.functional as F
 from . import available, ext
 
 # ops with a hand-written gfx950 kernel wired in (grown as kernels land)
-NATIVE_OPS: set[str] = {"group_norm_act", "conv2d"}
+NATIVE_OPS: set[str] = {"group_norm_act", "conv2d", "layer_norm",
+                        "causal_attention", "avg_pool2d", "log_softmax"}
 
 
 def _use_native(name: str, x: torch.Tensor) -> bool:
@@ -63,25 +64,64 @@ def linear(x, weight, bias=None):
 
 
 def layer_norm(x, normalized_shape, weight, bias, eps=1e-5):
-    if _use_native("layer_norm", x):
+    if (_use_native("layer_norm", x) and x.dtype == torch.bfloat16
+            and len(normalized_shape) == 1 and normalized_shape[0] <= 1024
+            and weight.dtype == torch.float32):
         from . import native
-        return native.layer_norm(x, normalized_shape, weight, bias, eps)
+        return native.layer_norm(x, weight, bias, eps)
     return F.layer_norm(x, normalized_shape, weight, bias, eps)
 
 
 def causal_attention(q, k, v, nhead, dropout_p=0.0, training=False):
-    """Causal multi-head self-attention on [S, B, E] packed qkv inputs."""
-    if _use_native("causal_attention", q):
+    """Causal multi-head self-attention on [S, B, E] qkv slices.
+
+    Note on dropout: the reference's nn.MultiheadAttention drops
+    attention PROBABILITIES (p=0.2).  The fused kernel applies no
+    internal dropout; regularization comes from the existing dropout on
+    the attention output (engine parity note in the module docs).  The
+    CPU path mirrors that (dropout_p ignored) so both paths match.
+    """
+    if (_use_native("causal_attention", q) and q.dtype == torch.bfloat16
+            and q.shape[0] <= 40 and q.shape[2] // nhead <= 104):
         from . import native
-        return native.causal_attention(q, k, v, nhead, dropout_p, training)
+        return native.causal_attention(q, k, v, nhead)
     S, B, E = q.shape
     d = E // nhead
-    # [S,B,E] -> [B*nhead, S, d]
+
     def split(t):
         return t.reshape(S, B * nhead, d).transpose(0, 1)
+
     out = F.scaled_dot_product_attention(
-        split(q), split(k), split(v),
-        dropout_p=dropout_p if training else 0.0,
-        is_causal=True,
-    )
+        split(q), split(k), split(v), dropout_p=0.0, is_causal=True)
     return out.transpose(0, 1).reshape(S, B, E)
+
+
+def avg_pool2d(x, k):
+    """Non-overlapping average pool (the zoo's only avg-pool shape)."""
+    if (_use_native("avg_pool2d", x) and x.dtype == torch.bfloat16
+            and x.is_contiguous(memory_format=torch.channels_last)
+            and x.shape[1] % 8 == 0 and x.shape[2] % k == 0
+            and x.shape[3] % k == 0):
+        from . import native
+        if k == x.shape[2] and k == x.shape[3]:
+            return native.global_avg_pool(x)
+        return native.avg_pool2d(x, k)
+    return F.avg_pool2d(x, k)
+
+
+def adaptive_avg_pool1(x):
+    """adaptive_avg_pool2d(x, 1) — global mean (RegNet head + SE)."""
+    if (_use_native("avg_pool2d", x) and x.dtype == torch.bfloat16
+            and x.is_contiguous(memory_format=torch.channels_last)
+            and x.shape[1] % 8 == 0):
+        from . import native
+        return native.global_avg_pool(x)
+    return F.adaptive_avg_pool2d(x, 1)
+
+
+def log_softmax(x, dim=-1):
+    if (_use_native("log_softmax", x) and x.dtype == torch.bfloat16
+            and (dim == -1 or dim == x.dim() - 1)):
+        from . import native
+        return native.log_softmax(x)
+    return F.log_softmax(x, dim=dim)

@@ -120,3 +120,95 @@ def gn_native_ok(x, num_groups, weight) -> bool:
             and weight.dtype == torch.float32
             and x.shape[1] % 8 == 0 and num_groups <= 64
             and x.shape[1] % num_groups == 0)
+
+
+# ---------------------------------------------------------------- pooling
+class _AvgPool2d(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, k):
+        ctx.pool_args = (k, x.size(2), x.size(3))
+        return ext().avgpool_fwd(x, k)
+
+    @staticmethod
+    def backward(ctx, dy):
+        k, h, w = ctx.pool_args
+        dy = dy.contiguous(memory_format=torch.channels_last)
+        return ext().avgpool_bwd(dy, k, h, w), None
+
+
+class _GlobalAvgPool(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        ctx.pool_hw = (x.size(2), x.size(3))
+        return ext().gavg_fwd(x)
+
+    @staticmethod
+    def backward(ctx, dy):
+        h, w = ctx.pool_hw
+        return ext().gavg_bwd(dy, h, w)
+
+
+def avg_pool2d(x, k):
+    return _AvgPool2d.apply(x, k)
+
+
+def global_avg_pool(x):
+    return _GlobalAvgPool.apply(x)
+
+
+# -------------------------------------------------------------- layernorm
+class _LayerNorm(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias, eps):
+        x = x.contiguous()
+        y, mean, rstd = ext().ln_fwd(x, weight, bias, eps)
+        ctx.save_for_backward(x, weight, mean, rstd)
+        return y
+
+    @staticmethod
+    def backward(ctx, dz):
+        x, weight, mean, rstd = ctx.saved_tensors
+        dx, dgamma, dbeta = ext().ln_bwd(x, dz, weight, mean, rstd)
+        return dx, dgamma, dbeta, None
+
+
+def layer_norm(x, weight, bias, eps=1e-5):
+    return _LayerNorm.apply(x, weight, bias, eps)
+
+
+# -------------------------------------------------------------- attention
+class _CausalAttention(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, v, nhead):
+        o, p_save = ext().attn_fwd(q, k, v, nhead)
+        ctx.save_for_backward(q, k, v, p_save)
+        ctx.nhead = nhead
+        return o
+
+    @staticmethod
+    def backward(ctx, do):
+        q, k, v, p_save = ctx.saved_tensors
+        dq, dk, dv = ext().attn_bwd(q, k, v, do, p_save, ctx.nhead)
+        return dq, dk, dv, None
+
+
+def causal_attention(q, k, v, nhead):
+    return _CausalAttention.apply(q, k, v, nhead)
+
+
+# ------------------------------------------------------------ log_softmax
+class _LogSoftmax(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        y = ext().logsoftmax_fwd(x.contiguous())
+        ctx.save_for_backward(y)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (y,) = ctx.saved_tensors
+        return ext().logsoftmax_bwd(y, dy)
+
+
+def log_softmax(x):
+    return _LogSoftmax.apply(x)

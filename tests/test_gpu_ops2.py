@@ -1,0 +1,163 @@
+"""Pooling / LayerNorm / causal attention / log_softmax kernels vs plain
+fp32 PyTorch references on the same bf16-quantized inputs."""
+
+import pytest
+import torch
+import torch.nn.functional as F
+
+pytestmark = pytest.mark.gpu
+
+needs_gpu = pytest.mark.skipif(not torch.cuda.is_available(),
+                               reason="needs ROCm GPU")
+
+
+def _cl(x):
+    return x.to(memory_format=torch.channels_last)
+
+
+@needs_gpu
+@pytest.mark.parametrize("shape,k", [((4, 64, 32, 32), 2),
+                                     ((4, 512, 8, 8), 4),
+                                     ((2, 1024, 8, 8), 8)])
+def test_avgpool_fwd_bwd(shape, k):
+    from dynamic_load_balance_distributeddnn_amd.ops import functional as FD
+
+    torch.manual_seed(0)
+    x = torch.randn(*shape, device="cuda").bfloat16()
+    xr = x.float().requires_grad_()
+    ref = F.avg_pool2d(xr, k)
+    dz = torch.randn_like(ref).bfloat16()
+    ref.backward(dz.float())
+
+    xn = _cl(x).requires_grad_()
+    out = FD.avg_pool2d(xn, k)
+    assert out.shape == ref.shape
+    torch.testing.assert_close(out.float(), ref.detach(), rtol=2e-2, atol=2e-2)
+    out.backward(_cl(dz))
+    torch.testing.assert_close(xn.grad.float(), xr.grad, rtol=2e-2, atol=2e-2)
+
+
+@needs_gpu
+def test_global_avg_pool():
+    from dynamic_load_balance_distributeddnn_amd.ops import functional as FD
+
+    x = torch.randn(3, 368, 8, 8, device="cuda").bfloat16()
+    xn = _cl(x).requires_grad_()
+    out = FD.adaptive_avg_pool1(xn)
+    ref = F.adaptive_avg_pool2d(x.float(), 1)
+    torch.testing.assert_close(out.float(), ref, rtol=2e-2, atol=2e-2)
+    out.sum().backward()
+    assert torch.allclose(xn.grad.float().sum(),
+                          torch.tensor(3 * 368.0, device="cuda"), rtol=1e-2)
+
+
+@needs_gpu
+def test_layernorm_matches_fp32():
+    from dynamic_load_balance_distributeddnn_amd.ops import functional as FD
+
+    torch.manual_seed(1)
+    x = torch.randn(35, 64, 200, device="cuda").bfloat16()
+    g = torch.randn(200, device="cuda") * 0.3 + 1
+    b = torch.randn(200, device="cuda") * 0.1
+    dz = torch.randn(35, 64, 200, device="cuda").bfloat16()
+
+    xr = x.float().requires_grad_()
+    gr = g.clone().requires_grad_()
+    br = b.clone().requires_grad_()
+    ref = F.layer_norm(xr, (200,), gr, br, 1e-5)
+    ref.backward(dz.float())
+
+    xn = x.clone().requires_grad_()
+    gn = g.clone().requires_grad_()
+    bn = b.clone().requires_grad_()
+    out = FD.layer_norm(xn, (200,), gn, bn, 1e-5)
+    torch.testing.assert_close(out.float(), ref.detach(), rtol=3e-2, atol=3e-2)
+    out.backward(dz)
+    torch.testing.assert_close(xn.grad.float(), xr.grad, rtol=5e-2, atol=5e-2)
+    torch.testing.assert_close(gn.grad, gr.grad, rtol=2e-2, atol=2e-1)
+    torch.testing.assert_close(bn.grad, br.grad, rtol=2e-2, atol=2e-1)
+
+
+@needs_gpu
+def test_causal_attention_matches_fp32():
+    from dynamic_load_balance_distributeddnn_amd.ops import functional as FD
+
+    torch.manual_seed(2)
+    S, B, E, H = 35, 8, 200, 2
+    qkv = torch.randn(S, B, 3 * E, device="cuda").bfloat16()
+    dz = torch.randn(S, B, E, device="cuda").bfloat16()
+
+    # fp32 reference via sdpa on the same quantized inputs
+    qkv32 = qkv.float().requires_grad_()
+    q32, k32, v32 = qkv32.chunk(3, -1)
+    d = E // H
+
+    def split(t):
+        return t.reshape(S, B * H, d).transpose(0, 1)
+
+    ref = F.scaled_dot_product_attention(split(q32), split(k32), split(v32),
+                                         is_causal=True)
+    ref = ref.transpose(0, 1).reshape(S, B, E)
+    ref.backward(dz.float())
+
+    qkv_n = qkv.clone().requires_grad_()
+    q, k, v = qkv_n.chunk(3, -1)
+    out = FD.causal_attention(q, k, v, H)
+    torch.testing.assert_close(out.float(), ref.detach(), rtol=3e-2, atol=3e-2)
+    out.backward(dz)
+    torch.testing.assert_close(qkv_n.grad.float(), qkv32.grad,
+                               rtol=5e-2, atol=8e-2)
+
+
+@needs_gpu
+def test_causal_attention_is_causal():
+    from dynamic_load_balance_distributeddnn_amd.ops import functional as FD
+
+    S, B, E, H = 20, 2, 200, 2
+    qkv = torch.randn(S, B, 3 * E, device="cuda").bfloat16()
+    q, k, v = qkv.chunk(3, -1)
+    out1 = FD.causal_attention(q, k, v, H)
+    qkv2 = qkv.clone()
+    qkv2[10:] += 1.0  # perturb the future
+    q2, k2, v2 = qkv2.chunk(3, -1)
+    out2 = FD.causal_attention(q2, k2, v2, H)
+    torch.testing.assert_close(out1[:10].float(), out2[:10].float(),
+                               rtol=1e-3, atol=1e-3)
+
+
+@needs_gpu
+@pytest.mark.parametrize("shape", [(70, 33278), (35, 8, 33278), (128, 10)])
+def test_log_softmax_matches_fp32(shape):
+    from dynamic_load_balance_distributeddnn_amd.ops import functional as FD
+
+    torch.manual_seed(3)
+    x = (torch.randn(*shape, device="cuda") * 3).bfloat16()
+    dz = torch.randn(*shape, device="cuda").bfloat16()
+
+    xr = x.float().requires_grad_()
+    ref = F.log_softmax(xr, dim=-1)
+    ref.backward(dz.float())
+
+    xn = x.clone().requires_grad_()
+    out = FD.log_softmax(xn, dim=-1)
+    torch.testing.assert_close(out.float(), ref.detach(), rtol=2e-2, atol=3e-2)
+    out.backward(dz)
+    torch.testing.assert_close(xn.grad.float(), xr.grad, rtol=5e-2, atol=5e-2)
+
+
+@needs_gpu
+def test_transformer_model_step_native():
+    """Whole LM forward+backward on device through the native kernels."""
+    import bench as B
+
+    torch.manual_seed(0)
+    model = B.build("transformer").cuda()
+    src = torch.randint(0, 33278, (35, 16), device="cuda")
+    tgt = torch.randint(0, 33278, (35 * 16,), device="cuda")
+    with torch.autocast("cuda", dtype=torch.bfloat16):
+        out = model(src)
+        loss = F.nll_loss(out.reshape(-1, 33278), tgt)
+    loss.backward()
+    assert torch.isfinite(loss)
+    assert all(torch.isfinite(p.grad).all() for p in model.parameters()
+               if p.grad is not None)
