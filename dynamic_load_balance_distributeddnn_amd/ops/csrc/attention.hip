@@ -101,9 +101,9 @@ struct AttnBwdParams {
   const bf16 *q, *k, *v;   // fwd inputs (slices, rowstride ld_qkv)
   const bf16* dout;        // [S, B, E] grad of O (rowstride ld_o)
   const float* p_save;     // [B*H, S, S]
-  bf16 *dq, *dk, *dv;      // grads, same slicing as q/k/v
+  bf16 *dq, *dk, *dv;      // grads: CONTIGUOUS [S,B,E] (row stride ld_g)
   int S, B, H, DH;
-  int ld_qkv, ld_o;
+  int ld_qkv, ld_o, ld_g;
   float scale;
 };
 
@@ -138,7 +138,7 @@ attn_bwd_kernel(const AttnBwdParams p) {
     float acc = 0.f;
     for (int i = j; i < p.S; ++i)
       acc += p_s[i * p.S + j] * __bfloat162float(a_s[i * p.DH + e]);
-    p.dv[((long)j * p.B + b) * p.ld_qkv + hoff + e] = __float2bfloat16(acc);
+    p.dv[((long)j * p.B + b) * p.ld_g + hoff + e] = __float2bfloat16(acc);
   }
   // dP[i][j] = dot(dO[i], V[j])
   for (int idx = t; idx < p.S * p.S; idx += AT_BLOCK) {
@@ -178,7 +178,7 @@ attn_bwd_kernel(const AttnBwdParams p) {
     float acc = 0.f;
     for (int j = 0; j <= i; ++j)
       acc += dp_s[i * p.S + j] * __bfloat162float(a_s[j * p.DH + e]);
-    p.dq[((long)i * p.B + b) * p.ld_qkv + hoff + e] = __float2bfloat16(acc);
+    p.dq[((long)i * p.B + b) * p.ld_g + hoff + e] = __float2bfloat16(acc);
   }
   // dK[j][e] = sum_{i>=j} dS[i][j] Q[i][e]
   for (int idx = t; idx < SD; idx += AT_BLOCK) {
@@ -186,7 +186,7 @@ attn_bwd_kernel(const AttnBwdParams p) {
     float acc = 0.f;
     for (int i = j; i < p.S; ++i)
       acc += dp_s[i * p.S + j] * __bfloat162float(b_s[i * p.DH + e]);
-    p.dk[((long)j * p.B + b) * p.ld_qkv + hoff + e] = __float2bfloat16(acc);
+    p.dk[((long)j * p.B + b) * p.ld_g + hoff + e] = __float2bfloat16(acc);
   }
 }
 
@@ -204,10 +204,11 @@ extern "C" void dlb_attn_fwd(const void* q, const void* k, const void* v,
 extern "C" void dlb_attn_bwd(const void* q, const void* k, const void* v,
                              const void* dout, const float* p_save, void* dq,
                              void* dk, void* dv, int S, int B, int H, int DH,
-                             int ld_qkv, int ld_o, hipStream_t stream) {
+                             int ld_qkv, int ld_o, int ld_g,
+                             hipStream_t stream) {
   AttnBwdParams p{(const bf16*)q, (const bf16*)k, (const bf16*)v,
                   (const bf16*)dout, p_save, (bf16*)dq, (bf16*)dk, (bf16*)dv,
-                  S, B, H, DH, ld_qkv, ld_o, 1.0f / sqrtf((float)DH)};
+                  S, B, H, DH, ld_qkv, ld_o, ld_g, 1.0f / sqrtf((float)DH)};
   hipLaunchKernelGGL(attn_bwd_kernel, dim3(B * H), dim3(AT_BLOCK), 0, stream,
                      p);
 }

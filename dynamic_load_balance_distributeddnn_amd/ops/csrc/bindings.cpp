@@ -175,7 +175,8 @@ extern "C" void dlb_attn_fwd(const void* q, const void* k, const void* v,
 extern "C" void dlb_attn_bwd(const void* q, const void* k, const void* v,
                              const void* dout, const float* p_save, void* dq,
                              void* dk, void* dv, int S, int B, int H, int DH,
-                             int ld_qkv, int ld_o, hipStream_t stream);
+                             int ld_qkv, int ld_o, int ld_g,
+                             hipStream_t stream);
 extern "C" void dlb_logsoftmax_fwd(const void* x, void* y, long R, int D,
                                    hipStream_t stream);
 extern "C" void dlb_logsoftmax_bwd(const void* y, const void* dy, void* dx,
@@ -290,7 +291,7 @@ static std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
   auto doc = do_.contiguous();
   dlb_attn_bwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), doc.data_ptr(),
                p_save.data_ptr<float>(), dq.data_ptr(), dk.data_ptr(),
-               dv.data_ptr(), S, B, (int)nhead, DH, ld, E,
+               dv.data_ptr(), S, B, (int)nhead, DH, ld, E, E,
                at::hip::getCurrentHIPStream().stream());
   return {dq, dk, dv};
 }
