@@ -142,7 +142,12 @@ class Trainer:
         t.backward_done()
         t.add_compute(self.fault.maybe_wait(epoch, steps_per_epoch))
         if self.is_lm:
-            torch.nn.utils.clip_grad_norm_(self.model.parameters(), 0.25)
+            # global-norm clip at 0.25 (dbs.py:274) — all grads live in ONE
+            # flat arena, so this is a norm + conditional scale on a single
+            # tensor instead of a 27-tensor walk.
+            norm = torch.linalg.vector_norm(self.sync.arena)
+            scale = (0.25 / (norm + 1e-6)).clamp(max=1.0)
+            self.sync.arena.mul_(scale)
         self.sync.finish()
         t.comm_done()
         self.optimizer.step()

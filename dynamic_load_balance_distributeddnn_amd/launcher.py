@@ -44,7 +44,13 @@ def init_process_group(args, rank: int, world_size: int,
     else:
         os.environ.setdefault("MASTER_PORT", "29500")
     device = resolve_device(args, rank)
-    backend = "gloo" if device.type == "cpu" else "nccl"
+    # RCCL requires one rank per GPU; the reference's straggler trick maps
+    # several ranks onto one GPU (-gpu 0,0,0,1 — README.md:23-28), which
+    # must then rendezvous over gloo (the reference's only backend).
+    oversubscribed = (isinstance(args.gpu, list)
+                      and len(set(args.gpu[:world_size])) < min(world_size,
+                                                                len(args.gpu)))
+    backend = "gloo" if (device.type == "cpu" or oversubscribed) else "nccl"
     if device.type == "cuda":
         torch.cuda.set_device(device)
     dist.init_process_group(backend, rank=rank, world_size=world_size)

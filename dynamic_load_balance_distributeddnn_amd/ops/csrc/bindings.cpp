@@ -317,6 +317,51 @@ static torch::Tensor logsoftmax_bwd(torch::Tensor y, torch::Tensor dy) {
   return dx;
 }
 
+extern "C" void dlb_gconv_fwd(const void* x, const void* w, void* y, int N,
+                              int IH, int IW, int C, int GW, int stride,
+                              hipStream_t stream);
+extern "C" void dlb_gconv_bwd(const void* dy, const void* w, void* dx, int N,
+                              int IH, int IW, int C, int GW, int stride,
+                              hipStream_t stream);
+extern "C" void dlb_gconv_wrw(const void* x, const void* dy, float* dw, int N,
+                              int IH, int IW, int C, int GW, int stride,
+                              hipStream_t stream);
+
+// grouped 3x3 (RegNet): x [N,C,H,W] cl bf16, w [C,GW,3,3] cl bf16
+static torch::Tensor gconv_fwd(torch::Tensor x, torch::Tensor w,
+                               int64_t stride) {
+  TORCH_CHECK(x.is_cuda() && is_cl(x) && is_cl(w));
+  const int N = x.size(0), C = x.size(1), IH = x.size(2), IW = x.size(3);
+  const int GW = w.size(1);
+  TORCH_CHECK(w.size(0) == C && w.size(2) == 3 && w.size(3) == 3);
+  const int OH = (IH + 2 - 3) / stride + 1, OW = (IW + 2 - 3) / stride + 1;
+  auto y = torch::empty({N, C, OH, OW},
+                        x.options().memory_format(torch::MemoryFormat::ChannelsLast));
+  dlb_gconv_fwd(x.data_ptr(), w.data_ptr(), y.data_ptr(), N, IH, IW, C, GW,
+                (int)stride, at::hip::getCurrentHIPStream().stream());
+  return y;
+}
+static torch::Tensor gconv_bwd(torch::Tensor dy, torch::Tensor w, int64_t IH,
+                               int64_t IW, int64_t stride) {
+  const int N = dy.size(0), C = dy.size(1);
+  const int GW = w.size(1);
+  auto dx = torch::empty({N, C, IH, IW},
+                         dy.options().memory_format(torch::MemoryFormat::ChannelsLast));
+  dlb_gconv_bwd(dy.data_ptr(), w.data_ptr(), dx.data_ptr(), N, (int)IH,
+                (int)IW, C, GW, (int)stride,
+                at::hip::getCurrentHIPStream().stream());
+  return dx;
+}
+static torch::Tensor gconv_wrw(torch::Tensor x, torch::Tensor dy, int64_t GW,
+                               int64_t stride) {
+  const int N = x.size(0), C = x.size(1), IH = x.size(2), IW = x.size(3);
+  auto dw = torch::zeros({C, 9 * GW}, x.options().dtype(torch::kFloat32));
+  dlb_gconv_wrw(x.data_ptr(), dy.data_ptr(), dw.data_ptr<float>(), N, IH, IW,
+                C, (int)GW, (int)stride,
+                at::hip::getCurrentHIPStream().stream());
+  return dw;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv_fwd", &conv_fwd, "Implicit-GEMM NHWC bf16 conv forward");
   m.def("conv_bwd_data", &conv_bwd_data, "NHWC bf16 conv input-grad");
@@ -333,6 +378,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ln_bwd", &ln_bwd);
   m.def("attn_fwd", &attn_fwd);
   m.def("attn_bwd", &attn_bwd);
+  m.def("gconv_fwd", &gconv_fwd);
+  m.def("gconv_bwd", &gconv_bwd);
+  m.def("gconv_wrw", &gconv_wrw);
   m.def("logsoftmax_fwd", &logsoftmax_fwd);
   m.def("logsoftmax_bwd", &logsoftmax_bwd);
 }

@@ -38,9 +38,12 @@ def conv2d(x, weight, bias=None, stride=1, padding=0, groups=1):
         x = x.to(torch.bfloat16)
     if _use_native("conv2d", x):
         from . import native
+        s = stride if isinstance(stride, int) else stride[0]
+        p = padding if isinstance(padding, int) else padding[0]
+        if groups > 1 and bias is None and \
+                native.gconv_native_ok(x, weight, s, p, groups):
+            return native.grouped_conv2d(x, weight, s)
         if native.conv_native_ok(x, weight, stride, padding, groups):
-            s = stride if isinstance(stride, int) else stride[0]
-            p = padding if isinstance(padding, int) else padding[0]
             return native.conv2d(x, weight, bias, s, p, groups)
     return F.conv2d(x, weight, bias, stride=stride, padding=padding, groups=groups)
 

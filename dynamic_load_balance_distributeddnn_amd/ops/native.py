@@ -212,3 +212,47 @@ class _LogSoftmax(torch.autograd.Function):
 
 def log_softmax(x):
     return _LogSoftmax.apply(x)
+
+
+# ------------------------------------------------------------ grouped conv
+class _GroupedConv2d(torch.autograd.Function):
+    """RegNet's grouped 3x3 (group_width channels per group, pad 1)."""
+
+    @staticmethod
+    def forward(ctx, x, weight, stride):
+        wcl = weight.detach().to(torch.bfloat16) \
+            .contiguous(memory_format=torch.channels_last)
+        y = ext().gconv_fwd(x, wcl, stride)
+        ctx.save_for_backward(x, wcl)
+        ctx.gconv_stride = stride
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, wcl = ctx.saved_tensors
+        stride = ctx.gconv_stride
+        dy = dy.contiguous(memory_format=torch.channels_last)
+        c, gw = wcl.shape[0], wcl.shape[1]
+        dx = None
+        if ctx.needs_input_grad[0]:
+            dx = ext().gconv_bwd(dy, wcl, x.size(2), x.size(3), stride)
+        dweight = None
+        if ctx.needs_input_grad[1]:
+            dwf = ext().gconv_wrw(x, dy, gw, stride)   # [C, 9*GW]
+            dweight = dwf.view(c, 3, 3, gw).permute(0, 3, 1, 2)
+        return dx, dweight, None
+
+
+def grouped_conv2d(x, weight, stride):
+    return _GroupedConv2d.apply(x, weight, stride)
+
+
+def gconv_native_ok(x, weight, stride, padding, groups) -> bool:
+    if x.dim() != 4 or x.dtype != torch.bfloat16:
+        return False
+    if not x.is_contiguous(memory_format=torch.channels_last):
+        return False
+    co, cig, r, s = weight.shape
+    gw = co // groups
+    return (r == 3 and s == 3 and padding == 1 and stride in (1, 2)
+            and cig == gw and gw in (8, 16) and co == x.shape[1])
