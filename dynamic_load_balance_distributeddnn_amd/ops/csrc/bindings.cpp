@@ -145,8 +145,20 @@ static torch::Tensor conv_wrw(torch::Tensor x, torch::Tensor dy, int64_t R,
     int v = atoi(env);
     if (v > 0 && v < splits) splits = v;
   }
-  auto part = torch::zeros({splits, Co, R * S * Ci},
-                           x.options().dtype(torch::kFloat32));
+  const long K = R * S * Ci;
+  // every slab element is written by exactly one block when the tiles
+  // cover Co and K exactly — skip the memset then (it measured 4% of
+  // the whole step as FillFunctor calls)
+  const bool halo3 = (R == 3 && S == 3 && Ci % 32 == 0 && Co % 8 == 0 &&
+                      OW <= 32 && (OW & (OW - 1)) == 0 &&
+                      (OH * OW) % 128 == 0 && stride == 1 && pad == 1);
+  const int bco = halo3 ? 32 : ((Co >= 64) ? 64 : ((Co >= 32) ? 32 : 16));
+  const long bkn = halo3 ? (9L * 32) : 128;  // halo covers K per ci-tile
+  const bool full = halo3 ? (Co % 32 == 0)
+                          : (Co % bco == 0 && K % bkn == 0);
+  auto part = full
+      ? torch::empty({splits, Co, K}, x.options().dtype(torch::kFloat32))
+      : torch::zeros({splits, Co, K}, x.options().dtype(torch::kFloat32));
   auto stream = at::hip::getCurrentHIPStream();
   dlb_conv_wrw(x.data_ptr(), dy.data_ptr(), part.data_ptr<float>(), N, IH,
                IW, Ci, OH, OW, Co, (int)R, (int)S, (int)stride, (int)pad,
