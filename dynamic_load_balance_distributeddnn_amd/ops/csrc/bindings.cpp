@@ -115,19 +115,19 @@ static torch::Tensor conv_fwd(torch::Tensor x, torch::Tensor w,
   return y;
 }
 
-// dy [N,Co,OH,OW] channels_last bf16; wt [Ci, R*S*Co] bf16 contiguous
-// (flipped kernel, built host-side).  Returns dx channels_last bf16.
-static torch::Tensor conv_bwd_data(torch::Tensor dy, torch::Tensor wt,
-                                   int64_t IH, int64_t IW, int64_t Ci,
-                                   int64_t R, int64_t S, int64_t stride,
+// dy [N,Co,OH,OW] channels_last bf16; w [Co,Ci,R,S] channels_last bf16
+// (the natural layout — the kernel gathers it directly, no host reshape).
+static torch::Tensor conv_bwd_data(torch::Tensor dy, torch::Tensor w,
+                                   int64_t IH, int64_t IW, int64_t stride,
                                    int64_t pad) {
-  TORCH_CHECK(dy.is_cuda() && is_cl(dy) && wt.is_contiguous());
+  TORCH_CHECK(dy.is_cuda() && is_cl(dy) && is_cl(w));
   const int N = dy.size(0), Co = dy.size(1), OH = dy.size(2), OW = dy.size(3);
+  const int Ci = w.size(1), R = w.size(2), S = w.size(3);
   auto dx = torch::empty({N, Ci, IH, IW},
                          dy.options().memory_format(torch::MemoryFormat::ChannelsLast));
   auto stream = at::hip::getCurrentHIPStream();
-  dlb_conv_bwd_data(dy.data_ptr(), wt.data_ptr(), dx.data_ptr(), N, (int)IH,
-                    (int)IW, (int)Ci, OH, OW, Co, (int)R, (int)S, (int)stride,
+  dlb_conv_bwd_data(dy.data_ptr(), w.data_ptr(), dx.data_ptr(), N, (int)IH,
+                    (int)IW, Ci, OH, OW, Co, R, S, (int)stride,
                     (int)pad, stream.stream());
   return dx;
 }
@@ -141,9 +141,10 @@ static torch::Tensor conv_wrw(torch::Tensor x, torch::Tensor dy, int64_t R,
   const int N = x.size(0), Ci = x.size(1), IH = x.size(2), IW = x.size(3);
   const int Co = dy.size(1), OH = dy.size(2), OW = dy.size(3);
   int splits = dlb_conv_wrw_nsplits(N, OH, OW, Ci, Co, (int)R, (int)S);
+  bool env_override = false;
   if (const char* env = getenv("DLB_WRW_SPLITS")) {
     int v = atoi(env);
-    if (v > 0 && v < splits) splits = v;
+    if (v > 0 && v < splits) { splits = v; env_override = true; }
   }
   const long K = R * S * Ci;
   // every slab element is written by exactly one block when the tiles
@@ -154,8 +155,9 @@ static torch::Tensor conv_wrw(torch::Tensor x, torch::Tensor dy, int64_t R,
                       (OH * OW) % 128 == 0 && stride == 1 && pad == 1);
   const int bco = halo3 ? 32 : ((Co >= 64) ? 64 : ((Co >= 32) ? 32 : 16));
   const long bkn = halo3 ? (9L * 32) : 128;  // halo covers K per ci-tile
-  const bool full = halo3 ? (Co % 32 == 0)
-                          : (Co % bco == 0 && K % bkn == 0);
+  const bool full = !env_override &&
+                    (halo3 ? (Co % 32 == 0)
+                           : (Co % bco == 0 && K % bkn == 0));
   auto part = full
       ? torch::empty({splits, Co, K}, x.options().dtype(torch::kFloat32))
       : torch::zeros({splits, Co, K}, x.options().dtype(torch::kFloat32));

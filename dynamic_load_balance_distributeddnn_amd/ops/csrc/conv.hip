@@ -221,10 +221,11 @@ conv_fwd_kernel(const ConvParams p) {
 // A-loader.
 struct ConvBwdParams {
   const bf16* dy;  // [N, OH, OW, Co]
-  const bf16* wt;  // [Ci][R*S*Co]
+  const bf16* w;   // [Co][R*S*Ci] — the NATURAL channels_last weight
   bf16* dx;        // [N, IH, IW, Ci]
   int N, IH, IW, Ci, OH, OW, Co, R, S, stride, pad;
-  int M, K;  // M = N*IH*IW, K = R*S*Co
+  int M, K;   // M = N*IH*IW, K = R*S*Co (gemm reduction)
+  int Kw;     // R*S*Ci (weight row length)
   ConvGeom g;  // fd_pix: /(IH*IW), fd_w: /IW, fd_c: /Co, fd_s: /S
 };
 
@@ -316,15 +317,22 @@ conv_bwd_data_kernel(const ConvBwdParams p) {
     for (int c = t; c < BCH; c += CONV_BLOCK) {
       const int nrow = c / (BK / 8);
       const int k8 = (c % (BK / 8)) * 8;
-      const int n = n0 + nrow;
+      // B[ci][k=(rs,co)] = w[co][rs*Ci + ci]; per-element gather (the
+      // weight is tiny and L2-hot; this removes the per-conv host-side
+      // transpose kernel the old layout needed)
+      const int n = n0 + nrow;  // ci
       const int k = kt + k8;
-      const bool ok = (n < p.Ci) & (k + 7 < p.K);
-      const long off = ok ? (long)n * p.K + k : 0;
-      bf16x8_t v = mask8(*reinterpret_cast<const bf16x8_t*>(p.wt + off), ok);
-      if (!ok && n < p.Ci) {
-        bf16* vv = reinterpret_cast<bf16*>(&v);
-        for (int j = 0; j < 8 && k + j < p.K; ++j)
-          vv[j] = p.wt[(long)n * p.K + k + j];
+      bf16x8_t v;
+      bf16* vv = reinterpret_cast<bf16*>(&v);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int kk = k + j;
+        unsigned rs, co;
+        p.g.fd_c.divmod(kk < p.K ? kk : 0, rs, co);
+        const bool e = (n < p.Ci) & (kk < p.K);
+        const long off = e ? (long)co * p.Kw + rs * p.Ci + n : 0;
+        bf16 t2 = p.w[off];
+        vv[j] = e ? t2 : (bf16)__float2bfloat16(0.f);
       }
       *reinterpret_cast<bf16x8_t*>(&b_lds[nrow * LDB + k8]) = v;
     }
@@ -572,12 +580,13 @@ extern "C" void dlb_conv_fwd(const void* x, const void* w, void* y,
                        dim3(CONV_BLOCK), 0, stream, p);
 }
 
-extern "C" void dlb_conv_bwd_data(const void* dy, const void* wt, void* dx,
+extern "C" void dlb_conv_bwd_data(const void* dy, const void* w, void* dx,
                                   int N, int IH, int IW, int Ci, int OH,
                                   int OW, int Co, int R, int S, int stride,
                                   int pad, hipStream_t stream) {
-  ConvBwdParams p{(const bf16*)dy, (const bf16*)wt, (bf16*)dx, N, IH, IW, Ci,
-                  OH, OW, Co, R, S, stride, pad, N * IH * IW, R * S * Co, {}};
+  ConvBwdParams p{(const bf16*)dy, (const bf16*)w, (bf16*)dx, N, IH, IW, Ci,
+                  OH, OW, Co, R, S, stride, pad, N * IH * IW, R * S * Co,
+                  R * S * Ci, {}};
   p.g.fd_pix.init(IH * IW);
   p.g.fd_w.init(IW);
   p.g.fd_c.init(Co);

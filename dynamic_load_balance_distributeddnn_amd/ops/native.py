@@ -71,12 +71,7 @@ class _Conv2d(torch.autograd.Function):
 
         dx = None
         if ctx.needs_input_grad[0]:
-            # B image for the data-grad GEMM: [Ci][R*S*Co].  No kernel flip:
-            # the device loader pairs dy(oh=(ih+pad-r)/stride) with the
-            # DIRECT tap (r,s), so B must be w[co][r][s][ci] as-is.
-            w4 = wcl.permute(0, 2, 3, 1)          # [Co,R,S,Ci] view
-            wt = w4.permute(3, 1, 2, 0).reshape(ci, r * s * co).contiguous()
-            dx = ext().conv_bwd_data(dy, wt, x.size(2), x.size(3), ci, r, s,
+            dx = ext().conv_bwd_data(dy, wcl, x.size(2), x.size(3),
                                      stride, padding)
 
         dweight = None
