@@ -27,6 +27,15 @@ def _cbr(cin, cout, k, groups_gn=8):
     )
 
 
+class _MaxPool(nn.Module):
+    def __init__(self, k, stride, pad):
+        super().__init__()
+        self.k, self.stride, self.pad = k, stride, pad
+
+    def forward(self, x):
+        return FD.max_pool2d(x, self.k, self.stride, self.pad)
+
+
 class _Inception(nn.Module):
     def __init__(self, cin, n1x1, n3x3red, n3x3, n5x5red, n5x5, pool_planes):
         super().__init__()
@@ -43,14 +52,15 @@ class _Inception(nn.Module):
             _cbr(n5x5red, n5x5, 3),
             _cbr(n5x5, n5x5, 3),
         )
+        self.branch4_pool = _MaxPool(3, 1, 1)
         self.branch4 = nn.Sequential(
-            nn.MaxPool2d(3, stride=1, padding=1),
             _cbr(cin, pool_planes, 1),
         )
 
     def forward(self, x):
         return torch.cat(
-            [self.branch1(x), self.branch2(x), self.branch3(x), self.branch4(x)],
+            [self.branch1(x), self.branch2(x), self.branch3(x),
+             self.branch4(self.branch4_pool(x))],
             dim=1,
         )
 
@@ -63,7 +73,7 @@ class GoogLeNet(nn.Module):
             _Inception(192, 64, 96, 128, 16, 32, 32),
             _Inception(256, 128, 128, 192, 32, 96, 64),
         )
-        self.pool = nn.MaxPool2d(3, stride=2, padding=1)
+        self.pool = _MaxPool(3, 2, 1)
         self.inc4 = nn.Sequential(
             _Inception(480, 192, 96, 208, 16, 48, 64),
             _Inception(512, 160, 112, 224, 24, 64, 64),

@@ -29,8 +29,8 @@ class MnistNet(nn.Module):
         self.fc2 = Linear(50, 10)
 
     def forward(self, x):
-        x = F.relu(F.max_pool2d(self.conv1(x), 2))
-        x = F.relu(F.max_pool2d(self.drop2d(self.conv2(x)), 2))
+        x = F.relu(FD.max_pool2d(self.conv1(x), 2))
+        x = F.relu(FD.max_pool2d(self.drop2d(self.conv2(x)), 2))
         x = x.flatten(1)
         x = F.dropout(F.relu(self.fc1(x)), training=self.training)
         return FD.log_softmax(self.fc2(x), dim=-1)

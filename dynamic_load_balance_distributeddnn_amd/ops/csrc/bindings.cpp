@@ -376,6 +376,40 @@ static torch::Tensor gconv_wrw(torch::Tensor x, torch::Tensor dy, int64_t GW,
   return dw;
 }
 
+extern "C" void dlb_maxpool_fwd(const void* x, void* y, unsigned char* idx,
+                                int N, int H, int W, int C, int k, int stride,
+                                int pad, hipStream_t stream);
+extern "C" void dlb_maxpool_bwd(const void* dy, const unsigned char* idx,
+                                void* dx, int N, int H, int W, int C, int k,
+                                int stride, int pad, hipStream_t stream);
+
+static std::vector<torch::Tensor> maxpool_fwd(torch::Tensor x, int64_t k,
+                                              int64_t stride, int64_t pad) {
+  TORCH_CHECK(x.is_cuda() && is_cl(x) && x.scalar_type() == torch::kBFloat16);
+  const int N = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  const int OH = (H + 2 * pad - k) / stride + 1;
+  const int OW = (W + 2 * pad - k) / stride + 1;
+  auto y = torch::empty({N, C, OH, OW},
+                        x.options().memory_format(torch::MemoryFormat::ChannelsLast));
+  auto idx = torch::empty({N, OH, OW, C}, x.options().dtype(torch::kByte));
+  dlb_maxpool_fwd(x.data_ptr(), y.data_ptr(), idx.data_ptr<unsigned char>(),
+                  N, H, W, C, (int)k, (int)stride, (int)pad,
+                  at::hip::getCurrentHIPStream().stream());
+  return {y, idx};
+}
+static torch::Tensor maxpool_bwd(torch::Tensor dy, torch::Tensor idx,
+                                 int64_t H, int64_t W, int64_t k,
+                                 int64_t stride, int64_t pad) {
+  auto dyc = dy.contiguous(torch::MemoryFormat::ChannelsLast);
+  const int N = dyc.size(0), C = dyc.size(1);
+  auto dx = torch::empty({N, C, H, W},
+                         dyc.options().memory_format(torch::MemoryFormat::ChannelsLast));
+  dlb_maxpool_bwd(dyc.data_ptr(), idx.data_ptr<unsigned char>(),
+                  dx.data_ptr(), N, (int)H, (int)W, C, (int)k, (int)stride,
+                  (int)pad, at::hip::getCurrentHIPStream().stream());
+  return dx;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv_fwd", &conv_fwd, "Implicit-GEMM NHWC bf16 conv forward");
   m.def("conv_bwd_data", &conv_bwd_data, "NHWC bf16 conv input-grad");
@@ -395,6 +429,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gconv_fwd", &gconv_fwd);
   m.def("gconv_bwd", &gconv_bwd);
   m.def("gconv_wrw", &gconv_wrw);
+  m.def("maxpool_fwd", &maxpool_fwd);
+  m.def("maxpool_bwd", &maxpool_bwd);
   m.def("logsoftmax_fwd", &logsoftmax_fwd);
   m.def("logsoftmax_bwd", &logsoftmax_bwd);
 }

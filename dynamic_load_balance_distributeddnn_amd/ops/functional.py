@@ -21,7 +21,8 @@ from . import available, ext
 
 # ops with a hand-written gfx950 kernel wired in (grown as kernels land)
 NATIVE_OPS: set[str] = {"group_norm_act", "conv2d", "layer_norm",
-                        "causal_attention", "avg_pool2d", "log_softmax"}
+                        "causal_attention", "avg_pool2d", "log_softmax",
+                        "max_pool2d"}
 
 
 def _use_native(name: str, x: torch.Tensor) -> bool:
@@ -128,3 +129,12 @@ def log_softmax(x, dim=-1):
         from . import native
         return native.log_softmax(x)
     return F.log_softmax(x, dim=dim)
+
+
+def max_pool2d(x, k, stride=None, padding=0):
+    stride = stride or k
+    if (_use_native("max_pool2d", x) and x.dtype == torch.bfloat16
+            and x.is_contiguous(memory_format=torch.channels_last)):
+        from . import native
+        return native.max_pool2d(x, k, stride, padding)
+    return F.max_pool2d(x, k, stride=stride, padding=padding)

@@ -251,3 +251,23 @@ def gconv_native_ok(x, weight, stride, padding, groups) -> bool:
     gw = co // groups
     return (r == 3 and s == 3 and padding == 1 and stride in (1, 2)
             and cig == gw and gw in (8, 16) and co == x.shape[1])
+
+
+# ---------------------------------------------------------------- maxpool
+class _MaxPool2d(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, k, stride, pad):
+        y, idx = ext().maxpool_fwd(x, k, stride, pad)
+        ctx.save_for_backward(idx)
+        ctx.mp_args = (k, stride, pad, x.size(2), x.size(3))
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (idx,) = ctx.saved_tensors
+        k, stride, pad, h, w = ctx.mp_args
+        return ext().maxpool_bwd(dy, idx, h, w, k, stride, pad), None, None, None
+
+
+def max_pool2d(x, k, stride, pad):
+    return _MaxPool2d.apply(x, k, stride, pad)

@@ -161,3 +161,30 @@ def test_transformer_model_step_native():
     assert torch.isfinite(loss)
     assert all(torch.isfinite(p.grad).all() for p in model.parameters()
                if p.grad is not None)
+
+
+@needs_gpu
+@pytest.mark.parametrize("cfg", [
+    (4, 192, 32, 3, 1, 1),    # GoogLeNet in-block pool (overlapping)
+    (4, 480, 32, 3, 2, 1),    # GoogLeNet downsample
+    (4, 16, 24, 2, 2, 0),     # MnistNet-style
+])
+def test_maxpool_fwd_bwd(cfg):
+    from dynamic_load_balance_distributeddnn_amd.ops import functional as FD
+
+    N, C, H, k, stride, pad = cfg
+    torch.manual_seed(4)
+    x = torch.randn(N, C, H, H, device="cuda").bfloat16()
+    xr = x.float().requires_grad_()
+    ref = F.max_pool2d(xr, k, stride=stride, padding=pad)
+    dz = torch.randn_like(ref).bfloat16()
+    ref.backward(dz.float())
+
+    xn = _cl(x).requires_grad_()
+    out = FD.max_pool2d(xn, k, stride, pad)
+    torch.testing.assert_close(out.float(), ref.detach(), rtol=0, atol=1e-3)
+    out.backward(_cl(dz))
+    # ties can route grads to a different (equal) argmax in bf16; compare sums
+    torch.testing.assert_close(xn.grad.float().sum(), xr.grad.sum(),
+                               rtol=2e-2, atol=1.0)
+    torch.testing.assert_close(xn.grad.float(), xr.grad, rtol=5e-2, atol=1e-1)
