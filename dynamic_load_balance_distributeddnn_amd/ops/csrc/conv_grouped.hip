@@ -1,13 +1,13 @@
 // Grouped 3x3 convolution (gfx950) — RegNet's only grouped shape
 // (SURVEY.md K3): channels-per-group == group_width (8 or 16), stride
-// 1/2, pad 1, square maps.  Near-depthwise: K per output is 9*GW
-// (72-144 MACs), far below the MFMA regime — direct VALU kernels, NHWC.
+// 1/2, pad 1, square maps.  Near-depthwise (K per output = 9*GW), far
+// below the MFMA regime — direct VALU kernels, NHWC.
 //
-// fwd: thread per (output pixel, co-octet); weights via L2 (<=110 KB).
-// bwd-data: thread per (input pixel, ci-octet), gathering matching taps.
-// wrw: block per (group, tap, m-split); dy/x chunks staged in LDS,
-//      thread per (co_local, ci_local) accumulator, atomics at the end
-//      (contention = split count <= 16 — cheap).
+// v2: v1 re-loaded every weight from global per output pixel (1152
+// scalar loads/thread — measured 3.7 ms/call, 53% of the RegNet step).
+// Now each block owns one channel-octet and stages its weight slice in
+// LDS once; weight reads are wave-uniform LDS broadcasts and x/dy reads
+// are bf16x8 vectors.
 
 #include "common.h"
 
@@ -26,136 +26,183 @@ struct GConvParams {
   int m_per_split;
 };
 
+// fwd: block = (co-octet, 256-pixel tile); weights for the octet in LDS.
 __global__ void __launch_bounds__(GBLOCK)
 gconv_fwd_kernel(const GConvParams p) {
-  const long total = (long)p.N * p.OH * p.OW * (p.C / 8);
-  for (long i = (long)blockIdx.x * GBLOCK + threadIdx.x; i < total;
-       i += (long)gridDim.x * GBLOCK) {
-    const int co8 = (int)(i % (p.C / 8)) * 8;
-    long rest = i / (p.C / 8);
-    const int ow = (int)(rest % p.OW); rest /= p.OW;
-    const int oh = (int)(rest % p.OH);
-    const int n = (int)(rest / p.OH);
-    const int g = co8 / p.GW;           // octet stays in one group (GW>=8)
-    const int ci0 = g * p.GW;
-    float acc[8] = {0, 0, 0, 0, 0, 0, 0, 0};
-    for (int r = 0; r < 3; ++r) {
-      const int ih = oh * p.stride - 1 + r;
-      if (ih < 0 || ih >= p.IH) continue;
-      for (int s = 0; s < 3; ++s) {
-        const int iw = ow * p.stride - 1 + s;
-        if (iw < 0 || iw >= p.IW) continue;
-        const bf16* xp = p.x + (((long)n * p.IH + ih) * p.IW + iw) * p.C + ci0;
-        for (int c = 0; c < p.GW; ++c) {
-          const float xv = __bfloat162float(xp[c]);
+  __shared__ bf16 wlds[8 * 9 * 16];  // [j][tap][ci], GW<=16
+
+  const int co8 = blockIdx.x * 8;
+  const int g = co8 / p.GW;
+  const int ci0 = g * p.GW;
+  const int t = threadIdx.x;
+
+  // stage the octet's weights: 8 rows of [9*GW] contiguous
+  for (int c = t; c < 8 * 9 * p.GW / 8; c += GBLOCK) {
+    const int j = c / (9 * p.GW / 8);
+    const int e8 = (c % (9 * p.GW / 8)) * 8;
+    *reinterpret_cast<bf16x8_t*>(&wlds[(j * 9 * p.GW) + e8]) =
+        *reinterpret_cast<const bf16x8_t*>(p.w + (long)(co8 + j) * 9 * p.GW +
+                                           e8);
+  }
+  __syncthreads();
+
+  const long M = (long)p.N * p.OH * p.OW;
+  const long m = (long)blockIdx.y * GBLOCK + t;
+  if (m >= M) return;
+  const int ow = (int)(m % p.OW);
+  const int oh = (int)((m / p.OW) % p.OH);
+  const int n = (int)(m / ((long)p.OH * p.OW));
+
+  float acc[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  for (int r = 0; r < 3; ++r) {
+    const int ih = oh * p.stride - 1 + r;
+    if (ih < 0 || ih >= p.IH) continue;
+    for (int s = 0; s < 3; ++s) {
+      const int iw = ow * p.stride - 1 + s;
+      if (iw < 0 || iw >= p.IW) continue;
+      const bf16* xp = p.x + (((long)n * p.IH + ih) * p.IW + iw) * p.C + ci0;
+      const bf16* wp = &wlds[(r * 3 + s) * p.GW];
+      for (int c8 = 0; c8 < p.GW; c8 += 8) {
+        bf16x8_t xv = *reinterpret_cast<const bf16x8_t*>(xp + c8);
+        const bf16* xe = reinterpret_cast<const bf16*>(&xv);
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          const float x1 = __bfloat162float(xe[e]);
 #pragma unroll
           for (int j = 0; j < 8; ++j)
-            acc[j] += xv * __bfloat162float(
-                p.w[(((long)(co8 + j) * 3 + r) * 3 + s) * p.GW + c]);
+            acc[j] += x1 * __bfloat162float(wp[j * 9 * p.GW + c8 + e]);
         }
       }
     }
-    bf16x8_t o;
-    bf16* ov = reinterpret_cast<bf16*>(&o);
-#pragma unroll
-    for (int j = 0; j < 8; ++j) ov[j] = __float2bfloat16(acc[j]);
-    *reinterpret_cast<bf16x8_t*>(
-        p.out + (((long)n * p.OH + oh) * p.OW + ow) * p.C + co8) = o;
   }
+  bf16x8_t o;
+  bf16* ov = reinterpret_cast<bf16*>(&o);
+#pragma unroll
+  for (int j = 0; j < 8; ++j) ov[j] = __float2bfloat16(acc[j]);
+  *reinterpret_cast<bf16x8_t*>(
+      p.out + (((long)n * p.OH + oh) * p.OW + ow) * p.C + co8) = o;
 }
 
+// bwd-data: block = (ci-octet, 256-pixel tile); the group's weight
+// columns for this ci-octet staged in LDS: [co_l][tap][8].
 __global__ void __launch_bounds__(GBLOCK)
 gconv_bwd_kernel(const GConvParams p) {
-  const long total = (long)p.N * p.IH * p.IW * (p.C / 8);
-  for (long i = (long)blockIdx.x * GBLOCK + threadIdx.x; i < total;
-       i += (long)gridDim.x * GBLOCK) {
-    const int ci8 = (int)(i % (p.C / 8)) * 8;
-    long rest = i / (p.C / 8);
-    const int iw = (int)(rest % p.IW); rest /= p.IW;
-    const int ih = (int)(rest % p.IH);
-    const int n = (int)(rest / p.IH);
-    const int g = ci8 / p.GW;
-    const int co0 = g * p.GW;
-    const int cl0 = ci8 - g * p.GW;     // ci position inside group
-    float acc[8] = {0, 0, 0, 0, 0, 0, 0, 0};
-    for (int r = 0; r < 3; ++r) {
-      const int ohn = ih + 1 - r;
-      if (ohn < 0 || ohn % p.stride) continue;
-      const int oh = ohn / p.stride;
-      if (oh >= p.OH) continue;
-      for (int s = 0; s < 3; ++s) {
-        const int own = iw + 1 - s;
-        if (own < 0 || own % p.stride) continue;
-        const int ow = own / p.stride;
-        if (ow >= p.OW) continue;
-        const bf16* dp =
-            p.dy + (((long)n * p.OH + oh) * p.OW + ow) * p.C + co0;
-        for (int c = 0; c < p.GW; ++c) {  // c = co within group
-          const float dv = __bfloat162float(dp[c]);
+  __shared__ bf16 wlds[16 * 9 * 8];  // [co_l][tap][jj]
+
+  const int ci8 = blockIdx.x * 8;
+  const int g = ci8 / p.GW;
+  const int co0 = g * p.GW;
+  const int cl0 = ci8 - g * p.GW;
+  const int t = threadIdx.x;
+
+  for (int c = t; c < p.GW * 9 * 8; c += GBLOCK) {
+    const int jj = c % 8;
+    const int tap = (c / 8) % 9;
+    const int col = c / 72;
+    wlds[c] = p.w[((long)(co0 + col) * 9 + tap) * p.GW + cl0 + jj];
+  }
+  __syncthreads();
+
+  const long M = (long)p.N * p.IH * p.IW;
+  const long m = (long)blockIdx.y * GBLOCK + t;
+  if (m >= M) return;
+  const int iw = (int)(m % p.IW);
+  const int ih = (int)((m / p.IW) % p.IH);
+  const int n = (int)(m / ((long)p.IH * p.IW));
+
+  float acc[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  for (int r = 0; r < 3; ++r) {
+    const int ohn = ih + 1 - r;
+    if (ohn < 0 || ohn % p.stride) continue;
+    const int oh = ohn / p.stride;
+    if (oh >= p.OH) continue;
+    for (int s = 0; s < 3; ++s) {
+      const int own = iw + 1 - s;
+      if (own < 0 || own % p.stride) continue;
+      const int ow = own / p.stride;
+      if (ow >= p.OW) continue;
+      const bf16* dp = p.dy + (((long)n * p.OH + oh) * p.OW + ow) * p.C + co0;
+      const int tap = r * 3 + s;
+      for (int c8 = 0; c8 < p.GW; c8 += 8) {
+        bf16x8_t dv = *reinterpret_cast<const bf16x8_t*>(dp + c8);
+        const bf16* de = reinterpret_cast<const bf16*>(&dv);
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          const float d1 = __bfloat162float(de[e]);
 #pragma unroll
           for (int j = 0; j < 8; ++j)
-            acc[j] += dv * __bfloat162float(
-                p.w[(((long)(co0 + c) * 3 + r) * 3 + s) * p.GW + cl0 + j]);
+            acc[j] += d1 * __bfloat162float(wlds[((c8 + e) * 9 + tap) * 8 + j]);
         }
       }
     }
-    bf16x8_t o;
-    bf16* ov = reinterpret_cast<bf16*>(&o);
-#pragma unroll
-    for (int j = 0; j < 8; ++j) ov[j] = __float2bfloat16(acc[j]);
-    *reinterpret_cast<bf16x8_t*>(
-        p.out + (((long)n * p.IH + ih) * p.IW + iw) * p.C + ci8) = o;
   }
+  bf16x8_t o;
+  bf16* ov = reinterpret_cast<bf16*>(&o);
+#pragma unroll
+  for (int j = 0; j < 8; ++j) ov[j] = __float2bfloat16(acc[j]);
+  *reinterpret_cast<bf16x8_t*>(
+      p.out + (((long)n * p.IH + ih) * p.IW + iw) * p.C + ci8) = o;
 }
 
-// wrw: grid (G, 9 taps, splits); thread (co_l, ci_l) pairs (GW*GW <= 256)
+// wrw: grid (G, 9 taps, splits); 128-m stages with vector loads; thread
+// = (co_l, ci_l, m-replica); LDS tree-reduce over replicas at the end.
 __global__ void __launch_bounds__(GBLOCK)
 gconv_wrw_kernel(const GConvParams p) {
-  __shared__ bf16 dy_s[32 * 16];  // [mm][co_l]
-  __shared__ bf16 x_s[32 * 16];   // [mm][ci_l]
+  __shared__ bf16 dy_s[128 * 16];  // [mm][co_l]
+  __shared__ bf16 x_s[128 * 16];   // [mm][ci_l]
+  __shared__ float red[GBLOCK];
 
   const int g = blockIdx.x;
   const int tap = blockIdx.y;
   const int r = tap / 3, s = tap % 3;
   const int co0 = g * p.GW, ci0 = g * p.GW;
   const int t = threadIdx.x;
-  const int col = t % p.GW;            // ci_l
-  const int row = t / p.GW;            // co_l
-  const bool active = row < p.GW;
+  const int pairs = p.GW * p.GW;          // 64 or 256
+  const int rep_n = GBLOCK / pairs;       // 4 or 1
+  const int pair = t % pairs;
+  const int rep = t / pairs;
+  const int col = pair % p.GW;            // ci_l
+  const int row = pair / p.GW;            // co_l
   const int M = p.N * p.OH * p.OW;
   const int mstart = blockIdx.z * p.m_per_split;
   const int mend = min(M, mstart + p.m_per_split);
 
   float acc = 0.f;
-  for (int mt = mstart; mt < mend; mt += 32) {
-    // stage dy[mm][co_l] and x(tap)[mm][ci_l]
-    for (int c = t; c < 32 * p.GW; c += GBLOCK) {
-      const int mm = c / p.GW, cl = c % p.GW;
+  for (int mt = mstart; mt < mend; mt += 128) {
+    for (int c = t; c < 128 * (p.GW / 8); c += GBLOCK) {
+      const int mm = c / (p.GW / 8);
+      const int c8 = (c % (p.GW / 8)) * 8;
       const int m = mt + mm;
-      bf16 dv = __float2bfloat16(0.f), xv = __float2bfloat16(0.f);
+      bf16x8_t dv = {0, 0, 0, 0, 0, 0, 0, 0};
+      bf16x8_t xv = {0, 0, 0, 0, 0, 0, 0, 0};
       if (m < mend) {
         const int n = m / (p.OH * p.OW);
         const int rem = m % (p.OH * p.OW);
         const int oh = rem / p.OW, ow = rem % p.OW;
-        dv = p.dy[(((long)n * p.OH + oh) * p.OW + ow) * p.C + co0 + cl];
+        dv = *reinterpret_cast<const bf16x8_t*>(
+            p.dy + (((long)n * p.OH + oh) * p.OW + ow) * p.C + co0 + c8);
         const int ih = oh * p.stride - 1 + r;
         const int iw = ow * p.stride - 1 + s;
         if (ih >= 0 && ih < p.IH && iw >= 0 && iw < p.IW)
-          xv = p.x[(((long)n * p.IH + ih) * p.IW + iw) * p.C + ci0 + cl];
+          xv = *reinterpret_cast<const bf16x8_t*>(
+              p.x + (((long)n * p.IH + ih) * p.IW + iw) * p.C + ci0 + c8);
       }
-      dy_s[mm * 16 + cl] = dv;
-      x_s[mm * 16 + cl] = xv;
+      *reinterpret_cast<bf16x8_t*>(&dy_s[mm * 16 + c8]) = dv;
+      *reinterpret_cast<bf16x8_t*>(&x_s[mm * 16 + c8]) = xv;
     }
     __syncthreads();
-    if (active) {
-      for (int mm = 0; mm < 32 && mt + mm < mend; ++mm)
-        acc += __bfloat162float(dy_s[mm * 16 + row]) *
-               __bfloat162float(x_s[mm * 16 + col]);
-    }
+    const int mlim = min(128, mend - mt);
+    for (int mm = rep; mm < mlim; mm += rep_n)
+      acc += __bfloat162float(dy_s[mm * 16 + row]) *
+             __bfloat162float(x_s[mm * 16 + col]);
     __syncthreads();
   }
-  if (active)
+  red[t] = acc;
+  __syncthreads();
+  if (rep == 0) {
+    for (int rr = 1; rr < rep_n; ++rr) acc += red[pair + rr * pairs];
     atomicAdd(&p.dw[(((long)(co0 + row) * 3 + r) * 3 + s) * p.GW + col], acc);
+  }
 }
 
 extern "C" void dlb_gconv_fwd(const void* x, const void* w, void* y, int N,
@@ -165,9 +212,8 @@ extern "C" void dlb_gconv_fwd(const void* x, const void* w, void* y, int N,
   const int OW = (IW + 2 - 3) / stride + 1;
   GConvParams p{(const bf16*)x, (const bf16*)w, nullptr, (bf16*)y, nullptr,
                 N, IH, IW, C, OH, OW, GW, stride, 0};
-  long total = (long)N * OH * OW * (C / 8);
-  int grid = (int)std::min<long>(cdiv(total, GBLOCK), 8192);
-  hipLaunchKernelGGL(gconv_fwd_kernel, dim3(grid), dim3(GBLOCK), 0, stream, p);
+  dim3 grid(C / 8, cdiv((long)N * OH * OW, GBLOCK));
+  hipLaunchKernelGGL(gconv_fwd_kernel, grid, dim3(GBLOCK), 0, stream, p);
 }
 
 extern "C" void dlb_gconv_bwd(const void* dy, const void* w, void* dx, int N,
@@ -177,9 +223,8 @@ extern "C" void dlb_gconv_bwd(const void* dy, const void* w, void* dx, int N,
   const int OW = (IW + 2 - 3) / stride + 1;
   GConvParams p{nullptr, (const bf16*)w, (const bf16*)dy, (bf16*)dx, nullptr,
                 N, IH, IW, C, OH, OW, GW, stride, 0};
-  long total = (long)N * IH * IW * (C / 8);
-  int grid = (int)std::min<long>(cdiv(total, GBLOCK), 8192);
-  hipLaunchKernelGGL(gconv_bwd_kernel, dim3(grid), dim3(GBLOCK), 0, stream, p);
+  dim3 grid(C / 8, cdiv((long)N * IH * IW, GBLOCK));
+  hipLaunchKernelGGL(gconv_bwd_kernel, grid, dim3(GBLOCK), 0, stream, p);
 }
 
 extern "C" void dlb_gconv_wrw(const void* x, const void* dy, float* dw, int N,
@@ -190,8 +235,10 @@ extern "C" void dlb_gconv_wrw(const void* x, const void* dy, float* dw, int N,
   GConvParams p{(const bf16*)x, nullptr, (const bf16*)dy, nullptr, dw,
                 N, IH, IW, C, OH, OW, GW, stride, 0};
   const int M = N * OH * OW;
-  int splits = std::min(16, std::max(1, M / (32 * 32)));
-  p.m_per_split = cdiv(cdiv(M, splits), 32) * 32;
+  const long tiles = (long)(C / GW) * 9;
+  int splits = (int)std::min<long>(std::max<long>(1, 2048 / tiles),
+                                   std::max<long>(1, M / (4 * 128)));
+  p.m_per_split = cdiv(cdiv(M, splits), 128) * 128;
   splits = cdiv(M, p.m_per_split);
   dim3 grid(C / GW, 9, splits);
   hipLaunchKernelGGL(gconv_wrw_kernel, grid, dim3(GBLOCK), 0, stream, p);
