@@ -83,13 +83,15 @@ class Trainer:
 
         # datasets built once; partitioned fresh each epoch
         if self.is_lm:
-            self.train_tokens = D.make_lm_tokens(train=True, seed=seed)
-            self.val_tokens = D.make_lm_tokens(train=False, seed=seed)
+            self.train_tokens = D.load_lm_tokens(train=True, seed=seed)
+            self.val_tokens = D.load_lm_tokens(train=False, seed=seed)
             self.bptt = LM_CONFIG["bptt"]
             self.ntokens = LM_CONFIG["ntokens"]
         else:
-            self.train_data = D.make_cv_dataset(args.dataset, train=True, seed=seed)
-            self.val_data = D.make_cv_dataset(args.dataset, train=False, seed=seed)
+            self.train_data = D.load_cv_dataset(args.dataset, train=True,
+                                                seed=seed)
+            self.val_data = D.load_cv_dataset(args.dataset, train=False,
+                                              seed=seed)
 
     # ------------------------------------------------------------------
     def _sync_initial_weights(self) -> None:

@@ -47,9 +47,11 @@ def init_process_group(args, rank: int, world_size: int,
     # RCCL requires one rank per GPU; the reference's straggler trick maps
     # several ranks onto one GPU (-gpu 0,0,0,1 — README.md:23-28), which
     # must then rendezvous over gloo (the reference's only backend).
-    oversubscribed = (isinstance(args.gpu, list)
-                      and len(set(args.gpu[:world_size])) < min(world_size,
-                                                                len(args.gpu)))
+    if isinstance(args.gpu, list):
+        distinct = len({args.gpu[r % len(args.gpu)] for r in range(world_size)})
+    else:
+        distinct = 1  # '-gpu 0' with ws>1: every rank on one GPU
+    oversubscribed = world_size > 1 and distinct < world_size
     backend = "gloo" if (device.type == "cpu" or oversubscribed) else "nccl"
     if device.type == "cuda":
         torch.cuda.set_device(device)
