@@ -39,6 +39,8 @@ def get_args():
     p.add_argument("--global-batch", type=int, default=512)
     p.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"])
     p.add_argument("--device", default=None, help="override (e.g. cpu for debug)")
+    p.add_argument("--no-graphs", action="store_true",
+                   help="disable hipGraph capture of the training step")
     p.add_argument("--no-channels-last", action="store_true",
                    help="disable NHWC layout (NHWC is the GPU default; the "
                         "gfx950 kernels are NHWC-native)")
@@ -142,13 +144,46 @@ def main():
     for _ in range(args.warmup):
         step()
 
+    # hipGraph capture: the zoo's steps are hundreds of small kernels, so
+    # replaying one captured graph removes launch overhead (decisive at
+    # small per-rank batches in the strong-scaling sweep).  Collectives
+    # and dynamic shapes fall back to eager if capture fails.
+    run_step = step
+    if device.type == "cuda" and not args.no_graphs:
+        try:
+            side = torch.cuda.Stream()
+            side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(side):
+                for _ in range(3):
+                    step()
+            torch.cuda.current_stream().wait_stream(side)
+            graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(graph):
+                step()
+            run_step = graph.replay
+        except Exception as e:  # capture unsupported (e.g. RCCL op) -> eager
+            if rank == 0:
+                print(f"# hipGraph capture unavailable ({type(e).__name__}); "
+                      "running eager", flush=True)
+            run_step = step
+        if world > 1:
+            # use the graph only if EVERY rank captured (divergent
+            # graph/eager ranks would desync the collectives)
+            ok = torch.tensor([1.0 if run_step is not step else 0.0],
+                              device=device)
+            dist.all_reduce(ok, op=dist.ReduceOp.MIN)
+            if ok.item() < 1:
+                run_step = step
+        if rank == 0 and run_step is not step:
+            print("# hipGraph capture active", flush=True)
+
     if world > 1:
         dist.barrier()
     if device.type == "cuda":
         torch.cuda.synchronize()
     t0 = time.perf_counter()
     for _ in range(args.steps):
-        step()
+        run_step()
     if device.type == "cuda":
         torch.cuda.synchronize()
     if world > 1:
