@@ -20,13 +20,13 @@ static void sgd_momentum(torch::Tensor p, torch::Tensor g, torch::Tensor m,
 
 extern "C" void dlb_gn_fwd(const void* x, void* y, const float* gamma,
                            const float* beta, float* mean, float* rstd,
-                           int N, int HW, int C, int G, float eps, int relu,
-                           hipStream_t stream);
+                           float* scratch, int N, int HW, int C, int G,
+                           float eps, int relu, hipStream_t stream);
 extern "C" void dlb_gn_bwd(const void* x, const void* dz, void* dx,
                            const float* gamma, const float* beta,
                            const float* mean, const float* rstd, float* dgamma,
-                           float* dbeta, int N, int HW, int C, int G, int relu,
-                           hipStream_t stream);
+                           float* dbeta, float* scratch, int N, int HW, int C,
+                           int G, int relu, hipStream_t stream);
 
 // x: [N, HW, C] bf16 contiguous (an NHWC view of a channels_last NCHW
 // tensor).  Returns (y, mean, rstd).
@@ -43,10 +43,17 @@ static std::vector<torch::Tensor> gn_fwd(torch::Tensor x, torch::Tensor gamma,
   auto mean = torch::empty({N, groups}, x.options().dtype(torch::kFloat32));
   auto rstd = torch::empty_like(mean);
   auto stream = at::hip::getCurrentHIPStream();
+  float* scratch = nullptr;
+  torch::Tensor scratch_t;
+  if (N < 192 && HW > 1) {  // small-batch path: split HW across blocks
+    scratch_t = torch::zeros({N, groups, 2},
+                             x.options().dtype(torch::kFloat32));
+    scratch = scratch_t.data_ptr<float>();
+  }
   dlb_gn_fwd(x.data_ptr(), y.data_ptr(), gamma.data_ptr<float>(),
              beta.data_ptr<float>(), mean.data_ptr<float>(),
-             rstd.data_ptr<float>(), N, HW, C, (int)groups, (float)eps,
-             relu ? 1 : 0, stream.stream());
+             rstd.data_ptr<float>(), scratch, N, HW, C, (int)groups,
+             (float)eps, relu ? 1 : 0, stream.stream());
   return {y, mean, rstd};
 }
 
@@ -63,11 +70,18 @@ static std::vector<torch::Tensor> gn_bwd(torch::Tensor x, torch::Tensor dz,
   auto dgamma = torch::zeros({C}, x.options().dtype(torch::kFloat32));
   auto dbeta = torch::zeros_like(dgamma);
   auto stream = at::hip::getCurrentHIPStream();
+  float* scratch = nullptr;
+  torch::Tensor scratch_t;
+  if (N < 192 && HW > 1) {
+    scratch_t = torch::zeros({N, groups, 2},
+                             x.options().dtype(torch::kFloat32));
+    scratch = scratch_t.data_ptr<float>();
+  }
   dlb_gn_bwd(x.data_ptr(), dz.data_ptr(), dx.data_ptr(),
              gamma.data_ptr<float>(), beta.data_ptr<float>(),
              mean.data_ptr<float>(), rstd.data_ptr<float>(),
-             dgamma.data_ptr<float>(), dbeta.data_ptr<float>(), N, HW, C,
-             (int)groups, relu ? 1 : 0, stream.stream());
+             dgamma.data_ptr<float>(), dbeta.data_ptr<float>(), scratch, N,
+             HW, C, (int)groups, relu ? 1 : 0, stream.stream());
   return {dx, dgamma, dbeta};
 }
 

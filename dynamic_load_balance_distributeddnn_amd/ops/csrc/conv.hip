@@ -617,7 +617,7 @@ extern "C" int dlb_conv_wrw_nsplits(int N, int OH, int OW, int Ci, int Co,
   const long tiles = (long)cdiv(Co, (Co >= 64) ? 64 : ((Co >= 32) ? 32 : 16))
                      * cdiv(K, 128);
   int splits = (int)std::min<long>(std::max<long>(1, 1024 / tiles),
-                                   std::max<long>(1, M / (16 * 64)));
+                                   std::max<long>(1, M / (4 * 64)));
   int mps = cdiv(cdiv(M, splits), 64) * 64;
   return cdiv(M, mps);
 }

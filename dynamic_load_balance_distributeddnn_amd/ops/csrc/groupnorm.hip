@@ -241,11 +241,284 @@ gn_bwd_kernel(const bf16* __restrict__ x, const bf16* __restrict__ dz,
   }
 }
 
+
+// ---------------- small-batch path (N blocks underfill 256 CUs) ----------
+// grid (N, SLICES): pass A accumulates partial group sums into a global
+// fp32 buffer [N][G][2] with atomics; pass B recomputes stats from it
+// in-block and normalizes its HW slice.  Launched when N < 192.
+
+extern "C" __global__ void __launch_bounds__(GN_BLOCK)
+gn_stats_part_kernel(const bf16* __restrict__ x, float* __restrict__ sums,
+                     const int HW, const int C, const int G,
+                     const int slices) {
+  const int n = blockIdx.x;
+  const int sl = blockIdx.y;
+  const int hw0 = (int)(((long)HW * sl) / slices);
+  const int hw1 = (int)(((long)HW * (sl + 1)) / slices);
+  const int TC = C >> 3;
+  const int TCe = TC < GN_BLOCK ? TC : GN_BLOCK;
+  const int TP = GN_BLOCK / TCe;
+  const int t = threadIdx.x;
+  const int tc = t % TCe, tp = t / TCe;
+  const int Cg = C / G;
+  const bool active = t < TCe * TP;
+
+  __shared__ float s_sum[GN_MAXG];
+  __shared__ float s_ssq[GN_MAXG];
+  for (int g = t; g < G; g += GN_BLOCK) { s_sum[g] = 0.f; s_ssq[g] = 0.f; }
+  __syncthreads();
+
+  const bf16* xb = x + (long)n * HW * C;
+  if (active) {
+    for (int oct = tc; oct < TC; oct += TCe) {
+      const int c0 = oct << 3;
+      float sacc[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+      float ss[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+      for (int p2 = hw0 + tp; p2 < hw1; p2 += TP) {
+        Bf16x8 chunk = *reinterpret_cast<const Bf16x8*>(xb + (long)p2 * C + c0);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float v = bf2f(chunk.v[j]);
+          sacc[j] += v;
+          ss[j] += v * v;
+        }
+      }
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        int g = (c0 + j) / Cg;
+        atomicAdd(&s_sum[g], sacc[j]);
+        atomicAdd(&s_ssq[g], ss[j]);
+      }
+    }
+  }
+  __syncthreads();
+  for (int g = t; g < G; g += GN_BLOCK) {
+    atomicAdd(&sums[((long)n * G + g) * 2 + 0], s_sum[g]);
+    atomicAdd(&sums[((long)n * G + g) * 2 + 1], s_ssq[g]);
+  }
+}
+
+extern "C" __global__ void __launch_bounds__(GN_BLOCK)
+gn_apply_kernel(const bf16* __restrict__ x, bf16* __restrict__ y,
+                const float* __restrict__ gamma,
+                const float* __restrict__ beta,
+                const float* __restrict__ sums, float* __restrict__ mean_out,
+                float* __restrict__ rstd_out, const int HW, const int C,
+                const int G, const float eps, const int relu,
+                const int slices) {
+  const int n = blockIdx.x;
+  const int sl = blockIdx.y;
+  const int hw0 = (int)(((long)HW * sl) / slices);
+  const int hw1 = (int)(((long)HW * (sl + 1)) / slices);
+  const int TC = C >> 3;
+  const int TCe = TC < GN_BLOCK ? TC : GN_BLOCK;
+  const int TP = GN_BLOCK / TCe;
+  const int t = threadIdx.x;
+  const int tc = t % TCe, tp = t / TCe;
+  const int Cg = C / G;
+  const bool active = t < TCe * TP;
+
+  __shared__ float s_mean[GN_MAXG];
+  __shared__ float s_rstd[GN_MAXG];
+  const float inv_m = 1.0f / ((float)HW * Cg);
+  for (int g = t; g < G; g += GN_BLOCK) {
+    float mu = sums[((long)n * G + g) * 2 + 0] * inv_m;
+    float var = sums[((long)n * G + g) * 2 + 1] * inv_m - mu * mu;
+    float r = rsqrtf(var + eps);
+    s_mean[g] = mu;
+    s_rstd[g] = r;
+    if (sl == 0) {
+      mean_out[(long)n * G + g] = mu;
+      rstd_out[(long)n * G + g] = r;
+    }
+  }
+  __syncthreads();
+  if (!active) return;
+
+  const bf16* xb = x + (long)n * HW * C;
+  bf16* yb = y + (long)n * HW * C;
+  for (int oct = tc; oct < TC; oct += TCe) {
+    const int c0 = oct << 3;
+    float ga[8], be[8], mu[8], rs[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int c = c0 + j, g = c / Cg;
+      ga[j] = gamma[c];
+      be[j] = beta[c];
+      mu[j] = s_mean[g];
+      rs[j] = s_rstd[g];
+    }
+    for (int p2 = hw0 + tp; p2 < hw1; p2 += TP) {
+      const long off = (long)p2 * C + c0;
+      Bf16x8 chunk = *reinterpret_cast<const Bf16x8*>(xb + off);
+      Bf16x8 out;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float v = (bf2f(chunk.v[j]) - mu[j]) * rs[j] * ga[j] + be[j];
+        if (relu) v = fmaxf(v, 0.f);
+        out.v[j] = f2bf(v);
+      }
+      *reinterpret_cast<Bf16x8*>(yb + off) = out;
+    }
+  }
+}
+
+// backward small-batch: pass A accumulates s1/s2 (+dgamma/dbeta) into
+// global buffers; pass B computes dx per HW slice.
+extern "C" __global__ void __launch_bounds__(GN_BLOCK)
+gn_bwd_part_kernel(const bf16* __restrict__ x, const bf16* __restrict__ dz,
+                   const float* __restrict__ gamma,
+                   const float* __restrict__ beta,
+                   const float* __restrict__ mean_in,
+                   const float* __restrict__ rstd_in,
+                   float* __restrict__ s12, float* __restrict__ dgamma,
+                   float* __restrict__ dbeta, const int HW, const int C,
+                   const int G, const int relu, const int slices) {
+  const int n = blockIdx.x;
+  const int sl = blockIdx.y;
+  const int hw0 = (int)(((long)HW * sl) / slices);
+  const int hw1 = (int)(((long)HW * (sl + 1)) / slices);
+  const int TC = C >> 3;
+  const int TCe = TC < GN_BLOCK ? TC : GN_BLOCK;
+  const int TP = GN_BLOCK / TCe;
+  const int t = threadIdx.x;
+  const int tc = t % TCe, tp = t / TCe;
+  const int Cg = C / G;
+  const bool active = t < TCe * TP;
+
+  __shared__ float s_s1[GN_MAXG];
+  __shared__ float s_s2[GN_MAXG];
+  extern __shared__ float s_dgb[];
+  for (int g = t; g < G; g += GN_BLOCK) { s_s1[g] = 0.f; s_s2[g] = 0.f; }
+  for (int c = t; c < 2 * C; c += GN_BLOCK) s_dgb[c] = 0.f;
+  __syncthreads();
+
+  const bf16* xb = x + (long)n * HW * C;
+  const bf16* db = dz + (long)n * HW * C;
+  if (active) {
+    for (int oct = tc; oct < TC; oct += TCe) {
+      const int c0 = oct << 3;
+      float ga[8], be[8], mu[8], rs[8];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        int c = c0 + j, g = c / Cg;
+        ga[j] = gamma[c];
+        be[j] = beta[c];
+        mu[j] = mean_in[(long)n * G + g];
+        rs[j] = rstd_in[(long)n * G + g];
+      }
+      float a1[8] = {0}, a2[8] = {0}, adg[8] = {0}, adb[8] = {0};
+      for (int p2 = hw0 + tp; p2 < hw1; p2 += TP) {
+        const long off = (long)p2 * C + c0;
+        Bf16x8 xc = *reinterpret_cast<const Bf16x8*>(xb + off);
+        Bf16x8 dc = *reinterpret_cast<const Bf16x8*>(db + off);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float xhat = (bf2f(xc.v[j]) - mu[j]) * rs[j];
+          float dy = bf2f(dc.v[j]);
+          if (relu) {
+            float yv = xhat * ga[j] + be[j];
+            dy = yv > 0.f ? dy : 0.f;
+          }
+          a1[j] += ga[j] * dy;
+          a2[j] += ga[j] * dy * xhat;
+          adg[j] += dy * xhat;
+          adb[j] += dy;
+        }
+      }
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        int g = (c0 + j) / Cg;
+        atomicAdd(&s_s1[g], a1[j]);
+        atomicAdd(&s_s2[g], a2[j]);
+        atomicAdd(&s_dgb[c0 + j], adg[j]);
+        atomicAdd(&s_dgb[C + c0 + j], adb[j]);
+      }
+    }
+  }
+  __syncthreads();
+  for (int g = t; g < G; g += GN_BLOCK) {
+    atomicAdd(&s12[((long)n * G + g) * 2 + 0], s_s1[g]);
+    atomicAdd(&s12[((long)n * G + g) * 2 + 1], s_s2[g]);
+  }
+  for (int c = t; c < C; c += GN_BLOCK) {
+    atomicAdd(&dgamma[c], s_dgb[c]);
+    atomicAdd(&dbeta[c], s_dgb[C + c]);
+  }
+}
+
+extern "C" __global__ void __launch_bounds__(GN_BLOCK)
+gn_bwd_apply_kernel(const bf16* __restrict__ x, const bf16* __restrict__ dz,
+                    bf16* __restrict__ dx, const float* __restrict__ gamma,
+                    const float* __restrict__ beta,
+                    const float* __restrict__ mean_in,
+                    const float* __restrict__ rstd_in,
+                    const float* __restrict__ s12, const int HW, const int C,
+                    const int G, const int relu, const int slices) {
+  const int n = blockIdx.x;
+  const int sl = blockIdx.y;
+  const int hw0 = (int)(((long)HW * sl) / slices);
+  const int hw1 = (int)(((long)HW * (sl + 1)) / slices);
+  const int TC = C >> 3;
+  const int TCe = TC < GN_BLOCK ? TC : GN_BLOCK;
+  const int TP = GN_BLOCK / TCe;
+  const int t = threadIdx.x;
+  const int tc = t % TCe, tp = t / TCe;
+  const int Cg = C / G;
+  if (t >= TCe * TP) return;
+
+  const float inv_m = 1.0f / ((float)HW * Cg);
+  const bf16* xb = x + (long)n * HW * C;
+  const bf16* db = dz + (long)n * HW * C;
+  bf16* dxb = dx + (long)n * HW * C;
+  for (int oct = tc; oct < TC; oct += TCe) {
+    const int c0 = oct << 3;
+    float ga[8], be[8], mu[8], rs[8], k1[8], k2[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int c = c0 + j, g = c / Cg;
+      ga[j] = gamma[c];
+      be[j] = beta[c];
+      mu[j] = mean_in[(long)n * G + g];
+      rs[j] = rstd_in[(long)n * G + g];
+      k1[j] = s12[((long)n * G + g) * 2 + 0] * inv_m;
+      k2[j] = s12[((long)n * G + g) * 2 + 1] * inv_m;
+    }
+    for (int p2 = hw0 + tp; p2 < hw1; p2 += TP) {
+      const long off = (long)p2 * C + c0;
+      Bf16x8 xc = *reinterpret_cast<const Bf16x8*>(xb + off);
+      Bf16x8 dc = *reinterpret_cast<const Bf16x8*>(db + off);
+      Bf16x8 out;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float xhat = (bf2f(xc.v[j]) - mu[j]) * rs[j];
+        float dy = bf2f(dc.v[j]);
+        if (relu) {
+          float yv = xhat * ga[j] + be[j];
+          dy = yv > 0.f ? dy : 0.f;
+        }
+        out.v[j] = f2bf(rs[j] * (ga[j] * dy - (k1[j] + xhat * k2[j])));
+      }
+      *reinterpret_cast<Bf16x8*>(dxb + off) = out;
+    }
+  }
+}
 // ---------------------------------------------------------------- launchers
 extern "C" void dlb_gn_fwd(const void* x, void* y, const float* gamma,
                            const float* beta, float* mean, float* rstd,
+                           float* scratch /* [N][G][2] zeroed, or null */,
                            int N, int HW, int C, int G, float eps, int relu,
                            hipStream_t stream) {
+  const int slices = scratch ? (int)std::min<long>((383 + N) / N, HW) : 1;
+  if (slices > 1) {
+    dim3 grid(N, slices);
+    hipLaunchKernelGGL(gn_stats_part_kernel, grid, dim3(GN_BLOCK), 0, stream,
+                       (const bf16*)x, scratch, HW, C, G, slices);
+    hipLaunchKernelGGL(gn_apply_kernel, grid, dim3(GN_BLOCK), 0, stream,
+                       (const bf16*)x, (bf16*)y, gamma, beta, scratch, mean,
+                       rstd, HW, C, G, eps, relu, slices);
+    return;
+  }
   hipLaunchKernelGGL(gn_fwd_kernel, dim3(N), dim3(GN_BLOCK), 0, stream,
                      (const bf16*)x, (bf16*)y, gamma, beta, mean, rstd, HW, C,
                      G, eps, relu);
@@ -254,9 +527,23 @@ extern "C" void dlb_gn_fwd(const void* x, void* y, const float* gamma,
 extern "C" void dlb_gn_bwd(const void* x, const void* dz, void* dx,
                            const float* gamma, const float* beta,
                            const float* mean, const float* rstd, float* dgamma,
-                           float* dbeta, int N, int HW, int C, int G, int relu,
+                           float* dbeta,
+                           float* scratch /* [N][G][2] zeroed, or null */,
+                           int N, int HW, int C, int G, int relu,
                            hipStream_t stream) {
   size_t shmem = 2 * (size_t)C * sizeof(float);
+  const int slices = scratch ? (int)std::min<long>((383 + N) / N, HW) : 1;
+  if (slices > 1) {
+    dim3 grid(N, slices);
+    hipLaunchKernelGGL(gn_bwd_part_kernel, grid, dim3(GN_BLOCK), shmem,
+                       stream, (const bf16*)x, (const bf16*)dz, gamma, beta,
+                       mean, rstd, scratch, dgamma, dbeta, HW, C, G, relu,
+                       slices);
+    hipLaunchKernelGGL(gn_bwd_apply_kernel, grid, dim3(GN_BLOCK), 0, stream,
+                       (const bf16*)x, (const bf16*)dz, (bf16*)dx, gamma,
+                       beta, mean, rstd, scratch, HW, C, G, relu, slices);
+    return;
+  }
   hipLaunchKernelGGL(gn_bwd_kernel, dim3(N), dim3(GN_BLOCK), shmem, stream,
                      (const bf16*)x, (const bf16*)dz, (bf16*)dx, gamma, beta,
                      mean, rstd, dgamma, dbeta, HW, C, G, relu);
