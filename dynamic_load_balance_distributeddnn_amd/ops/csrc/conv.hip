@@ -396,6 +396,32 @@ struct WrwParams {
   ConvGeom g;
 };
 
+#define LDS3 __attribute__((address_space(3)))
+typedef __attribute__((__vector_size__(4 * sizeof(__bf16)))) __bf16 trvec;
+
+// ds_read_b64_tr_b16 fragment read from a NATURAL row-major [m][cols]
+// LDS image (row stride LROW elements): returns the 8 m-elements of
+// column (colbase + lane&15) for m in [mbase + (lane>>4)*8, +8).
+// Lane map verified on hardware by tools/mfma_probe/tr_probe.hip:
+// result-lane i gets element (i&3) of the row addressed by lane
+// ((i>>2) + 4k) of its 16-lane group, so address-lane j points at
+// row (mbase + q*8 + half*4 + (j>>2)), col (colbase + 4*(j&3)).
+template <int LROW>
+__device__ inline bf16x8_t tr_frag(const bf16* img, int mbase, int colbase,
+                                   int lane) {
+  const int j15 = lane & 15, q = lane >> 4;
+  const int row = mbase + q * 8 + (j15 >> 2);
+  const int col = colbase + 4 * (j15 & 3);
+  auto p0 = (LDS3 trvec*)((LDS3 bf16*)img + (long)row * LROW + col);
+  auto p1 = (LDS3 trvec*)((LDS3 bf16*)img + (long)(row + 4) * LROW + col);
+  trvec lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p0);
+  trvec hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p1);
+  union { struct { trvec a, b; } t; bf16x8_t v; } u;
+  u.t.a = lo;
+  u.t.b = hi;
+  return u.v;
+}
+
 template <int BCO, int BKN>
 __global__ void __launch_bounds__(CONV_BLOCK)
 conv_wrw_kernel(const WrwParams p) {
@@ -403,14 +429,18 @@ conv_wrw_kernel(const WrwParams p) {
   constexpr int WTN = BKN / 4;        // wave k-columns
   constexpr int FA = BCO / 16;
   constexpr int FB = WTN / 16;
-  constexpr int LMD = BM + 8;         // 16B-aligned [row][m] strides
+  constexpr int LD = BCO + 8;         // dy image row stride (16B-aligned)
+  constexpr int LX = BKN + 8;         // x image row stride
   constexpr int DCH = BM * (BCO / 8);
   constexpr int XCH = BM * (BKN / 8);
   constexpr int DPT = (DCH + CONV_BLOCK - 1) / CONV_BLOCK;
   constexpr int XPT = (XCH + CONV_BLOCK - 1) / CONV_BLOCK;
 
-  __shared__ bf16 dy_t[2][BCO * LMD];
-  __shared__ bf16 x_t[2][BKN * LMD];
+  // natural [m][*] images, double-buffered; staged with VECTOR writes and
+  // read as MFMA fragments via hardware transpose reads (tr_frag) — v4's
+  // per-element transposed ds_writes were the measured bottleneck.
+  __shared__ bf16 dy_t[2][BM * LD];
+  __shared__ bf16 x_t[2][BM * LX];
 
   const int t = threadIdx.x;
   const int lane = t & 63;
@@ -467,9 +497,7 @@ conv_wrw_kernel(const WrwParams p) {
       if (c < DCH) {
         const int mm = c / (BCO / 8);
         const int c8 = (c % (BCO / 8)) * 8;
-        const bf16* vv = reinterpret_cast<const bf16*>(&dreg[u]);
-#pragma unroll
-        for (int j = 0; j < 8; ++j) dy_t[buf][(c8 + j) * LMD + mm] = vv[j];
+        *reinterpret_cast<bf16x8_t*>(&dy_t[buf][mm * LD + c8]) = dreg[u];
       }
     }
 #pragma unroll
@@ -478,9 +506,7 @@ conv_wrw_kernel(const WrwParams p) {
       if (c < XCH) {
         const int mm = c / (BKN / 8);
         const int k8 = (c % (BKN / 8)) * 8;
-        const bf16* vv = reinterpret_cast<const bf16*>(&xreg[u]);
-#pragma unroll
-        for (int j = 0; j < 8; ++j) x_t[buf][(k8 + j) * LMD + mm] = vv[j];
+        *reinterpret_cast<bf16x8_t*>(&x_t[buf][mm * LX + k8]) = xreg[u];
       }
     }
   };
@@ -498,17 +524,11 @@ conv_wrw_kernel(const WrwParams p) {
     for (int sub = 0; sub < BM / 32; ++sub) {
       bf16x8_t afrag[FA], bfrag[FB];
 #pragma unroll
-      for (int i = 0; i < FA; ++i) {
-        const int co = i * 16 + (lane & 15);
-        afrag[i] = *reinterpret_cast<const bf16x8_t*>(
-            &dy_t[buf][co * LMD + sub * 32 + (lane >> 4) * 8]);
-      }
+      for (int i = 0; i < FA; ++i)
+        afrag[i] = tr_frag<LD>(dy_t[buf], sub * 32, i * 16, lane);
 #pragma unroll
-      for (int j = 0; j < FB; ++j) {
-        const int kk = wave * WTN + j * 16 + (lane & 15);
-        bfrag[j] = *reinterpret_cast<const bf16x8_t*>(
-            &x_t[buf][kk * LMD + sub * 32 + (lane >> 4) * 8]);
-      }
+      for (int j = 0; j < FB; ++j)
+        bfrag[j] = tr_frag<LX>(x_t[buf], sub * 32, wave * WTN + j * 16, lane);
 #pragma unroll
       for (int i = 0; i < FA; ++i)
 #pragma unroll
