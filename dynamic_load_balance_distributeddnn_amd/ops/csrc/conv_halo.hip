@@ -19,6 +19,8 @@ typedef __attribute__((ext_vector_type(8))) short bf16x8_t;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 
 #define HBLOCK 256
+#define LDS3H __attribute__((address_space(3)))
+typedef __attribute__((__vector_size__(4 * sizeof(__bf16)))) __bf16 trvec_h;
 #define CI_CHUNK 32
 #define HPAD 40  // padded ci stride in halo LDS rows (conflict-free, 16B-aligned)
 
@@ -161,7 +163,7 @@ conv3x3_wrw_halo(const bf16* __restrict__ x, const bf16* __restrict__ dy,
                  const int m_per_split, const int wshift) {
   constexpr int BCO = 32;
   constexpr int BM = 128;             // reduction chunk (whole rows)
-  constexpr int LMD = BM + 8;         // 16B-aligned [co][m] rows
+  constexpr int LMD = BCO + 8;        // natural [m][co] rows (16B-aligned)
   const int t = threadIdx.x;
   const int lane = t & 63;
   const int wave = t >> 6;
@@ -169,7 +171,7 @@ conv3x3_wrw_halo(const bf16* __restrict__ x, const bf16* __restrict__ dy,
   const int fb_count = (wave < 2) ? 5 : 4;
   const int fb_base = (wave < 2) ? wave * 5 : 10 + (wave - 2) * 4;
 
-  __shared__ bf16 dy_t[BCO * LMD];        // [co][m]
+  __shared__ bf16 dy_t[BM * LMD];         // natural [m][co] image
   __shared__ bf16 halo[204 * CI_CHUNK];   // band: (BM/W+2) x (W+2) pixels
 
   const int co0 = blockIdx.x * BCO;
@@ -205,9 +207,7 @@ conv3x3_wrw_halo(const bf16* __restrict__ x, const bf16* __restrict__ dy,
         for (int j = 0; j < 8 && co0 + c8 + j < Co; ++j)
           vv[j] = dy[(long)m * Co + co0 + c8 + j];
       }
-      const bf16* vv = reinterpret_cast<const bf16*>(&v);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) dy_t[(c8 + j) * LMD + mm] = vv[j];
+      *reinterpret_cast<bf16x8_t*>(&dy_t[mm * LMD + c8]) = v;
     }
     // ---- stage x band: rows [py0-1 .. py0+rows], cols [-1..W]
     const int n = mt / (H * W);
@@ -231,26 +231,48 @@ conv3x3_wrw_halo(const bf16* __restrict__ x, const bf16* __restrict__ dy,
     for (int sub = 0; sub < 4; ++sub) {
       const int msub = sub * 32;
       bf16x8_t afrag[2];
+      {
+        const int j15 = lane & 15, q = lane >> 4;
+        const int mrow = msub + q * 8 + (j15 >> 2);
 #pragma unroll
-      for (int i = 0; i < 2; ++i) {
-        const int co = i * 16 + (lane & 15);
-        afrag[i] = *reinterpret_cast<const bf16x8_t*>(
-            &dy_t[co * LMD + msub + (lane >> 4) * 8]);
+        for (int i = 0; i < 2; ++i) {
+          const int col = i * 16 + 4 * (j15 & 3);
+          auto p0 = (LDS3H trvec_h*)((LDS3H bf16*)dy_t + mrow * LMD + col);
+          auto p1 = (LDS3H trvec_h*)((LDS3H bf16*)dy_t +
+                                     (mrow + 4) * LMD + col);
+          trvec_h lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p0);
+          trvec_h hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p1);
+          union { struct { trvec_h a, b; } t2; bf16x8_t v; } u;
+          u.t2.a = lo;
+          u.t2.b = hi;
+          afrag[i] = u.v;
+        }
       }
 #pragma unroll
       for (int j = 0; j < 5; ++j) {
         if (j >= fb_count) continue;
-        const int kk = (fb_base + j) * 16 + (lane & 15);
-        const int tap = kk / CI_CHUNK, cil = kk % CI_CHUNK;
+        // fragment columns are 16 consecutive k = (tap, cil); a 4-col
+        // address run never crosses a tap boundary (16 | frag base)
+        const int base16 = (fb_base + j) * 16;
+        const int tap = base16 / CI_CHUNK;
         const int r = tap / 3, sxx = tap % 3;
-        bf16x8_t bfrag;
-        bf16* bp = reinterpret_cast<bf16*>(&bfrag);
-#pragma unroll
-        for (int q = 0; q < 8; ++q) {
-          const int mm = msub + (lane >> 4) * 8 + q;
-          const int py = mm >> wshift, px = mm & wmask;
-          bp[q] = halo[((py + r) * hw_cols + px + sxx) * CI_CHUNK + cil];
-        }
+        const int j15 = lane & 15, q = lane >> 4;
+        const int mm = msub + q * 8 + (j15 >> 2);
+        const int py = mm >> wshift, px = mm & wmask;
+        const int cil = (base16 % CI_CHUNK) + 4 * (j15 & 3);
+        const long rowoff =
+            ((long)(py + r) * hw_cols + px + sxx) * CI_CHUNK + cil;
+        const long rowoff4 =
+            ((long)((mm + 4) >> wshift) + r) * hw_cols * CI_CHUNK +
+            (long)(((mm + 4) & wmask) + sxx) * CI_CHUNK + cil;
+        auto p0 = (LDS3H trvec_h*)((LDS3H bf16*)halo + rowoff);
+        auto p1 = (LDS3H trvec_h*)((LDS3H bf16*)halo + rowoff4);
+        trvec_h lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p0);
+        trvec_h hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p1);
+        union { struct { trvec_h a, b; } t2; bf16x8_t v; } u;
+        u.t2.a = lo;
+        u.t2.b = hi;
+        bf16x8_t bfrag = u.v;
 #pragma unroll
         for (int i = 0; i < 2; ++i)
           acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
