@@ -114,6 +114,10 @@ __device__ inline bf16x8_t im2col_load8(const bf16* __restrict__ x, int m,
 }
 
 // ---------------------------------------------------------------- forward
+// Double-buffered register-staged pipeline: chunk t+1's global loads are
+// issued before chunk t's MFMAs, written to the alternate LDS buffer
+// after them — ONE barrier per K-step (the 2-barrier form left the SIMDs
+// idle on every PMC).
 template <int BM, int BN, int WM, int WN>
 __global__ void __launch_bounds__(CONV_BLOCK)
 conv_fwd_kernel(const ConvParams p) {
@@ -123,9 +127,13 @@ conv_fwd_kernel(const ConvParams p) {
   constexpr int FB = WTN / 16;
   constexpr int LDA = BK + 8;         // padded LDS row (elements)
   constexpr int LDB = BK + 8;
+  constexpr int ACH = BM * (BK / 8);
+  constexpr int BCH = BN * (BK / 8);
+  constexpr int APT = (ACH + CONV_BLOCK - 1) / CONV_BLOCK;
+  constexpr int BPT = (BCH + CONV_BLOCK - 1) / CONV_BLOCK;
 
-  __shared__ bf16 a_lds[BM * LDA];
-  __shared__ bf16 b_lds[BN * LDB];    // [n][k] image
+  __shared__ bf16 a_lds[2][BM * LDA];
+  __shared__ bf16 b_lds[2][BN * LDB];    // [n][k] image
 
   const int t = threadIdx.x;
   const int lane = t & 63;
@@ -141,50 +149,79 @@ conv_fwd_kernel(const ConvParams p) {
 #pragma unroll
     for (int j = 0; j < FB; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
-  constexpr int ACH = BM * (BK / 8);
-  constexpr int BCH = BN * (BK / 8);
+  bf16x8_t areg[APT], breg[BPT];
 
-  for (int kt = 0; kt < p.K; kt += BK) {
-    for (int c = t; c < ACH; c += CONV_BLOCK) {
+  auto load_tile = [&](int kt) {
+#pragma unroll
+    for (int u = 0; u < APT; ++u) {
+      const int c = t + u * CONV_BLOCK;
       const int row = c / (BK / 8);
       const int k8 = (c % (BK / 8)) * 8;
-      const int m = m0 + row;
-      const int k = kt + k8;
-      bf16x8_t v = im2col_load8(p.x, m, k, p.IH, p.IW, p.Ci, p.OH, p.OW,
-                                p.S, p.stride, p.pad, p.K, p.M, vec, p.g);
-      *reinterpret_cast<bf16x8_t*>(&a_lds[row * LDA + k8]) = v;
+      areg[u] = im2col_load8(p.x, m0 + row, kt + k8, p.IH, p.IW, p.Ci, p.OH,
+                             p.OW, p.S, p.stride, p.pad, p.K, p.M, vec, p.g);
+      if (c >= ACH) areg[u] = zero8();
     }
-    for (int c = t; c < BCH; c += CONV_BLOCK) {
+#pragma unroll
+    for (int u = 0; u < BPT; ++u) {
+      const int c = t + u * CONV_BLOCK;
       const int nrow = c / (BK / 8);
       const int k8 = (c % (BK / 8)) * 8;
       const int n = n0 + nrow;
       const int k = kt + k8;
-      // K % 8 == 0 whenever Ci % 8 == 0; clamp+mask keeps it branchless,
-      // the generic tail only triggers for the tiny non-vec stems.
-      const bool ok = (n < p.Co) & (k + 7 < p.K);
+      const bool ok = (c < BCH) & (n < p.Co) & (k + 7 < p.K);
       const long off = ok ? (long)n * p.K + k : 0;
       bf16x8_t v = mask8(*reinterpret_cast<const bf16x8_t*>(p.w + off), ok);
-      if (!ok && n < p.Co) {
+      if (!ok && c < BCH && n < p.Co) {
         bf16* vv = reinterpret_cast<bf16*>(&v);
         for (int j = 0; j < 8 && k + j < p.K; ++j)
           vv[j] = p.w[(long)n * p.K + k + j];
       }
-      *reinterpret_cast<bf16x8_t*>(&b_lds[nrow * LDB + k8]) = v;
+      breg[u] = v;
     }
-    __syncthreads();
+  };
+
+  auto write_tile = [&](int buf) {
+#pragma unroll
+    for (int u = 0; u < APT; ++u) {
+      const int c = t + u * CONV_BLOCK;
+      if (c < ACH) {
+        const int row = c / (BK / 8);
+        const int k8 = (c % (BK / 8)) * 8;
+        *reinterpret_cast<bf16x8_t*>(&a_lds[buf][row * LDA + k8]) = areg[u];
+      }
+    }
+#pragma unroll
+    for (int u = 0; u < BPT; ++u) {
+      const int c = t + u * CONV_BLOCK;
+      if (c < BCH) {
+        const int nrow = c / (BK / 8);
+        const int k8 = (c % (BK / 8)) * 8;
+        *reinterpret_cast<bf16x8_t*>(&b_lds[buf][nrow * LDB + k8]) = breg[u];
+      }
+    }
+  };
+
+  load_tile(0);
+  write_tile(0);
+  __syncthreads();
+
+  int buf = 0;
+  for (int kt = 0; kt < p.K; kt += BK) {
+    const bool more = kt + BK < p.K;
+    if (more) load_tile(kt + BK);
 
     bf16x8_t afrag[FA], bfrag[FB];
 #pragma unroll
     for (int i = 0; i < FA; ++i) {
       const int row = wr * WTM + i * 16 + (lane & 15);
       afrag[i] = *reinterpret_cast<const bf16x8_t*>(
-          &a_lds[row * LDA + (lane >> 4) * 8]);
+          &a_lds[buf][row * LDA + (lane >> 4) * 8]);
     }
 #pragma unroll
     for (int j = 0; j < FB; ++j) {
       const int col = wc * WTN + j * 16 + (lane & 15);
       bfrag[j] = *reinterpret_cast<const bf16x8_t*>(
-          &b_lds[col * LDB + (lane >> 4) * 8]);
+          &b_lds[buf][col * LDB + (lane >> 4) * 8]);
     }
 #pragma unroll
     for (int i = 0; i < FA; ++i)
@@ -192,7 +229,9 @@ conv_fwd_kernel(const ConvParams p) {
       for (int j = 0; j < FB; ++j)
         acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
+    if (more) write_tile(buf ^ 1);
     __syncthreads();
+    buf ^= 1;
   }
 
 #pragma unroll
@@ -284,9 +323,13 @@ conv_bwd_data_kernel(const ConvBwdParams p) {
   constexpr int FB = WTN / 16;
   constexpr int LDA = BK + 8;
   constexpr int LDB = BK + 8;
+  constexpr int ACH = BM * (BK / 8);
+  constexpr int BCH = BN * (BK / 8);
+  constexpr int APT = (ACH + CONV_BLOCK - 1) / CONV_BLOCK;
+  constexpr int BPT = (BCH + CONV_BLOCK - 1) / CONV_BLOCK;
 
-  __shared__ bf16 a_lds[BM * LDA];
-  __shared__ bf16 b_lds[BN * LDB];
+  __shared__ bf16 a_lds[2][BM * LDA];
+  __shared__ bf16 b_lds[2][BN * LDB];
 
   const int t = threadIdx.x;
   const int lane = t & 63;
@@ -302,26 +345,26 @@ conv_bwd_data_kernel(const ConvBwdParams p) {
 #pragma unroll
     for (int j = 0; j < FB; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
-  constexpr int ACH = BM * (BK / 8);
-  constexpr int BCH = BN * (BK / 8);
+  bf16x8_t areg[APT], breg[BPT];
 
-  for (int kt = 0; kt < p.K; kt += BK) {
-    for (int c = t; c < ACH; c += CONV_BLOCK) {
+  auto load_tile = [&](int kt) {
+#pragma unroll
+    for (int u = 0; u < APT; ++u) {
+      const int c = t + u * CONV_BLOCK;
       const int row = c / (BK / 8);
       const int k8 = (c % (BK / 8)) * 8;
-      const int m = m0 + row;
-      const int k = kt + k8;
-      bf16x8_t v = dcol_load8(p, m, k, vec);
-      *reinterpret_cast<bf16x8_t*>(&a_lds[row * LDA + k8]) = v;
+      areg[u] = dcol_load8(p, m0 + row, kt + k8, vec);
+      if (c >= ACH) areg[u] = zero8();
     }
-    for (int c = t; c < BCH; c += CONV_BLOCK) {
-      const int nrow = c / (BK / 8);
+#pragma unroll
+    for (int u = 0; u < BPT; ++u) {
+      const int c = t + u * CONV_BLOCK;
+      const int nrow = c / (BK / 8);  // ci
       const int k8 = (c % (BK / 8)) * 8;
-      // B[ci][k=(rs,co)] = w[co][rs*Ci + ci]; per-element gather (the
-      // weight is tiny and L2-hot; this removes the per-conv host-side
-      // transpose kernel the old layout needed)
-      const int n = n0 + nrow;  // ci
+      const int n = n0 + nrow;
       const int k = kt + k8;
+      // B[ci][k=(rs,co)] = w[co][rs*Ci + ci] gathered from the natural
+      // channels_last weight (tiny, L2-hot)
       bf16x8_t v;
       bf16* vv = reinterpret_cast<bf16*>(&v);
 #pragma unroll
@@ -329,27 +372,57 @@ conv_bwd_data_kernel(const ConvBwdParams p) {
         const int kk = k + j;
         unsigned rs, co;
         p.g.fd_c.divmod(kk < p.K ? kk : 0, rs, co);
-        const bool e = (n < p.Ci) & (kk < p.K);
+        const bool e = (c < BCH) & (n < p.Ci) & (kk < p.K);
         const long off = e ? (long)co * p.Kw + rs * p.Ci + n : 0;
         bf16 t2 = p.w[off];
         vv[j] = e ? t2 : (bf16)__float2bfloat16(0.f);
       }
-      *reinterpret_cast<bf16x8_t*>(&b_lds[nrow * LDB + k8]) = v;
+      breg[u] = v;
     }
-    __syncthreads();
+  };
+
+  auto write_tile = [&](int buf) {
+#pragma unroll
+    for (int u = 0; u < APT; ++u) {
+      const int c = t + u * CONV_BLOCK;
+      if (c < ACH) {
+        const int row = c / (BK / 8);
+        const int k8 = (c % (BK / 8)) * 8;
+        *reinterpret_cast<bf16x8_t*>(&a_lds[buf][row * LDA + k8]) = areg[u];
+      }
+    }
+#pragma unroll
+    for (int u = 0; u < BPT; ++u) {
+      const int c = t + u * CONV_BLOCK;
+      if (c < BCH) {
+        const int nrow = c / (BK / 8);
+        const int k8 = (c % (BK / 8)) * 8;
+        *reinterpret_cast<bf16x8_t*>(&b_lds[buf][nrow * LDB + k8]) = breg[u];
+      }
+    }
+  };
+
+  load_tile(0);
+  write_tile(0);
+  __syncthreads();
+
+  int buf = 0;
+  for (int kt = 0; kt < p.K; kt += BK) {
+    const bool more = kt + BK < p.K;
+    if (more) load_tile(kt + BK);
 
     bf16x8_t afrag[FA], bfrag[FB];
 #pragma unroll
     for (int i = 0; i < FA; ++i) {
       const int row = wr * WTM + i * 16 + (lane & 15);
       afrag[i] = *reinterpret_cast<const bf16x8_t*>(
-          &a_lds[row * LDA + (lane >> 4) * 8]);
+          &a_lds[buf][row * LDA + (lane >> 4) * 8]);
     }
 #pragma unroll
     for (int j = 0; j < FB; ++j) {
       const int col = wc * WTN + j * 16 + (lane & 15);
       bfrag[j] = *reinterpret_cast<const bf16x8_t*>(
-          &b_lds[col * LDB + (lane >> 4) * 8]);
+          &b_lds[buf][col * LDB + (lane >> 4) * 8]);
     }
 #pragma unroll
     for (int i = 0; i < FA; ++i)
@@ -357,7 +430,9 @@ conv_bwd_data_kernel(const ConvBwdParams p) {
       for (int j = 0; j < FB; ++j)
         acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
+    if (more) write_tile(buf ^ 1);
     __syncthreads();
+    buf ^= 1;
   }
 
 #pragma unroll
