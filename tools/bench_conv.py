@@ -41,12 +41,8 @@ def main():
         OH = (H + 2 * pad - R) // stride + 1
         dy = torch.randn(N, Co, OH, OH, device="cuda").bfloat16() \
             .to(memory_format=torch.channels_last)
-        wt = w.permute(0, 2, 3, 1).permute(3, 1, 2, 0) \
-            .reshape(Ci, R * R * Co).contiguous()
-
         t_fwd = timeit(lambda: ext().conv_fwd(x, w, None, stride, pad))
-        t_bwd = timeit(lambda: ext().conv_bwd_data(dy, wt, H, H, Ci, R, R,
-                                                   stride, pad))
+        t_bwd = timeit(lambda: ext().conv_bwd_data(dy, w, H, H, stride, pad))
         t_wrw = timeit(lambda: ext().conv_wrw(x, dy, R, R, stride, pad))
 
         wf = w.float()
