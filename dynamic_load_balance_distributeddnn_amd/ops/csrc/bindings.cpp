@@ -45,7 +45,7 @@ static std::vector<torch::Tensor> gn_fwd(torch::Tensor x, torch::Tensor gamma,
   auto stream = at::hip::getCurrentHIPStream();
   float* scratch = nullptr;
   torch::Tensor scratch_t;
-  if (N < 1024 && HW > 1) {  // underfilled grid: split HW across blocks
+  if (N < 192 && HW > 1) {  // small-batch path: split HW across blocks
     scratch_t = torch::zeros({N, groups, 2},
                              x.options().dtype(torch::kFloat32));
     scratch = scratch_t.data_ptr<float>();
@@ -72,7 +72,7 @@ static std::vector<torch::Tensor> gn_bwd(torch::Tensor x, torch::Tensor dz,
   auto stream = at::hip::getCurrentHIPStream();
   float* scratch = nullptr;
   torch::Tensor scratch_t;
-  if (N < 1024 && HW > 1) {
+  if (N < 192 && HW > 1) {
     scratch_t = torch::zeros({N, groups, 2},
                              x.options().dtype(torch::kFloat32));
     scratch = scratch_t.data_ptr<float>();

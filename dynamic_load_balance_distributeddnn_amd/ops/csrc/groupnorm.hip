@@ -509,7 +509,7 @@ extern "C" void dlb_gn_fwd(const void* x, void* y, const float* gamma,
                            float* scratch /* [N][G][2] zeroed, or null */,
                            int N, int HW, int C, int G, float eps, int relu,
                            hipStream_t stream) {
-  const int slices = scratch ? (int)std::min<long>((1023 + N) / N, HW) : 1;
+  const int slices = scratch ? (int)std::min<long>((383 + N) / N, HW) : 1;
   if (slices > 1) {
     dim3 grid(N, slices);
     hipLaunchKernelGGL(gn_stats_part_kernel, grid, dim3(GN_BLOCK), 0, stream,
@@ -532,7 +532,7 @@ extern "C" void dlb_gn_bwd(const void* x, const void* dz, void* dx,
                            int N, int HW, int C, int G, int relu,
                            hipStream_t stream) {
   size_t shmem = 2 * (size_t)C * sizeof(float);
-  const int slices = scratch ? (int)std::min<long>((1023 + N) / N, HW) : 1;
+  const int slices = scratch ? (int)std::min<long>((383 + N) / N, HW) : 1;
   if (slices > 1) {
     dim3 grid(N, slices);
     hipLaunchKernelGGL(gn_bwd_part_kernel, grid, dim3(GN_BLOCK), shmem,
