@@ -149,7 +149,10 @@ def main():
     # small per-rank batches in the strong-scaling sweep).  Collectives
     # and dynamic shapes fall back to eager if capture fails.
     run_step = step
-    if device.type == "cuda" and not args.no_graphs:
+    # capture only single-rank: RCCL collectives inside hipGraph capture
+    # are not validated on this stack, and a capture hang on one rank
+    # would stall the whole job
+    if device.type == "cuda" and world == 1 and not args.no_graphs:
         try:
             side = torch.cuda.Stream()
             side.wait_stream(torch.cuda.current_stream())
@@ -166,14 +169,6 @@ def main():
                 print(f"# hipGraph capture unavailable ({type(e).__name__}); "
                       "running eager", flush=True)
             run_step = step
-        if world > 1:
-            # use the graph only if EVERY rank captured (divergent
-            # graph/eager ranks would desync the collectives)
-            ok = torch.tensor([1.0 if run_step is not step else 0.0],
-                              device=device)
-            dist.all_reduce(ok, op=dist.ReduceOp.MIN)
-            if ok.item() < 1:
-                run_step = step
         if rank == 0 and run_step is not step:
             print("# hipGraph capture active", flush=True)
 
