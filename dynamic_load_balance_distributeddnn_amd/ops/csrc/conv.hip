@@ -113,6 +113,32 @@ __device__ inline bf16x8_t im2col_load8(const bf16* __restrict__ x, int m,
   return v;
 }
 
+#define LDS3 __attribute__((address_space(3)))
+typedef __attribute__((__vector_size__(4 * sizeof(__bf16)))) __bf16 trvec;
+
+// ds_read_b64_tr_b16 fragment read from a NATURAL row-major [m][cols]
+// LDS image (row stride LROW elements): returns the 8 m-elements of
+// column (colbase + lane&15) for m in [mbase + (lane>>4)*8, +8).
+// Lane map verified on hardware by tools/mfma_probe/tr_probe.hip:
+// result-lane i gets element (i&3) of the row addressed by lane
+// ((i>>2) + 4k) of its 16-lane group, so address-lane j points at
+// row (mbase + q*8 + half*4 + (j>>2)), col (colbase + 4*(j&3)).
+template <int LROW>
+__device__ inline bf16x8_t tr_frag(const bf16* img, int mbase, int colbase,
+                                   int lane) {
+  const int j15 = lane & 15, q = lane >> 4;
+  const int row = mbase + q * 8 + (j15 >> 2);
+  const int col = colbase + 4 * (j15 & 3);
+  auto p0 = (LDS3 trvec*)((LDS3 bf16*)img + (long)row * LROW + col);
+  auto p1 = (LDS3 trvec*)((LDS3 bf16*)img + (long)(row + 4) * LROW + col);
+  trvec lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p0);
+  trvec hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p1);
+  union { struct { trvec a, b; } t; bf16x8_t v; } u;
+  u.t.a = lo;
+  u.t.b = hi;
+  return u.v;
+}
+
 // ---------------------------------------------------------------- forward
 // Double-buffered register-staged pipeline: chunk t+1's global loads are
 // issued before chunk t's MFMAs, written to the alternate LDS buffer
@@ -142,6 +168,7 @@ conv_fwd_kernel(const ConvParams p) {
   const int m0 = blockIdx.x * BM;
   const int n0 = blockIdx.y * BN;
   const bool vec = (p.Ci & 7) == 0;
+  constexpr bool wtr = false;  // fwd stages B naturally; no transpose path
 
   f32x4 acc[FA][FB];
 #pragma unroll
@@ -193,7 +220,14 @@ conv_fwd_kernel(const ConvParams p) {
 #pragma unroll
     for (int u = 0; u < BPT; ++u) {
       const int c = t + u * CONV_BLOCK;
-      if (c < BCH) {
+      if (wtr) {
+        if (c < 32 * (BN / 8)) {
+          const int corow = c / (BN / 8);
+          const int ci8 = (c % (BN / 8)) * 8;
+          *reinterpret_cast<bf16x8_t*>(
+              &b_lds[buf][corow * (BN + 8) + ci8]) = breg[u];
+        }
+      } else if (c < BCH) {
         const int nrow = c / (BK / 8);
         const int k8 = (c % (BK / 8)) * 8;
         *reinterpret_cast<bf16x8_t*>(&b_lds[buf][nrow * LDB + k8]) = breg[u];
@@ -217,11 +251,17 @@ conv_fwd_kernel(const ConvParams p) {
       afrag[i] = *reinterpret_cast<const bf16x8_t*>(
           &a_lds[buf][row * LDA + (lane >> 4) * 8]);
     }
+    if (wtr) {
 #pragma unroll
-    for (int j = 0; j < FB; ++j) {
-      const int col = wc * WTN + j * 16 + (lane & 15);
-      bfrag[j] = *reinterpret_cast<const bf16x8_t*>(
-          &b_lds[buf][col * LDB + (lane >> 4) * 8]);
+      for (int j = 0; j < FB; ++j)
+        bfrag[j] = tr_frag<BN + 8>(b_lds[buf], 0, wc * WTN + j * 16, lane);
+    } else {
+#pragma unroll
+      for (int j = 0; j < FB; ++j) {
+        const int col = wc * WTN + j * 16 + (lane & 15);
+        bfrag[j] = *reinterpret_cast<const bf16x8_t*>(
+            &b_lds[buf][col * LDB + (lane >> 4) * 8]);
+      }
     }
 #pragma unroll
     for (int i = 0; i < FA; ++i)
@@ -327,9 +367,10 @@ conv_bwd_data_kernel(const ConvBwdParams p) {
   constexpr int BCH = BN * (BK / 8);
   constexpr int APT = (ACH + CONV_BLOCK - 1) / CONV_BLOCK;
   constexpr int BPT = (BCH + CONV_BLOCK - 1) / CONV_BLOCK;
+  constexpr int BSZ = (BN * LDB > 32 * (BN + 8)) ? BN * LDB : 32 * (BN + 8);
 
   __shared__ bf16 a_lds[2][BM * LDA];
-  __shared__ bf16 b_lds[2][BN * LDB];
+  __shared__ bf16 b_lds[2][BSZ];
 
   const int t = threadIdx.x;
   const int lane = t & 63;
@@ -338,6 +379,7 @@ conv_bwd_data_kernel(const ConvBwdParams p) {
   const int m0 = blockIdx.x * BM;
   const int n0 = blockIdx.y * BN;
   const bool vec = (p.Co & 7) == 0;
+  const bool wtr = (p.Co % 32) == 0;  // k-tile = 32 co of one tap
 
   f32x4 acc[FA][FB];
 #pragma unroll
@@ -359,6 +401,27 @@ conv_bwd_data_kernel(const ConvBwdParams p) {
 #pragma unroll
     for (int u = 0; u < BPT; ++u) {
       const int c = t + u * CONV_BLOCK;
+      if (wtr) {
+        // Co % 32 == 0: a BK=32 k-tile is 32 consecutive co of ONE tap.
+        // Stage the weight rows NATURALLY ([co-row][ci]) with vector
+        // loads; fragments use transpose reads (tr_frag) like the wrw.
+        const int corow = c / (BN / 8);       // 0..31 within the k-tile
+        const int ci8 = (c % (BN / 8)) * 8;
+        const int kk = kt + corow;            // global k = rs*Co + co
+        unsigned rs, co;
+        p.g.fd_c.divmod(kk < p.K ? kk : 0, rs, co);
+        const bool ok = (c < 32 * (BN / 8)) & (kk < p.K) &
+                        (n0 + ci8 + 7 < p.Ci);
+        const long off = ok ? (long)co * p.Kw + rs * p.Ci + n0 + ci8 : 0;
+        bf16x8_t v = mask8(*reinterpret_cast<const bf16x8_t*>(p.w + off), ok);
+        if (!ok && c < 32 * (BN / 8) && kk < p.K && n0 + ci8 < p.Ci) {
+          bf16* vv = reinterpret_cast<bf16*>(&v);
+          for (int j = 0; j < 8 && n0 + ci8 + j < p.Ci; ++j)
+            vv[j] = p.w[(long)co * p.Kw + rs * p.Ci + n0 + ci8 + j];
+        }
+        breg[u] = v;
+        continue;
+      }
       const int nrow = c / (BK / 8);  // ci
       const int k8 = (c % (BK / 8)) * 8;
       const int n = n0 + nrow;
@@ -394,7 +457,14 @@ conv_bwd_data_kernel(const ConvBwdParams p) {
 #pragma unroll
     for (int u = 0; u < BPT; ++u) {
       const int c = t + u * CONV_BLOCK;
-      if (c < BCH) {
+      if (wtr) {
+        if (c < 32 * (BN / 8)) {
+          const int corow = c / (BN / 8);
+          const int ci8 = (c % (BN / 8)) * 8;
+          *reinterpret_cast<bf16x8_t*>(
+              &b_lds[buf][corow * (BN + 8) + ci8]) = breg[u];
+        }
+      } else if (c < BCH) {
         const int nrow = c / (BK / 8);
         const int k8 = (c % (BK / 8)) * 8;
         *reinterpret_cast<bf16x8_t*>(&b_lds[buf][nrow * LDB + k8]) = breg[u];
@@ -418,11 +488,17 @@ conv_bwd_data_kernel(const ConvBwdParams p) {
       afrag[i] = *reinterpret_cast<const bf16x8_t*>(
           &a_lds[buf][row * LDA + (lane >> 4) * 8]);
     }
+    if (wtr) {
 #pragma unroll
-    for (int j = 0; j < FB; ++j) {
-      const int col = wc * WTN + j * 16 + (lane & 15);
-      bfrag[j] = *reinterpret_cast<const bf16x8_t*>(
-          &b_lds[buf][col * LDB + (lane >> 4) * 8]);
+      for (int j = 0; j < FB; ++j)
+        bfrag[j] = tr_frag<BN + 8>(b_lds[buf], 0, wc * WTN + j * 16, lane);
+    } else {
+#pragma unroll
+      for (int j = 0; j < FB; ++j) {
+        const int col = wc * WTN + j * 16 + (lane & 15);
+        bfrag[j] = *reinterpret_cast<const bf16x8_t*>(
+            &b_lds[buf][col * LDB + (lane >> 4) * 8]);
+      }
     }
 #pragma unroll
     for (int i = 0; i < FA; ++i)
@@ -470,32 +546,6 @@ struct WrwParams {
   int m_per_split;
   ConvGeom g;
 };
-
-#define LDS3 __attribute__((address_space(3)))
-typedef __attribute__((__vector_size__(4 * sizeof(__bf16)))) __bf16 trvec;
-
-// ds_read_b64_tr_b16 fragment read from a NATURAL row-major [m][cols]
-// LDS image (row stride LROW elements): returns the 8 m-elements of
-// column (colbase + lane&15) for m in [mbase + (lane>>4)*8, +8).
-// Lane map verified on hardware by tools/mfma_probe/tr_probe.hip:
-// result-lane i gets element (i&3) of the row addressed by lane
-// ((i>>2) + 4k) of its 16-lane group, so address-lane j points at
-// row (mbase + q*8 + half*4 + (j>>2)), col (colbase + 4*(j&3)).
-template <int LROW>
-__device__ inline bf16x8_t tr_frag(const bf16* img, int mbase, int colbase,
-                                   int lane) {
-  const int j15 = lane & 15, q = lane >> 4;
-  const int row = mbase + q * 8 + (j15 >> 2);
-  const int col = colbase + 4 * (j15 & 3);
-  auto p0 = (LDS3 trvec*)((LDS3 bf16*)img + (long)row * LROW + col);
-  auto p1 = (LDS3 trvec*)((LDS3 bf16*)img + (long)(row + 4) * LROW + col);
-  trvec lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p0);
-  trvec hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p1);
-  union { struct { trvec a, b; } t; bf16x8_t v; } u;
-  u.t.a = lo;
-  u.t.b = hi;
-  return u.v;
-}
 
 template <int BCO, int BKN>
 __global__ void __launch_bounds__(CONV_BLOCK)
