@@ -63,6 +63,7 @@ gn_fwd_kernel(const bf16* __restrict__ x, bf16* __restrict__ y,
       const int c0 = oct << 3;
       float s[8] = {0, 0, 0, 0, 0, 0, 0, 0};
       float ss[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+      #pragma unroll 2
       for (int p = tp; p < HW; p += TP) {
         Bf16x8 chunk = *reinterpret_cast<const Bf16x8*>(xb + (long)p * C + c0);
 #pragma unroll
@@ -108,7 +109,8 @@ gn_fwd_kernel(const bf16* __restrict__ x, bf16* __restrict__ y,
       mu[j] = s_mean[g];
       rs[j] = s_rstd[g];
     }
-    for (int p = tp; p < HW; p += TP) {
+    #pragma unroll 2
+      for (int p = tp; p < HW; p += TP) {
       const long off = (long)p * C + c0;
       Bf16x8 chunk = *reinterpret_cast<const Bf16x8*>(xb + off);
       Bf16x8 out;
@@ -168,6 +170,7 @@ gn_bwd_kernel(const bf16* __restrict__ x, const bf16* __restrict__ dz,
         rs[j] = rstd_in[(long)n * G + g];
       }
       float a1[8] = {0}, a2[8] = {0}, adg[8] = {0}, adb[8] = {0};
+      #pragma unroll 2
       for (int p = tp; p < HW; p += TP) {
         const long off = (long)p * C + c0;
         Bf16x8 xc = *reinterpret_cast<const Bf16x8*>(xb + off);
@@ -221,7 +224,8 @@ gn_bwd_kernel(const bf16* __restrict__ x, const bf16* __restrict__ dz,
       k1[j] = s_s1[g] * inv_m;
       k2[j] = s_s2[g] * inv_m;
     }
-    for (int p = tp; p < HW; p += TP) {
+    #pragma unroll 2
+      for (int p = tp; p < HW; p += TP) {
       const long off = (long)p * C + c0;
       Bf16x8 xc = *reinterpret_cast<const Bf16x8*>(xb + off);
       Bf16x8 dc = *reinterpret_cast<const Bf16x8*>(db + off);
@@ -274,6 +278,7 @@ gn_stats_part_kernel(const bf16* __restrict__ x, float* __restrict__ sums,
       const int c0 = oct << 3;
       float sacc[8] = {0, 0, 0, 0, 0, 0, 0, 0};
       float ss[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+      #pragma unroll 2
       for (int p2 = hw0 + tp; p2 < hw1; p2 += TP) {
         Bf16x8 chunk = *reinterpret_cast<const Bf16x8*>(xb + (long)p2 * C + c0);
 #pragma unroll
@@ -348,7 +353,8 @@ gn_apply_kernel(const bf16* __restrict__ x, bf16* __restrict__ y,
       mu[j] = s_mean[g];
       rs[j] = s_rstd[g];
     }
-    for (int p2 = hw0 + tp; p2 < hw1; p2 += TP) {
+    #pragma unroll 2
+      for (int p2 = hw0 + tp; p2 < hw1; p2 += TP) {
       const long off = (long)p2 * C + c0;
       Bf16x8 chunk = *reinterpret_cast<const Bf16x8*>(xb + off);
       Bf16x8 out;
@@ -408,6 +414,7 @@ gn_bwd_part_kernel(const bf16* __restrict__ x, const bf16* __restrict__ dz,
         rs[j] = rstd_in[(long)n * G + g];
       }
       float a1[8] = {0}, a2[8] = {0}, adg[8] = {0}, adb[8] = {0};
+      #pragma unroll 2
       for (int p2 = hw0 + tp; p2 < hw1; p2 += TP) {
         const long off = (long)p2 * C + c0;
         Bf16x8 xc = *reinterpret_cast<const Bf16x8*>(xb + off);
@@ -484,7 +491,8 @@ gn_bwd_apply_kernel(const bf16* __restrict__ x, const bf16* __restrict__ dz,
       k1[j] = s12[((long)n * G + g) * 2 + 0] * inv_m;
       k2[j] = s12[((long)n * G + g) * 2 + 1] * inv_m;
     }
-    for (int p2 = hw0 + tp; p2 < hw1; p2 += TP) {
+    #pragma unroll 2
+      for (int p2 = hw0 + tp; p2 < hw1; p2 += TP) {
       const long off = (long)p2 * C + c0;
       Bf16x8 xc = *reinterpret_cast<const Bf16x8*>(xb + off);
       Bf16x8 dc = *reinterpret_cast<const Bf16x8*>(db + off);
