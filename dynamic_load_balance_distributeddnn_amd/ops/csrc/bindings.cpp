@@ -18,71 +18,105 @@ static void sgd_momentum(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                    stream.stream());
 }
 
-extern "C" void dlb_gn_fwd(const void* x, void* y, const float* gamma,
-                           const float* beta, float* mean, float* rstd,
-                           float* scratch, int N, int HW, int C, int G,
-                           float eps, int relu, hipStream_t stream);
-extern "C" void dlb_gn_bwd(const void* x, const void* dz, void* dx,
-                           const float* gamma, const float* beta,
-                           const float* mean, const float* rstd, float* dgamma,
-                           float* dbeta, float* scratch, int N, int HW, int C,
-                           int G, int relu, hipStream_t stream);
+extern "C" void dlb_gn_fwd_segs(const void* const* xs, const int* starts,
+                                int nseg, void* y, const float* gamma,
+                                const float* beta, float* mean, float* rstd,
+                                float* scratch, int N, int HW, int C, int G,
+                                float eps, int relu, hipStream_t stream);
+extern "C" void dlb_gn_bwd_segs(const void* const* xs, const int* starts,
+                                int nseg, const void* dz, void* const* dxs,
+                                const float* gamma, const float* beta,
+                                const float* mean, const float* rstd,
+                                float* dgamma, float* dbeta, float* scratch,
+                                int N, int HW, int C, int G, int relu,
+                                hipStream_t stream);
 
-// x: [N, HW, C] bf16 contiguous (an NHWC view of a channels_last NCHW
-// tensor).  Returns (y, mean, rstd).
-static std::vector<torch::Tensor> gn_fwd(torch::Tensor x, torch::Tensor gamma,
+// Segments: [N, HW, Ci] bf16 contiguous views of channels_last tensors
+// forming a virtual channel-concat.  Returns (y packed, mean, rstd).
+static std::vector<torch::Tensor> gn_fwd(std::vector<torch::Tensor> xs,
+                                         torch::Tensor gamma,
                                          torch::Tensor beta, int64_t groups,
                                          double eps, bool relu) {
-  TORCH_CHECK(x.is_cuda() && x.dim() == 3 && x.is_contiguous());
-  TORCH_CHECK(x.scalar_type() == torch::kBFloat16, "bf16 activations only");
+  TORCH_CHECK(!xs.empty() && xs.size() <= 56);
+  int C = 0;
+  const void* ptrs[56];
+  int starts[57];
+  for (size_t i = 0; i < xs.size(); ++i) {
+    auto& x = xs[i];
+    TORCH_CHECK(x.is_cuda() && x.dim() == 3 && x.is_contiguous());
+    TORCH_CHECK(x.scalar_type() == torch::kBFloat16, "bf16 activations only");
+    TORCH_CHECK(x.size(2) % 8 == 0);
+    ptrs[i] = x.data_ptr();
+    starts[i] = C;
+    C += (int)x.size(2);
+  }
+  starts[xs.size()] = C;
   TORCH_CHECK(gamma.scalar_type() == torch::kFloat32);
-  const int N = x.size(0), HW = x.size(1), C = x.size(2);
-  TORCH_CHECK(C % 8 == 0, "C must be a multiple of 8");
+  const int N = xs[0].size(0), HW = xs[0].size(1);
   TORCH_CHECK(groups <= 64 && C % groups == 0);
-  auto y = torch::empty_like(x);
-  auto mean = torch::empty({N, groups}, x.options().dtype(torch::kFloat32));
+  auto y = torch::empty({N, HW, C}, xs[0].options());
+  auto mean = torch::empty({N, groups},
+                           xs[0].options().dtype(torch::kFloat32));
   auto rstd = torch::empty_like(mean);
   auto stream = at::hip::getCurrentHIPStream();
   float* scratch = nullptr;
   torch::Tensor scratch_t;
-  if (N < 192 && HW > 1) {  // small-batch path: split HW across blocks
+  if (N < 192 && HW > 1 && xs.size() == 1) {
     scratch_t = torch::zeros({N, groups, 2},
-                             x.options().dtype(torch::kFloat32));
+                             xs[0].options().dtype(torch::kFloat32));
     scratch = scratch_t.data_ptr<float>();
   }
-  dlb_gn_fwd(x.data_ptr(), y.data_ptr(), gamma.data_ptr<float>(),
-             beta.data_ptr<float>(), mean.data_ptr<float>(),
-             rstd.data_ptr<float>(), scratch, N, HW, C, (int)groups,
-             (float)eps, relu ? 1 : 0, stream.stream());
+  dlb_gn_fwd_segs(ptrs, starts, (int)xs.size(), y.data_ptr(),
+                  gamma.data_ptr<float>(), beta.data_ptr<float>(),
+                  mean.data_ptr<float>(), rstd.data_ptr<float>(), scratch, N,
+                  HW, C, (int)groups, (float)eps, relu ? 1 : 0,
+                  stream.stream());
   return {y, mean, rstd};
 }
 
-static std::vector<torch::Tensor> gn_bwd(torch::Tensor x, torch::Tensor dz,
+static std::vector<torch::Tensor> gn_bwd(std::vector<torch::Tensor> xs,
+                                         torch::Tensor dz,
                                          torch::Tensor gamma,
                                          torch::Tensor beta,
                                          torch::Tensor mean,
                                          torch::Tensor rstd, int64_t groups,
                                          bool relu) {
-  TORCH_CHECK(x.is_cuda() && x.dim() == 3 && x.is_contiguous());
-  TORCH_CHECK(dz.is_contiguous() && dz.sizes() == x.sizes());
-  const int N = x.size(0), HW = x.size(1), C = x.size(2);
-  auto dx = torch::empty_like(x);
-  auto dgamma = torch::zeros({C}, x.options().dtype(torch::kFloat32));
+  TORCH_CHECK(!xs.empty() && xs.size() <= 56);
+  int C = 0;
+  const void* ptrs[56];
+  void* dptrs[56];
+  int starts[57];
+  std::vector<torch::Tensor> out;
+  for (size_t i = 0; i < xs.size(); ++i) {
+    auto& x = xs[i];
+    TORCH_CHECK(x.is_cuda() && x.dim() == 3 && x.is_contiguous());
+    ptrs[i] = x.data_ptr();
+    starts[i] = C;
+    C += (int)x.size(2);
+    out.push_back(torch::empty_like(x));
+    dptrs[i] = out.back().data_ptr();
+  }
+  starts[xs.size()] = C;
+  const int N = xs[0].size(0), HW = xs[0].size(1);
+  TORCH_CHECK(dz.is_contiguous() && dz.size(2) == C);
+  auto dgamma = torch::zeros({C}, xs[0].options().dtype(torch::kFloat32));
   auto dbeta = torch::zeros_like(dgamma);
   auto stream = at::hip::getCurrentHIPStream();
   float* scratch = nullptr;
   torch::Tensor scratch_t;
-  if (N < 192 && HW > 1) {
+  if (N < 192 && HW > 1 && xs.size() == 1) {
     scratch_t = torch::zeros({N, groups, 2},
-                             x.options().dtype(torch::kFloat32));
+                             xs[0].options().dtype(torch::kFloat32));
     scratch = scratch_t.data_ptr<float>();
   }
-  dlb_gn_bwd(x.data_ptr(), dz.data_ptr(), dx.data_ptr(),
-             gamma.data_ptr<float>(), beta.data_ptr<float>(),
-             mean.data_ptr<float>(), rstd.data_ptr<float>(),
-             dgamma.data_ptr<float>(), dbeta.data_ptr<float>(), scratch, N,
-             HW, C, (int)groups, relu ? 1 : 0, stream.stream());
-  return {dx, dgamma, dbeta};
+  dlb_gn_bwd_segs(ptrs, starts, (int)xs.size(), dz.data_ptr(), dptrs,
+                  gamma.data_ptr<float>(), beta.data_ptr<float>(),
+                  mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                  dgamma.data_ptr<float>(), dbeta.data_ptr<float>(), scratch,
+                  N, HW, C, (int)groups, relu ? 1 : 0, stream.stream());
+  out.push_back(dgamma);
+  out.push_back(dbeta);
+  return out;  // [dx_0..dx_{k-1}, dgamma, dbeta]
 }
 
 extern "C" void dlb_conv_fwd(const void* x, const void* w, void* y,

@@ -138,3 +138,20 @@ def max_pool2d(x, k, stride=None, padding=0):
         from . import native
         return native.max_pool2d(x, k, stride, padding)
     return F.max_pool2d(x, k, stride=stride, padding=padding)
+
+
+def group_norm_act_cat(segs, num_groups, weight, bias, eps=1e-5, relu=False):
+    """GroupNorm(+ReLU) over a virtual channel-concat (DenseNet stream)."""
+    if (segs[0].is_cuda and _use_native("group_norm_act", segs[0])
+            and len(segs) <= 56
+            and all(s.dtype == torch.bfloat16 and s.shape[1] % 8 == 0
+                    for s in segs)
+            and weight.dtype == torch.float32
+            and sum(s.shape[1] for s in segs) % num_groups == 0
+            and num_groups <= 64):
+        from . import native
+        return native.group_norm_act_cat(segs, num_groups, weight, bias,
+                                         eps, relu)
+    x = torch.cat(list(segs), dim=1) if len(segs) > 1 else segs[0]
+    out = F.group_norm(x, num_groups, weight, bias, eps)
+    return F.relu(out, inplace=True) if relu else out

@@ -23,7 +23,8 @@ class _GroupNormAct(torch.autograd.Function):
     def forward(ctx, x, num_groups, weight, bias, eps, relu):
         n, c, h, w = x.shape
         x3 = _to_nhwc3(x)
-        y3, mean, rstd = ext().gn_fwd(x3, weight, bias, num_groups, eps, relu)
+        y3, mean, rstd = ext().gn_fwd([x3], weight, bias, num_groups, eps,
+                                      relu)
         ctx.save_for_backward(x3, weight, bias, mean, rstd)
         ctx.gn_dims = (n, c, h, w, num_groups, relu)
         return y3.view(n, h, w, c).permute(0, 3, 1, 2)
@@ -33,14 +34,50 @@ class _GroupNormAct(torch.autograd.Function):
         x3, weight, bias, mean, rstd = ctx.saved_tensors
         n, c, h, w, groups, relu = ctx.gn_dims
         dz3 = _to_nhwc3(dz)
-        dx3, dgamma, dbeta = ext().gn_bwd(x3, dz3, weight, bias, mean, rstd,
-                                          groups, relu)
+        dx3, dgamma, dbeta = ext().gn_bwd([x3], dz3, weight, bias, mean,
+                                          rstd, groups, relu)
         dx = dx3.view(n, h, w, c).permute(0, 3, 1, 2)
         return dx, None, dgamma, dbeta, None, None
 
 
 def group_norm_act(x, num_groups, weight, bias, eps=1e-5, relu=False):
     return _GroupNormAct.apply(x, num_groups, weight, bias, eps, relu)
+
+
+class _GroupNormActCat(torch.autograd.Function):
+    """GroupNorm(+ReLU) over a VIRTUAL channel-concat of segments.
+
+    DenseNet's residual stream is a concat of every previous layer's
+    output (reference Net/Densenet.py:20); reading the segments in place
+    removes the per-layer cat copy (and the grad-slice copies in
+    backward).  Output is packed NHWC; grads come back packed per
+    segment."""
+
+    @staticmethod
+    def forward(ctx, num_groups, weight, bias, eps, relu, *segs):
+        n, _, h, w = segs[0].shape
+        segs3 = [_to_nhwc3(s) for s in segs]
+        y3, mean, rstd = ext().gn_fwd(segs3, weight, bias, num_groups, eps,
+                                      relu)
+        ctx.save_for_backward(weight, bias, mean, rstd, *segs3)
+        ctx.gn_dims = (n, h, w, num_groups, relu)
+        c = y3.shape[-1]
+        return y3.view(n, h, w, c).permute(0, 3, 1, 2)
+
+    @staticmethod
+    def backward(ctx, dz):
+        weight, bias, mean, rstd, *segs3 = ctx.saved_tensors
+        n, h, w, groups, relu = ctx.gn_dims
+        dz3 = _to_nhwc3(dz)
+        outs = ext().gn_bwd(list(segs3), dz3, weight, bias, mean, rstd,
+                            groups, relu)
+        dxs, dgamma, dbeta = outs[:-2], outs[-2], outs[-1]
+        dx4 = [d.view(n, h, w, -1).permute(0, 3, 1, 2) for d in dxs]
+        return (None, dgamma, dbeta, None, None, *dx4)
+
+
+def group_norm_act_cat(segs, num_groups, weight, bias, eps=1e-5, relu=False):
+    return _GroupNormActCat.apply(num_groups, weight, bias, eps, relu, *segs)
 
 
 class _Conv2d(torch.autograd.Function):

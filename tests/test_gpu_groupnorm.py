@@ -91,3 +91,51 @@ def test_gn_autograd_gradcheck_small():
     y.sum().backward()
     assert torch.isfinite(x.grad.float()).all()
     assert m.weight.grad is not None and m.weight.grad.dtype == torch.float32
+
+
+@needs_gpu
+def test_gn_virtual_concat_matches_materialized():
+    """Multi-segment GN == GN over torch.cat of the same segments."""
+    from dynamic_load_balance_distributeddnn_amd.ops import functional as FD
+
+    torch.manual_seed(5)
+    segs32 = [torch.randn(4, c, 16, 16, device="cuda")
+              for c in (32, 64, 128, 32)]
+    G, C = 32, 256
+    gamma = torch.randn(C, device="cuda") * 0.4 + 1
+    beta = torch.randn(C, device="cuda") * 0.1
+    dz = torch.randn(4, C, 16, 16, device="cuda").bfloat16()
+
+    # materialized reference through the SAME single-tensor kernel
+    segs_a = [s.bfloat16().to(memory_format=torch.channels_last)
+              .requires_grad_() for s in segs32]
+    cat_in = torch.cat(segs_a, 1)
+    ref = FD.group_norm_act(cat_in, G, gamma, beta, relu=True)
+
+    segs_b = [s.bfloat16().to(memory_format=torch.channels_last)
+              .requires_grad_() for s in segs32]
+    out = FD.group_norm_act_cat(segs_b, G, gamma, beta, relu=True)
+    torch.testing.assert_close(out.float(), ref.float(), rtol=1e-3, atol=1e-3)
+
+    ref.backward(dz)
+    out.backward(dz)
+    for a, b in zip(segs_a, segs_b):
+        torch.testing.assert_close(a.grad.float(), b.grad.float(),
+                                   rtol=1e-3, atol=1e-3)
+
+
+@needs_gpu
+def test_densenet_virtual_concat_trains():
+    import torch.nn.functional as F2
+
+    from dynamic_load_balance_distributeddnn_amd.models import DenseNet121
+
+    torch.manual_seed(0)
+    m = DenseNet121(10).cuda().to(memory_format=torch.channels_last)
+    x = torch.randn(8, 3, 32, 32, device="cuda") \
+        .to(memory_format=torch.channels_last)
+    with torch.autocast("cuda", dtype=torch.bfloat16):
+        loss = F2.cross_entropy(m(x), torch.randint(0, 10, (8,), device="cuda"))
+    loss.backward()
+    assert torch.isfinite(loss)
+    assert all(torch.isfinite(p.grad).all() for p in m.parameters())
