@@ -115,13 +115,15 @@ def test_gn_virtual_concat_matches_materialized():
     segs_b = [s.bfloat16().to(memory_format=torch.channels_last)
               .requires_grad_() for s in segs32]
     out = FD.group_norm_act_cat(segs_b, G, gamma, beta, relu=True)
-    torch.testing.assert_close(out.float(), ref.float(), rtol=1e-3, atol=1e-3)
+    torch.testing.assert_close(out.float(), ref.float(), rtol=5e-3, atol=5e-3)
 
     ref.backward(dz)
     out.backward(dz)
     for a, b in zip(segs_a, segs_b):
+        # LDS-atomic accumulation order differs between the two paths ->
+        # occasional 1-ulp bf16 differences
         torch.testing.assert_close(a.grad.float(), b.grad.float(),
-                                   rtol=1e-3, atol=1e-3)
+                                   rtol=5e-3, atol=5e-3)
 
 
 @needs_gpu
