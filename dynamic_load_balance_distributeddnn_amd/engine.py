@@ -266,3 +266,29 @@ class Trainer:
             recorder.save()
         if self.logger:
             self.logger.info(f"Rank {self.rank} finished; total {wallclock:.1f}s")
+
+    # ------------------------------------------------------------------
+    # Checkpointing — an extension beyond the reference (which persists
+    # only logs + the rank-0 .npy recorder).  Weights live
+    # in the flat parameter arena, so a checkpoint is three flat tensors
+    # plus the scheduler state.
+    def save_checkpoint(self, path: str) -> None:
+        torch.save({
+            "params": self.optimizer.param_arena.cpu(),
+            "momentum": self.optimizer.momentum_buf.cpu(),
+            "batches": self.sched.batches,
+            "nodes_time": np.asarray(self.nodes_time),
+            "lr": self.optimizer.param_groups[0]["lr"],
+            "model": self.args.model,
+        }, path)
+
+    def load_checkpoint(self, path: str) -> None:
+        state = torch.load(path, map_location="cpu", weights_only=False)
+        assert state["model"] == self.args.model, "checkpoint/model mismatch"
+        with torch.no_grad():
+            self.optimizer.param_arena.copy_(state["params"].to(self.device))
+            self.optimizer.momentum_buf.copy_(
+                state["momentum"].to(self.device))
+        self.sched.batches = np.asarray(state["batches"])
+        self.nodes_time = np.asarray(state["nodes_time"])
+        self.optimizer.param_groups[0]["lr"] = float(state["lr"])

@@ -91,3 +91,26 @@ def test_recorder_layout(tmp_path):
     assert set(loaded.keys()) == {
         "epoch", "train_loss", "train_time", "sync_time", "val_loss",
         "accuracy", "partition", "node_time", "wallclock_time"}
+
+
+def test_checkpoint_roundtrip(tmp_path, monkeypatch):
+    monkeypatch.setenv("DLB_SYNTH_SCALE", "0.002")
+    import torch
+
+    from dynamic_load_balance_distributeddnn_amd.cli import get_parser
+    from dynamic_load_balance_distributeddnn_amd.engine import Trainer
+
+    args = get_parser().parse_args(
+        ["-d", "true", "-ws", "1", "-b", "8", "-e", "1", "-ds", "mnist",
+         "-m", "mnistnet"])
+    tr = Trainer(args, 0, 1, torch.device("cpu"), logger=None)
+    tr.train_epoch(0)
+    path = str(tmp_path / "ck.pt")
+    tr.save_checkpoint(path)
+    ref = tr.optimizer.param_arena.clone()
+
+    tr2 = Trainer(args, 0, 1, torch.device("cpu"), logger=None)
+    assert not torch.allclose(tr2.optimizer.param_arena, ref)
+    tr2.load_checkpoint(path)
+    assert torch.allclose(tr2.optimizer.param_arena, ref)
+    assert (tr2.sched.batches == tr.sched.batches).all()

@@ -150,3 +150,33 @@ def test_straggler_shifts_partition(free_port):
     assert batches[1] < batches[0]
     # converged near the 2:1 speed ratio -> ~(43, 21)
     assert batches[1] <= 24
+
+
+def _deferred_sync_worker(rank, world):
+    """LM-style deferred bucket launch: clip-then-reduce ordering."""
+    from dynamic_load_balance_distributeddnn_amd.parallel import \
+        GradientSynchronizer
+
+    torch.manual_seed(7)
+    model = torch.nn.Linear(16, 16)
+    sync = GradientSynchronizer(model, defer=True)
+    sync.set_weight(0.5)
+    torch.manual_seed(rank + 50)
+    x = torch.randn(4, 16)
+    sync.zero()
+    (model(x) ** 2).mean().backward()
+    # something between backward and reduce (the LM grad clip) must see
+    # LOCAL grads — no bucket may have launched yet
+    assert all(not b.launched for b in sync.buckets)
+    local = [p.grad.clone() for p in model.parameters()]
+    sync.finish()
+    return local, [p.grad.clone() for p in model.parameters()]
+
+
+def test_deferred_sync_launches_at_finish(free_port):
+    res = run_distributed(_deferred_sync_worker, 2, free_port)
+    local0, reduced0 = res[0]
+    local1, reduced1 = res[1]
+    for l0, l1, r0, r1 in zip(local0, local1, reduced0, reduced1):
+        assert torch.allclose(r0, r1, atol=1e-6)       # ranks agree
+        assert torch.allclose(r0, 0.5 * l0 + 0.5 * l1, atol=1e-6)
