@@ -88,7 +88,7 @@ gn_fwd_kernel(const GnSegs segs, bf16* __restrict__ y,
       const bf16* xb = sb + (long)n * HW * cs + cloc;
       float s[8] = {0, 0, 0, 0, 0, 0, 0, 0};
       float ss[8] = {0, 0, 0, 0, 0, 0, 0, 0};
-      #pragma unroll 2
+      #pragma unroll 4
       for (int p = tp; p < HW; p += TP) {
         Bf16x8 chunk = *reinterpret_cast<const Bf16x8*>(xb + (long)p * cs);
 #pragma unroll
@@ -137,7 +137,7 @@ gn_fwd_kernel(const GnSegs segs, bf16* __restrict__ y,
       mu[j] = s_mean[g];
       rs[j] = s_rstd[g];
     }
-    #pragma unroll 2
+    #pragma unroll 4
       for (int p = tp; p < HW; p += TP) {
       Bf16x8 chunk = *reinterpret_cast<const Bf16x8*>(xb + (long)p * cs);
       Bf16x8 out;
@@ -199,7 +199,7 @@ gn_bwd_kernel(const GnSegs segs, const bf16* __restrict__ dz,
         rs[j] = rstd_in[(long)n * G + g];
       }
       float a1[8] = {0}, a2[8] = {0}, adg[8] = {0}, adb[8] = {0};
-      #pragma unroll 2
+      #pragma unroll 4
       for (int p = tp; p < HW; p += TP) {
         Bf16x8 xc = *reinterpret_cast<const Bf16x8*>(xb + (long)p * cs);
         Bf16x8 dc = *reinterpret_cast<const Bf16x8*>(db + (long)p * C + c0);
@@ -257,7 +257,7 @@ gn_bwd_kernel(const GnSegs segs, const bf16* __restrict__ dz,
       k1[j] = s_s1[g] * inv_m;
       k2[j] = s_s2[g] * inv_m;
     }
-    #pragma unroll 2
+    #pragma unroll 4
       for (int p = tp; p < HW; p += TP) {
       Bf16x8 xc = *reinterpret_cast<const Bf16x8*>(xb + (long)p * cs);
       Bf16x8 dc = *reinterpret_cast<const Bf16x8*>(db + (long)p * C + c0);
@@ -310,7 +310,7 @@ gn_stats_part_kernel(const bf16* __restrict__ x, float* __restrict__ sums,
       const int c0 = oct << 3;
       float sacc[8] = {0, 0, 0, 0, 0, 0, 0, 0};
       float ss[8] = {0, 0, 0, 0, 0, 0, 0, 0};
-      #pragma unroll 2
+      #pragma unroll 4
       for (int p2 = hw0 + tp; p2 < hw1; p2 += TP) {
         Bf16x8 chunk = *reinterpret_cast<const Bf16x8*>(xb + (long)p2 * C + c0);
 #pragma unroll
@@ -385,7 +385,7 @@ gn_apply_kernel(const bf16* __restrict__ x, bf16* __restrict__ y,
       mu[j] = s_mean[g];
       rs[j] = s_rstd[g];
     }
-    #pragma unroll 2
+    #pragma unroll 4
       for (int p2 = hw0 + tp; p2 < hw1; p2 += TP) {
       const long off = (long)p2 * C + c0;
       Bf16x8 chunk = *reinterpret_cast<const Bf16x8*>(xb + off);
@@ -446,7 +446,7 @@ gn_bwd_part_kernel(const bf16* __restrict__ x, const bf16* __restrict__ dz,
         rs[j] = rstd_in[(long)n * G + g];
       }
       float a1[8] = {0}, a2[8] = {0}, adg[8] = {0}, adb[8] = {0};
-      #pragma unroll 2
+      #pragma unroll 4
       for (int p2 = hw0 + tp; p2 < hw1; p2 += TP) {
         const long off = (long)p2 * C + c0;
         Bf16x8 xc = *reinterpret_cast<const Bf16x8*>(xb + off);
@@ -523,7 +523,7 @@ gn_bwd_apply_kernel(const bf16* __restrict__ x, const bf16* __restrict__ dz,
       k1[j] = s12[((long)n * G + g) * 2 + 0] * inv_m;
       k2[j] = s12[((long)n * G + g) * 2 + 1] * inv_m;
     }
-    #pragma unroll 2
+    #pragma unroll 4
       for (int p2 = hw0 + tp; p2 < hw1; p2 += TP) {
       const long off = (long)p2 * C + c0;
       Bf16x8 xc = *reinterpret_cast<const Bf16x8*>(xb + off);
