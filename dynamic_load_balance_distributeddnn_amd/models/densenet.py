@@ -20,7 +20,6 @@ import math
 
 import torch
 import torch.nn as nn
-import torch.nn.functional as F
 
 from ..ops import functional as FD
 from ..ops.layers import Conv2d, GroupNormAct, Linear

@@ -13,7 +13,6 @@ from __future__ import annotations
 
 import torch
 import torch.nn as nn
-import torch.nn.functional as F
 
 from ..ops import functional as FD
 from ..ops.layers import Conv2d, GroupNormAct, Linear

@@ -34,7 +34,6 @@ conv3x3_fwd_halo(const bf16* __restrict__ x, const bf16* __restrict__ w,
                  const int N, const int H, const int W, const int Ci,
                  const int Co) {
   constexpr int TH = 8, TW = 16;
-  constexpr int NPIX = TH * TW;       // 128
   constexpr int HH = TH + 2, HW = TW + 2;  // 10 x 18 halo
   constexpr int FA = 2;               // 32 pixels per wave (4 waves x 32)
   constexpr int FB = BN / 16;
