@@ -274,18 +274,43 @@ conv_fwd_kernel(const ConvParams p) {
     buf ^= 1;
   }
 
+  // Epilogue through LDS, 32 output rows per bounce: the C/D fragment
+  // layout stores 2-byte values at 4 different rows per lane (~50% HBM
+  // write efficiency); LDS-bouncing turns the global writes into
+  // coalesced 16-byte rows.  a_lds[0] holds one 32-row chunk.
+  constexpr int LDO = BN + 8;
+  static_assert(BM * LDA >= 32 * LDO, "epilogue chunk must fit a_lds[0]");
+  bf16* o_lds = a_lds[0];
 #pragma unroll
-  for (int i = 0; i < FA; ++i) {
+  for (int ch = 0; ch < BM / 32; ++ch) {
+    __syncthreads();
 #pragma unroll
-    for (int j = 0; j < FB; ++j) {
-      const int col = n0 + wc * WTN + j * 16 + (lane & 15);
-      if (col >= p.Co) continue;
-      const float b = p.bias ? p.bias[col] : 0.f;
+    for (int i = 0; i < FA; ++i) {
+      if ((wr * WTM + i * 16) / 32 != ch) continue;  // wave-uniform guard
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int m = m0 + wr * WTM + i * 16 + (lane >> 4) * 4 + r;
-        if (m < p.M)
-          p.y[(long)m * p.Co + col] = __float2bfloat16(acc[i][j][r] + b);
+      for (int j = 0; j < FB; ++j) {
+        const int col = wc * WTN + j * 16 + (lane & 15);
+        const float b = p.bias ? (n0 + col < p.Co ? p.bias[n0 + col] : 0.f)
+                               : 0.f;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int row = (wr * WTM + i * 16 + (lane >> 4) * 4 + r) & 31;
+          o_lds[row * LDO + col] = __float2bfloat16(acc[i][j][r] + b);
+        }
+      }
+    }
+    __syncthreads();
+    for (int c = t; c < 32 * (BN / 8); c += CONV_BLOCK) {
+      const int row = c / (BN / 8);
+      const int c8 = (c % (BN / 8)) * 8;
+      const int m = m0 + ch * 32 + row;
+      if (m >= p.M) continue;
+      if (n0 + c8 + 7 < p.Co) {
+        *reinterpret_cast<bf16x8_t*>(p.y + (long)m * p.Co + n0 + c8) =
+            *reinterpret_cast<const bf16x8_t*>(&o_lds[row * LDO + c8]);
+      } else {
+        for (int j = 0; j < 8 && n0 + c8 + j < p.Co; ++j)
+          p.y[(long)m * p.Co + n0 + c8 + j] = o_lds[row * LDO + c8 + j];
       }
     }
   }
@@ -511,17 +536,38 @@ conv_bwd_data_kernel(const ConvBwdParams p) {
     buf ^= 1;
   }
 
+  // LDS-bounced epilogue, 32-row chunks (see the forward kernel)
+  constexpr int LDO = BN + 8;
+  static_assert(BM * LDA >= 32 * LDO, "epilogue chunk must fit a_lds[0]");
+  bf16* o_lds = a_lds[0];
 #pragma unroll
-  for (int i = 0; i < FA; ++i) {
+  for (int ch = 0; ch < BM / 32; ++ch) {
+    __syncthreads();
 #pragma unroll
-    for (int j = 0; j < FB; ++j) {
-      const int col = n0 + wc * WTN + j * 16 + (lane & 15);
-      if (col >= p.Ci) continue;
+    for (int i = 0; i < FA; ++i) {
+      if ((wr * WTM + i * 16) / 32 != ch) continue;
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int m = m0 + wr * WTM + i * 16 + (lane >> 4) * 4 + r;
-        if (m < p.M)
-          p.dx[(long)m * p.Ci + col] = __float2bfloat16(acc[i][j][r]);
+      for (int j = 0; j < FB; ++j) {
+        const int col = wc * WTN + j * 16 + (lane & 15);
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int row = (wr * WTM + i * 16 + (lane >> 4) * 4 + r) & 31;
+          o_lds[row * LDO + col] = __float2bfloat16(acc[i][j][r]);
+        }
+      }
+    }
+    __syncthreads();
+    for (int c = t; c < 32 * (BN / 8); c += CONV_BLOCK) {
+      const int row = c / (BN / 8);
+      const int c8 = (c % (BN / 8)) * 8;
+      const int m = m0 + ch * 32 + row;
+      if (m >= p.M) continue;
+      if (n0 + c8 + 7 < p.Ci) {
+        *reinterpret_cast<bf16x8_t*>(p.dx + (long)m * p.Ci + n0 + c8) =
+            *reinterpret_cast<const bf16x8_t*>(&o_lds[row * LDO + c8]);
+      } else {
+        for (int j = 0; j < 8 && n0 + c8 + j < p.Ci; ++j)
+          p.dx[(long)m * p.Ci + n0 + c8 + j] = o_lds[row * LDO + c8 + j];
       }
     }
   }
