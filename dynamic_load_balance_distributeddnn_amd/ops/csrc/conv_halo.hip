@@ -125,23 +125,39 @@ conv3x3_fwd_halo(const bf16* __restrict__ x, const bf16* __restrict__ w,
     __syncthreads();
   }
 
-  // ---- epilogue
+  // ---- epilogue through LDS (coalesced 16-byte stores; the fragment
+  // layout's native stores are 2-byte at 4 rows per lane).  wlds is
+  // free after the last MFMA: BN * WLD >= NPIXELS(128) * (BN + 8).
+  constexpr int LDO = BN + 8;
+  static_assert(BN * WLD >= 128 * LDO, "epilogue tile must fit wlds");
+  bf16* o_lds = wlds;
+  __syncthreads();
 #pragma unroll
   for (int i = 0; i < FA; ++i) {
 #pragma unroll
     for (int j = 0; j < FB; ++j) {
-      const int co = co0 + j * 16 + (lane & 15);
-      if (co >= Co) continue;
-      const float b = bias ? bias[co] : 0.f;
+      const int col = j * 16 + (lane & 15);
+      const float b = bias ? (co0 + col < Co ? bias[co0 + col] : 0.f) : 0.f;
 #pragma unroll
       for (int rr = 0; rr < 4; ++rr) {
         const int p = wave * 32 + i * 16 + (lane >> 4) * 4 + rr;
-        const int oh = oh0 + p / TW, ow = ow0 + p % TW;
-        if (oh < H && ow < W)
-          y[(((long)n * H + oh) * W + ow) * Co + co] =
-              __float2bfloat16(acc[i][j][rr] + b);
+        o_lds[p * LDO + col] = __float2bfloat16(acc[i][j][rr] + b);
       }
     }
+  }
+  __syncthreads();
+  for (int c = t; c < 128 * (BN / 8); c += HBLOCK) {
+    const int p = c / (BN / 8);
+    const int c8 = (c % (BN / 8)) * 8;
+    const int oh = oh0 + p / TW, ow = ow0 + p % TW;
+    if (oh >= H || ow >= W) continue;
+    bf16* dst = y + (((long)n * H + oh) * W + ow) * Co + co0 + c8;
+    if (co0 + c8 + 7 < Co)
+      *reinterpret_cast<bf16x8_t*>(dst) =
+          *reinterpret_cast<const bf16x8_t*>(&o_lds[p * LDO + c8]);
+    else
+      for (int j = 0; j < 8 && co0 + c8 + j < Co; ++j)
+        dst[j] = o_lds[p * LDO + c8 + j];
   }
 }
 
