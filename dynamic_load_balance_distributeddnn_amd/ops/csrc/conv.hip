@@ -805,7 +805,9 @@ extern "C" int dlb_conv_wrw_nsplits(int N, int OH, int OW, int Ci, int Co,
                                     int R, int S) {
   const int M = N * OH * OW;
   const int K = R * S * Ci;
-  const long tiles = (long)cdiv(Co, (Co >= 64) ? 64 : ((Co >= 32) ? 32 : 16))
+  const long tiles = (long)cdiv(Co, (Co >= 128) ? 128
+                                    : ((Co >= 64) ? 64 : ((Co >= 32) ? 32
+                                                                     : 16)))
                      * cdiv(K, 128);
   int splits = (int)std::min<long>(std::max<long>(1, 1024 / tiles),
                                    std::max<long>(1, M / (4 * 64)));
@@ -833,7 +835,11 @@ extern "C" void dlb_conv_wrw(const void* x, const void* dy, float* dw, int N,
   p.g.flat = (R == 1 && S == 1 && stride == 1 && pad == 0) ? 1 : 0;
   p.m_per_split = cdiv(cdiv(p.M, splits), 64) * 64;
   splits = cdiv(p.M, p.m_per_split);
-  if (Co >= 64) {
+  if (Co >= 128) {
+    dim3 grid(cdiv(Co, 128), cdiv(p.K, 128), splits);
+    hipLaunchKernelGGL((conv_wrw_kernel<128, 128>), grid, dim3(CONV_BLOCK), 0,
+                       stream, p);
+  } else if (Co >= 64) {
     dim3 grid(cdiv(Co, 64), cdiv(p.K, 128), splits);
     hipLaunchKernelGGL((conv_wrw_kernel<64, 128>), grid, dim3(CONV_BLOCK), 0,
                        stream, p);
