@@ -91,11 +91,8 @@ class DenseNet(nn.Module):
     def forward(self, x):
         segs = [self.stem(x)]
         for i, block in enumerate(self.blocks):
-            for layer in block:
-                fresh = layer(segs)
-                segs.insert(0, fresh)  # cat([fresh, x]) order
-            if i < len(self.transitions):
-                segs = [self.transitions[i](segs)]
+            trans = self.transitions[i] if i < len(self.transitions) else None
+            segs = FD.dense_block(block, trans, segs)
         out = self.final_norm(segs)
         out = FD.avg_pool2d(out, 4).flatten(1)
         return self.head(out)

@@ -29,7 +29,7 @@ extern "C" void dlb_gn_bwd_segs(const void* const* xs, const int* starts,
                                 const float* mean, const float* rstd,
                                 float* dgamma, float* dbeta, float* scratch,
                                 int N, int HW, int C, int G, int relu,
-                                hipStream_t stream);
+                                int accumulate, hipStream_t stream);
 
 // Segments: [N, HW, Ci] bf16 contiguous views of channels_last tensors
 // forming a virtual channel-concat.  Returns (y packed, mean, rstd).
@@ -80,12 +80,18 @@ static std::vector<torch::Tensor> gn_bwd(std::vector<torch::Tensor> xs,
                                          torch::Tensor beta,
                                          torch::Tensor mean,
                                          torch::Tensor rstd, int64_t groups,
-                                         bool relu) {
+                                         bool relu,
+                                         c10::optional<std::vector<torch::Tensor>>
+                                             dx_accum = c10::nullopt) {
+  // dx_accum: preallocated per-segment grad buffers — the kernel ADDS
+  // into them (the dense-stream manual backward), instead of allocating
+  // fresh outputs for autograd to sum pairwise.
   TORCH_CHECK(!xs.empty() && xs.size() <= 56);
   int C = 0;
   const void* ptrs[56];
   void* dptrs[56];
   int starts[57];
+  const bool acc = dx_accum.has_value();
   std::vector<torch::Tensor> out;
   for (size_t i = 0; i < xs.size(); ++i) {
     auto& x = xs[i];
@@ -93,7 +99,9 @@ static std::vector<torch::Tensor> gn_bwd(std::vector<torch::Tensor> xs,
     ptrs[i] = x.data_ptr();
     starts[i] = C;
     C += (int)x.size(2);
-    out.push_back(torch::empty_like(x));
+    out.push_back(acc ? (*dx_accum)[i] : torch::empty_like(x));
+    TORCH_CHECK(out.back().is_contiguous() &&
+                out.back().numel() == x.numel());
     dptrs[i] = out.back().data_ptr();
   }
   starts[xs.size()] = C;
@@ -113,7 +121,8 @@ static std::vector<torch::Tensor> gn_bwd(std::vector<torch::Tensor> xs,
                   gamma.data_ptr<float>(), beta.data_ptr<float>(),
                   mean.data_ptr<float>(), rstd.data_ptr<float>(),
                   dgamma.data_ptr<float>(), dbeta.data_ptr<float>(), scratch,
-                  N, HW, C, (int)groups, relu ? 1 : 0, stream.stream());
+                  N, HW, C, (int)groups, relu ? 1 : 0, acc ? 1 : 0,
+                  stream.stream());
   out.push_back(dgamma);
   out.push_back(dbeta);
   return out;  // [dx_0..dx_{k-1}, dgamma, dbeta]
@@ -465,7 +474,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sgd_momentum", &sgd_momentum,
         "Fused SGD momentum step over flat arenas (gfx950)");
   m.def("gn_fwd", &gn_fwd, "Fused GroupNorm(+ReLU) forward, NHWC bf16");
-  m.def("gn_bwd", &gn_bwd, "Fused GroupNorm(+ReLU) backward, NHWC bf16");
+  m.def("gn_bwd", &gn_bwd, "Fused GroupNorm(+ReLU) backward, NHWC bf16",
+        py::arg("xs"), py::arg("dz"), py::arg("gamma"), py::arg("beta"),
+        py::arg("mean"), py::arg("rstd"), py::arg("groups"), py::arg("relu"),
+        py::arg("dx_accum") = py::none());
   m.def("avgpool_fwd", &avgpool_fwd);
   m.def("avgpool_bwd", &avgpool_bwd);
   m.def("gavg_fwd", &gavg_fwd);

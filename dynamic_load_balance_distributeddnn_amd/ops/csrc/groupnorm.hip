@@ -164,7 +164,8 @@ gn_bwd_kernel(const GnSegs segs, const bf16* __restrict__ dz,
               const float* __restrict__ mean_in,
               const float* __restrict__ rstd_in,
               float* __restrict__ dgamma, float* __restrict__ dbeta,
-              const int HW, const int C, const int G, const int relu) {
+              const int HW, const int C, const int G, const int relu,
+              const int accumulate) {
   const int n = blockIdx.x;
   const int TC = C >> 3;
   const int TCe = TC < GN_BLOCK ? TC : GN_BLOCK;
@@ -262,6 +263,9 @@ gn_bwd_kernel(const GnSegs segs, const bf16* __restrict__ dz,
       Bf16x8 xc = *reinterpret_cast<const Bf16x8*>(xb + (long)p * cs);
       Bf16x8 dc = *reinterpret_cast<const Bf16x8*>(db + (long)p * C + c0);
       Bf16x8 out;
+      Bf16x8 prev;
+      if (accumulate)
+        prev = *reinterpret_cast<const Bf16x8*>(dxb + (long)p * cs);
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         float xhat = (bf2f(xc.v[j]) - mu[j]) * rs[j];
@@ -270,7 +274,9 @@ gn_bwd_kernel(const GnSegs segs, const bf16* __restrict__ dz,
           float yv = xhat * ga[j] + be[j];
           dy = yv > 0.f ? dy : 0.f;
         }
-        out.v[j] = f2bf(rs[j] * (ga[j] * dy - (k1[j] + xhat * k2[j])));
+        float v = rs[j] * (ga[j] * dy - (k1[j] + xhat * k2[j]));
+        if (accumulate) v += bf2f(prev.v[j]);
+        out.v[j] = f2bf(v);
       }
       *reinterpret_cast<Bf16x8*>(dxb + (long)p * cs) = out;
     }
@@ -579,7 +585,7 @@ extern "C" void dlb_gn_bwd_segs(const void* const* xs, const int* starts,
                                 const float* mean, const float* rstd,
                                 float* dgamma, float* dbeta, float* scratch,
                                 int N, int HW, int C, int G, int relu,
-                                hipStream_t stream) {
+                                int accumulate, hipStream_t stream) {
   GnSegs sg{};
   GnSegsMut dsg{};
   sg.nseg = nseg;
@@ -590,7 +596,7 @@ extern "C" void dlb_gn_bwd_segs(const void* const* xs, const int* starts,
   }
   sg.start[nseg] = starts[nseg];
   size_t shmem = 2 * (size_t)C * sizeof(float);
-  const int slices = (scratch && nseg == 1)
+  const int slices = (scratch && nseg == 1 && !accumulate)
                          ? (int)std::min<long>((383 + N) / N, HW) : 1;
   if (slices > 1) {
     dim3 grid(N, slices);
@@ -605,5 +611,5 @@ extern "C" void dlb_gn_bwd_segs(const void* const* xs, const int* starts,
   }
   hipLaunchKernelGGL(gn_bwd_kernel, dim3(N), dim3(GN_BLOCK), shmem, stream,
                      sg, (const bf16*)dz, dsg, gamma, beta,
-                     mean, rstd, dgamma, dbeta, HW, C, G, relu);
+                     mean, rstd, dgamma, dbeta, HW, C, G, relu, accumulate);
 }

@@ -1,0 +1,195 @@
+"""Manual backward over a whole DenseNet block (+ its transition).
+
+Why this exists: in the autograd virtual-concat path every segment of the
+dense residual stream is consumed by up to L later GroupNorms, so
+autograd materializes a pairwise bf16 `add` chain per segment per step
+(~530 extra elementwise kernels on DenseNet-121).  Here backward walks
+the block in reverse itself and the fused GroupNorm backward kernel
+ACCUMULATES each consumer's contribution directly into one grad buffer
+per segment (`gn_bwd(..., dx_accum=...)`), eliminating the adds.  The
+transition (or the external grads, for the last block) is processed
+first, so the buffers are written fresh — no zero-fill pass either.
+
+Numerics: the same kernels as the autograd path; per-segment grads add
+the same bf16 terms in a slightly different association order, so
+results match the autograd path to bf16 rounding.
+
+Parity anchor: reference Net/Densenet.py:14-33 (_DenseLayer / _Transition
+composition); the math is GN->ReLU->1x1 -> GN->ReLU->3x3 per layer with
+cat([out, x], 1), transitions GN->ReLU->1x1 -> avg_pool(2).
+"""
+
+from __future__ import annotations
+
+import torch
+
+from . import ext
+from .native import _to_nhwc3
+
+_EPS = 1e-5
+
+
+def _as4(x3, n, h, w):
+    return x3.view(n, h, w, -1).permute(0, 3, 1, 2)
+
+
+def _wcl(weight):
+    return weight.detach().to(torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last)
+
+
+def _conv_bwd_data(dy4, wcl, h, w, stride, pad):
+    co, ci, r, s = wcl.shape
+    if (r == 3 and s == 3 and stride == 1 and pad == 1 and co % 32 == 0
+            and ci >= 32):
+        # dx of 3x3/s1/p1 == same conv of dy with flip-transposed weights
+        wt = wcl.permute(0, 2, 3, 1).flip(1, 2).permute(3, 1, 2, 0) \
+            .contiguous().permute(0, 3, 1, 2)
+        return ext().conv_fwd(dy4, wt, None, 1, 1)
+    return ext().conv_bwd_data(dy4, wcl, h, w, stride, pad)
+
+
+class _DenseBlockFn(torch.autograd.Function):
+    """One dense block, optionally ending in its transition.
+
+    inputs:  meta=(nlayers, groups, has_transition), seg0 (4D cl bf16),
+             then per layer (g1, b1, w1, g2, b2, w2), then transition
+             (gt, bt, wt) when present.
+    outputs: the pooled transition tensor, or (when no transition) the
+             final segment list newest-first — each consumed exactly once
+             by the model's final norm, so no external grad adds appear.
+    """
+
+    @staticmethod
+    def forward(ctx, meta, seg0, *params):
+        nlayers, groups, has_trans = meta
+        n, _, h, w = seg0.shape
+        segs3 = [_to_nhwc3(seg0)]
+        saves = []
+        for li in range(nlayers):
+            g1, b1, w1, g2, b2, w2 = params[6 * li:6 * li + 6]
+            y1, m1, r1 = ext().gn_fwd(segs3, g1, b1, groups, _EPS, True)
+            w1c = _wcl(w1)
+            h1 = ext().conv_fwd(_as4(y1, n, h, w), w1c, None, 1, 0)
+            h13 = _to_nhwc3(h1)
+            y2, m2, r2 = ext().gn_fwd([h13], g2, b2, groups, _EPS, True)
+            w2c = _wcl(w2)
+            fresh = ext().conv_fwd(_as4(y2, n, h, w), w2c, None, 1, 1)
+            segs3.insert(0, _to_nhwc3(fresh))
+            saves += [y1, m1, r1, w1c, h13, m2, r2, y2, w2c]
+        if has_trans:
+            gt, bt, wt = params[6 * nlayers:6 * nlayers + 3]
+            yt, mt, rt = ext().gn_fwd(segs3, gt, bt, groups, _EPS, True)
+            wtc = _wcl(wt)
+            ht = ext().conv_fwd(_as4(yt, n, h, w), wtc, None, 1, 0)
+            out = ext().avgpool_fwd(ht, 2)
+            saves += [yt, mt, rt, wtc]
+        ctx.save_for_backward(*segs3, *saves, *params)
+        ctx.blk = (nlayers, groups, has_trans, n, h, w)
+        if has_trans:
+            return out
+        return tuple(_as4(s, n, h, w) for s in segs3)
+
+    @staticmethod
+    def backward(ctx, *douts):
+        nlayers, groups, has_trans, n, h, w = ctx.blk
+        nseg = nlayers + 1
+        segs3 = list(ctx.saved_tensors[:nseg])
+        saves = ctx.saved_tensors[nseg:]
+        params = saves[len(saves) - (6 * nlayers + (3 if has_trans else 0)):]
+        pgrads = [None] * len(params)
+
+        if has_trans:
+            yt, mt, rt, wtc = saves[9 * nlayers:9 * nlayers + 4]
+            gt, bt = params[6 * nlayers], params[6 * nlayers + 1]
+            dht = ext().avgpool_bwd(
+                douts[0].contiguous(memory_format=torch.channels_last),
+                2, h, w)
+            yt4 = _as4(yt, n, h, w)
+            dyt = _conv_bwd_data(dht, wtc, h, w, 1, 0)
+            dwt = ext().conv_wrw(yt4, dht, 1, 1, 1, 0)
+            co, ci = wtc.shape[0], wtc.shape[1]
+            pgrads[6 * nlayers + 2] = dwt.view(co, 1, 1, ci) \
+                .permute(0, 3, 1, 2)
+            # the transition norm is every segment's LAST consumer: its
+            # backward writes the per-segment grad buffers fresh
+            outs = ext().gn_bwd(segs3, _to_nhwc3(dyt), gt, bt, mt, rt,
+                                groups, True)
+            dsegs = list(outs[:-2])
+            pgrads[6 * nlayers] = outs[-2]
+            pgrads[6 * nlayers + 1] = outs[-1]
+        else:
+            # external grads arrive per segment (one consumer each);
+            # clone into owned buffers the kernels then accumulate into
+            dsegs = [_to_nhwc3(d).clone() for d in douts]
+
+        for li in range(nlayers - 1, -1, -1):
+            y1, m1, r1, w1c, h13, m2, r2, y2, w2c = saves[9 * li:9 * li + 9]
+            g1, b1 = params[6 * li], params[6 * li + 1]
+            g2, b2 = params[6 * li + 3], params[6 * li + 4]
+            in_segs = segs3[nlayers - li:]
+            dfresh4 = _as4(dsegs[nlayers - 1 - li], n, h, w)
+            y24 = _as4(y2, n, h, w)
+            dy2 = _conv_bwd_data(dfresh4, w2c, h, w, 1, 1)
+            dw2 = ext().conv_wrw(y24, dfresh4, 3, 3, 1, 1)
+            co2, ci2 = w2c.shape[0], w2c.shape[1]
+            pgrads[6 * li + 5] = dw2.view(co2, 3, 3, ci2).permute(0, 3, 1, 2)
+            dh1, dg2, db2 = ext().gn_bwd([h13], _to_nhwc3(dy2), g2, b2,
+                                         m2, r2, groups, True)
+            pgrads[6 * li + 3], pgrads[6 * li + 4] = dg2, db2
+            dh14 = _as4(dh1, n, h, w)
+            y14 = _as4(y1, n, h, w)
+            dy1 = _conv_bwd_data(dh14, w1c, h, w, 1, 0)
+            dw1 = ext().conv_wrw(y14, dh14, 1, 1, 1, 0)
+            co1, ci1 = w1c.shape[0], w1c.shape[1]
+            pgrads[6 * li + 2] = dw1.view(co1, 1, 1, ci1).permute(0, 3, 1, 2)
+            outs = ext().gn_bwd(in_segs, _to_nhwc3(dy1), g1, b1, m1, r1,
+                                groups, True,
+                                dx_accum=dsegs[nlayers - li:])
+            pgrads[6 * li], pgrads[6 * li + 1] = outs[-2], outs[-1]
+
+        dseg0 = _as4(dsegs[-1], n, h, w)
+        return (None, dseg0, *pgrads)
+
+
+def dense_block_forward(block, transition, segs):
+    """Run `block` (list of _DenseLayer) + optional `transition` through
+    the manual-backward Function.  `segs` is the incoming segment list
+    (length 1 in practice: stem output or previous transition)."""
+    params = []
+    for layer in block:
+        params += [layer.norm1.weight, layer.norm1.bias, layer.conv1.weight,
+                   layer.norm2.weight, layer.norm2.bias, layer.conv2.weight]
+    has_trans = transition is not None
+    if has_trans:
+        params += [transition.norm.weight, transition.norm.bias,
+                   transition.conv.weight]
+    meta = (len(block), block[0].norm1.num_groups, has_trans)
+    out = _DenseBlockFn.apply(meta, segs[0], *params)
+    return out if has_trans else list(out)
+
+
+def block_fn_ok(block, segs) -> bool:
+    """Envelope: single incoming bf16 channels_last segment, channel
+    counts in the fused-GN octet envelope, segment count within GN_MAXSEG,
+    all norms sharing one group count."""
+    if len(segs) != 1:
+        return False
+    x = segs[0]
+    if not (x.is_cuda and x.dtype == torch.bfloat16 and x.dim() == 4
+            and x.is_contiguous(memory_format=torch.channels_last)):
+        return False
+    if len(block) + 1 > 56:
+        return False
+    g = block[0].norm1.num_groups
+    growth = block[0].conv2.weight.shape[0]
+    mid = block[0].conv1.weight.shape[0]
+    cin0 = x.shape[1]
+    if growth % 8 or mid % 8 or cin0 % 8:
+        return False
+    for layer in block:
+        if layer.norm1.num_groups != g or layer.norm2.num_groups != g:
+            return False
+    # every layer's norm1 sees cin0 + k*growth channels; divisibility by g
+    # for all k needs both terms divisible (norm2 sees mid)
+    return cin0 % g == 0 and growth % g == 0 and mid % g == 0 and g <= 64

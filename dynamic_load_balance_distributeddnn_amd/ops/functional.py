@@ -139,6 +139,27 @@ def max_pool2d(x, k, stride=None, padding=0):
     return F.max_pool2d(x, k, stride=stride, padding=padding)
 
 
+def dense_block(block, transition, segs):
+    """One DenseNet block (+ optional transition) over the segment list.
+
+    GPU path: a single autograd Function with a manual reverse walk whose
+    GroupNorm backward ACCUMULATES per-segment grads inside the kernel
+    (ops/denseblock.py) — removes the ~530 autograd `add` kernels the
+    virtual-concat stream otherwise costs per step.  Fallback: the
+    per-layer module path (autograd composes the same kernels).
+    """
+    import os
+    if (_use_native("group_norm_act", segs[0])
+            and not os.environ.get("DLB_NO_BLOCK_FN")):
+        from . import denseblock
+        if denseblock.block_fn_ok(block, segs):
+            return denseblock.dense_block_forward(block, transition, segs)
+    segs = list(segs)
+    for layer in block:
+        segs.insert(0, layer(segs))
+    return [transition(segs)] if transition is not None else segs
+
+
 def group_norm_act_cat(segs, num_groups, weight, bias, eps=1e-5, relu=False):
     """GroupNorm(+ReLU) over a virtual channel-concat (DenseNet stream)."""
     if (segs[0].is_cuda and _use_native("group_norm_act", segs[0])
