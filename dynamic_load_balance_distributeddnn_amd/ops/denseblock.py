@@ -166,7 +166,7 @@ def dense_block_forward(block, transition, segs):
                    transition.conv.weight]
     meta = (len(block), block[0].norm1.num_groups, has_trans)
     out = _DenseBlockFn.apply(meta, segs[0], *params)
-    return out if has_trans else list(out)
+    return [out] if has_trans else list(out)
 
 
 def block_fn_ok(block, segs) -> bool:
