@@ -30,6 +30,7 @@ extern "C" void dlb_gn_bwd_segs(const void* const* xs, const int* starts,
                                 float* dgamma, float* dbeta, float* scratch,
                                 int N, int HW, int C, int G, int relu,
                                 int accumulate, hipStream_t stream);
+extern "C" int dlb_gn_nslices(int N, int HW, int bwd);
 
 // Segments: [N, HW, Ci] bf16 contiguous views of channels_last tensors
 // forming a virtual channel-concat.  Returns (y packed, mean, rstd).
@@ -61,8 +62,11 @@ static std::vector<torch::Tensor> gn_fwd(std::vector<torch::Tensor> xs,
   auto stream = at::hip::getCurrentHIPStream();
   float* scratch = nullptr;
   torch::Tensor scratch_t;
-  if (N < 192 && HW > 1 && xs.size() == 1) {
-    scratch_t = torch::zeros({N, groups, 2},
+  const int nsl_f = dlb_gn_nslices(N, HW, 0);
+  if (nsl_f > 1) {
+    // per-(sample, slice) partials, written without atomics — no
+    // zero-fill needed
+    scratch_t = torch::empty({N, nsl_f, groups, 2},
                              xs[0].options().dtype(torch::kFloat32));
     scratch = scratch_t.data_ptr<float>();
   }
@@ -107,13 +111,17 @@ static std::vector<torch::Tensor> gn_bwd(std::vector<torch::Tensor> xs,
   starts[xs.size()] = C;
   const int N = xs[0].size(0), HW = xs[0].size(1);
   TORCH_CHECK(dz.is_contiguous() && dz.size(2) == C);
-  auto dgamma = torch::zeros({C}, xs[0].options().dtype(torch::kFloat32));
-  auto dbeta = torch::zeros_like(dgamma);
+  // one zero-fill kernel for both reductions (they were 2 of the ~288
+  // fill launches per DenseNet step)
+  auto gbuf = torch::zeros({2, C}, xs[0].options().dtype(torch::kFloat32));
+  auto dgamma = gbuf[0];
+  auto dbeta = gbuf[1];
   auto stream = at::hip::getCurrentHIPStream();
   float* scratch = nullptr;
   torch::Tensor scratch_t;
-  if (N < 192 && HW > 1 && xs.size() == 1) {
-    scratch_t = torch::zeros({N, groups, 2},
+  const int nsl_b = dlb_gn_nslices(N, HW, 1);
+  if (nsl_b > 1) {
+    scratch_t = torch::empty({N, nsl_b, groups, 2},
                              xs[0].options().dtype(torch::kFloat32));
     scratch = scratch_t.data_ptr<float>();
   }
@@ -320,8 +328,9 @@ static std::vector<torch::Tensor> ln_bwd(torch::Tensor x, torch::Tensor dz,
   const long R = x.numel() / D;
   auto dzc = dz.contiguous();
   auto dx = torch::empty_like(x);
-  auto dgamma = torch::zeros({D}, x.options().dtype(torch::kFloat32));
-  auto dbeta = torch::zeros_like(dgamma);
+  auto gbuf = torch::zeros({2, D}, x.options().dtype(torch::kFloat32));
+  auto dgamma = gbuf[0];
+  auto dbeta = gbuf[1];
   dlb_ln_bwd(x.data_ptr(), dzc.data_ptr(), dx.data_ptr(),
              gamma.data_ptr<float>(), mean.data_ptr<float>(),
              rstd.data_ptr<float>(), dgamma.data_ptr<float>(),

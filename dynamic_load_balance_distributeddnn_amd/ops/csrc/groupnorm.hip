@@ -284,13 +284,18 @@ gn_bwd_kernel(const GnSegs segs, const bf16* __restrict__ dz,
 }
 
 
-// ---------------- small-batch path (N blocks underfill 256 CUs) ----------
-// grid (N, SLICES): pass A accumulates partial group sums into a global
-// fp32 buffer [N][G][2] with atomics; pass B recomputes stats from it
-// in-block and normalizes its HW slice.  Launched when N < 192.
+// ------------------- HW-sliced path (occupancy for any N) ---------------
+// One workgroup per sample gives only N blocks; at DenseNet's N=512 that
+// is 2 blocks/CU and the stream runs at ~25% of HBM speed (profiles/).
+// Slicing HW across grid.y multiplies blocks to DLB_GN_TARGET (~2048).
+// Pass A writes per-(sample, slice) group partials to scratch
+// [N][S][G][2] WITHOUT atomics (deterministic across runs, no zero-fill);
+// pass B reduces the S partials in-block and streams its slice.  Total
+// DRAM traffic is identical to the fused kernel (it also re-reads x for
+// the apply sweep); only the tiny stats buffer is extra.
 
 extern "C" __global__ void __launch_bounds__(GN_BLOCK)
-gn_stats_part_kernel(const bf16* __restrict__ x, float* __restrict__ sums,
+gn_stats_part_kernel(const GnSegs segs, float* __restrict__ part,
                      const int HW, const int C, const int G,
                      const int slices) {
   const int n = blockIdx.x;
@@ -310,15 +315,17 @@ gn_stats_part_kernel(const bf16* __restrict__ x, float* __restrict__ sums,
   for (int g = t; g < G; g += GN_BLOCK) { s_sum[g] = 0.f; s_ssq[g] = 0.f; }
   __syncthreads();
 
-  const bf16* xb = x + (long)n * HW * C;
   if (active) {
     for (int oct = tc; oct < TC; oct += TCe) {
       const int c0 = oct << 3;
+      int cloc, cs;
+      const bf16* sb = seg_locate(segs, c0, cloc, cs);
+      const bf16* xb = sb + (long)n * HW * cs + cloc;
       float sacc[8] = {0, 0, 0, 0, 0, 0, 0, 0};
       float ss[8] = {0, 0, 0, 0, 0, 0, 0, 0};
       #pragma unroll 4
       for (int p2 = hw0 + tp; p2 < hw1; p2 += TP) {
-        Bf16x8 chunk = *reinterpret_cast<const Bf16x8*>(xb + (long)p2 * C + c0);
+        Bf16x8 chunk = *reinterpret_cast<const Bf16x8*>(xb + (long)p2 * cs);
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           float v = bf2f(chunk.v[j]);
@@ -335,17 +342,18 @@ gn_stats_part_kernel(const bf16* __restrict__ x, float* __restrict__ sums,
     }
   }
   __syncthreads();
+  float* out = part + (((long)n * slices + sl) * G) * 2;
   for (int g = t; g < G; g += GN_BLOCK) {
-    atomicAdd(&sums[((long)n * G + g) * 2 + 0], s_sum[g]);
-    atomicAdd(&sums[((long)n * G + g) * 2 + 1], s_ssq[g]);
+    out[g * 2 + 0] = s_sum[g];
+    out[g * 2 + 1] = s_ssq[g];
   }
 }
 
 extern "C" __global__ void __launch_bounds__(GN_BLOCK)
-gn_apply_kernel(const bf16* __restrict__ x, bf16* __restrict__ y,
+gn_apply_kernel(const GnSegs segs, bf16* __restrict__ y,
                 const float* __restrict__ gamma,
                 const float* __restrict__ beta,
-                const float* __restrict__ sums, float* __restrict__ mean_out,
+                const float* __restrict__ part, float* __restrict__ mean_out,
                 float* __restrict__ rstd_out, const int HW, const int C,
                 const int G, const float eps, const int relu,
                 const int slices) {
@@ -364,9 +372,15 @@ gn_apply_kernel(const bf16* __restrict__ x, bf16* __restrict__ y,
   __shared__ float s_mean[GN_MAXG];
   __shared__ float s_rstd[GN_MAXG];
   const float inv_m = 1.0f / ((float)HW * Cg);
+  const float* pb = part + ((long)n * slices * G) * 2;
   for (int g = t; g < G; g += GN_BLOCK) {
-    float mu = sums[((long)n * G + g) * 2 + 0] * inv_m;
-    float var = sums[((long)n * G + g) * 2 + 1] * inv_m - mu * mu;
+    float su = 0.f, sq = 0.f;
+    for (int s2 = 0; s2 < slices; ++s2) {
+      su += pb[((long)s2 * G + g) * 2 + 0];
+      sq += pb[((long)s2 * G + g) * 2 + 1];
+    }
+    float mu = su * inv_m;
+    float var = sq * inv_m - mu * mu;
     float r = rsqrtf(var + eps);
     s_mean[g] = mu;
     s_rstd[g] = r;
@@ -378,10 +392,12 @@ gn_apply_kernel(const bf16* __restrict__ x, bf16* __restrict__ y,
   __syncthreads();
   if (!active) return;
 
-  const bf16* xb = x + (long)n * HW * C;
   bf16* yb = y + (long)n * HW * C;
   for (int oct = tc; oct < TC; oct += TCe) {
     const int c0 = oct << 3;
+    int cloc, cs;
+    const bf16* sb = seg_locate(segs, c0, cloc, cs);
+    const bf16* xb = sb + (long)n * HW * cs + cloc;
     float ga[8], be[8], mu[8], rs[8];
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
@@ -392,9 +408,8 @@ gn_apply_kernel(const bf16* __restrict__ x, bf16* __restrict__ y,
       rs[j] = s_rstd[g];
     }
     #pragma unroll 4
-      for (int p2 = hw0 + tp; p2 < hw1; p2 += TP) {
-      const long off = (long)p2 * C + c0;
-      Bf16x8 chunk = *reinterpret_cast<const Bf16x8*>(xb + off);
+    for (int p2 = hw0 + tp; p2 < hw1; p2 += TP) {
+      Bf16x8 chunk = *reinterpret_cast<const Bf16x8*>(xb + (long)p2 * cs);
       Bf16x8 out;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
@@ -402,20 +417,21 @@ gn_apply_kernel(const bf16* __restrict__ x, bf16* __restrict__ y,
         if (relu) v = fmaxf(v, 0.f);
         out.v[j] = f2bf(v);
       }
-      *reinterpret_cast<Bf16x8*>(yb + off) = out;
+      *reinterpret_cast<Bf16x8*>(yb + (long)p2 * C + c0) = out;
     }
   }
 }
 
-// backward small-batch: pass A accumulates s1/s2 (+dgamma/dbeta) into
-// global buffers; pass B computes dx per HW slice.
+// backward sliced: pass A writes per-(n, slice) s1/s2 partials (again no
+// atomics / no zero-fill) and publishes dgamma/dbeta via global atomics;
+// pass B reduces the partials and streams dx for its slice.
 extern "C" __global__ void __launch_bounds__(GN_BLOCK)
-gn_bwd_part_kernel(const bf16* __restrict__ x, const bf16* __restrict__ dz,
+gn_bwd_part_kernel(const GnSegs segs, const bf16* __restrict__ dz,
                    const float* __restrict__ gamma,
                    const float* __restrict__ beta,
                    const float* __restrict__ mean_in,
                    const float* __restrict__ rstd_in,
-                   float* __restrict__ s12, float* __restrict__ dgamma,
+                   float* __restrict__ part, float* __restrict__ dgamma,
                    float* __restrict__ dbeta, const int HW, const int C,
                    const int G, const int relu, const int slices) {
   const int n = blockIdx.x;
@@ -437,11 +453,13 @@ gn_bwd_part_kernel(const bf16* __restrict__ x, const bf16* __restrict__ dz,
   for (int c = t; c < 2 * C; c += GN_BLOCK) s_dgb[c] = 0.f;
   __syncthreads();
 
-  const bf16* xb = x + (long)n * HW * C;
   const bf16* db = dz + (long)n * HW * C;
   if (active) {
     for (int oct = tc; oct < TC; oct += TCe) {
       const int c0 = oct << 3;
+      int cloc, cs;
+      const bf16* sb = seg_locate(segs, c0, cloc, cs);
+      const bf16* xb = sb + (long)n * HW * cs + cloc;
       float ga[8], be[8], mu[8], rs[8];
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
@@ -454,9 +472,8 @@ gn_bwd_part_kernel(const bf16* __restrict__ x, const bf16* __restrict__ dz,
       float a1[8] = {0}, a2[8] = {0}, adg[8] = {0}, adb[8] = {0};
       #pragma unroll 4
       for (int p2 = hw0 + tp; p2 < hw1; p2 += TP) {
-        const long off = (long)p2 * C + c0;
-        Bf16x8 xc = *reinterpret_cast<const Bf16x8*>(xb + off);
-        Bf16x8 dc = *reinterpret_cast<const Bf16x8*>(db + off);
+        Bf16x8 xc = *reinterpret_cast<const Bf16x8*>(xb + (long)p2 * cs);
+        Bf16x8 dc = *reinterpret_cast<const Bf16x8*>(db + (long)p2 * C + c0);
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           float xhat = (bf2f(xc.v[j]) - mu[j]) * rs[j];
@@ -482,9 +499,10 @@ gn_bwd_part_kernel(const bf16* __restrict__ x, const bf16* __restrict__ dz,
     }
   }
   __syncthreads();
+  float* out = part + (((long)n * slices + sl) * G) * 2;
   for (int g = t; g < G; g += GN_BLOCK) {
-    atomicAdd(&s12[((long)n * G + g) * 2 + 0], s_s1[g]);
-    atomicAdd(&s12[((long)n * G + g) * 2 + 1], s_s2[g]);
+    out[g * 2 + 0] = s_s1[g];
+    out[g * 2 + 1] = s_s2[g];
   }
   for (int c = t; c < C; c += GN_BLOCK) {
     atomicAdd(&dgamma[c], s_dgb[c]);
@@ -493,13 +511,14 @@ gn_bwd_part_kernel(const bf16* __restrict__ x, const bf16* __restrict__ dz,
 }
 
 extern "C" __global__ void __launch_bounds__(GN_BLOCK)
-gn_bwd_apply_kernel(const bf16* __restrict__ x, const bf16* __restrict__ dz,
-                    bf16* __restrict__ dx, const float* __restrict__ gamma,
+gn_bwd_apply_kernel(const GnSegs segs, const bf16* __restrict__ dz,
+                    const GnSegsMut dxs, const float* __restrict__ gamma,
                     const float* __restrict__ beta,
                     const float* __restrict__ mean_in,
                     const float* __restrict__ rstd_in,
-                    const float* __restrict__ s12, const int HW, const int C,
-                    const int G, const int relu, const int slices) {
+                    const float* __restrict__ part, const int HW,
+                    const int C, const int G, const int relu,
+                    const int slices, const int accumulate) {
   const int n = blockIdx.x;
   const int sl = blockIdx.y;
   const int hw0 = (int)(((long)HW * sl) / slices);
@@ -513,11 +532,16 @@ gn_bwd_apply_kernel(const bf16* __restrict__ x, const bf16* __restrict__ dz,
   if (t >= TCe * TP) return;
 
   const float inv_m = 1.0f / ((float)HW * Cg);
-  const bf16* xb = x + (long)n * HW * C;
+  const float* pb = part + ((long)n * slices * G) * 2;
   const bf16* db = dz + (long)n * HW * C;
-  bf16* dxb = dx + (long)n * HW * C;
   for (int oct = tc; oct < TC; oct += TCe) {
     const int c0 = oct << 3;
+    int cloc, cs;
+    const bf16* sb = seg_locate(segs, c0, cloc, cs);
+    int si = 0;
+    while (si + 1 < segs.nseg && c0 >= segs.start[si + 1]) ++si;
+    const bf16* xb = sb + (long)n * HW * cs + cloc;
+    bf16* dxb = dxs.p[si] + (long)n * HW * cs + cloc;
     float ga[8], be[8], mu[8], rs[8], k1[8], k2[8];
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
@@ -526,15 +550,22 @@ gn_bwd_apply_kernel(const bf16* __restrict__ x, const bf16* __restrict__ dz,
       be[j] = beta[c];
       mu[j] = mean_in[(long)n * G + g];
       rs[j] = rstd_in[(long)n * G + g];
-      k1[j] = s12[((long)n * G + g) * 2 + 0] * inv_m;
-      k2[j] = s12[((long)n * G + g) * 2 + 1] * inv_m;
+      float su = 0.f, sq = 0.f;
+      for (int s2 = 0; s2 < slices; ++s2) {
+        su += pb[((long)s2 * G + g) * 2 + 0];
+        sq += pb[((long)s2 * G + g) * 2 + 1];
+      }
+      k1[j] = su * inv_m;
+      k2[j] = sq * inv_m;
     }
     #pragma unroll 4
-      for (int p2 = hw0 + tp; p2 < hw1; p2 += TP) {
-      const long off = (long)p2 * C + c0;
-      Bf16x8 xc = *reinterpret_cast<const Bf16x8*>(xb + off);
-      Bf16x8 dc = *reinterpret_cast<const Bf16x8*>(db + off);
+    for (int p2 = hw0 + tp; p2 < hw1; p2 += TP) {
+      Bf16x8 xc = *reinterpret_cast<const Bf16x8*>(xb + (long)p2 * cs);
+      Bf16x8 dc = *reinterpret_cast<const Bf16x8*>(db + (long)p2 * C + c0);
       Bf16x8 out;
+      Bf16x8 prev;
+      if (accumulate)
+        prev = *reinterpret_cast<const Bf16x8*>(dxb + (long)p2 * cs);
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         float xhat = (bf2f(xc.v[j]) - mu[j]) * rs[j];
@@ -543,13 +574,37 @@ gn_bwd_apply_kernel(const bf16* __restrict__ x, const bf16* __restrict__ dz,
           float yv = xhat * ga[j] + be[j];
           dy = yv > 0.f ? dy : 0.f;
         }
-        out.v[j] = f2bf(rs[j] * (ga[j] * dy - (k1[j] + xhat * k2[j])));
+        float v = rs[j] * (ga[j] * dy - (k1[j] + xhat * k2[j]));
+        if (accumulate) v += bf2f(prev.v[j]);
+        out.v[j] = f2bf(v);
       }
-      *reinterpret_cast<Bf16x8*>(dxb + off) = out;
+      *reinterpret_cast<Bf16x8*>(dxb + (long)p2 * cs) = out;
     }
   }
 }
+
 // ---------------------------------------------------------------- launchers
+static int gn_target_blocks(int bwd) {
+  static int cached[2] = {-1, -1};
+  if (cached[bwd] < 0) {
+    const char* e = getenv(bwd ? "DLB_GN_TARGET_BWD" : "DLB_GN_TARGET");
+    cached[bwd] = e ? atoi(e) : 512;
+    if (cached[bwd] < 1) cached[bwd] = 1;
+  }
+  return cached[bwd];
+}
+
+// slice count so that N*slices ~ target blocks (separate fwd/bwd knobs:
+// the bwd sliced pass multiplies dgamma/dbeta global-atomic traffic by
+// the slice count, so its profitable range is narrower)
+extern "C" int dlb_gn_nslices(int N, int HW, int bwd) {
+  int s = (gn_target_blocks(bwd) + N - 1) / N;
+  if (s > HW) s = HW;
+  if (s > 32) s = 32;
+  if (s < 1) s = 1;
+  return s;
+}
+
 extern "C" void dlb_gn_fwd_segs(const void* const* xs, const int* starts,
                                 int nseg, void* y, const float* gamma,
                                 const float* beta, float* mean, float* rstd,
@@ -562,15 +617,13 @@ extern "C" void dlb_gn_fwd_segs(const void* const* xs, const int* starts,
     sg.start[i] = starts[i];
   }
   sg.start[nseg] = starts[nseg];
-  // HW-sliced small-batch path handles the 1-segment case only
-  const int slices = (scratch && nseg == 1)
-                         ? (int)std::min<long>((383 + N) / N, HW) : 1;
+  const int slices = scratch ? dlb_gn_nslices(N, HW, 0) : 1;
   if (slices > 1) {
     dim3 grid(N, slices);
     hipLaunchKernelGGL(gn_stats_part_kernel, grid, dim3(GN_BLOCK), 0, stream,
-                       sg.p[0], scratch, HW, C, G, slices);
+                       sg, scratch, HW, C, G, slices);
     hipLaunchKernelGGL(gn_apply_kernel, grid, dim3(GN_BLOCK), 0, stream,
-                       sg.p[0], (bf16*)y, gamma, beta, scratch, mean,
+                       sg, (bf16*)y, gamma, beta, scratch, mean,
                        rstd, HW, C, G, eps, relu, slices);
     return;
   }
@@ -596,17 +649,17 @@ extern "C" void dlb_gn_bwd_segs(const void* const* xs, const int* starts,
   }
   sg.start[nseg] = starts[nseg];
   size_t shmem = 2 * (size_t)C * sizeof(float);
-  const int slices = (scratch && nseg == 1 && !accumulate)
-                         ? (int)std::min<long>((383 + N) / N, HW) : 1;
+  const int slices = scratch ? dlb_gn_nslices(N, HW, 1) : 1;
   if (slices > 1) {
     dim3 grid(N, slices);
     hipLaunchKernelGGL(gn_bwd_part_kernel, grid, dim3(GN_BLOCK), shmem,
-                       stream, sg.p[0], (const bf16*)dz, gamma, beta,
+                       stream, sg, (const bf16*)dz, gamma, beta,
                        mean, rstd, scratch, dgamma, dbeta, HW, C, G, relu,
                        slices);
     hipLaunchKernelGGL(gn_bwd_apply_kernel, grid, dim3(GN_BLOCK), 0, stream,
-                       sg.p[0], (const bf16*)dz, dsg.p[0], gamma,
-                       beta, mean, rstd, scratch, HW, C, G, relu, slices);
+                       sg, (const bf16*)dz, dsg, gamma,
+                       beta, mean, rstd, scratch, HW, C, G, relu, slices,
+                       accumulate);
     return;
   }
   hipLaunchKernelGGL(gn_bwd_kernel, dim3(N), dim3(GN_BLOCK), shmem, stream,
