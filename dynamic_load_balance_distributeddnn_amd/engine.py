@@ -289,6 +289,7 @@ class Trainer:
             self.optimizer.param_arena.copy_(state["params"].to(self.device))
             self.optimizer.momentum_buf.copy_(
                 state["momentum"].to(self.device))
+            self.optimizer.refresh_mirror()
         self.sched.batches = np.asarray(state["batches"])
         self.nodes_time = np.asarray(state["nodes_time"])
         self.optimizer.param_groups[0]["lr"] = float(state["lr"])

@@ -3,18 +3,25 @@
 #include <ATen/hip/HIPContext.h>
 
 extern "C" void dlb_sgd_momentum(float* p, const float* g, float* m,
-                                 float lr, float mu, long n,
+                                 void* q, float lr, float mu, long n,
                                  hipStream_t stream);
 
 static void sgd_momentum(torch::Tensor p, torch::Tensor g, torch::Tensor m,
-                         double lr, double mu) {
+                         double lr, double mu,
+                         c10::optional<torch::Tensor> mirror = c10::nullopt) {
   TORCH_CHECK(p.is_cuda() && g.is_cuda() && m.is_cuda(), "expects GPU tensors");
   TORCH_CHECK(p.is_contiguous() && g.is_contiguous() && m.is_contiguous());
   TORCH_CHECK(p.scalar_type() == torch::kFloat32, "fp32 master weights only");
   TORCH_CHECK(p.numel() == g.numel() && p.numel() == m.numel());
+  void* q = nullptr;
+  if (mirror.has_value()) {
+    TORCH_CHECK(mirror->scalar_type() == torch::kBFloat16 &&
+                mirror->is_contiguous() && mirror->numel() == p.numel());
+    q = mirror->data_ptr();
+  }
   auto stream = at::hip::getCurrentHIPStream();
   dlb_sgd_momentum(p.data_ptr<float>(), g.data_ptr<float>(),
-                   m.data_ptr<float>(), (float)lr, (float)mu, p.numel(),
+                   m.data_ptr<float>(), q, (float)lr, (float)mu, p.numel(),
                    stream.stream());
 }
 
@@ -481,6 +488,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv_bwd_data", &conv_bwd_data, "NHWC bf16 conv input-grad");
   m.def("conv_wrw", &conv_wrw, "NHWC bf16 conv weight-grad (fp32 out)");
   m.def("sgd_momentum", &sgd_momentum,
+        py::arg("p"), py::arg("g"), py::arg("m"), py::arg("lr"),
+        py::arg("mu"), py::arg("mirror") = py::none(),
         "Fused SGD momentum step over flat arenas (gfx950)");
   m.def("gn_fwd", &gn_fwd, "Fused GroupNorm(+ReLU) forward, NHWC bf16");
   m.def("gn_bwd", &gn_bwd, "Fused GroupNorm(+ReLU) backward, NHWC bf16",

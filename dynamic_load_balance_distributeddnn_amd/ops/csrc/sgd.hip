@@ -9,10 +9,15 @@
 
 #include "common.h"
 
+typedef __hip_bfloat16 bf16;
+
 extern "C" __global__ void __launch_bounds__(256)
 sgd_momentum_kernel(float* __restrict__ p, const float* __restrict__ g,
-                    float* __restrict__ m, const float lr, const float mu,
-                    const long n) {
+                    float* __restrict__ m, bf16* __restrict__ q,
+                    const float lr, const float mu, const long n) {
+  // q (optional): bf16 mirror of the updated parameters, written in the
+  // same pass — the conv forwards read weights from this mirror, which
+  // removes the ~120 per-tensor fp32->bf16 cast kernels per step.
   const long stride = (long)gridDim.x * blockDim.x * 4;
   for (long i = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 4; i < n;
        i += stride) {
@@ -30,23 +35,32 @@ sgd_momentum_kernel(float* __restrict__ p, const float* __restrict__ g,
       pv.w -= lr * mv.w;
       *reinterpret_cast<float4*>(m + i) = mv;
       *reinterpret_cast<float4*>(p + i) = pv;
+      if (q) {
+        union { bf16 h[4]; unsigned long long u; } qv;
+        qv.h[0] = __float2bfloat16(pv.x);
+        qv.h[1] = __float2bfloat16(pv.y);
+        qv.h[2] = __float2bfloat16(pv.z);
+        qv.h[3] = __float2bfloat16(pv.w);
+        *reinterpret_cast<unsigned long long*>(q + i) = qv.u;
+      }
     } else {
       for (long j = i; j < n; ++j) {
         float mv = mu * m[j] + g[j];
         m[j] = mv;
         p[j] -= lr * mv;
+        if (q) q[j] = __float2bfloat16(p[j]);
       }
     }
   }
 }
 
 extern "C" void dlb_sgd_momentum(float* p, const float* g, float* m,
-                                 float lr, float mu, long n,
+                                 void* q, float lr, float mu, long n,
                                  hipStream_t stream) {
   const int block = 256;
   // >> 256 workgroups to fill 8 XCDs; cap so the grid-stride loop amortizes
   int grid = (int)std::min<long>(cdiv((long)n, block * 4), 8192);
   if (grid == 0) grid = 1;
   hipLaunchKernelGGL(sgd_momentum_kernel, dim3(grid), dim3(block), 0, stream,
-                     p, g, m, lr, mu, n);
+                     p, g, m, (bf16*)q, lr, mu, n);
 }

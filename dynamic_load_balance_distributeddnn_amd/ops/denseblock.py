@@ -24,7 +24,7 @@ from __future__ import annotations
 import torch
 
 from . import ext
-from .native import _to_nhwc3
+from .native import _to_nhwc3, weight_bf16
 
 _EPS = 1e-5
 
@@ -33,9 +33,7 @@ def _as4(x3, n, h, w):
     return x3.view(n, h, w, -1).permute(0, 3, 1, 2)
 
 
-def _wcl(weight):
-    return weight.detach().to(torch.bfloat16) \
-        .contiguous(memory_format=torch.channels_last)
+_wcl = weight_bf16
 
 
 def _conv_bwd_data(dy4, wcl, h, w, stride, pad):

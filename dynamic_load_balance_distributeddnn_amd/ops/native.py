@@ -18,6 +18,18 @@ def _to_nhwc3(x: torch.Tensor) -> torch.Tensor:
     return x3 if x3.is_contiguous() else x3.contiguous()
 
 
+def weight_bf16(weight: torch.Tensor) -> torch.Tensor:
+    """bf16 channels_last view of a conv weight.  When the parameter
+    lives in the flat arena with a bf16 mirror (FlatSGD), this is a free
+    view of the mirror (the SGD kernel keeps it in sync); otherwise a
+    per-call cast copy."""
+    m = getattr(weight, "_dlb_bf16", None)
+    if m is not None:
+        return m
+    return weight.detach().to(torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last)
+
+
 class _GroupNormAct(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, num_groups, weight, bias, eps, relu):
@@ -91,8 +103,7 @@ class _Conv2d(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, weight, bias, stride, padding):
-        wcl = weight.detach().to(torch.bfloat16) \
-            .contiguous(memory_format=torch.channels_last)
+        wcl = weight_bf16(weight)
         b32 = bias.detach().float() if bias is not None else None
         y = ext().conv_fwd(x, wcl, b32, stride, padding)
         ctx.save_for_backward(x, wcl)
@@ -260,8 +271,7 @@ class _GroupedConv2d(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, weight, stride):
-        wcl = weight.detach().to(torch.bfloat16) \
-            .contiguous(memory_format=torch.channels_last)
+        wcl = weight_bf16(weight)
         y = ext().gconv_fwd(x, wcl, stride)
         ctx.save_for_backward(x, wcl)
         ctx.gconv_stride = stride

@@ -33,12 +33,29 @@ class FlatSGD:
         self.momentum_buf = torch.zeros(total, dtype=torch.float32, device=device)
         from .grad_sync import _strided_view
 
+        # bf16 weight mirror: the SGD kernel writes updated params as
+        # bf16 into this arena in the same pass, and conv forwards read
+        # weights from per-param views of it — removing the ~120
+        # per-tensor fp32->bf16 cast kernels per step (profiles/).
+        self.bf16_mirror = (torch.empty(total, dtype=torch.bfloat16,
+                                        device=device)
+                            if self.param_arena.is_cuda else None)
         with torch.no_grad():
             for p in sync.params:
                 off, n = sync.offsets[id(p)]
                 view = _strided_view(self.param_arena, off, p)
                 view.copy_(p.data)
                 p.data = view  # re-home the parameter into the arena
+                if self.bf16_mirror is not None and p.dim() == 4:
+                    p._dlb_bf16 = _strided_view(self.bf16_mirror, off, p)
+            self.refresh_mirror()
+
+    @torch.no_grad()
+    def refresh_mirror(self) -> None:
+        """Re-sync the bf16 weight mirror after any out-of-band write to
+        the parameter arena (initial sync, checkpoint load)."""
+        if self.bf16_mirror is not None:
+            self.bf16_mirror.copy_(self.param_arena)
 
     @torch.no_grad()
     def step(self) -> None:
@@ -48,7 +65,7 @@ class FlatSGD:
             from ..ops import ext
 
             ext().sgd_momentum(self.param_arena, g, self.momentum_buf,
-                               lr, self.momentum)
+                               lr, self.momentum, self.bf16_mirror)
         else:
             self.momentum_buf.mul_(self.momentum).add_(g)
             self.param_arena.add_(self.momentum_buf, alpha=-lr)
