@@ -147,6 +147,9 @@ extern "C" void dlb_conv_fwd(const void* x, const void* w, void* y,
                              const float* bias, int N, int IH, int IW, int Ci,
                              int OH, int OW, int Co, int R, int S, int stride,
                              int pad, hipStream_t stream);
+extern "C" bool dlb_conv3x3_bwd_halo(const void* dy, const void* w, void* dx,
+                                     int N, int H, int W, int Ci, int Co,
+                                     hipStream_t stream);
 extern "C" void dlb_conv_bwd_data(const void* dy, const void* wt, void* dx,
                                   int N, int IH, int IW, int Ci, int OH,
                                   int OW, int Co, int R, int S, int stride,
@@ -198,6 +201,12 @@ static torch::Tensor conv_bwd_data(torch::Tensor dy, torch::Tensor w,
   auto dx = torch::empty({N, Ci, IH, IW},
                          dy.options().memory_format(torch::MemoryFormat::ChannelsLast));
   auto stream = at::hip::getCurrentHIPStream();
+  // 3x3/s1/p1: the LDS-halo kernel in transpose-read mode reads the
+  // weight unmodified (no flip+copy transform, ~2x the generic kernel)
+  if (R == 3 && S == 3 && stride == 1 && pad == 1 && IH == OH && IW == OW &&
+      dlb_conv3x3_bwd_halo(dy.data_ptr(), w.data_ptr(), dx.data_ptr(), N,
+                           (int)IH, (int)IW, Ci, Co, stream.stream()))
+    return dx;
   dlb_conv_bwd_data(dy.data_ptr(), w.data_ptr(), dx.data_ptr(), N, (int)IH,
                     (int)IW, Ci, OH, OW, Co, R, S, (int)stride,
                     (int)pad, stream.stream());

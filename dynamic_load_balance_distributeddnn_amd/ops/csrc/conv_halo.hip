@@ -21,13 +21,43 @@ typedef __attribute__((ext_vector_type(4))) float f32x4;
 #define HBLOCK 256
 #define LDS3H __attribute__((address_space(3)))
 typedef __attribute__((__vector_size__(4 * sizeof(__bf16)))) __bf16 trvec_h;
+
+// MFMA fragment via hardware transpose reads from an LDS image laid out
+// [k rows][n cols] with row stride LROW (same recipe as conv.hip; lane
+// semantics verified on hardware, tools/mfma_probe/).
+template <int LROW>
+__device__ inline bf16x8_t tr_frag_h(const __hip_bfloat16* img, int mbase,
+                                     int colbase, int lane) {
+  const int j15 = lane & 15, q = lane >> 4;
+  const int row = mbase + q * 8 + (j15 >> 2);
+  const int col = colbase + 4 * (j15 & 3);
+  auto p0 = (LDS3H trvec_h*)((LDS3H __hip_bfloat16*)img + (long)row * LROW + col);
+  auto p1 = (LDS3H trvec_h*)((LDS3H __hip_bfloat16*)img +
+                             (long)(row + 4) * LROW + col);
+  trvec_h lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p0);
+  trvec_h hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p1);
+  union { struct { trvec_h a, b; } t; bf16x8_t v; } u;
+  u.t.a = lo;
+  u.t.b = hi;
+  return u.v;
+}
 #define CI_CHUNK 32
 #define HPAD 40  // padded ci stride in halo LDS rows (conflict-free, 16B-aligned)
 
 // ---------------------------------------------------------------- forward
 // Output tile: TH x TW pixels (8x16) for one sample, BN output channels.
 // K-loop over ci chunks of 32; inner loop over the 9 taps.
-template <int BN>
+// WTR=false: w is the conv weight in channels_last memory [Co][3][3][Ci]
+//            (B rows staged naturally, k = ci contiguous).
+// WTR=true:  computes the DATA GRADIENT of a 3x3/s1/p1 conv as a halo
+//            forward over dy: roles swap (kernel Ci = original Co is the
+//            reduction, kernel Co = original Ci is the output), w stays
+//            in NATURAL channels_last memory [k=Co_orig][3][3][n=Ci_orig]
+//            — rows are staged k-major into LDS and B fragments use
+//            hardware transpose reads with the tap FLIPPED (r'=2-r,
+//            s'=2-s).  This replaces the flip+copy weight transform the
+//            python path used to materialize per call.
+template <int BN, bool WTR>
 __global__ void __launch_bounds__(HBLOCK)
 conv3x3_fwd_halo(const bf16* __restrict__ x, const bf16* __restrict__ w,
                  bf16* __restrict__ y, const float* __restrict__ bias,
@@ -39,9 +69,11 @@ conv3x3_fwd_halo(const bf16* __restrict__ x, const bf16* __restrict__ w,
   constexpr int FB = BN / 16;
   constexpr int WK = 9 * CI_CHUNK;    // staged weight k-extent (288)
   constexpr int WLD = WK + 8;
+  constexpr int LW = BN + 8;          // WTR row stride ([tap*32+k][n])
+  constexpr int WSZ = WTR ? 9 * CI_CHUNK * LW : BN * WLD;
 
   __shared__ bf16 halo[HH * HW * HPAD];
-  __shared__ bf16 wlds[BN * WLD];
+  __shared__ bf16 wlds[WSZ];
 
   const int t = threadIdx.x;
   const int lane = t & 63;
@@ -81,6 +113,25 @@ conv3x3_fwd_halo(const bf16* __restrict__ x, const bf16* __restrict__ w,
       u2.q.z = ok ? u2.q.z : 0; u2.q.w = ok ? u2.q.w : 0;
       *reinterpret_cast<bf16x8_t*>(&halo[(hh * HW + ww) * HPAD + c8]) = u2.h;
     }
+    if (WTR) {
+      // ---- stage natural weight rows [tap*32 + k][n] (n contiguous)
+      constexpr int WCH2 = 9 * CI_CHUNK * (BN / 8);
+      for (int c = t; c < WCH2; c += HBLOCK) {
+        const int c8 = (c % (BN / 8)) * 8;           // n (= original ci)
+        const int rest = c / (BN / 8);
+        const int tap = rest % 9;
+        const int kl = rest / 9;                     // k (= original co)
+        const bool ok = co0 + c8 < Co;
+        const long off =
+            ok ? ((long)(ci0 + kl) * 9 + tap) * Co + co0 + c8 : 0;
+        bf16x8_t v = *reinterpret_cast<const bf16x8_t*>(w + off);
+        union { bf16x8_t h; int4 q; } u2; u2.h = v;
+        u2.q.x = ok ? u2.q.x : 0; u2.q.y = ok ? u2.q.y : 0;
+        u2.q.z = ok ? u2.q.z : 0; u2.q.w = ok ? u2.q.w : 0;
+        *reinterpret_cast<bf16x8_t*>(
+            &wlds[(tap * CI_CHUNK + kl) * LW + c8]) = u2.h;
+      }
+    } else {
     // ---- stage weights [co][tap*32+ci] from w[co][tap*Ci + ci]
     constexpr int WCH = BN * 9 * (CI_CHUNK / 8);
     for (int c = t; c < WCH; c += HBLOCK) {
@@ -95,6 +146,7 @@ conv3x3_fwd_halo(const bf16* __restrict__ x, const bf16* __restrict__ w,
       u2.q.x = ok ? u2.q.x : 0; u2.q.y = ok ? u2.q.y : 0;
       u2.q.z = ok ? u2.q.z : 0; u2.q.w = ok ? u2.q.w : 0;
       *reinterpret_cast<bf16x8_t*>(&wlds[co * WLD + tap * CI_CHUNK + c8]) = u2.h;
+    }
     }
     __syncthreads();
 
@@ -111,9 +163,13 @@ conv3x3_fwd_halo(const bf16* __restrict__ x, const bf16* __restrict__ w,
       }
 #pragma unroll
       for (int j = 0; j < FB; ++j) {
-        const int co = j * 16 + (lane & 15);
-        bfrag[j] = *reinterpret_cast<const bf16x8_t*>(
-            &wlds[co * WLD + tap * CI_CHUNK + (lane >> 4) * 8]);
+        if (WTR) {
+          bfrag[j] = tr_frag_h<LW>(wlds, (8 - tap) * CI_CHUNK, j * 16, lane);
+        } else {
+          const int co = j * 16 + (lane & 15);
+          bfrag[j] = *reinterpret_cast<const bf16x8_t*>(
+              &wlds[co * WLD + tap * CI_CHUNK + (lane >> 4) * 8]);
+        }
       }
 #pragma unroll
       for (int i = 0; i < FA; ++i)
@@ -129,7 +185,7 @@ conv3x3_fwd_halo(const bf16* __restrict__ x, const bf16* __restrict__ w,
   // layout's native stores are 2-byte at 4 rows per lane).  wlds is
   // free after the last MFMA: BN * WLD >= NPIXELS(128) * (BN + 8).
   constexpr int LDO = BN + 8;
-  static_assert(BN * WLD >= 128 * LDO, "epilogue tile must fit wlds");
+  static_assert(WSZ >= 128 * LDO, "epilogue tile must fit wlds");
   bf16* o_lds = wlds;
   __syncthreads();
 #pragma unroll
@@ -325,14 +381,37 @@ extern "C" bool dlb_conv3x3_fwd_halo(const void* x, const void* w, void* y,
   if (Co % 32 != 0 && Co < 32) return false;
   if (Co % 64 == 0 && Co >= 64) {
     dim3 grid(tiles, cdiv(Co, 64));
-    hipLaunchKernelGGL((conv3x3_fwd_halo<64>), grid, dim3(HBLOCK), 0, stream,
-                       (const bf16*)x, (const bf16*)w, (bf16*)y, bias, N, H,
-                       W, Ci, Co);
+    hipLaunchKernelGGL((conv3x3_fwd_halo<64, false>), grid, dim3(HBLOCK), 0,
+                       stream, (const bf16*)x, (const bf16*)w, (bf16*)y,
+                       bias, N, H, W, Ci, Co);
   } else {
     dim3 grid(tiles, cdiv(Co, 32));
-    hipLaunchKernelGGL((conv3x3_fwd_halo<32>), grid, dim3(HBLOCK), 0, stream,
-                       (const bf16*)x, (const bf16*)w, (bf16*)y, bias, N, H,
-                       W, Ci, Co);
+    hipLaunchKernelGGL((conv3x3_fwd_halo<32, false>), grid, dim3(HBLOCK), 0,
+                       stream, (const bf16*)x, (const bf16*)w, (bf16*)y,
+                       bias, N, H, W, Ci, Co);
+  }
+  return true;
+}
+
+// Data gradient of a 3x3/s1/p1 conv via the halo kernel in WTR mode:
+// dx[N,H,W,Ci] from dy[N,H,W,Co] and the UNMODIFIED channels_last weight.
+extern "C" bool dlb_conv3x3_bwd_halo(const void* dy, const void* w, void* dx,
+                                     int N, int H, int W, int Ci, int Co,
+                                     hipStream_t stream) {
+  // kernel roles: "Ci" = Co (reduction), "Co" = Ci (output channels)
+  if (Co % CI_CHUNK != 0) return false;
+  if (Ci % 8 != 0) return false;
+  const int tiles = N * cdiv(H, 8) * cdiv(W, 16);
+  if (Ci % 64 == 0 && Ci >= 64) {
+    dim3 grid(tiles, cdiv(Ci, 64));
+    hipLaunchKernelGGL((conv3x3_fwd_halo<64, true>), grid, dim3(HBLOCK), 0,
+                       stream, (const bf16*)dy, (const bf16*)w, (bf16*)dx,
+                       (const float*)nullptr, N, H, W, Co, Ci);
+  } else {
+    dim3 grid(tiles, cdiv(Ci, 32));
+    hipLaunchKernelGGL((conv3x3_fwd_halo<32, true>), grid, dim3(HBLOCK), 0,
+                       stream, (const bf16*)dy, (const bf16*)w, (bf16*)dx,
+                       (const float*)nullptr, N, H, W, Co, Ci);
   }
   return true;
 }

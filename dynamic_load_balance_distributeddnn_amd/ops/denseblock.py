@@ -37,13 +37,8 @@ _wcl = weight_bf16
 
 
 def _conv_bwd_data(dy4, wcl, h, w, stride, pad):
-    co, ci, r, s = wcl.shape
-    if (r == 3 and s == 3 and stride == 1 and pad == 1 and co % 32 == 0
-            and ci >= 32):
-        # dx of 3x3/s1/p1 == same conv of dy with flip-transposed weights
-        wt = wcl.permute(0, 2, 3, 1).flip(1, 2).permute(3, 1, 2, 0) \
-            .contiguous().permute(0, 3, 1, 2)
-        return ext().conv_fwd(dy4, wt, None, 1, 1)
+    # 3x3/s1/p1 dispatches to the halo kernel's transpose-read mode
+    # inside the binding (flip-free data gradient)
     return ext().conv_bwd_data(dy4, wcl, h, w, stride, pad)
 
 

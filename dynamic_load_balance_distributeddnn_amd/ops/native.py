@@ -119,16 +119,10 @@ class _Conv2d(torch.autograd.Function):
 
         dx = None
         if ctx.needs_input_grad[0]:
-            if (r == 3 and s == 3 and stride == 1 and padding == 1
-                    and co % 32 == 0 and (ci % 32 == 0 or ci >= 32)):
-                # dx of a 3x3/s1/p1 conv IS a 3x3/s1/p1 conv of dy with the
-                # flip-transposed weight — use the fast halo forward kernel
-                wt = wcl.permute(0, 2, 3, 1).flip(1, 2)                     .permute(3, 1, 2, 0).contiguous()      # [Ci,R,S,Co]
-                wt = wt.permute(0, 3, 1, 2)                # cl [Ci,Co,R,S]
-                dx = ext().conv_fwd(dy, wt, None, 1, 1)
-            else:
-                dx = ext().conv_bwd_data(dy, wcl, x.size(2), x.size(3),
-                                         stride, padding)
+            # 3x3/s1/p1 shapes dispatch to the halo kernel's
+            # transpose-read mode inside the binding (flip-free)
+            dx = ext().conv_bwd_data(dy, wcl, x.size(2), x.size(3),
+                                     stride, padding)
 
         dweight = None
         if ctx.needs_input_grad[1]:
