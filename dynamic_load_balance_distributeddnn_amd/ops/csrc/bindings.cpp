@@ -38,6 +38,22 @@ extern "C" void dlb_gn_bwd_segs(const void* const* xs, const int* starts,
                                 int N, int HW, int C, int G, int relu,
                                 int accumulate, hipStream_t stream);
 extern "C" int dlb_gn_nslices(int N, int HW, int bwd);
+extern "C" void dlb_gn_stats_segs(const void* const* xs, const int* starts,
+                                  int nseg, float* mean, float* rstd, int N,
+                                  int HW, int C, int G, float eps,
+                                  hipStream_t stream);
+extern "C" void dlb_gnconv1x1_fwd(const void* const* xs, const int* starts,
+                                  int nseg, const float* mean,
+                                  const float* rstd, const float* gamma,
+                                  const float* beta, const void* w, void* y,
+                                  int N, int HW, int C, int G, int Co,
+                                  int relu, hipStream_t stream);
+extern "C" int dlb_gnconv1x1_wrw(const void* const* xs, const int* starts,
+                                 int nseg, const float* mean,
+                                 const float* rstd, const float* gamma,
+                                 const float* beta, const void* dy, float* dw,
+                                 int N, int HW, int C, int G, int Co,
+                                 int relu, int splits, hipStream_t stream);
 
 // Segments: [N, HW, Ci] bf16 contiguous views of channels_last tensors
 // forming a virtual channel-concat.  Returns (y packed, mean, rstd).
@@ -143,6 +159,36 @@ static std::vector<torch::Tensor> gn_bwd(std::vector<torch::Tensor> xs,
   return out;  // [dx_0..dx_{k-1}, dgamma, dbeta]
 }
 
+// Stats-only GroupNorm over the virtual concat (mean/rstd for the fused
+// GN->1x1-conv kernels; the normalized activation is never materialized).
+static std::vector<torch::Tensor> gn_stats(std::vector<torch::Tensor> xs,
+                                           int64_t groups, double eps) {
+  TORCH_CHECK(!xs.empty() && xs.size() <= 56);
+  int C = 0;
+  const void* ptrs[56];
+  int starts[57];
+  for (size_t i = 0; i < xs.size(); ++i) {
+    auto& x = xs[i];
+    TORCH_CHECK(x.is_cuda() && x.dim() == 3 && x.is_contiguous());
+    TORCH_CHECK(x.scalar_type() == torch::kBFloat16);
+    TORCH_CHECK(x.size(2) % 8 == 0);
+    ptrs[i] = x.data_ptr();
+    starts[i] = C;
+    C += (int)x.size(2);
+  }
+  starts[xs.size()] = C;
+  const int N = xs[0].size(0), HW = xs[0].size(1);
+  TORCH_CHECK(groups <= 64 && C % groups == 0);
+  auto mean = torch::empty({N, groups},
+                           xs[0].options().dtype(torch::kFloat32));
+  auto rstd = torch::empty_like(mean);
+  auto stream = at::hip::getCurrentHIPStream();
+  dlb_gn_stats_segs(ptrs, starts, (int)xs.size(),
+                    mean.data_ptr<float>(), rstd.data_ptr<float>(), N, HW, C,
+                    (int)groups, (float)eps, stream.stream());
+  return {mean, rstd};
+}
+
 extern "C" void dlb_conv_fwd(const void* x, const void* w, void* y,
                              const float* bias, int N, int IH, int IW, int Ci,
                              int OH, int OW, int Co, int R, int S, int stride,
@@ -167,6 +213,81 @@ static inline bool is_cl(const torch::Tensor& t) {
 
 // x [N,Ci,H,W] channels_last bf16; w [Co,Ci,R,S] channels_last bf16;
 // bias fp32 [Co] or empty.  Returns y [N,Co,OH,OW] channels_last bf16.
+// y[N,HW,Co] = (GN(concat xs) with given stats) @ w^T   (1x1 conv)
+static torch::Tensor gn_conv1x1_fwd(std::vector<torch::Tensor> xs,
+                                    torch::Tensor mean, torch::Tensor rstd,
+                                    torch::Tensor gamma, torch::Tensor beta,
+                                    bool relu, torch::Tensor w) {
+  TORCH_CHECK(!xs.empty() && xs.size() <= 56);
+  int C = 0;
+  const void* ptrs[56];
+  int starts[57];
+  for (size_t i = 0; i < xs.size(); ++i) {
+    auto& x = xs[i];
+    TORCH_CHECK(x.is_cuda() && x.dim() == 3 && x.is_contiguous());
+    TORCH_CHECK(x.size(2) % 8 == 0);
+    ptrs[i] = x.data_ptr();
+    starts[i] = C;
+    C += (int)x.size(2);
+  }
+  starts[xs.size()] = C;
+  const int N = xs[0].size(0), HW = xs[0].size(1);
+  TORCH_CHECK(is_cl(w) && w.size(1) == C && w.size(2) == 1 && w.size(3) == 1);
+  TORCH_CHECK(w.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK((reinterpret_cast<uintptr_t>(gamma.data_ptr()) & 15) == 0 &&
+              (reinterpret_cast<uintptr_t>(beta.data_ptr()) & 15) == 0,
+              "gamma/beta must be 16B-aligned (vector loads)");
+  TORCH_CHECK(C / mean.size(1) >= 2, "fused path needs >=2 channels/group");
+  const int Co = w.size(0);
+  const int G = mean.size(1);
+  auto y = torch::empty({N, HW, Co}, xs[0].options());
+  auto stream = at::hip::getCurrentHIPStream();
+  dlb_gnconv1x1_fwd(ptrs, starts, (int)xs.size(), mean.data_ptr<float>(),
+                    rstd.data_ptr<float>(), gamma.data_ptr<float>(),
+                    beta.data_ptr<float>(), w.data_ptr(), y.data_ptr(), N,
+                    HW, C, G, Co, relu ? 1 : 0, stream.stream());
+  return y;
+}
+
+// dW[Co, C] fp32 = dy^T @ GN(concat xs)  (1x1 conv weight grad)
+static torch::Tensor gn_conv1x1_wrw(std::vector<torch::Tensor> xs,
+                                    torch::Tensor mean, torch::Tensor rstd,
+                                    torch::Tensor gamma, torch::Tensor beta,
+                                    bool relu, torch::Tensor dy) {
+  TORCH_CHECK(!xs.empty() && xs.size() <= 56);
+  int C = 0;
+  const void* ptrs[56];
+  int starts[57];
+  for (size_t i = 0; i < xs.size(); ++i) {
+    auto& x = xs[i];
+    TORCH_CHECK(x.is_cuda() && x.dim() == 3 && x.is_contiguous());
+    ptrs[i] = x.data_ptr();
+    starts[i] = C;
+    C += (int)x.size(2);
+  }
+  starts[xs.size()] = C;
+  const int N = xs[0].size(0), HW = xs[0].size(1);
+  TORCH_CHECK(is_cl(dy) && dy.size(0) == N && dy.size(2) * dy.size(3) == HW);
+  const int Co = dy.size(1);
+  const int G = mean.size(1);
+  TORCH_CHECK((reinterpret_cast<uintptr_t>(gamma.data_ptr()) & 15) == 0 &&
+              (reinterpret_cast<uintptr_t>(beta.data_ptr()) & 15) == 0,
+              "gamma/beta must be 16B-aligned (vector loads)");
+  TORCH_CHECK(C / G >= 2, "fused path needs >=2 channels/group");
+  int splits = dlb_conv_wrw_nsplits(N, 1, HW, C, Co, 1, 1);
+  const int mps = ((N * HW + splits - 1) / splits + 63) / 64 * 64;
+  splits = (N * HW + mps - 1) / mps;
+  auto part = torch::empty({splits, Co, C},
+                           xs[0].options().dtype(torch::kFloat32));
+  auto stream = at::hip::getCurrentHIPStream();
+  dlb_gnconv1x1_wrw(ptrs, starts, (int)xs.size(), mean.data_ptr<float>(),
+                    rstd.data_ptr<float>(), gamma.data_ptr<float>(),
+                    beta.data_ptr<float>(), dy.data_ptr(),
+                    part.data_ptr<float>(), N, HW, C, G, Co, relu ? 1 : 0,
+                    splits, stream.stream());
+  return splits == 1 ? part.squeeze(0) : part.sum(0);
+}
+
 static torch::Tensor conv_fwd(torch::Tensor x, torch::Tensor w,
                               c10::optional<torch::Tensor> bias,
                               int64_t stride, int64_t pad) {
@@ -505,6 +626,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("xs"), py::arg("dz"), py::arg("gamma"), py::arg("beta"),
         py::arg("mean"), py::arg("rstd"), py::arg("groups"), py::arg("relu"),
         py::arg("dx_accum") = py::none());
+  m.def("gn_stats", &gn_stats, "Stats-only GroupNorm over virtual concat");
+  m.def("gn_conv1x1_fwd", &gn_conv1x1_fwd,
+        "Fused GroupNorm(+ReLU) -> 1x1 conv forward (stream never packed)");
+  m.def("gn_conv1x1_wrw", &gn_conv1x1_wrw,
+        "Fused GroupNorm(+ReLU) -> 1x1 conv weight grad");
   m.def("avgpool_fwd", &avgpool_fwd);
   m.def("avgpool_bwd", &avgpool_bwd);
   m.def("gavg_fwd", &gavg_fwd);

@@ -21,7 +21,9 @@
 
 #include "common.h"
 
+#ifndef GN_BLOCK
 #define GN_BLOCK 256
+#endif
 #define GN_MAXG 64  // num_groups <= 64 covers the zoo (8/16/32)
 
 typedef __hip_bfloat16 bf16;
@@ -149,6 +151,65 @@ gn_fwd_kernel(const GnSegs segs, bf16* __restrict__ y,
       }
       *reinterpret_cast<Bf16x8*>(yb + (long)p * C + c0) = out;
     }
+  }
+}
+
+// ------------------------------------------------------------- stats-only
+// First half of the fused forward: per-(sample, group) mean/rstd over the
+// virtual concat, with NO normalize pass.  Used by the fused
+// GN->1x1-conv kernels (conv_gn.hip), which normalize at operand-load
+// time — the packed y activation is never materialized.
+extern "C" __global__ void __launch_bounds__(GN_BLOCK)
+gn_stats_kernel(const GnSegs segs, float* __restrict__ mean_out,
+                float* __restrict__ rstd_out, const int HW, const int C,
+                const int G, const float eps) {
+  const int n = blockIdx.x;
+  const int TC = C >> 3;
+  const int TCe = TC < GN_BLOCK ? TC : GN_BLOCK;
+  const int TP = GN_BLOCK / TCe;
+  const int t = threadIdx.x;
+  const int tc = t % TCe, tp = t / TCe;
+  const int Cg = C / G;
+  const bool active = t < TCe * TP;
+
+  __shared__ float s_sum[GN_MAXG];
+  __shared__ float s_ssq[GN_MAXG];
+  for (int g = t; g < G; g += GN_BLOCK) { s_sum[g] = 0.f; s_ssq[g] = 0.f; }
+  __syncthreads();
+
+  if (active) {
+    for (int oct = tc; oct < TC; oct += TCe) {
+      const int c0 = oct << 3;
+      int cloc, cs;
+      const bf16* sb = seg_locate(segs, c0, cloc, cs);
+      const bf16* xb = sb + (long)n * HW * cs + cloc;
+      float s[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+      float ss[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+      #pragma unroll 4
+      for (int p = tp; p < HW; p += TP) {
+        Bf16x8 chunk = *reinterpret_cast<const Bf16x8*>(xb + (long)p * cs);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float v = bf2f(chunk.v[j]);
+          s[j] += v;
+          ss[j] += v * v;
+        }
+      }
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        int g = (c0 + j) / Cg;
+        atomicAdd(&s_sum[g], s[j]);
+        atomicAdd(&s_ssq[g], ss[j]);
+      }
+    }
+  }
+  __syncthreads();
+  const float inv_m = 1.0f / ((float)HW * Cg);
+  for (int g = t; g < G; g += GN_BLOCK) {
+    float mu = s_sum[g] * inv_m;
+    float var = s_ssq[g] * inv_m - mu * mu;
+    mean_out[(long)n * G + g] = mu;
+    rstd_out[(long)n * G + g] = rsqrtf(var + eps);
   }
 }
 
@@ -665,4 +726,19 @@ extern "C" void dlb_gn_bwd_segs(const void* const* xs, const int* starts,
   hipLaunchKernelGGL(gn_bwd_kernel, dim3(N), dim3(GN_BLOCK), shmem, stream,
                      sg, (const bf16*)dz, dsg, gamma, beta,
                      mean, rstd, dgamma, dbeta, HW, C, G, relu, accumulate);
+}
+
+extern "C" void dlb_gn_stats_segs(const void* const* xs, const int* starts,
+                                  int nseg, float* mean, float* rstd, int N,
+                                  int HW, int C, int G, float eps,
+                                  hipStream_t stream) {
+  GnSegs sg{};
+  sg.nseg = nseg;
+  for (int i = 0; i < nseg; ++i) {
+    sg.p[i] = (const bf16*)xs[i];
+    sg.start[i] = starts[i];
+  }
+  sg.start[nseg] = starts[nseg];
+  hipLaunchKernelGGL(gn_stats_kernel, dim3(N), dim3(GN_BLOCK), 0, stream,
+                     sg, mean, rstd, HW, C, G, eps);
 }

@@ -61,22 +61,23 @@ class _DenseBlockFn(torch.autograd.Function):
         saves = []
         for li in range(nlayers):
             g1, b1, w1, g2, b2, w2 = params[6 * li:6 * li + 6]
-            y1, m1, r1 = ext().gn_fwd(segs3, g1, b1, groups, _EPS, True)
+            # fused GN->1x1: stats in one pass, normalize at the conv's
+            # operand load — the packed norm1 output never exists
+            m1, r1 = ext().gn_stats(segs3, groups, _EPS)
             w1c = _wcl(w1)
-            h1 = ext().conv_fwd(_as4(y1, n, h, w), w1c, None, 1, 0)
-            h13 = _to_nhwc3(h1)
+            h13 = ext().gn_conv1x1_fwd(segs3, m1, r1, g1, b1, True, w1c)
             y2, m2, r2 = ext().gn_fwd([h13], g2, b2, groups, _EPS, True)
             w2c = _wcl(w2)
             fresh = ext().conv_fwd(_as4(y2, n, h, w), w2c, None, 1, 1)
             segs3.insert(0, _to_nhwc3(fresh))
-            saves += [y1, m1, r1, w1c, h13, m2, r2, y2, w2c]
+            saves += [m1, r1, w1c, h13, m2, r2, y2, w2c]
         if has_trans:
             gt, bt, wt = params[6 * nlayers:6 * nlayers + 3]
-            yt, mt, rt = ext().gn_fwd(segs3, gt, bt, groups, _EPS, True)
+            mt, rt = ext().gn_stats(segs3, groups, _EPS)
             wtc = _wcl(wt)
-            ht = ext().conv_fwd(_as4(yt, n, h, w), wtc, None, 1, 0)
-            out = ext().avgpool_fwd(ht, 2)
-            saves += [yt, mt, rt, wtc]
+            ht3 = ext().gn_conv1x1_fwd(segs3, mt, rt, gt, bt, True, wtc)
+            out = ext().avgpool_fwd(_as4(ht3, n, h, w), 2)
+            saves += [mt, rt, wtc]
         ctx.save_for_backward(*segs3, *saves, *params)
         ctx.blk = (nlayers, groups, has_trans, n, h, w)
         if has_trans:
@@ -93,14 +94,13 @@ class _DenseBlockFn(torch.autograd.Function):
         pgrads = [None] * len(params)
 
         if has_trans:
-            yt, mt, rt, wtc = saves[9 * nlayers:9 * nlayers + 4]
+            mt, rt, wtc = saves[8 * nlayers:8 * nlayers + 3]
             gt, bt = params[6 * nlayers], params[6 * nlayers + 1]
             dht = ext().avgpool_bwd(
                 douts[0].contiguous(memory_format=torch.channels_last),
                 2, h, w)
-            yt4 = _as4(yt, n, h, w)
             dyt = _conv_bwd_data(dht, wtc, h, w, 1, 0)
-            dwt = ext().conv_wrw(yt4, dht, 1, 1, 1, 0)
+            dwt = ext().gn_conv1x1_wrw(segs3, mt, rt, gt, bt, True, dht)
             co, ci = wtc.shape[0], wtc.shape[1]
             pgrads[6 * nlayers + 2] = dwt.view(co, 1, 1, ci) \
                 .permute(0, 3, 1, 2)
@@ -117,7 +117,7 @@ class _DenseBlockFn(torch.autograd.Function):
             dsegs = [_to_nhwc3(d).clone() for d in douts]
 
         for li in range(nlayers - 1, -1, -1):
-            y1, m1, r1, w1c, h13, m2, r2, y2, w2c = saves[9 * li:9 * li + 9]
+            m1, r1, w1c, h13, m2, r2, y2, w2c = saves[8 * li:8 * li + 8]
             g1, b1 = params[6 * li], params[6 * li + 1]
             g2, b2 = params[6 * li + 3], params[6 * li + 4]
             in_segs = segs3[nlayers - li:]
@@ -131,9 +131,10 @@ class _DenseBlockFn(torch.autograd.Function):
                                          m2, r2, groups, True)
             pgrads[6 * li + 3], pgrads[6 * li + 4] = dg2, db2
             dh14 = _as4(dh1, n, h, w)
-            y14 = _as4(y1, n, h, w)
             dy1 = _conv_bwd_data(dh14, w1c, h, w, 1, 0)
-            dw1 = ext().conv_wrw(y14, dh14, 1, 1, 1, 0)
+            # fused weight grad re-normalizes segments at load time from
+            # the saved stats (the packed norm1 output was never saved)
+            dw1 = ext().gn_conv1x1_wrw(in_segs, m1, r1, g1, b1, True, dh14)
             co1, ci1 = w1c.shape[0], w1c.shape[1]
             pgrads[6 * li + 2] = dw1.view(co1, 1, 1, ci1).permute(0, 3, 1, 2)
             outs = ext().gn_bwd(in_segs, _to_nhwc3(dy1), g1, b1, m1, r1,

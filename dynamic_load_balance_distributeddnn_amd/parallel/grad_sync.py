@@ -61,7 +61,7 @@ class GradientSynchronizer:
         self.defer = defer
         self.params = [p for p in model.parameters() if p.requires_grad]
         device = self.params[0].device
-        total = sum(p.numel() for p in self.params)
+        total = sum((p.numel() + 7) & ~7 for p in self.params)
         self.arena = torch.zeros(total, dtype=grad_dtype, device=device)
         self.weight = 1.0
         self._works: list = []
@@ -79,6 +79,9 @@ class GradientSynchronizer:
         elem = self.arena.element_size()
         for p in ordered:
             n = p.numel()
+            # 8-element alignment so fp32 views are 16-byte aligned and
+            # the fused kernels can use vector parameter loads
+            offset = (offset + 7) & ~7
             self.offsets[id(p)] = (offset, n)
             p.grad = _strided_view(self.arena, offset, p)
             cur_params.append(p)

@@ -20,7 +20,7 @@ HIPCC = shutil.which("hipcc")
 
 
 @pytest.mark.skipif(HIPCC is None, reason="hipcc not on PATH")
-@pytest.mark.parametrize("src", ["conv.hip", "conv_halo.hip",
+@pytest.mark.parametrize("src", ["conv.hip", "conv_halo.hip", "conv_gn.hip",
                                  "conv_grouped.hip", "groupnorm.hip",
                                  "layernorm.hip", "attention.hip",
                                  "softmax.hip", "pool.hip", "maxpool.hip",
