@@ -238,6 +238,8 @@ static torch::Tensor gn_conv1x1_fwd(std::vector<torch::Tensor> xs,
               (reinterpret_cast<uintptr_t>(beta.data_ptr()) & 15) == 0,
               "gamma/beta must be 16B-aligned (vector loads)");
   TORCH_CHECK(C / mean.size(1) >= 2, "fused path needs >=2 channels/group");
+  TORCH_CHECK(HW >= 32 || (HW >= 16 && 128 % HW == 0),
+              "row-chunk sample span exceeds the staged table");
   const int Co = w.size(0);
   const int G = mean.size(1);
   auto y = torch::empty({N, HW, Co}, xs[0].options());
@@ -274,6 +276,8 @@ static torch::Tensor gn_conv1x1_wrw(std::vector<torch::Tensor> xs,
               (reinterpret_cast<uintptr_t>(beta.data_ptr()) & 15) == 0,
               "gamma/beta must be 16B-aligned (vector loads)");
   TORCH_CHECK(C / G >= 2, "fused path needs >=2 channels/group");
+  TORCH_CHECK(HW >= 32 || (HW >= 16 && 64 % HW == 0),
+              "row-chunk sample span exceeds the staged table");
   int splits = dlb_conv_wrw_nsplits(N, 1, HW, C, Co, 1, 1);
   const int mps = ((N * HW + splits - 1) / splits + 63) / 64 * 64;
   splits = (N * HW + mps - 1) / mps;

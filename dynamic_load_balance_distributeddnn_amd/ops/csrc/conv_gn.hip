@@ -68,12 +68,43 @@ __device__ inline bf16x8_t gnc_zero8() {
   return u.v;
 }
 
+// Per-(sample, channel) affine tables: gn(x) = x*a + b with
+//   a[n][c] = rstd[n][g]*gamma[c],  b[n][c] = beta[c] - mean[n][g]*a
+// (algebraically equal to the unfused form; fp32-rounds one ulp apart).
+// A chunk of rows spans a bounded run of consecutive samples (chunk
+// starts are BM-aligned, so at HW=16 the span is exactly BM/HW), and the
+// tables are tiny — staged in LDS one pipeline chunk ahead, making the
+// hot loader one FMA per element.
+#define GNC_NSPAN_FWD 8   // BM=128: HW>=32 spans <=4; HW=16 exactly 8
+#define GNC_NSPAN_WRW 4   // BM=64
+
+// stage a/b for channels [c0, c0+W) and samples [n_base, n_base+nspan)
+__device__ inline void gnc_stage_ab(const GnCParams& p, float* sa, float* sb,
+                                    int c0, int W, int n_base, int nspan,
+                                    int t, int nthreads) {
+  const int NT = nspan * W;
+  for (int idx = t; idx < NT; idx += nthreads) {
+    const int nl = idx / W;
+    const int c = c0 + idx % W;
+    const long n = n_base + nl;
+    float a = 0.f, b = 0.f;
+    if (c < p.C && n * p.HW < p.M) {
+      const int g = (int)p.fd_cg.div((unsigned)c);
+      const float rs = p.rstd[n * p.G + g];
+      a = rs * p.gamma[c];
+      b = p.beta[c] - p.mean[n * p.G + g] * a;
+    }
+    sa[nl * W + idx % W] = a;
+    sb[nl * W + idx % W] = b;
+  }
+}
+
 // Normalized 8-channel load at (row m, channel k): one LDS segment
-// lookup, vectorized gamma/beta (k is 8-aligned so float4 is exact),
-// 8 channels span at most 4 groups (cg >= 2) tracked incrementally —
-// no per-element division or serial segment walk.
+// lookup + 8 FMAs against the staged tables.
 __device__ inline bf16x8_t gn_load8(const GnCParams& p, const short* soct,
-                                    int m, int k, int mbound) {
+                                    const float* sa, const float* sb, int W,
+                                    int koff, int n_base, int m, int k,
+                                    int mbound) {
   const bool ok = (m < mbound) & (k < p.C);
   if (!ok) return gnc_zero8();
   const unsigned n = p.fd_hw.div((unsigned)m);
@@ -83,40 +114,14 @@ __device__ inline bf16x8_t gn_load8(const GnCParams& p, const short* soct,
   const bf16* ptr = p.segs.p[si] +
                     ((long)n * p.HW + pix) * cs + (k - p.segs.start[si]);
   bf16x8_t x8 = *reinterpret_cast<const bf16x8_t*>(ptr);
-  const int g0 = (int)p.fd_cg.div((unsigned)k);
-  const float* mu = p.mean + (long)n * p.G + g0;
-  const float* rs = p.rstd + (long)n * p.G + g0;
-  float muv[4], rsv[4];
-#pragma unroll
-  for (int q = 0; q < 4; ++q) {
-    // guard is buffer-bounds only: gi below never selects a group the
-    // 8 channels don't touch, and touched groups are always < G
-    const bool in_r = g0 + q < p.G;
-    muv[q] = in_r ? mu[q] : 0.f;
-    rsv[q] = in_r ? rs[q] : 0.f;
-  }
-  const float4 gaA = *reinterpret_cast<const float4*>(p.gamma + k);
-  const float4 gaB = *reinterpret_cast<const float4*>(p.gamma + k + 4);
-  const float4 beA = *reinterpret_cast<const float4*>(p.beta + k);
-  const float4 beB = *reinterpret_cast<const float4*>(p.beta + k + 4);
-  const float* ga = reinterpret_cast<const float*>(&gaA);  // [8] static idx
-  const float* be = reinterpret_cast<const float*>(&beA);
+  const int base = ((int)n - n_base) * W + koff;
   union { bf16x8_t v; bf16 h[8]; } in, out;
   in.v = x8;
-  int rj = k - g0 * p.cg;   // offset within group g0
-  int gi = 0;
 #pragma unroll
   for (int j = 0; j < 8; ++j) {
-    const float gaj = j < 4 ? ga[j] : reinterpret_cast<const float*>(&gaB)[j - 4];
-    const float bej = j < 4 ? be[j] : reinterpret_cast<const float*>(&beB)[j - 4];
-    const float m0 = gi == 0 ? muv[0] : gi == 1 ? muv[1]
-                     : gi == 2 ? muv[2] : muv[3];
-    const float r0 = gi == 0 ? rsv[0] : gi == 1 ? rsv[1]
-                     : gi == 2 ? rsv[2] : rsv[3];
-    float v = (__bfloat162float(in.h[j]) - m0) * r0 * gaj + bej;
+    float v = __bfloat162float(in.h[j]) * sa[base + j] + sb[base + j];
     if (p.relu) v = fmaxf(v, 0.f);
     out.h[j] = __float2bfloat16(v);
-    if (++rj == p.cg) { rj = 0; ++gi; }
   }
   return out.v;
 }
@@ -167,6 +172,8 @@ gnconv1x1_fwd_kernel(const GnCParams p) {
   __shared__ bf16 a_lds[2][BM * LDA];
   __shared__ bf16 b_lds[2][BN * LDB];
   __shared__ short s_oct[GNC_MAXOCT];
+  __shared__ float s_ga[2][GNC_NSPAN_FWD * BK];  // per k-chunk parity
+  __shared__ float s_gb[2][GNC_NSPAN_FWD * BK];
 
   const int t = threadIdx.x;
   const int lane = t & 63;
@@ -174,8 +181,8 @@ gnconv1x1_fwd_kernel(const GnCParams p) {
   const int wr = wave / WN, wc = wave % WN;
   const int m0 = blockIdx.x * BM;
   const int n0 = blockIdx.y * BN;
+  const int nb0 = (int)p.fd_hw.div((unsigned)m0);  // block rows' first n
   gnc_stage_oct(p, s_oct, t, CONV_BLOCK);
-  __syncthreads();
 
   f32x4 acc[FA][FB];
 #pragma unroll
@@ -191,7 +198,9 @@ gnconv1x1_fwd_kernel(const GnCParams p) {
       const int c = t + u * CONV_BLOCK;
       const int row = c / (BK / 8);
       const int k8 = (c % (BK / 8)) * 8;
-      areg[u] = gn_load8(p, s_oct, m0 + row, kt + k8, p.M);
+      const int par = (kt / BK) & 1;
+      areg[u] = gn_load8(p, s_oct, s_ga[par], s_gb[par], BK, k8, nb0,
+                         m0 + row, kt + k8, p.M);
       if (c >= ACH) areg[u] = gnc_zero8();
     }
 #pragma unroll
@@ -228,8 +237,16 @@ gnconv1x1_fwd_kernel(const GnCParams p) {
     }
   };
 
+  auto stage_ab = [&](int kt) {
+    gnc_stage_ab(p, s_ga[(kt / BK) & 1], s_gb[(kt / BK) & 1], kt, BK, nb0,
+                 GNC_NSPAN_FWD, t, CONV_BLOCK);
+  };
+
+  stage_ab(0);
+  __syncthreads();
   load_tile(0);
   write_tile(0);
+  stage_ab(BK);
   __syncthreads();
 
   int buf = 0;
@@ -257,6 +274,7 @@ gnconv1x1_fwd_kernel(const GnCParams p) {
         acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
     if (more) write_tile(buf ^ 1);
+    if (kt + 2 * BK < p.C) stage_ab(kt + 2 * BK);
     __syncthreads();
     buf ^= 1;
   }
@@ -317,6 +335,8 @@ gnconv1x1_wrw_kernel(const GnCParams p) {
   __shared__ bf16 dy_t[2][BM * LD];
   __shared__ bf16 x_t[2][BM * LX];
   __shared__ short s_oct[GNC_MAXOCT];
+  __shared__ float s_ga[2][GNC_NSPAN_WRW * BKN];
+  __shared__ float s_gb[2][GNC_NSPAN_WRW * BKN];
 
   const int t = threadIdx.x;
   const int lane = t & 63;
@@ -357,7 +377,10 @@ gnconv1x1_wrw_kernel(const GnCParams p) {
       const int c = t + u * CONV_BLOCK;
       const int mm = c / (BKN / 8);
       const int k8 = (c % (BKN / 8)) * 8;
-      xreg[u] = gn_load8(p, s_oct, mt + mm, k0 + k8, mend);
+      const int par = (mt / BM) & 1;
+      xreg[u] = gn_load8(p, s_oct, s_ga[par], s_gb[par], BKN, k8,
+                         (int)p.fd_hw.div((unsigned)mt), mt + mm, k0 + k8,
+                         mend);
       if (c >= XCH) xreg[u] = gnc_zero8();
     }
   };
@@ -383,10 +406,18 @@ gnconv1x1_wrw_kernel(const GnCParams p) {
     }
   };
 
+  auto stage_ab = [&](int mt) {
+    gnc_stage_ab(p, s_ga[(mt / BM) & 1], s_gb[(mt / BM) & 1], k0, BKN,
+                 (int)p.fd_hw.div((unsigned)mt), GNC_NSPAN_WRW, t,
+                 CONV_BLOCK);
+  };
+
   gnc_stage_oct(p, s_oct, t, CONV_BLOCK);
+  stage_ab(mstart);
   __syncthreads();
   load_chunk(mstart);
   write_chunk(0);
+  stage_ab(mstart + BM);
   __syncthreads();
 
   int buf = 0;
@@ -412,6 +443,7 @@ gnconv1x1_wrw_kernel(const GnCParams p) {
               afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
     }
     if (more) write_chunk(buf ^ 1);
+    if (mt + 2 * BM < mend) stage_ab(mt + 2 * BM);
     __syncthreads();
     buf ^= 1;
   }

@@ -57,50 +57,74 @@ class _DenseBlockFn(torch.autograd.Function):
     def forward(ctx, meta, seg0, *params):
         nlayers, groups, has_trans = meta
         n, _, h, w = seg0.shape
+        # fused GN->1x1 kernels bound the samples a row chunk spans
+        hw = h * w
+        fused = hw >= 32 or (hw >= 16 and 128 % hw == 0)
         segs3 = [_to_nhwc3(seg0)]
         saves = []
         for li in range(nlayers):
             g1, b1, w1, g2, b2, w2 = params[6 * li:6 * li + 6]
-            # fused GN->1x1: stats in one pass, normalize at the conv's
-            # operand load — the packed norm1 output never exists
-            m1, r1 = ext().gn_stats(segs3, groups, _EPS)
             w1c = _wcl(w1)
-            h13 = ext().gn_conv1x1_fwd(segs3, m1, r1, g1, b1, True, w1c)
+            if fused:
+                # fused GN->1x1: stats in one pass, normalize at the
+                # conv's operand load — the packed norm1 never exists
+                m1, r1 = ext().gn_stats(segs3, groups, _EPS)
+                h13 = ext().gn_conv1x1_fwd(segs3, m1, r1, g1, b1, True, w1c)
+                lay = [m1, r1, w1c, h13]
+            else:
+                y1, m1, r1 = ext().gn_fwd(segs3, g1, b1, groups, _EPS, True)
+                h1 = ext().conv_fwd(_as4(y1, n, h, w), w1c, None, 1, 0)
+                h13 = _to_nhwc3(h1)
+                lay = [y1, m1, r1, w1c, h13]
             y2, m2, r2 = ext().gn_fwd([h13], g2, b2, groups, _EPS, True)
             w2c = _wcl(w2)
             fresh = ext().conv_fwd(_as4(y2, n, h, w), w2c, None, 1, 1)
             segs3.insert(0, _to_nhwc3(fresh))
-            saves += [m1, r1, w1c, h13, m2, r2, y2, w2c]
+            saves += lay + [m2, r2, y2, w2c]
         if has_trans:
             gt, bt, wt = params[6 * nlayers:6 * nlayers + 3]
-            mt, rt = ext().gn_stats(segs3, groups, _EPS)
             wtc = _wcl(wt)
-            ht3 = ext().gn_conv1x1_fwd(segs3, mt, rt, gt, bt, True, wtc)
-            out = ext().avgpool_fwd(_as4(ht3, n, h, w), 2)
-            saves += [mt, rt, wtc]
+            if fused:
+                mt, rt = ext().gn_stats(segs3, groups, _EPS)
+                ht3 = ext().gn_conv1x1_fwd(segs3, mt, rt, gt, bt, True, wtc)
+                ht4 = _as4(ht3, n, h, w)
+                saves += [mt, rt, wtc]
+            else:
+                yt, mt, rt = ext().gn_fwd(segs3, gt, bt, groups, _EPS, True)
+                ht4 = ext().conv_fwd(_as4(yt, n, h, w), wtc, None, 1, 0)
+                saves += [yt, mt, rt, wtc]
+            out = ext().avgpool_fwd(ht4, 2)
         ctx.save_for_backward(*segs3, *saves, *params)
-        ctx.blk = (nlayers, groups, has_trans, n, h, w)
+        ctx.blk = (nlayers, groups, has_trans, n, h, w, fused)
         if has_trans:
             return out
         return tuple(_as4(s, n, h, w) for s in segs3)
 
     @staticmethod
     def backward(ctx, *douts):
-        nlayers, groups, has_trans, n, h, w = ctx.blk
+        nlayers, groups, has_trans, n, h, w, fused = ctx.blk
         nseg = nlayers + 1
+        LW = 8 if fused else 9       # saves per layer
         segs3 = list(ctx.saved_tensors[:nseg])
         saves = ctx.saved_tensors[nseg:]
         params = saves[len(saves) - (6 * nlayers + (3 if has_trans else 0)):]
         pgrads = [None] * len(params)
 
         if has_trans:
-            mt, rt, wtc = saves[8 * nlayers:8 * nlayers + 3]
+            tbase = LW * nlayers
+            if fused:
+                mt, rt, wtc = saves[tbase:tbase + 3]
+            else:
+                yt, mt, rt, wtc = saves[tbase:tbase + 4]
             gt, bt = params[6 * nlayers], params[6 * nlayers + 1]
             dht = ext().avgpool_bwd(
                 douts[0].contiguous(memory_format=torch.channels_last),
                 2, h, w)
             dyt = _conv_bwd_data(dht, wtc, h, w, 1, 0)
-            dwt = ext().gn_conv1x1_wrw(segs3, mt, rt, gt, bt, True, dht)
+            if fused:
+                dwt = ext().gn_conv1x1_wrw(segs3, mt, rt, gt, bt, True, dht)
+            else:
+                dwt = ext().conv_wrw(_as4(yt, n, h, w), dht, 1, 1, 1, 0)
             co, ci = wtc.shape[0], wtc.shape[1]
             pgrads[6 * nlayers + 2] = dwt.view(co, 1, 1, ci) \
                 .permute(0, 3, 1, 2)
@@ -117,7 +141,12 @@ class _DenseBlockFn(torch.autograd.Function):
             dsegs = [_to_nhwc3(d).clone() for d in douts]
 
         for li in range(nlayers - 1, -1, -1):
-            m1, r1, w1c, h13, m2, r2, y2, w2c = saves[8 * li:8 * li + 8]
+            if fused:
+                m1, r1, w1c, h13, m2, r2, y2, w2c = \
+                    saves[LW * li:LW * li + LW]
+            else:
+                y1, m1, r1, w1c, h13, m2, r2, y2, w2c = \
+                    saves[LW * li:LW * li + LW]
             g1, b1 = params[6 * li], params[6 * li + 1]
             g2, b2 = params[6 * li + 3], params[6 * li + 4]
             in_segs = segs3[nlayers - li:]
@@ -132,9 +161,13 @@ class _DenseBlockFn(torch.autograd.Function):
             pgrads[6 * li + 3], pgrads[6 * li + 4] = dg2, db2
             dh14 = _as4(dh1, n, h, w)
             dy1 = _conv_bwd_data(dh14, w1c, h, w, 1, 0)
-            # fused weight grad re-normalizes segments at load time from
-            # the saved stats (the packed norm1 output was never saved)
-            dw1 = ext().gn_conv1x1_wrw(in_segs, m1, r1, g1, b1, True, dh14)
+            if fused:
+                # fused weight grad re-normalizes segments at load time
+                # from the saved stats (norm1's output was never saved)
+                dw1 = ext().gn_conv1x1_wrw(in_segs, m1, r1, g1, b1, True,
+                                           dh14)
+            else:
+                dw1 = ext().conv_wrw(_as4(y1, n, h, w), dh14, 1, 1, 1, 0)
             co1, ci1 = w1c.shape[0], w1c.shape[1]
             pgrads[6 * li + 2] = dw1.view(co1, 1, 1, ci1).permute(0, 3, 1, 2)
             outs = ext().gn_bwd(in_segs, _to_nhwc3(dy1), g1, b1, m1, r1,
