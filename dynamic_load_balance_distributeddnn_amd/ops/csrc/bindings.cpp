@@ -328,7 +328,12 @@ static torch::Tensor conv_bwd_data(torch::Tensor dy, torch::Tensor w,
   auto stream = at::hip::getCurrentHIPStream();
   // 3x3/s1/p1: the LDS-halo kernel in transpose-read mode reads the
   // weight unmodified (no flip+copy transform, ~2x the generic kernel)
-  if (R == 3 && S == 3 && stride == 1 && pad == 1 && IH == OH && IW == OW &&
+  // measured: the halo-WTR route wins for shallow reductions (DenseNet
+  // growth convs, Co=32) and loses to the generic tr-read kernel at
+  // ResNet depths — gate on the reduction size
+  static const bool no_bwd_halo = getenv("DLB_NO_BWD_HALO") != nullptr;
+  if (!no_bwd_halo && Co <= 64 &&
+      R == 3 && S == 3 && stride == 1 && pad == 1 && IH == OH && IW == OW &&
       dlb_conv3x3_bwd_halo(dy.data_ptr(), w.data_ptr(), dx.data_ptr(), N,
                            (int)IH, (int)IW, Ci, Co, stream.stream()))
     return dx;
