@@ -109,10 +109,16 @@ static std::vector<torch::Tensor> gn_bwd(std::vector<torch::Tensor> xs,
                                          torch::Tensor rstd, int64_t groups,
                                          bool relu,
                                          c10::optional<std::vector<torch::Tensor>>
-                                             dx_accum = c10::nullopt) {
+                                             dx_accum = c10::nullopt,
+                                         c10::optional<torch::Tensor>
+                                             dgamma_out = c10::nullopt,
+                                         c10::optional<torch::Tensor>
+                                             dbeta_out = c10::nullopt) {
   // dx_accum: preallocated per-segment grad buffers — the kernel ADDS
   // into them (the dense-stream manual backward), instead of allocating
   // fresh outputs for autograd to sum pairwise.
+  // dgamma_out/dbeta_out: pre-ZEROED [C] fp32 buffers (flat-arena grad
+  // views; the kernel accumulates into them with atomics).
   TORCH_CHECK(!xs.empty() && xs.size() <= 56);
   int C = 0;
   const void* ptrs[56];
@@ -134,11 +140,20 @@ static std::vector<torch::Tensor> gn_bwd(std::vector<torch::Tensor> xs,
   starts[xs.size()] = C;
   const int N = xs[0].size(0), HW = xs[0].size(1);
   TORCH_CHECK(dz.is_contiguous() && dz.size(2) == C);
-  // one zero-fill kernel for both reductions (they were 2 of the ~288
-  // fill launches per DenseNet step)
-  auto gbuf = torch::zeros({2, C}, xs[0].options().dtype(torch::kFloat32));
-  auto dgamma = gbuf[0];
-  auto dbeta = gbuf[1];
+  torch::Tensor dgamma, dbeta;
+  if (dgamma_out.has_value()) {
+    TORCH_CHECK(dgamma_out->is_contiguous() && dgamma_out->numel() == C &&
+                dbeta_out.has_value() && dbeta_out->is_contiguous() &&
+                dbeta_out->numel() == C);
+    dgamma = *dgamma_out;
+    dbeta = *dbeta_out;
+  } else {
+    // one zero-fill kernel for both reductions (they were 2 of the ~288
+    // fill launches per DenseNet step)
+    auto gbuf = torch::zeros({2, C}, xs[0].options().dtype(torch::kFloat32));
+    dgamma = gbuf[0];
+    dbeta = gbuf[1];
+  }
   auto stream = at::hip::getCurrentHIPStream();
   float* scratch = nullptr;
   torch::Tensor scratch_t;
@@ -255,7 +270,8 @@ static torch::Tensor gn_conv1x1_fwd(std::vector<torch::Tensor> xs,
 static torch::Tensor gn_conv1x1_wrw(std::vector<torch::Tensor> xs,
                                     torch::Tensor mean, torch::Tensor rstd,
                                     torch::Tensor gamma, torch::Tensor beta,
-                                    bool relu, torch::Tensor dy) {
+                                    bool relu, torch::Tensor dy,
+                                    c10::optional<torch::Tensor> out = c10::nullopt) {
   TORCH_CHECK(!xs.empty() && xs.size() <= 56);
   int C = 0;
   const void* ptrs[56];
@@ -289,6 +305,13 @@ static torch::Tensor gn_conv1x1_wrw(std::vector<torch::Tensor> xs,
                     beta.data_ptr<float>(), dy.data_ptr(),
                     part.data_ptr<float>(), N, HW, C, G, Co, relu ? 1 : 0,
                     splits, stream.stream());
+  if (out.has_value()) {
+    TORCH_CHECK(out->numel() == (long)Co * C &&
+                out->scalar_type() == torch::kFloat32);
+    auto o = out->view({Co, C});
+    torch::sum_out(o, part, 0);
+    return *out;
+  }
   return splits == 1 ? part.squeeze(0) : part.sum(0);
 }
 
@@ -347,7 +370,10 @@ static torch::Tensor conv_bwd_data(torch::Tensor dy, torch::Tensor w,
 // Split-K partials go to per-split slabs (no atomic contention) and are
 // reduced with one sum kernel here.
 static torch::Tensor conv_wrw(torch::Tensor x, torch::Tensor dy, int64_t R,
-                              int64_t S, int64_t stride, int64_t pad) {
+                              int64_t S, int64_t stride, int64_t pad,
+                              c10::optional<torch::Tensor> out = c10::nullopt) {
+  // out: a [Co, R*S*Ci] fp32 buffer the slab sum is written into (the
+  // flat-arena grad view of the weight) — skips the separate grad add.
   TORCH_CHECK(x.is_cuda() && is_cl(x) && is_cl(dy));
   const int N = x.size(0), Ci = x.size(1), IH = x.size(2), IW = x.size(3);
   const int Co = dy.size(1), OH = dy.size(2), OW = dy.size(3);
@@ -376,6 +402,13 @@ static torch::Tensor conv_wrw(torch::Tensor x, torch::Tensor dy, int64_t R,
   dlb_conv_wrw(x.data_ptr(), dy.data_ptr(), part.data_ptr<float>(), N, IH,
                IW, Ci, OH, OW, Co, (int)R, (int)S, (int)stride, (int)pad,
                splits, stream.stream());
+  if (out.has_value()) {
+    TORCH_CHECK(out->numel() == (long)Co * K &&
+                out->scalar_type() == torch::kFloat32);
+    auto o = out->view({Co, K});
+    torch::sum_out(o, part, 0);
+    return *out;
+  }
   return splits == 1 ? part.squeeze(0) : part.sum(0);
 }
 
@@ -625,7 +658,10 @@ static torch::Tensor maxpool_bwd(torch::Tensor dy, torch::Tensor idx,
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv_fwd", &conv_fwd, "Implicit-GEMM NHWC bf16 conv forward");
   m.def("conv_bwd_data", &conv_bwd_data, "NHWC bf16 conv input-grad");
-  m.def("conv_wrw", &conv_wrw, "NHWC bf16 conv weight-grad (fp32 out)");
+  m.def("conv_wrw", &conv_wrw,
+        py::arg("x"), py::arg("dy"), py::arg("R"), py::arg("S"),
+        py::arg("stride"), py::arg("pad"), py::arg("out") = py::none(),
+        "NHWC bf16 conv weight-grad (fp32 out)");
   m.def("sgd_momentum", &sgd_momentum,
         py::arg("p"), py::arg("g"), py::arg("m"), py::arg("lr"),
         py::arg("mu"), py::arg("mirror") = py::none(),
@@ -634,11 +670,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gn_bwd", &gn_bwd, "Fused GroupNorm(+ReLU) backward, NHWC bf16",
         py::arg("xs"), py::arg("dz"), py::arg("gamma"), py::arg("beta"),
         py::arg("mean"), py::arg("rstd"), py::arg("groups"), py::arg("relu"),
-        py::arg("dx_accum") = py::none());
+        py::arg("dx_accum") = py::none(),
+        py::arg("dgamma_out") = py::none(), py::arg("dbeta_out") = py::none());
   m.def("gn_stats", &gn_stats, "Stats-only GroupNorm over virtual concat");
   m.def("gn_conv1x1_fwd", &gn_conv1x1_fwd,
         "Fused GroupNorm(+ReLU) -> 1x1 conv forward (stream never packed)");
   m.def("gn_conv1x1_wrw", &gn_conv1x1_wrw,
+        py::arg("xs"), py::arg("mean"), py::arg("rstd"), py::arg("gamma"),
+        py::arg("beta"), py::arg("relu"), py::arg("dy"),
+        py::arg("out") = py::none(),
         "Fused GroupNorm(+ReLU) -> 1x1 conv weight grad");
   m.def("avgpool_fwd", &avgpool_fwd);
   m.def("avgpool_bwd", &avgpool_bwd);

@@ -110,6 +110,25 @@ class _DenseBlockFn(torch.autograd.Function):
         params = saves[len(saves) - (6 * nlayers + (3 if has_trans else 0)):]
         pgrads = [None] * len(params)
 
+        # When the params live in a GradientSynchronizer arena, write
+        # grads straight into their arena views and notify the bucket
+        # machinery, returning None to autograd — removes one small add
+        # kernel per parameter per step (~364 launches on DenseNet-121).
+        sink = getattr(params[0], "_dlb_sink", None)
+        direct = sink is not None and params[0].grad is not None
+
+        def norm_outs(p_g, p_b):
+            if direct:
+                return dict(dgamma_out=p_g.grad, dbeta_out=p_b.grad)
+            return {}
+
+        def put(idx, grad):
+            # grad already landed in the arena on the direct path
+            if direct:
+                sink.mark_ready(params[idx])
+            else:
+                pgrads[idx] = grad
+
         if has_trans:
             tbase = LW * nlayers
             if fused:
@@ -121,20 +140,27 @@ class _DenseBlockFn(torch.autograd.Function):
                 douts[0].contiguous(memory_format=torch.channels_last),
                 2, h, w)
             dyt = _conv_bwd_data(dht, wtc, h, w, 1, 0)
-            if fused:
-                dwt = ext().gn_conv1x1_wrw(segs3, mt, rt, gt, bt, True, dht)
-            else:
-                dwt = ext().conv_wrw(_as4(yt, n, h, w), dht, 1, 1, 1, 0)
             co, ci = wtc.shape[0], wtc.shape[1]
-            pgrads[6 * nlayers + 2] = dwt.view(co, 1, 1, ci) \
-                .permute(0, 3, 1, 2)
+            wt_p = params[6 * nlayers + 2]
+            wt_out = dict(out=wt_p.grad.permute(0, 2, 3, 1).reshape(co, ci)) \
+                if direct else {}
+            if fused:
+                dwt = ext().gn_conv1x1_wrw(segs3, mt, rt, gt, bt, True, dht,
+                                           **wt_out)
+            else:
+                dwt = ext().conv_wrw(_as4(yt, n, h, w), dht, 1, 1, 1, 0,
+                                     **wt_out)
+            put(6 * nlayers + 2, None if direct else
+                dwt.view(co, 1, 1, ci).permute(0, 3, 1, 2))
             # the transition norm is every segment's LAST consumer: its
             # backward writes the per-segment grad buffers fresh
             outs = ext().gn_bwd(segs3, _to_nhwc3(dyt), gt, bt, mt, rt,
-                                groups, True)
+                                groups, True,
+                                **norm_outs(params[6 * nlayers],
+                                            params[6 * nlayers + 1]))
             dsegs = list(outs[:-2])
-            pgrads[6 * nlayers] = outs[-2]
-            pgrads[6 * nlayers + 1] = outs[-1]
+            put(6 * nlayers, outs[-2])
+            put(6 * nlayers + 1, outs[-1])
         else:
             # external grads arrive per segment (one consumer each);
             # clone into owned buffers the kernels then accumulate into
@@ -153,27 +179,43 @@ class _DenseBlockFn(torch.autograd.Function):
             dfresh4 = _as4(dsegs[nlayers - 1 - li], n, h, w)
             y24 = _as4(y2, n, h, w)
             dy2 = _conv_bwd_data(dfresh4, w2c, h, w, 1, 1)
-            dw2 = ext().conv_wrw(y24, dfresh4, 3, 3, 1, 1)
             co2, ci2 = w2c.shape[0], w2c.shape[1]
-            pgrads[6 * li + 5] = dw2.view(co2, 3, 3, ci2).permute(0, 3, 1, 2)
-            dh1, dg2, db2 = ext().gn_bwd([h13], _to_nhwc3(dy2), g2, b2,
-                                         m2, r2, groups, True)
-            pgrads[6 * li + 3], pgrads[6 * li + 4] = dg2, db2
+            w2_p = params[6 * li + 5]
+            w2_out = dict(out=w2_p.grad.permute(0, 2, 3, 1)
+                          .reshape(co2, 9 * ci2)) if direct else {}
+            dw2 = ext().conv_wrw(y24, dfresh4, 3, 3, 1, 1, **w2_out)
+            put(6 * li + 5, None if direct else
+                dw2.view(co2, 3, 3, ci2).permute(0, 3, 1, 2))
+            outs2 = ext().gn_bwd([h13], _to_nhwc3(dy2), g2, b2,
+                                 m2, r2, groups, True,
+                                 **norm_outs(params[6 * li + 3],
+                                             params[6 * li + 4]))
+            dh1, dg2, db2 = outs2
+            put(6 * li + 3, dg2)
+            put(6 * li + 4, db2)
             dh14 = _as4(dh1, n, h, w)
             dy1 = _conv_bwd_data(dh14, w1c, h, w, 1, 0)
+            co1, ci1 = w1c.shape[0], w1c.shape[1]
+            w1_p = params[6 * li + 2]
+            w1_out = dict(out=w1_p.grad.permute(0, 2, 3, 1)
+                          .reshape(co1, ci1)) if direct else {}
             if fused:
                 # fused weight grad re-normalizes segments at load time
                 # from the saved stats (norm1's output was never saved)
                 dw1 = ext().gn_conv1x1_wrw(in_segs, m1, r1, g1, b1, True,
-                                           dh14)
+                                           dh14, **w1_out)
             else:
-                dw1 = ext().conv_wrw(_as4(y1, n, h, w), dh14, 1, 1, 1, 0)
-            co1, ci1 = w1c.shape[0], w1c.shape[1]
-            pgrads[6 * li + 2] = dw1.view(co1, 1, 1, ci1).permute(0, 3, 1, 2)
+                dw1 = ext().conv_wrw(_as4(y1, n, h, w), dh14, 1, 1, 1, 0,
+                                     **w1_out)
+            put(6 * li + 2, None if direct else
+                dw1.view(co1, 1, 1, ci1).permute(0, 3, 1, 2))
             outs = ext().gn_bwd(in_segs, _to_nhwc3(dy1), g1, b1, m1, r1,
                                 groups, True,
-                                dx_accum=dsegs[nlayers - li:])
-            pgrads[6 * li], pgrads[6 * li + 1] = outs[-2], outs[-1]
+                                dx_accum=dsegs[nlayers - li:],
+                                **norm_outs(params[6 * li],
+                                            params[6 * li + 1]))
+            put(6 * li, outs[-2])
+            put(6 * li + 1, outs[-1])
 
         dseg0 = _as4(dsegs[-1], n, h, w)
         return (None, dseg0, *pgrads)

@@ -95,6 +95,10 @@ class GradientSynchronizer:
         for p in self.params:
             h = p.register_post_accumulate_grad_hook(self._on_grad_ready)
             self._hooks.append(h)
+            # manual-backward Functions (ops/denseblock.py) write grads
+            # straight into the arena views and call mark_ready instead
+            # of returning them to autograd (no per-tensor add kernels)
+            p._dlb_sink = self
 
     def _seal(self, start, end, params):
         b = _Bucket(start, end, list(params))
@@ -125,6 +129,9 @@ class GradientSynchronizer:
         if b.pending == 0:
             self._launch(b)
 
+    # public entry for manual-backward code that wrote the grad directly
+    mark_ready = _on_grad_ready
+
     def _launch(self, b: _Bucket) -> None:
         if b.launched:
             return
@@ -152,3 +159,6 @@ class GradientSynchronizer:
         for h in self._hooks:
             h.remove()
         self._hooks.clear()
+        for p in self.params:
+            if hasattr(p, "_dlb_sink"):
+                del p._dlb_sink
