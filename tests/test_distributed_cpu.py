@@ -180,3 +180,54 @@ def test_deferred_sync_launches_at_finish(free_port):
     for l0, l1, r0, r1 in zip(local0, local1, reduced0, reduced1):
         assert torch.allclose(r0, r1, atol=1e-6)       # ranks agree
         assert torch.allclose(r0, 0.5 * l0 + 0.5 * l1, atol=1e-6)
+
+
+def _direct_write_worker(rank, world):
+    """The manual-backward grad path: grads written straight into the
+    arena views + mark_ready, never returned to autograd (the GPU
+    dense-block backward uses this; semantics pinned here on gloo)."""
+    from dynamic_load_balance_distributeddnn_amd.parallel import \
+        GradientSynchronizer
+
+    torch.manual_seed(21)
+    model = torch.nn.Sequential(torch.nn.Linear(8, 16), torch.nn.Linear(16, 4))
+    sync = GradientSynchronizer(model, bucket_bytes=128)
+    weights = [0.6, 0.4]
+    sync.set_weight(weights[rank])
+    assert all(getattr(p, "_dlb_sink", None) is sync
+               for p in model.parameters())
+
+    torch.manual_seed(100 + rank)
+    fake = {n: torch.randn_like(p) for n, p in model.named_parameters()}
+    sync.zero()
+    with torch.no_grad():
+        for n, p in model.named_parameters():
+            p.grad.copy_(fake[n])          # direct write into arena view
+            sync.mark_ready(p)             # bucket notification
+    sync.finish()
+    return fake, {n: p.grad.clone() for n, p in model.named_parameters()}
+
+
+def test_direct_grad_write_and_mark_ready(free_port):
+    res = run_distributed(_direct_write_worker, 2, free_port)
+    fake0, red0 = res[0]
+    fake1, red1 = res[1]
+    for n in red0:
+        assert torch.allclose(red0[n], red1[n], atol=1e-6)
+        want = 0.6 * fake0[n] + 0.4 * fake1[n]
+        assert torch.allclose(red0[n], want, atol=1e-5), n
+
+
+def test_arena_views_are_16B_aligned():
+    """The fused kernels vector-load gamma/beta from arena views; the
+    8-element offset alignment guarantees 16-byte pointers."""
+    from dynamic_load_balance_distributeddnn_amd.parallel import \
+        GradientSynchronizer
+
+    model = torch.nn.Sequential(torch.nn.Linear(5, 7), torch.nn.Linear(7, 3),
+                                torch.nn.Linear(3, 11))
+    sync = GradientSynchronizer(model)
+    for p in model.parameters():
+        off, _ = sync.offsets[id(p)]
+        assert off % 8 == 0
+        assert p.grad.data_ptr() % 16 == 0
