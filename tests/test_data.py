@@ -82,3 +82,32 @@ def test_lm_tokens_in_vocab():
     tokens = make_lm_tokens(train=True)
     assert tokens.min() >= 0
     assert tokens.max() < 33278
+
+
+def test_corpus_tokenize_real_files(tmp_path):
+    """End-to-end Corpus over real text files (the path used when
+    rnn_data/wikitext-2 is present — reference dataloader.py:120-163)."""
+    from dynamic_load_balance_distributeddnn_amd.data.corpus import (
+        Corpus, corpus_available)
+
+    d = tmp_path / "wikitext-2"
+    d.mkdir()
+    (d / "train.txt").write_text("the cat sat\nthe dog ran\n")
+    (d / "valid.txt").write_text("the cat ran\n")
+    (d / "test.txt").write_text("a dog\n")
+    assert corpus_available(str(d))
+    c = Corpus(str(d))
+    # every line ends with <eos>; words are interned in first-seen order
+    assert c.train.tolist()[:4] == [0, 1, 2, 3]          # the cat sat <eos>
+    assert c.train[-1].item() == c.dictionary.word2idx["<eos>"]
+    assert len(c.dictionary) == 8  # the cat sat <eos> dog ran a + none dup
+    # valid/test reuse the shared dictionary (reference behavior)
+    assert c.valid[0].item() == c.dictionary.word2idx["the"]
+    assert c.test[0].item() == c.dictionary.word2idx["a"]
+
+
+def test_corpus_available_false_when_missing(tmp_path):
+    from dynamic_load_balance_distributeddnn_amd.data.corpus import \
+        corpus_available
+
+    assert not corpus_available(str(tmp_path / "nope"))

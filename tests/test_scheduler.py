@@ -86,3 +86,52 @@ def test_scheduler_disabled_keeps_partition():
     sched.step(np.array([1.0, 5.0, 1.0, 1.0]))
     assert (sched.batches == before).all()
     assert before.sum() == 100  # 25 each
+
+
+# ------------------------------------------------------- property tests
+try:
+    from hypothesis import given, settings
+    from hypothesis import strategies as st
+
+    @settings(max_examples=200, deadline=None)
+    @given(
+        times=st.lists(st.floats(min_value=1e-3, max_value=1e3,
+                                 allow_nan=False), min_size=1, max_size=16),
+        batch=st.integers(min_value=1, max_value=4096),
+    )
+    def test_solve_partition_properties(times, batch):
+        """Exact sum, floor respected (when feasible), deterministic,
+        replicated-decision safe for ANY positive time vector."""
+        from dynamic_load_balance_distributeddnn_amd.scheduler import \
+            solve_partition
+
+        n = len(times)
+        t = np.asarray(times)
+        frac = np.full(n, 1.0 / n)
+        if batch < n:  # documented contract: every rank needs >= 1 sample
+            with pytest.raises(ValueError):
+                solve_partition(t, frac, batch)
+            return
+        out = solve_partition(t, frac, batch)
+        assert out.sum() == batch
+        assert out.dtype.kind == "i"
+        assert (out >= 1).all()
+        # determinism: same inputs -> same outputs (replicated decision)
+        again = solve_partition(t.copy(), frac.copy(), batch)
+        assert (out == again).all()
+
+    @settings(max_examples=100, deadline=None)
+    @given(
+        ratio=st.floats(min_value=1.5, max_value=50.0),
+        batch=st.integers(min_value=8, max_value=2048),
+    )
+    def test_slower_rank_gets_fewer(ratio, batch):
+        from dynamic_load_balance_distributeddnn_amd.scheduler import \
+            solve_partition
+
+        t = np.array([1.0, float(ratio)])
+        out = solve_partition(t, np.array([0.5, 0.5]), batch)
+        assert out.sum() == batch
+        assert out[1] <= out[0]
+except ImportError:  # pragma: no cover - hypothesis always in this image
+    pass
