@@ -100,7 +100,7 @@ def test_corpus_tokenize_real_files(tmp_path):
     # every line ends with <eos>; words are interned in first-seen order
     assert c.train.tolist()[:4] == [0, 1, 2, 3]          # the cat sat <eos>
     assert c.train[-1].item() == c.dictionary.word2idx["<eos>"]
-    assert len(c.dictionary) == 8  # the cat sat <eos> dog ran a + none dup
+    assert len(c.dictionary) == 7  # the cat sat <eos> dog ran a
     # valid/test reuse the shared dictionary (reference behavior)
     assert c.valid[0].item() == c.dictionary.word2idx["the"]
     assert c.test[0].item() == c.dictionary.word2idx["a"]
