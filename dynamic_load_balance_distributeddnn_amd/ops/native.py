@@ -95,10 +95,12 @@ def group_norm_act_cat(segs, num_groups, weight, bias, eps=1e-5, relu=False):
 class _Conv2d(torch.autograd.Function):
     """NHWC bf16 implicit-GEMM conv (gfx950 MFMA kernels).
 
-    Weight/bias parameters stay fp32 masters; the forward casts the
-    weight to a channels_last bf16 copy itself (no autocast wrapping),
-    and the backward returns an fp32 weight grad directly — matching the
-    bf16-compute / fp32-master-grad regime of the whole framework.
+    Weight/bias parameters stay fp32 masters; the forward obtains the
+    bf16 channels_last weight via ``weight_bf16`` — a free view of the
+    SGD-maintained mirror when the param lives in the flat arena, or a
+    per-call cast otherwise — and the backward returns an fp32 weight
+    grad directly, matching the bf16-compute / fp32-master-grad regime
+    of the whole framework.
     """
 
     @staticmethod
