@@ -50,3 +50,25 @@ def test_bench_torchrun_two_ranks(free_port):
     rec = json.loads(line)
     assert rec["n_gpus"] == 2
     assert rec["config"]["parallelism"] == "dbs-dp2"
+
+
+def test_bench_flagship_two_ranks_cpu(free_port):
+    """The driver's SCALE run shape, on gloo: flagship model, 2 ranks,
+    split global batch, MAX-over-ranks aggregation."""
+    env = dict(os.environ, MASTER_ADDR="127.0.0.1",
+               MASTER_PORT=str(free_port))
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", str(free_port), "bench.py", "--gpus", "2",
+         "--steps", "2", "--warmup", "1", "--device", "cpu",
+         "--global-batch", "16", "--model", "densenet"],
+        capture_output=True, text=True, timeout=600, cwd=REPO, env=env)
+    assert out.returncode == 0, out.stderr[-1500:]
+    line = [l for l in out.stdout.strip().splitlines()
+            if l.startswith("{")][-1]
+    rec = json.loads(line)
+    assert rec["n_gpus"] == 2
+    assert rec["config"]["model"] == "DenseNet-121"
+    assert rec["config"]["global_batch"] == 16
+    assert rec["value"] > 0
