@@ -10,8 +10,11 @@
 // materialized (and never saved for backward: the weight-grad kernel
 // re-normalizes at load time from the same saved stats).
 //
-// Numerics are bit-identical to the unfused pair: the same fp32
-// normalize expression rounded to bf16 feeds the same MFMA tiling.
+// Numerics: gn(x) is evaluated as x*a + b with per-(sample, channel)
+// precomputed a = rstd*gamma and b = beta - mean*a — algebraically equal
+// to the unfused kernel's ((x-mu)*rstd)*gamma + beta, fp32-rounded one
+// ulp apart; the result feeds the same MFMA tiling (validated against
+// the unfused pair in tests/test_gpu_gnconv.py).
 
 #include "common.h"
 
