@@ -5,7 +5,7 @@ Usage (on a GPU box):
     python tools/gn_bwd_bench.py [--iters 50] [--accumulate]
 
 Prints per-shape kernel time and effective HBM throughput assuming the
-kernel's 6 logical tensor passes (x и dz twice each, dx write, dx read
+kernel's 6 logical tensor passes (x and dz twice each, dx write, dx read
 in accumulate mode), so variants can be compared against the ~8 TB/s
 roof.  Env knobs DLB_GN_TARGET_BWD etc. apply.
 """
