@@ -231,3 +231,43 @@ def test_arena_views_are_16B_aligned():
         off, _ = sync.offsets[id(p)]
         assert off % 8 == 0
         assert p.grad.data_ptr() % 16 == 0
+
+
+def _four_rank_worker(rank, world):
+    """4-way: time exchange ordering + weighted reduce + partition step
+    (the mechanics the driver's 4/8-GPU scaling run exercises)."""
+    from dynamic_load_balance_distributeddnn_amd.parallel import \
+        GradientSynchronizer
+    from dynamic_load_balance_distributeddnn_amd.scheduler import (
+        DBSScheduler, exchange_times)
+
+    times = exchange_times(1.0 + rank * 0.5)
+    sched = DBSScheduler(world, global_batch=512)
+    sched.step(times)
+
+    torch.manual_seed(3)
+    model = torch.nn.Linear(16, 8)
+    sync = GradientSynchronizer(model, bucket_bytes=64)
+    sync.set_weight(float(sched.weights[rank]))
+    torch.manual_seed(200 + rank)
+    x = torch.randn(4, 16)
+    sync.zero()
+    (model(x) ** 2).mean().backward()
+    sync.finish()
+    return dict(times=times.tolist(), batches=sched.batches.tolist(),
+                weights=float(sched.weights[rank]),
+                grad=model.weight.grad.clone())
+
+
+def test_four_rank_mechanics(free_port):
+    res = run_distributed(_four_rank_worker, 4, free_port)
+    t0 = res[0]["times"]
+    assert t0 == [1.0, 1.5, 2.0, 2.5]
+    for r in range(4):
+        assert res[r]["times"] == t0
+        assert res[r]["batches"] == res[0]["batches"]
+        assert torch.allclose(res[r]["grad"], res[0]["grad"], atol=1e-6)
+    assert sum(res[0]["batches"]) == 512
+    b = res[0]["batches"]
+    assert b[0] >= b[1] >= b[2] >= b[3]     # faster ranks get more
+    assert abs(sum(res[r]["weights"] for r in range(4)) - 1.0) < 1e-6
