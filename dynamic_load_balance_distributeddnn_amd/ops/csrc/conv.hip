@@ -621,7 +621,6 @@ conv_wrw_kernel(const WrwParams p) {
   const int mstart = blockIdx.z * p.m_per_split;
   const int mend = min(p.M, mstart + p.m_per_split);
   const bool xvec = (p.Ci & 7) == 0;
-  const bool dvec = (p.Co & 7) == 0;
 
   f32x4 acc[FA][FB];
 #pragma unroll
