@@ -8,6 +8,8 @@ shapes (the benchmarked configuration — BASELINE.json) are used.
 
 from __future__ import annotations
 
+import logging
+
 import torch
 
 from .corpus import Corpus, corpus_available
@@ -40,7 +42,13 @@ def load_cv_dataset(name: str, train: bool, seed: int = 1234):
         tfm = transforms.Compose(aug if train else aug[2:])
         ctor = datasets.CIFAR10 if name == "cifar10" else datasets.CIFAR100
         return ctor("./data", train=train, download=False, transform=tfm)
-    except Exception:
+    except (ImportError, FileNotFoundError, RuntimeError) as e:
+        # Only the "dataset not present / torchvision absent" class of
+        # failures falls back to synthetic — and loudly, so a corrupt
+        # ./data tree can't silently train on random tensors.
+        logging.getLogger(__name__).warning(
+            "real %s unavailable (%s: %s); using synthetic data",
+            name, type(e).__name__, e)
         return make_cv_dataset(name, train, seed)
 
 

@@ -164,6 +164,7 @@ class Trainer:
                      one_cycle_lr(args.learning_rate, epoch, args.epoch_size))
 
         source, steps = self.repartition(epoch)
+        self.steps_per_epoch = steps
         self.model.train()
         self.timer.reset()
         if dist.is_initialized():
@@ -197,7 +198,11 @@ class Trainer:
     @torch.no_grad()
     def validate_epoch(self, epoch: int):
         """Full validation every epoch on every rank (reference
-        dbs.py:141-181 semantics, including its normalization quirks)."""
+        dbs.py:141-181 semantics, including its normalization quirks:
+        the returned/recorded val_loss carries the reference's extra
+        division by the TRAIN steps-per-epoch — dbs.py:160-161, 180-181 —
+        while the LM 'accuracy' = 1 − val_loss uses the UNdivided loss)."""
+        num_batches = max(1, getattr(self, "steps_per_epoch", 1))
         self.model.eval()
         if self.is_lm:
             sheet = D.batchify(self.val_tokens, 10).to(self.device)
@@ -210,6 +215,7 @@ class Trainer:
                 denom += len(inputs)
             val_loss /= max(1, denom)
             accuracy = 1.0 - val_loss  # reference's LM "accuracy" (dbs.py:181)
+            val_loss /= num_batches   # recorded scale quirk (dbs.py:180-181)
             if self.logger:
                 self.logger.info(f"Rank {self.rank}, epoch {epoch}, "
                                  f"val_loss {val_loss:.4f}")
@@ -228,6 +234,7 @@ class Trainer:
             correct += (out.argmax(1) == target).sum().item()
             total += target.numel()
         val_loss /= max(1, total)
+        val_loss /= num_batches  # recorded scale quirk (dbs.py:160-161)
         accuracy = 100.0 * correct / max(1, total)
         if self.logger:
             self.logger.info(f"Rank {self.rank}, epoch {epoch}, "

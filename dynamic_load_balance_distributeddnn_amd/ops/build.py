@@ -19,8 +19,13 @@ def build(verbose: bool = False):
     os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
     from torch.utils.cpp_extension import load
 
-    sources = sorted(glob.glob(os.path.join(CSRC, "*.cpp")) +
-                     glob.glob(os.path.join(CSRC, "*.hip")))
+    # Exclude "*_hip.hip": torch's extension builder writes hipify
+    # passthrough copies of each .hip next to the original; globbing them
+    # on a rebuild in a used tree would double-compile every kernel.
+    sources = sorted(
+        s for s in (glob.glob(os.path.join(CSRC, "*.cpp")) +
+                    glob.glob(os.path.join(CSRC, "*.hip")))
+        if not s.endswith("_hip.hip"))
     mod = load(
         name="_dlb_kernels",
         sources=sources,

@@ -56,6 +56,7 @@ class StepTimer:
 
     # -------------------------------------------------- per-iteration
     def iter_start(self):
+        self._cur_inject = 0.0
         if self.is_cuda:
             e = self._event(); e.record()
             self._cur = [e]
@@ -70,6 +71,7 @@ class StepTimer:
 
     def step_done(self):
         self._mark()
+        self._cur.append(self._cur_inject)
         self._marks.append(self._cur)
 
     def _mark(self):
@@ -80,8 +82,15 @@ class StepTimer:
             self._cur.append(time.perf_counter())
 
     def add_compute(self, seconds: float) -> None:
-        """Fold host-side injected delay (fault injector) into compute."""
+        """Fold host-side injected delay (fault injector) into compute.
+
+        The sleep happens between backward_done and comm_done, so the same
+        gap also lands in this iteration's e1→e2 interval; it is remembered
+        here and subtracted from that interval in epoch_totals so the
+        injected delay is attributed exactly once (to compute)."""
         self.compute_s += seconds
+        if hasattr(self, "_cur_inject"):
+            self._cur_inject += seconds
 
     # -------------------------------------------------- epoch close
     def epoch_totals(self) -> tuple[float, float]:
@@ -91,13 +100,13 @@ class StepTimer:
         """
         if self.is_cuda:
             torch.cuda.synchronize()
-            for e0, e1, e2, e3 in self._marks:
+            for e0, e1, e2, e3, inj in self._marks:
                 self.compute_s += (e0.elapsed_time(e1) + e2.elapsed_time(e3)) / 1e3
-                self.sync_s += e1.elapsed_time(e2) / 1e3
+                self.sync_s += max(0.0, e1.elapsed_time(e2) / 1e3 - inj)
         else:
-            for t0, t1, t2, t3 in self._marks:
+            for t0, t1, t2, t3, inj in self._marks:
                 self.compute_s += (t1 - t0) + (t3 - t2)
-                self.sync_s += t2 - t1
+                self.sync_s += max(0.0, (t2 - t1) - inj)
         self._marks.clear()
         self._used = 0
         return self.compute_s, self.sync_s
