@@ -185,11 +185,20 @@ def main():
         dist.barrier()
     elapsed = time.perf_counter() - t0
 
-    # MAX over ranks (slowest rank defines the job)
+    # Gather every rank's elapsed: MAX defines the job (slowest rank),
+    # and the spread gives the straggler idle % — the second half of the
+    # BASELINE metric (Σ(max−t_i)/(N·max), see scheduler.straggler_idle_pct).
+    idle_pct = 0.0
     if world > 1:
-        t = torch.tensor([elapsed], device=device if device.type == "cuda" else "cpu")
-        dist.all_reduce(t, op=dist.ReduceOp.MAX)
-        elapsed = float(t.item())
+        t = torch.tensor([elapsed],
+                         device=device if device.type == "cuda" else "cpu")
+        allt = torch.empty(world, dtype=t.dtype, device=t.device)
+        dist.all_gather_into_tensor(allt, t)
+        per_rank = allt.cpu().numpy()
+        elapsed = float(per_rank.max())
+        from dynamic_load_balance_distributeddnn_amd.scheduler import \
+            straggler_idle_pct
+        idle_pct = straggler_idle_pct(per_rank)
 
     if rank == 0:
         value = items_per_step * args.steps / elapsed
@@ -206,6 +215,7 @@ def main():
             "vs_baseline": None,
             "dtype": args.dtype,
             "data": "synthetic",
+            "straggler_idle_pct": round(idle_pct, 3),
             "config": {
                 "model": "DenseNet-121" if args.model == "densenet" else args.model,
                 "global_batch": args.global_batch,

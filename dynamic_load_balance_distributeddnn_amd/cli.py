@@ -75,6 +75,16 @@ def get_parser() -> argparse.ArgumentParser:
                    help="Enable the one-cycle learning-rate policy.")
     p.add_argument("-de", "--disable_enhancements", type=str2bool, default=False,
                    help="Ablation: disable one-cycle LR and weighted averaging.")
+    # ---- extension flags (beyond the reference's frozen 13; not part of
+    # the base_filename schema) -------------------------------------------
+    p.add_argument("-dbsi", "--dbs_interval", type=int, default=0,
+                   help="Re-partition every N iterations instead of every "
+                        "epoch (0 = per-epoch, the reference's cadence). "
+                        "Per-iteration hipEvent times feed an EMA that "
+                        "drives the solver mid-epoch. CV models only; the "
+                        "LM path keeps per-epoch cadence (its batchified "
+                        "token sheet fixes the batch width for sequence "
+                        "continuity).")
     return p
 
 

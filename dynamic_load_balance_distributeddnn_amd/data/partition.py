@@ -23,7 +23,8 @@ import numpy as np
 import torch
 from torch.utils.data import DataLoader, Dataset, Subset
 
-__all__ = ["partition_cv", "partition_lm", "batchify", "bptt_batch"]
+__all__ = ["partition_cv", "partition_lm", "batchify", "bptt_batch",
+           "GlobalBatchStream"]
 
 
 def _shard_bounds(batches: np.ndarray, steps: int) -> np.ndarray:
@@ -67,6 +68,47 @@ def partition_cv(
         pin_memory=torch.cuda.is_available(),
     )
     return loader, steps
+
+
+class GlobalBatchStream:
+    """Iteration-granularity CV data source (the `-dbsi` mode).
+
+    The per-epoch sharding above fixes each rank's slice for the whole
+    epoch, so the batch split can only change at epoch boundaries — the
+    reference's cadence (dbs.py:385-395).  The north star asks for
+    re-partitioning *every iteration*; this stream makes that sound:
+    one replicated global permutation per epoch, step ``s`` consumes
+    indices ``[s*GB, (s+1)*GB)``, and each rank takes the contiguous
+    sub-slice its CURRENT split assigns.  The split may change between
+    any two iterations without skewing sample coverage (each sample is
+    seen exactly once per epoch) or iteration counts
+    (``steps = len(dataset) // GB`` on every rank regardless of split).
+    """
+
+    def __init__(self, dataset: Dataset, global_batch: int, seed: int,
+                 epoch: int):
+        self.dataset = dataset
+        self.global_batch = int(global_batch)
+        data_len = len(dataset)
+        self.steps = data_len // self.global_batch
+        if self.steps == 0:
+            raise ValueError(
+                f"dataset of {data_len} samples < global batch {global_batch}")
+        g = torch.Generator().manual_seed(seed * 100_003 + epoch)
+        self.perm = torch.randperm(data_len, generator=g)
+
+    def batch(self, step: int, batches: np.ndarray, rank: int):
+        """(inputs, targets) for ``rank`` at global step ``step`` under the
+        current integer split ``batches`` (sum == global_batch)."""
+        batches = np.asarray(batches, dtype=np.int64)
+        off = int(batches[:rank].sum())
+        base = step * self.global_batch
+        idx = self.perm[base + off: base + off + int(batches[rank])]
+        tensors = getattr(self.dataset, "tensors", None)
+        if tensors is not None:  # TensorDataset fast path (synthetic data)
+            return tuple(t[idx] for t in tensors)
+        items = [self.dataset[int(i)] for i in idx]
+        return torch.utils.data.default_collate(items)
 
 
 def batchify(tokens: torch.Tensor, width: int) -> torch.Tensor:

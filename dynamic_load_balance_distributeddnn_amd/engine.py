@@ -29,7 +29,7 @@ from . import data as D
 from .models import LM_CONFIG, build_model
 from .parallel import GradientSynchronizer, StepTimer
 from .parallel.optim import FlatSGD
-from .scheduler import DBSScheduler, exchange_times
+from .scheduler import DBSScheduler, exchange_times, straggler_idle_pct
 from .utils import FaultInjector, StatsRecorder
 from .utils.lr_policy import apply_lr, one_cycle_lr
 
@@ -80,6 +80,15 @@ class Trainer:
                                    args.fault_tolerance_chance, rank,
                                    logger=logger)
         self.nodes_time = np.ones(world_size)
+        # Iteration-granularity DBS (`-dbsi N`): re-partition every N
+        # iterations from an EMA of per-iteration hipEvent compute times
+        # (the north star's "every iteration" cadence; N=0 keeps the
+        # reference's per-epoch cadence).  CV only — the LM token sheet
+        # fixes its batch width for sequence continuity.
+        self.dbs_interval = int(getattr(args, "dbs_interval", 0) or 0)
+        if self.is_lm:
+            self.dbs_interval = 0
+        self._ema_iter_s: float | None = None
 
         # datasets built once; partitioned fresh each epoch
         if self.is_lm:
@@ -127,8 +136,28 @@ class Trainer:
         if self.is_lm:
             return D.partition_lm(self.train_tokens, batches, self.rank,
                                   self.bptt)
+        if self.dbs_interval > 0 and self.args.dynamic_batch_size:
+            stream = D.GlobalBatchStream(self.train_data, self.args.batch_size,
+                                         self.seed, epoch)
+            return stream, stream.steps
         return D.partition_cv(self.train_data, batches, self.rank,
                               self.seed, epoch)
+
+    def _interval_repartition(self, drained_iters: int) -> None:
+        """Mid-epoch DBS step: drain the hipEvent timer, EMA the
+        per-iteration compute time, exchange, re-solve the split."""
+        dc, _ = self.timer.drain()
+        per_iter = dc / max(1, drained_iters)
+        self._ema_iter_s = (per_iter if self._ema_iter_s is None
+                            else 0.5 * self._ema_iter_s + 0.5 * per_iter)
+        if dist.is_initialized() and dist.get_world_size() > 1:
+            times = exchange_times(self._ema_iter_s, self.device)
+        else:
+            times = np.asarray([self._ema_iter_s])
+        self.sched.step(times)
+        w = (1.0 / self.world_size if self.args.disable_enhancements
+             else self.sched.weights[self.rank])
+        self.sync.set_weight(w)
 
     # ------------------------------------------------------------------
     def _step(self, inputs, target, epoch, steps_per_epoch):
@@ -177,6 +206,21 @@ class Trainer:
                 inputs, target = D.bptt_batch(sheet, i, self.bptt)
                 loss = self._step(inputs, target, epoch, steps)
                 epoch_loss += loss.detach()
+        elif isinstance(source, D.GlobalBatchStream):
+            k = self.dbs_interval
+            since = 0
+            for s in range(steps):
+                if since >= k:
+                    self._interval_repartition(since)
+                    since = 0
+                inputs, target = source.batch(s, self.sched.batches, self.rank)
+                inputs = inputs.to(self.device, non_blocking=True)
+                if self.device.type == "cuda":
+                    inputs = inputs.contiguous(memory_format=torch.channels_last)
+                target = target.to(self.device, non_blocking=True)
+                loss = self._step(inputs, target, epoch, steps)
+                epoch_loss += loss.detach()
+                since += 1
         else:
             for inputs, target in source:
                 inputs = inputs.to(self.device, non_blocking=True)
@@ -255,11 +299,20 @@ class Trainer:
             wallclock += time.time() - t0
             val_loss, accuracy = self.validate_epoch(epoch)
 
-            if args.dynamic_batch_size:
+            # Exchange per-rank compute times every epoch regardless of
+            # -dbs: the DBS solver consumes them when enabled, and the
+            # straggler idle % (half of the BASELINE metric) needs them
+            # either way — the -dbs false A/B run is exactly where the
+            # idle number is expected to stay high.
+            if dist.is_initialized() and dist.get_world_size() > 1:
                 self.nodes_time = exchange_times(compute_s, self.device)
-                if self.logger:
-                    self.logger.info(
-                        f"Rank {self.rank}: node times {self.nodes_time.tolist()}")
+            else:
+                self.nodes_time = np.asarray([compute_s])
+            idle_pct = straggler_idle_pct(self.nodes_time)
+            if self.logger:
+                self.logger.info(
+                    f"Rank {self.rank}: node times {self.nodes_time.tolist()}"
+                    f", straggler idle {idle_pct:.2f}%")
 
             if recorder is not None:
                 recorder.append(
@@ -267,7 +320,8 @@ class Trainer:
                     sync_time=sync_s, val_loss=val_loss, accuracy=accuracy,
                     partition=self.sched.fractions.copy(),
                     node_time=np.asarray(self.nodes_time).copy(),
-                    wallclock_time=wallclock)
+                    wallclock_time=wallclock,
+                    straggler_idle_pct=idle_pct)
 
         if recorder is not None:
             recorder.save()

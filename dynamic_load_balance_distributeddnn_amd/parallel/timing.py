@@ -93,11 +93,15 @@ class StepTimer:
             self._cur_inject += seconds
 
     # -------------------------------------------------- epoch close
-    def epoch_totals(self) -> tuple[float, float]:
-        """(compute seconds, sync seconds) for all iterations since reset.
+    def drain(self) -> tuple[float, float]:
+        """Consume completed iteration marks into the running totals;
+        returns the (compute_s, sync_s) DELTA for the drained span.
 
-        On GPU this synchronizes once (events must have completed).
+        On GPU this synchronizes once.  Used mid-epoch by the
+        iteration-granularity DBS mode (the per-interval sensor) and by
+        epoch_totals at epoch close.
         """
+        c0, s0 = self.compute_s, self.sync_s
         if self.is_cuda:
             torch.cuda.synchronize()
             for e0, e1, e2, e3, inj in self._marks:
@@ -109,4 +113,9 @@ class StepTimer:
                 self.sync_s += max(0.0, (t2 - t1) - inj)
         self._marks.clear()
         self._used = 0
+        return self.compute_s - c0, self.sync_s - s0
+
+    def epoch_totals(self) -> tuple[float, float]:
+        """(compute seconds, sync seconds) for all iterations since reset."""
+        self.drain()
         return self.compute_s, self.sync_s

@@ -34,8 +34,31 @@ import torch.distributed as dist
 __all__ = [
     "solve_partition",
     "exchange_times",
+    "straggler_idle_pct",
     "DBSScheduler",
 ]
+
+
+def straggler_idle_pct(nodes_time: np.ndarray) -> float:
+    """Straggler idle percentage — the second half of the BASELINE metric.
+
+    With synchronous SGD every rank waits for the slowest, so the fraction
+    of whole-node compute capacity lost to stragglers over a window is
+
+        idle% = 100 * Σ_rank (max_t − t_rank) / (N * max_t)
+
+    computed from the rank-ordered pure-compute vector the DBS loop
+    already exchanges (the reference's node_time, dbs.py:425).  0 means
+    perfectly balanced; DBS drives this toward 0 while a fixed split
+    under a straggler pins it high.
+    """
+    t = np.asarray(nodes_time, dtype=np.float64)
+    if t.size == 0:
+        return 0.0
+    m = float(t.max())
+    if m <= 0:
+        return 0.0
+    return float(100.0 * (m - t).sum() / (t.size * m))
 
 
 def solve_partition(

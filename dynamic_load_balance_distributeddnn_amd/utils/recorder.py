@@ -21,7 +21,12 @@ import numpy as np
 __all__ = ["init_logger", "StatsRecorder"]
 
 FIELDS = ("epoch", "train_loss", "train_time", "sync_time", "val_loss",
-          "accuracy", "partition", "node_time", "wallclock_time")
+          "accuracy", "partition", "node_time", "wallclock_time",
+          # extension beyond the reference's 9 lists: the BASELINE
+          # metric's "straggler idle %" per epoch, derived from node_time
+          # (scheduler.straggler_idle_pct) — the 9 reference lists above
+          # keep their exact names and order.
+          "straggler_idle_pct")
 
 
 def init_logger(args, rank: int, base_filename: str, output_dir: str = "./logs"):
