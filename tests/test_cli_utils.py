@@ -88,9 +88,11 @@ def test_recorder_layout(tmp_path):
     assert os.path.basename(path) == "m-ds-node0-x.npy"
     loaded = np.load(path, allow_pickle=True).item()
     assert loaded["epoch"] == [0]
+    # the reference's 9 lists plus the straggler_idle_pct extension
     assert set(loaded.keys()) == {
         "epoch", "train_loss", "train_time", "sync_time", "val_loss",
-        "accuracy", "partition", "node_time", "wallclock_time"}
+        "accuracy", "partition", "node_time", "wallclock_time",
+        "straggler_idle_pct"}
 
 
 def test_checkpoint_roundtrip(tmp_path, monkeypatch):
