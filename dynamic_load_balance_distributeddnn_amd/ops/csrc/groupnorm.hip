@@ -10,14 +10,20 @@
 //        recomputed from saved stats so no mask tensor is stored.
 //
 // Layout: NHWC ([N, HW, C] contiguous in C) — the layout the whole CV
-// path runs in (channels_last).  One workgroup per sample n.  Threads are
-// organized as (pixel, channel-octet): within an octet sweep, a thread
-// owns a FIXED run of 8 channels across pixels strided by TP, so partial
-// sums live in 8 statically-indexed registers (no scratch — CDNA guide
-// rule 20) and the group merge is 8 LDS atomics per thread per sweep.
-// All loads are 16-byte bf16x8, fully coalesced.  An outer octet loop
-// covers C > 8*blockDim (e.g. DenseNet-161's C=2208).  Requires C%8==0
-// (every channel count in the zoo satisfies this).
+// path runs in (channels_last).  Grid = (N, channel-chunks): GroupNorm's
+// reductions are per-(sample, group), so a chunk of WHOLE groups is
+// fully independent of every other chunk — slicing C (at group-aligned
+// octet granularity) multiplies blocks without scratch buffers, extra
+// passes, or any cross-block coordination.  Round-1 ran one block per
+// sample (N=512 -> 2 blocks/CU) and the small-HW DenseNet layers were
+// latency-bound at 0.4-1.6 TB/s (profiles/, gn_bwd_bench); round-1's
+// alternative (HW slicing + scratch reduce) measured slower at every
+// target and is gone.  Threads within a block are (pixel, channel-octet)
+// as before: a thread owns a FIXED run of 8 channels across pixels
+// strided by TP, so partial sums live in 8 statically-indexed registers
+// (no scratch — CDNA guide rule 20) and the group merge is 8 LDS atomics
+// per thread per sweep.  All loads are 16-byte bf16x8, fully coalesced.
+// Requires C%8==0 (every channel count in the zoo satisfies this).
 
 #include "common.h"
 
@@ -59,31 +65,53 @@ __device__ inline const bf16* seg_locate(const GnSegs& sg, int c0, int& cloc,
   return sg.p[si];
 }
 
+// This block's octet range [o0, o1) and thread mapping for it.
+// chunk_oct is group-aligned, so [o0*8, o1*8) covers whole groups.
+struct ChunkMap {
+  int o0, o1;    // octet range
+  int TCe, TP;   // channel-octet threads, pixel stride
+  int tc, tp;    // this thread's coordinates
+  bool active;
+};
+
+__device__ inline ChunkMap chunk_map(int TC, int chunk_oct) {
+  ChunkMap m;
+  m.o0 = blockIdx.y * chunk_oct;
+  m.o1 = min(m.o0 + chunk_oct, TC);
+  const int span = m.o1 - m.o0;
+  m.TCe = span < GN_BLOCK ? span : GN_BLOCK;
+  m.TP = GN_BLOCK / m.TCe;
+  const int t = threadIdx.x;
+  m.tc = t % m.TCe;
+  m.tp = t / m.TCe;
+  m.active = t < m.TCe * m.TP;
+  return m;
+}
+
 // ---------------------------------------------------------------- forward
 extern "C" __global__ void __launch_bounds__(GN_BLOCK)
 gn_fwd_kernel(const GnSegs segs, bf16* __restrict__ y,
               const float* __restrict__ gamma, const float* __restrict__ beta,
               float* __restrict__ mean_out, float* __restrict__ rstd_out,
               const int HW, const int C, const int G, const float eps,
-              const int relu) {
+              const int relu, const int chunk_oct) {
   const int n = blockIdx.x;
-  const int TC = C >> 3;                     // channel-octets per pixel
-  const int TCe = TC < GN_BLOCK ? TC : GN_BLOCK;
-  const int TP = GN_BLOCK / TCe;             // pixels per sweep
+  const int TC = C >> 3;
+  const ChunkMap m = chunk_map(TC, chunk_oct);
   const int t = threadIdx.x;
-  const int tc = t % TCe, tp = t / TCe;
   const int Cg = C / G;
-  const bool active = t < TCe * TP;
+  // groups owned by this chunk (chunk boundaries are group-aligned)
+  const int g0 = (m.o0 << 3) / Cg, g1 = (m.o1 << 3) / Cg;
 
   __shared__ float s_sum[GN_MAXG];
   __shared__ float s_ssq[GN_MAXG];
   __shared__ float s_mean[GN_MAXG];
   __shared__ float s_rstd[GN_MAXG];
-  for (int g = t; g < G; g += GN_BLOCK) { s_sum[g] = 0.f; s_ssq[g] = 0.f; }
+  for (int g = g0 + t; g < g1; g += GN_BLOCK) { s_sum[g] = 0.f; s_ssq[g] = 0.f; }
   __syncthreads();
 
-  if (active) {
-    for (int oct = tc; oct < TC; oct += TCe) {
+  if (m.active) {
+    for (int oct = m.o0 + m.tc; oct < m.o1; oct += m.TCe) {
       const int c0 = oct << 3;
       int cloc, cs;
       const bf16* sb = seg_locate(segs, c0, cloc, cs);
@@ -91,7 +119,7 @@ gn_fwd_kernel(const GnSegs segs, bf16* __restrict__ y,
       float s[8] = {0, 0, 0, 0, 0, 0, 0, 0};
       float ss[8] = {0, 0, 0, 0, 0, 0, 0, 0};
       #pragma unroll 4
-      for (int p = tp; p < HW; p += TP) {
+      for (int p = m.tp; p < HW; p += m.TP) {
         Bf16x8 chunk = *reinterpret_cast<const Bf16x8*>(xb + (long)p * cs);
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
@@ -111,7 +139,7 @@ gn_fwd_kernel(const GnSegs segs, bf16* __restrict__ y,
   __syncthreads();
 
   const float inv_m = 1.0f / ((float)HW * Cg);
-  for (int g = t; g < G; g += GN_BLOCK) {
+  for (int g = g0 + t; g < g1; g += GN_BLOCK) {
     float mu = s_sum[g] * inv_m;
     float var = s_ssq[g] * inv_m - mu * mu;
     float r = rsqrtf(var + eps);
@@ -122,10 +150,10 @@ gn_fwd_kernel(const GnSegs segs, bf16* __restrict__ y,
   }
   __syncthreads();
 
-  if (!active) return;
+  if (!m.active) return;
 
   bf16* yb = y + (long)n * HW * C;
-  for (int oct = tc; oct < TC; oct += TCe) {
+  for (int oct = m.o0 + m.tc; oct < m.o1; oct += m.TCe) {
     const int c0 = oct << 3;
     int cloc, cs;
     const bf16* sb = seg_locate(segs, c0, cloc, cs);
@@ -140,7 +168,7 @@ gn_fwd_kernel(const GnSegs segs, bf16* __restrict__ y,
       rs[j] = s_rstd[g];
     }
     #pragma unroll 4
-      for (int p = tp; p < HW; p += TP) {
+      for (int p = m.tp; p < HW; p += m.TP) {
       Bf16x8 chunk = *reinterpret_cast<const Bf16x8*>(xb + (long)p * cs);
       Bf16x8 out;
 #pragma unroll
@@ -162,23 +190,21 @@ gn_fwd_kernel(const GnSegs segs, bf16* __restrict__ y,
 extern "C" __global__ void __launch_bounds__(GN_BLOCK)
 gn_stats_kernel(const GnSegs segs, float* __restrict__ mean_out,
                 float* __restrict__ rstd_out, const int HW, const int C,
-                const int G, const float eps) {
+                const int G, const float eps, const int chunk_oct) {
   const int n = blockIdx.x;
   const int TC = C >> 3;
-  const int TCe = TC < GN_BLOCK ? TC : GN_BLOCK;
-  const int TP = GN_BLOCK / TCe;
+  const ChunkMap m = chunk_map(TC, chunk_oct);
   const int t = threadIdx.x;
-  const int tc = t % TCe, tp = t / TCe;
   const int Cg = C / G;
-  const bool active = t < TCe * TP;
+  const int g0 = (m.o0 << 3) / Cg, g1 = (m.o1 << 3) / Cg;
 
   __shared__ float s_sum[GN_MAXG];
   __shared__ float s_ssq[GN_MAXG];
-  for (int g = t; g < G; g += GN_BLOCK) { s_sum[g] = 0.f; s_ssq[g] = 0.f; }
+  for (int g = g0 + t; g < g1; g += GN_BLOCK) { s_sum[g] = 0.f; s_ssq[g] = 0.f; }
   __syncthreads();
 
-  if (active) {
-    for (int oct = tc; oct < TC; oct += TCe) {
+  if (m.active) {
+    for (int oct = m.o0 + m.tc; oct < m.o1; oct += m.TCe) {
       const int c0 = oct << 3;
       int cloc, cs;
       const bf16* sb = seg_locate(segs, c0, cloc, cs);
@@ -186,7 +212,7 @@ gn_stats_kernel(const GnSegs segs, float* __restrict__ mean_out,
       float s[8] = {0, 0, 0, 0, 0, 0, 0, 0};
       float ss[8] = {0, 0, 0, 0, 0, 0, 0, 0};
       #pragma unroll 4
-      for (int p = tp; p < HW; p += TP) {
+      for (int p = m.tp; p < HW; p += m.TP) {
         Bf16x8 chunk = *reinterpret_cast<const Bf16x8*>(xb + (long)p * cs);
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
@@ -205,7 +231,7 @@ gn_stats_kernel(const GnSegs segs, float* __restrict__ mean_out,
   }
   __syncthreads();
   const float inv_m = 1.0f / ((float)HW * Cg);
-  for (int g = t; g < G; g += GN_BLOCK) {
+  for (int g = g0 + t; g < g1; g += GN_BLOCK) {
     float mu = s_sum[g] * inv_m;
     float var = s_ssq[g] * inv_m - mu * mu;
     mean_out[(long)n * G + g] = mu;
@@ -218,6 +244,7 @@ gn_stats_kernel(const GnSegs segs, float* __restrict__ mean_out,
 //   s1 = sum(g_c * dy),  s2 = sum(g_c * dy * xhat)
 // dgamma_c = sum_{n,p} dy*xhat ; dbeta_c = sum_{n,p} dy  (global atomics,
 // caller zero-fills).  ReLU mask recomputed as (xhat*g+b) > 0.
+// Dynamic LDS: [2 * chunk_oct*8] dgamma/dbeta partials, local channels.
 extern "C" __global__ void __launch_bounds__(GN_BLOCK)
 gn_bwd_kernel(const GnSegs segs, const bf16* __restrict__ dz,
               const GnSegsMut dxs, const float* __restrict__ gamma,
@@ -226,27 +253,27 @@ gn_bwd_kernel(const GnSegs segs, const bf16* __restrict__ dz,
               const float* __restrict__ rstd_in,
               float* __restrict__ dgamma, float* __restrict__ dbeta,
               const int HW, const int C, const int G, const int relu,
-              const int accumulate) {
+              const int accumulate, const int chunk_oct) {
   const int n = blockIdx.x;
   const int TC = C >> 3;
-  const int TCe = TC < GN_BLOCK ? TC : GN_BLOCK;
-  const int TP = GN_BLOCK / TCe;
+  const ChunkMap m = chunk_map(TC, chunk_oct);
   const int t = threadIdx.x;
-  const int tc = t % TCe, tp = t / TCe;
   const int Cg = C / G;
-  const bool active = t < TCe * TP;
+  const int g0 = (m.o0 << 3) / Cg, g1 = (m.o1 << 3) / Cg;
+  const int cbase = m.o0 << 3;                  // first channel of chunk
+  const int cspan = (m.o1 - m.o0) << 3;         // channels in chunk
 
   __shared__ float s_s1[GN_MAXG];
   __shared__ float s_s2[GN_MAXG];
-  extern __shared__ float s_dgb[];  // [2*C]: dgamma then dbeta partials
-  for (int g = t; g < G; g += GN_BLOCK) { s_s1[g] = 0.f; s_s2[g] = 0.f; }
-  for (int c = t; c < 2 * C; c += GN_BLOCK) s_dgb[c] = 0.f;
+  extern __shared__ float s_dgb[];  // [2*cspan]: dgamma then dbeta partials
+  for (int g = g0 + t; g < g1; g += GN_BLOCK) { s_s1[g] = 0.f; s_s2[g] = 0.f; }
+  for (int c = t; c < 2 * cspan; c += GN_BLOCK) s_dgb[c] = 0.f;
   __syncthreads();
 
   const bf16* db = dz + (long)n * HW * C;
 
-  if (active) {
-    for (int oct = tc; oct < TC; oct += TCe) {
+  if (m.active) {
+    for (int oct = m.o0 + m.tc; oct < m.o1; oct += m.TCe) {
       const int c0 = oct << 3;
       int cloc, cs;
       const bf16* sb = seg_locate(segs, c0, cloc, cs);
@@ -262,7 +289,7 @@ gn_bwd_kernel(const GnSegs segs, const bf16* __restrict__ dz,
       }
       float a1[8] = {0}, a2[8] = {0}, adg[8] = {0}, adb[8] = {0};
       #pragma unroll 4
-      for (int p = tp; p < HW; p += TP) {
+      for (int p = m.tp; p < HW; p += m.TP) {
         Bf16x8 xc = *reinterpret_cast<const Bf16x8*>(xb + (long)p * cs);
         Bf16x8 dc = *reinterpret_cast<const Bf16x8*>(db + (long)p * C + c0);
 #pragma unroll
@@ -284,23 +311,23 @@ gn_bwd_kernel(const GnSegs segs, const bf16* __restrict__ dz,
         int g = (c0 + j) / Cg;
         atomicAdd(&s_s1[g], a1[j]);
         atomicAdd(&s_s2[g], a2[j]);
-        atomicAdd(&s_dgb[c0 + j], adg[j]);
-        atomicAdd(&s_dgb[C + c0 + j], adb[j]);
+        atomicAdd(&s_dgb[c0 - cbase + j], adg[j]);
+        atomicAdd(&s_dgb[cspan + c0 - cbase + j], adb[j]);
       }
     }
   }
   __syncthreads();
 
   // publish per-channel param grads (one global atomic per channel)
-  for (int c = t; c < C; c += GN_BLOCK) {
-    atomicAdd(&dgamma[c], s_dgb[c]);
-    atomicAdd(&dbeta[c], s_dgb[C + c]);
+  for (int c = t; c < cspan; c += GN_BLOCK) {
+    atomicAdd(&dgamma[cbase + c], s_dgb[c]);
+    atomicAdd(&dbeta[cbase + c], s_dgb[cspan + c]);
   }
 
-  if (!active) return;
+  if (!m.active) return;
 
   const float inv_m = 1.0f / ((float)HW * Cg);
-  for (int oct = tc; oct < TC; oct += TCe) {
+  for (int oct = m.o0 + m.tc; oct < m.o1; oct += m.TCe) {
     const int c0 = oct << 3;
     int cloc, cs;
     const bf16* sb = seg_locate(segs, c0, cloc, cs);
@@ -320,7 +347,7 @@ gn_bwd_kernel(const GnSegs segs, const bf16* __restrict__ dz,
       k2[j] = s_s2[g] * inv_m;
     }
     #pragma unroll 4
-      for (int p = tp; p < HW; p += TP) {
+      for (int p = m.tp; p < HW; p += m.TP) {
       Bf16x8 xc = *reinterpret_cast<const Bf16x8*>(xb + (long)p * cs);
       Bf16x8 dc = *reinterpret_cast<const Bf16x8*>(db + (long)p * C + c0);
       Bf16x8 out;
@@ -344,332 +371,39 @@ gn_bwd_kernel(const GnSegs segs, const bf16* __restrict__ dz,
   }
 }
 
-
-// ------------------- HW-sliced path (occupancy for any N) ---------------
-// One workgroup per sample gives only N blocks; at DenseNet's N=512 that
-// is 2 blocks/CU and the stream runs at ~25% of HBM speed (profiles/).
-// Slicing HW across grid.y multiplies blocks to DLB_GN_TARGET (~2048).
-// Pass A writes per-(sample, slice) group partials to scratch
-// [N][S][G][2] WITHOUT atomics (deterministic across runs, no zero-fill);
-// pass B reduces the S partials in-block and streams its slice.  Total
-// DRAM traffic is identical to the fused kernel (it also re-reads x for
-// the apply sweep); only the tiny stats buffer is extra.
-
-extern "C" __global__ void __launch_bounds__(GN_BLOCK)
-gn_stats_part_kernel(const GnSegs segs, float* __restrict__ part,
-                     const int HW, const int C, const int G,
-                     const int slices) {
-  const int n = blockIdx.x;
-  const int sl = blockIdx.y;
-  const int hw0 = (int)(((long)HW * sl) / slices);
-  const int hw1 = (int)(((long)HW * (sl + 1)) / slices);
-  const int TC = C >> 3;
-  const int TCe = TC < GN_BLOCK ? TC : GN_BLOCK;
-  const int TP = GN_BLOCK / TCe;
-  const int t = threadIdx.x;
-  const int tc = t % TCe, tp = t / TCe;
-  const int Cg = C / G;
-  const bool active = t < TCe * TP;
-
-  __shared__ float s_sum[GN_MAXG];
-  __shared__ float s_ssq[GN_MAXG];
-  for (int g = t; g < G; g += GN_BLOCK) { s_sum[g] = 0.f; s_ssq[g] = 0.f; }
-  __syncthreads();
-
-  if (active) {
-    for (int oct = tc; oct < TC; oct += TCe) {
-      const int c0 = oct << 3;
-      int cloc, cs;
-      const bf16* sb = seg_locate(segs, c0, cloc, cs);
-      const bf16* xb = sb + (long)n * HW * cs + cloc;
-      float sacc[8] = {0, 0, 0, 0, 0, 0, 0, 0};
-      float ss[8] = {0, 0, 0, 0, 0, 0, 0, 0};
-      #pragma unroll 4
-      for (int p2 = hw0 + tp; p2 < hw1; p2 += TP) {
-        Bf16x8 chunk = *reinterpret_cast<const Bf16x8*>(xb + (long)p2 * cs);
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          float v = bf2f(chunk.v[j]);
-          sacc[j] += v;
-          ss[j] += v * v;
-        }
-      }
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        int g = (c0 + j) / Cg;
-        atomicAdd(&s_sum[g], sacc[j]);
-        atomicAdd(&s_ssq[g], ss[j]);
-      }
-    }
-  }
-  __syncthreads();
-  float* out = part + (((long)n * slices + sl) * G) * 2;
-  for (int g = t; g < G; g += GN_BLOCK) {
-    out[g * 2 + 0] = s_sum[g];
-    out[g * 2 + 1] = s_ssq[g];
-  }
-}
-
-extern "C" __global__ void __launch_bounds__(GN_BLOCK)
-gn_apply_kernel(const GnSegs segs, bf16* __restrict__ y,
-                const float* __restrict__ gamma,
-                const float* __restrict__ beta,
-                const float* __restrict__ part, float* __restrict__ mean_out,
-                float* __restrict__ rstd_out, const int HW, const int C,
-                const int G, const float eps, const int relu,
-                const int slices) {
-  const int n = blockIdx.x;
-  const int sl = blockIdx.y;
-  const int hw0 = (int)(((long)HW * sl) / slices);
-  const int hw1 = (int)(((long)HW * (sl + 1)) / slices);
-  const int TC = C >> 3;
-  const int TCe = TC < GN_BLOCK ? TC : GN_BLOCK;
-  const int TP = GN_BLOCK / TCe;
-  const int t = threadIdx.x;
-  const int tc = t % TCe, tp = t / TCe;
-  const int Cg = C / G;
-  const bool active = t < TCe * TP;
-
-  __shared__ float s_mean[GN_MAXG];
-  __shared__ float s_rstd[GN_MAXG];
-  const float inv_m = 1.0f / ((float)HW * Cg);
-  const float* pb = part + ((long)n * slices * G) * 2;
-  for (int g = t; g < G; g += GN_BLOCK) {
-    float su = 0.f, sq = 0.f;
-    for (int s2 = 0; s2 < slices; ++s2) {
-      su += pb[((long)s2 * G + g) * 2 + 0];
-      sq += pb[((long)s2 * G + g) * 2 + 1];
-    }
-    float mu = su * inv_m;
-    float var = sq * inv_m - mu * mu;
-    float r = rsqrtf(var + eps);
-    s_mean[g] = mu;
-    s_rstd[g] = r;
-    if (sl == 0) {
-      mean_out[(long)n * G + g] = mu;
-      rstd_out[(long)n * G + g] = r;
-    }
-  }
-  __syncthreads();
-  if (!active) return;
-
-  bf16* yb = y + (long)n * HW * C;
-  for (int oct = tc; oct < TC; oct += TCe) {
-    const int c0 = oct << 3;
-    int cloc, cs;
-    const bf16* sb = seg_locate(segs, c0, cloc, cs);
-    const bf16* xb = sb + (long)n * HW * cs + cloc;
-    float ga[8], be[8], mu[8], rs[8];
-#pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      int c = c0 + j, g = c / Cg;
-      ga[j] = gamma[c];
-      be[j] = beta[c];
-      mu[j] = s_mean[g];
-      rs[j] = s_rstd[g];
-    }
-    #pragma unroll 4
-    for (int p2 = hw0 + tp; p2 < hw1; p2 += TP) {
-      Bf16x8 chunk = *reinterpret_cast<const Bf16x8*>(xb + (long)p2 * cs);
-      Bf16x8 out;
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        float v = (bf2f(chunk.v[j]) - mu[j]) * rs[j] * ga[j] + be[j];
-        if (relu) v = fmaxf(v, 0.f);
-        out.v[j] = f2bf(v);
-      }
-      *reinterpret_cast<Bf16x8*>(yb + (long)p2 * C + c0) = out;
-    }
-  }
-}
-
-// backward sliced: pass A writes per-(n, slice) s1/s2 partials (again no
-// atomics / no zero-fill) and publishes dgamma/dbeta via global atomics;
-// pass B reduces the partials and streams dx for its slice.
-extern "C" __global__ void __launch_bounds__(GN_BLOCK)
-gn_bwd_part_kernel(const GnSegs segs, const bf16* __restrict__ dz,
-                   const float* __restrict__ gamma,
-                   const float* __restrict__ beta,
-                   const float* __restrict__ mean_in,
-                   const float* __restrict__ rstd_in,
-                   float* __restrict__ part, float* __restrict__ dgamma,
-                   float* __restrict__ dbeta, const int HW, const int C,
-                   const int G, const int relu, const int slices) {
-  const int n = blockIdx.x;
-  const int sl = blockIdx.y;
-  const int hw0 = (int)(((long)HW * sl) / slices);
-  const int hw1 = (int)(((long)HW * (sl + 1)) / slices);
-  const int TC = C >> 3;
-  const int TCe = TC < GN_BLOCK ? TC : GN_BLOCK;
-  const int TP = GN_BLOCK / TCe;
-  const int t = threadIdx.x;
-  const int tc = t % TCe, tp = t / TCe;
-  const int Cg = C / G;
-  const bool active = t < TCe * TP;
-
-  __shared__ float s_s1[GN_MAXG];
-  __shared__ float s_s2[GN_MAXG];
-  extern __shared__ float s_dgb[];
-  for (int g = t; g < G; g += GN_BLOCK) { s_s1[g] = 0.f; s_s2[g] = 0.f; }
-  for (int c = t; c < 2 * C; c += GN_BLOCK) s_dgb[c] = 0.f;
-  __syncthreads();
-
-  const bf16* db = dz + (long)n * HW * C;
-  if (active) {
-    for (int oct = tc; oct < TC; oct += TCe) {
-      const int c0 = oct << 3;
-      int cloc, cs;
-      const bf16* sb = seg_locate(segs, c0, cloc, cs);
-      const bf16* xb = sb + (long)n * HW * cs + cloc;
-      float ga[8], be[8], mu[8], rs[8];
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        int c = c0 + j, g = c / Cg;
-        ga[j] = gamma[c];
-        be[j] = beta[c];
-        mu[j] = mean_in[(long)n * G + g];
-        rs[j] = rstd_in[(long)n * G + g];
-      }
-      float a1[8] = {0}, a2[8] = {0}, adg[8] = {0}, adb[8] = {0};
-      #pragma unroll 4
-      for (int p2 = hw0 + tp; p2 < hw1; p2 += TP) {
-        Bf16x8 xc = *reinterpret_cast<const Bf16x8*>(xb + (long)p2 * cs);
-        Bf16x8 dc = *reinterpret_cast<const Bf16x8*>(db + (long)p2 * C + c0);
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          float xhat = (bf2f(xc.v[j]) - mu[j]) * rs[j];
-          float dy = bf2f(dc.v[j]);
-          if (relu) {
-            float yv = xhat * ga[j] + be[j];
-            dy = yv > 0.f ? dy : 0.f;
-          }
-          a1[j] += ga[j] * dy;
-          a2[j] += ga[j] * dy * xhat;
-          adg[j] += dy * xhat;
-          adb[j] += dy;
-        }
-      }
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        int g = (c0 + j) / Cg;
-        atomicAdd(&s_s1[g], a1[j]);
-        atomicAdd(&s_s2[g], a2[j]);
-        atomicAdd(&s_dgb[c0 + j], adg[j]);
-        atomicAdd(&s_dgb[C + c0 + j], adb[j]);
-      }
-    }
-  }
-  __syncthreads();
-  float* out = part + (((long)n * slices + sl) * G) * 2;
-  for (int g = t; g < G; g += GN_BLOCK) {
-    out[g * 2 + 0] = s_s1[g];
-    out[g * 2 + 1] = s_s2[g];
-  }
-  for (int c = t; c < C; c += GN_BLOCK) {
-    atomicAdd(&dgamma[c], s_dgb[c]);
-    atomicAdd(&dbeta[c], s_dgb[C + c]);
-  }
-}
-
-extern "C" __global__ void __launch_bounds__(GN_BLOCK)
-gn_bwd_apply_kernel(const GnSegs segs, const bf16* __restrict__ dz,
-                    const GnSegsMut dxs, const float* __restrict__ gamma,
-                    const float* __restrict__ beta,
-                    const float* __restrict__ mean_in,
-                    const float* __restrict__ rstd_in,
-                    const float* __restrict__ part, const int HW,
-                    const int C, const int G, const int relu,
-                    const int slices, const int accumulate) {
-  const int n = blockIdx.x;
-  const int sl = blockIdx.y;
-  const int hw0 = (int)(((long)HW * sl) / slices);
-  const int hw1 = (int)(((long)HW * (sl + 1)) / slices);
-  const int TC = C >> 3;
-  const int TCe = TC < GN_BLOCK ? TC : GN_BLOCK;
-  const int TP = GN_BLOCK / TCe;
-  const int t = threadIdx.x;
-  const int tc = t % TCe, tp = t / TCe;
-  const int Cg = C / G;
-  if (t >= TCe * TP) return;
-
-  const float inv_m = 1.0f / ((float)HW * Cg);
-  const float* pb = part + ((long)n * slices * G) * 2;
-  const bf16* db = dz + (long)n * HW * C;
-  for (int oct = tc; oct < TC; oct += TCe) {
-    const int c0 = oct << 3;
-    int cloc, cs;
-    const bf16* sb = seg_locate(segs, c0, cloc, cs);
-    int si = 0;
-    while (si + 1 < segs.nseg && c0 >= segs.start[si + 1]) ++si;
-    const bf16* xb = sb + (long)n * HW * cs + cloc;
-    bf16* dxb = dxs.p[si] + (long)n * HW * cs + cloc;
-    float ga[8], be[8], mu[8], rs[8], k1[8], k2[8];
-#pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      int c = c0 + j, g = c / Cg;
-      ga[j] = gamma[c];
-      be[j] = beta[c];
-      mu[j] = mean_in[(long)n * G + g];
-      rs[j] = rstd_in[(long)n * G + g];
-      float su = 0.f, sq = 0.f;
-      for (int s2 = 0; s2 < slices; ++s2) {
-        su += pb[((long)s2 * G + g) * 2 + 0];
-        sq += pb[((long)s2 * G + g) * 2 + 1];
-      }
-      k1[j] = su * inv_m;
-      k2[j] = sq * inv_m;
-    }
-    #pragma unroll 4
-    for (int p2 = hw0 + tp; p2 < hw1; p2 += TP) {
-      Bf16x8 xc = *reinterpret_cast<const Bf16x8*>(xb + (long)p2 * cs);
-      Bf16x8 dc = *reinterpret_cast<const Bf16x8*>(db + (long)p2 * C + c0);
-      Bf16x8 out;
-      Bf16x8 prev;
-      if (accumulate)
-        prev = *reinterpret_cast<const Bf16x8*>(dxb + (long)p2 * cs);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        float xhat = (bf2f(xc.v[j]) - mu[j]) * rs[j];
-        float dy = bf2f(dc.v[j]);
-        if (relu) {
-          float yv = xhat * ga[j] + be[j];
-          dy = yv > 0.f ? dy : 0.f;
-        }
-        float v = rs[j] * (ga[j] * dy - (k1[j] + xhat * k2[j]));
-        if (accumulate) v += bf2f(prev.v[j]);
-        out.v[j] = f2bf(v);
-      }
-      *reinterpret_cast<Bf16x8*>(dxb + (long)p2 * cs) = out;
-    }
-  }
-}
-
 // ---------------------------------------------------------------- launchers
 static int gn_target_blocks(int bwd) {
   static int cached[2] = {-1, -1};
   if (cached[bwd] < 0) {
     const char* e = getenv(bwd ? "DLB_GN_TARGET_BWD" : "DLB_GN_TARGET");
-    cached[bwd] = e ? atoi(e) : 512;
+    cached[bwd] = e ? atoi(e) : 2048;
     if (cached[bwd] < 1) cached[bwd] = 1;
   }
   return cached[bwd];
 }
 
-// slice count so that N*slices ~ target blocks (separate fwd/bwd knobs:
-// the bwd sliced pass multiplies dgamma/dbeta global-atomic traffic by
-// the slice count, so its profitable range is narrower)
-extern "C" int dlb_gn_nslices(int N, int HW, int bwd) {
-  int s = (gn_target_blocks(bwd) + N - 1) / N;
-  if (s > HW) s = HW;
-  if (s > 32) s = 32;
-  if (s < 1) s = 1;
-  return s;
+static int gcd_i(int a, int b) { while (b) { int t = a % b; a = b; b = t; } return a; }
+
+// Chunk size in octets: group-aligned (a chunk covers whole groups) and
+// sized so N * nchunks ~ the block target.  Returns (chunk_oct, nchunks).
+static void gn_chunking(int N, int C, int G, int bwd, int* chunk_oct,
+                        int* nchunks) {
+  const int TC = C >> 3;
+  const int Cg = C / G;
+  const int u = Cg / gcd_i(Cg, 8);       // octets per group-aligned unit
+  const int units = TC / u;              // always exact (C = G*Cg, C%8==0)
+  int want = gn_target_blocks(bwd) / (N > 0 ? N : 1);
+  if (want < 1) want = 1;
+  if (want > units) want = units;
+  int cu = (units + want - 1) / want;    // units per chunk
+  *chunk_oct = cu * u;
+  *nchunks = (TC + *chunk_oct - 1) / *chunk_oct;
 }
 
 extern "C" void dlb_gn_fwd_segs(const void* const* xs, const int* starts,
                                 int nseg, void* y, const float* gamma,
                                 const float* beta, float* mean, float* rstd,
-                                float* scratch, int N, int HW, int C, int G,
+                                int N, int HW, int C, int G,
                                 float eps, int relu, hipStream_t stream) {
   GnSegs sg{};
   sg.nseg = nseg;
@@ -678,26 +412,18 @@ extern "C" void dlb_gn_fwd_segs(const void* const* xs, const int* starts,
     sg.start[i] = starts[i];
   }
   sg.start[nseg] = starts[nseg];
-  const int slices = scratch ? dlb_gn_nslices(N, HW, 0) : 1;
-  if (slices > 1) {
-    dim3 grid(N, slices);
-    hipLaunchKernelGGL(gn_stats_part_kernel, grid, dim3(GN_BLOCK), 0, stream,
-                       sg, scratch, HW, C, G, slices);
-    hipLaunchKernelGGL(gn_apply_kernel, grid, dim3(GN_BLOCK), 0, stream,
-                       sg, (bf16*)y, gamma, beta, scratch, mean,
-                       rstd, HW, C, G, eps, relu, slices);
-    return;
-  }
-  hipLaunchKernelGGL(gn_fwd_kernel, dim3(N), dim3(GN_BLOCK), 0, stream,
-                     sg, (bf16*)y, gamma, beta, mean, rstd, HW, C,
-                     G, eps, relu);
+  int chunk_oct, nchunks;
+  gn_chunking(N, C, G, 0, &chunk_oct, &nchunks);
+  hipLaunchKernelGGL(gn_fwd_kernel, dim3(N, nchunks), dim3(GN_BLOCK), 0,
+                     stream, sg, (bf16*)y, gamma, beta, mean, rstd, HW, C,
+                     G, eps, relu, chunk_oct);
 }
 
 extern "C" void dlb_gn_bwd_segs(const void* const* xs, const int* starts,
                                 int nseg, const void* dz, void* const* dxs,
                                 const float* gamma, const float* beta,
                                 const float* mean, const float* rstd,
-                                float* dgamma, float* dbeta, float* scratch,
+                                float* dgamma, float* dbeta,
                                 int N, int HW, int C, int G, int relu,
                                 int accumulate, hipStream_t stream) {
   GnSegs sg{};
@@ -709,23 +435,13 @@ extern "C" void dlb_gn_bwd_segs(const void* const* xs, const int* starts,
     dsg.p[i] = (bf16*)dxs[i];
   }
   sg.start[nseg] = starts[nseg];
-  size_t shmem = 2 * (size_t)C * sizeof(float);
-  const int slices = scratch ? dlb_gn_nslices(N, HW, 1) : 1;
-  if (slices > 1) {
-    dim3 grid(N, slices);
-    hipLaunchKernelGGL(gn_bwd_part_kernel, grid, dim3(GN_BLOCK), shmem,
-                       stream, sg, (const bf16*)dz, gamma, beta,
-                       mean, rstd, scratch, dgamma, dbeta, HW, C, G, relu,
-                       slices);
-    hipLaunchKernelGGL(gn_bwd_apply_kernel, grid, dim3(GN_BLOCK), 0, stream,
-                       sg, (const bf16*)dz, dsg, gamma,
-                       beta, mean, rstd, scratch, HW, C, G, relu, slices,
-                       accumulate);
-    return;
-  }
-  hipLaunchKernelGGL(gn_bwd_kernel, dim3(N), dim3(GN_BLOCK), shmem, stream,
-                     sg, (const bf16*)dz, dsg, gamma, beta,
-                     mean, rstd, dgamma, dbeta, HW, C, G, relu, accumulate);
+  int chunk_oct, nchunks;
+  gn_chunking(N, C, G, 1, &chunk_oct, &nchunks);
+  size_t shmem = 2 * (size_t)(chunk_oct * 8) * sizeof(float);
+  hipLaunchKernelGGL(gn_bwd_kernel, dim3(N, nchunks), dim3(GN_BLOCK), shmem,
+                     stream, sg, (const bf16*)dz, dsg, gamma, beta,
+                     mean, rstd, dgamma, dbeta, HW, C, G, relu, accumulate,
+                     chunk_oct);
 }
 
 extern "C" void dlb_gn_stats_segs(const void* const* xs, const int* starts,
@@ -739,6 +455,8 @@ extern "C" void dlb_gn_stats_segs(const void* const* xs, const int* starts,
     sg.start[i] = starts[i];
   }
   sg.start[nseg] = starts[nseg];
-  hipLaunchKernelGGL(gn_stats_kernel, dim3(N), dim3(GN_BLOCK), 0, stream,
-                     sg, mean, rstd, HW, C, G, eps);
+  int chunk_oct, nchunks;
+  gn_chunking(N, C, G, 0, &chunk_oct, &nchunks);
+  hipLaunchKernelGGL(gn_stats_kernel, dim3(N, nchunks), dim3(GN_BLOCK), 0,
+                     stream, sg, mean, rstd, HW, C, G, eps, chunk_oct);
 }
