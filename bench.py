@@ -129,10 +129,17 @@ def main():
     def step():
         sync.zero()
         with amp:
-            out = model(x)
-            if is_lm:
-                out = out.reshape(-1, ntokens)
-            loss = criterion(out, y)
+            if is_lm and device.type == "cuda":
+                from dynamic_load_balance_distributeddnn_amd.ops import \
+                    functional as FD
+                h = model.forward_features(x)
+                loss = FD.lm_loss(h, model.decoder.weight,
+                                  model.decoder.bias, y)
+            else:
+                out = model(x)
+                if is_lm:
+                    out = out.reshape(-1, ntokens)
+                loss = criterion(out, y)
         loss.backward()
         if is_lm:
             torch.nn.utils.clip_grad_norm_(model.parameters(), 0.25)

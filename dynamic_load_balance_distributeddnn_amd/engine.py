@@ -27,6 +27,7 @@ import torch.nn.functional as F
 
 from . import data as D
 from .models import LM_CONFIG, build_model
+from .ops import functional as FD
 from .parallel import GradientSynchronizer, StepTimer
 from .parallel.optim import FlatSGD
 from .scheduler import DBSScheduler, exchange_times, straggler_idle_pct
@@ -165,10 +166,17 @@ class Trainer:
         t.iter_start()
         self.sync.zero()
         with self._autocast():
-            output = self.model(inputs)
-            if self.is_lm:
-                output = output.reshape(-1, self.ntokens)
-            loss = self.criterion(output, target)
+            if self.is_lm and self.device.type == "cuda":
+                # fused decoder->log_softmax->NLL head: the [T, ntokens]
+                # logits never materialize (ops.functional.lm_loss)
+                h = self.model.forward_features(inputs)
+                loss = FD.lm_loss(h, self.model.decoder.weight,
+                                  self.model.decoder.bias, target)
+            else:
+                output = self.model(inputs)
+                if self.is_lm:
+                    output = output.reshape(-1, self.ntokens)
+                loss = self.criterion(output, target)
         loss.backward()
         t.backward_done()
         t.add_compute(self.fault.maybe_wait(epoch, steps_per_epoch))

@@ -79,7 +79,9 @@ class TransformerModel(nn.Module):
         nn.init.uniform_(self.decoder.weight, -0.1, 0.1)
         nn.init.zeros_(self.decoder.bias)
 
-    def forward(self, src):  # [S, B] int64
+    def forward_features(self, src):  # [S, B] int64 -> [S, B, d]
+        """Encoder output BEFORE the decoder — the input of the fused
+        LM loss head (ops.functional.lm_loss)."""
         x = self.embed(src) * math.sqrt(self.d_model)
         x = self.pos(x)
         if x.is_cuda and torch.is_autocast_enabled("cuda"):
@@ -88,4 +90,8 @@ class TransformerModel(nn.Module):
             x = x.to(torch.bfloat16)
         for layer in self.layers:
             x = layer(x)
-        return FD.log_softmax(self.decoder(x), dim=-1)
+        return x
+
+    def forward(self, src):  # [S, B] int64
+        return FD.log_softmax(self.decoder(self.forward_features(src)),
+                              dim=-1)

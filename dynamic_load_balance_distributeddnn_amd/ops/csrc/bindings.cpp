@@ -34,9 +34,12 @@ extern "C" void dlb_gn_bwd_segs(const void* const* xs, const int* starts,
                                 int nseg, const void* dz, void* const* dxs,
                                 const float* gamma, const float* beta,
                                 const float* mean, const float* rstd,
-                                float* dgamma, float* dbeta,
+                                float* dgb_part,
                                 int N, int HW, int C, int G, int relu,
                                 int accumulate, hipStream_t stream);
+extern "C" void dlb_gn_dgb_reduce(const float* part, float* mid, int N,
+                                  int groups, int C, float* dgamma,
+                                  float* dbeta, hipStream_t stream);
 extern "C" void dlb_gn_stats_segs(const void* const* xs, const int* starts,
                                   int nseg, float* mean, float* rstd, int N,
                                   int HW, int C, int G, float eps,
@@ -144,12 +147,23 @@ static std::vector<torch::Tensor> gn_bwd(std::vector<torch::Tensor> xs,
     dbeta = gbuf[1];
   }
   auto stream = at::hip::getCurrentHIPStream();
+  // dgamma/dbeta path: per-sample partial rows [N, 2C] (plain stores in
+  // the kernel) + a deterministic two-level column reduction — replaces
+  // the contended global-atomic publish (see reduce.hip).
+  auto part = torch::empty({N, 2 * C},
+                           xs[0].options().dtype(torch::kFloat32));
+  const int rgroups = N >= 64 ? 16 : (N >= 8 ? 4 : 1);
+  auto midb = torch::empty({rgroups, 2 * C},
+                           xs[0].options().dtype(torch::kFloat32));
   dlb_gn_bwd_segs(ptrs, starts, (int)xs.size(), dz.data_ptr(), dptrs,
                   gamma.data_ptr<float>(), beta.data_ptr<float>(),
                   mean.data_ptr<float>(), rstd.data_ptr<float>(),
-                  dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),
+                  part.data_ptr<float>(),
                   N, HW, C, (int)groups, relu ? 1 : 0, acc ? 1 : 0,
                   stream.stream());
+  dlb_gn_dgb_reduce(part.data_ptr<float>(), midb.data_ptr<float>(), N,
+                    rgroups, C, dgamma.data_ptr<float>(),
+                    dbeta.data_ptr<float>(), stream.stream());
   out.push_back(dgamma);
   out.push_back(dbeta);
   return out;  // [dx_0..dx_{k-1}, dgamma, dbeta]
@@ -577,6 +591,77 @@ static torch::Tensor logsoftmax_bwd(torch::Tensor y, torch::Tensor dy) {
   return dx;
 }
 
+extern "C" void dlb_lmloss_fwd(const void* h, const void* w,
+                               const float* bias, const int* tgt,
+                               float* part, float* ztgt, float* lse,
+                               float* loss, long T, int d, int V, int P,
+                               hipStream_t stream);
+extern "C" void dlb_lmloss_bwd(const void* h, const void* w,
+                               const float* bias, const int* tgt,
+                               const float* lse, const float* go, float* dh,
+                               float* dw, float* db, long T, int d, int V,
+                               int P, hipStream_t stream);
+
+static int lmloss_partitions(long T) {
+  // enough (row-block x partition) blocks to fill 256 CUs
+  const long rb = (T + 63) / 64;
+  long p = (512 + rb - 1) / rb;
+  if (p < 1) p = 1;
+  if (p > 16) p = 16;
+  return (int)p;
+}
+
+// h [T,d] bf16, w [V,d] bf16, bias [V] fp32, tgt [T] int32.
+// Returns {loss (0-dim fp32), lse [T] fp32}.
+static std::vector<torch::Tensor> lmloss_fwd(torch::Tensor h, torch::Tensor w,
+                                             torch::Tensor bias,
+                                             torch::Tensor tgt) {
+  TORCH_CHECK(h.is_cuda() && h.dim() == 2 && h.is_contiguous() &&
+              h.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(w.is_contiguous() && w.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(bias.is_contiguous() && bias.scalar_type() == torch::kFloat32);
+  TORCH_CHECK(tgt.is_contiguous() && tgt.scalar_type() == torch::kInt32);
+  const long T = h.size(0);
+  const int d = h.size(1), V = w.size(0);
+  TORCH_CHECK(w.size(1) == d && d <= 224 && d % 8 == 0);
+  TORCH_CHECK(tgt.numel() == T && bias.numel() == V);
+  const int P = lmloss_partitions(T);
+  auto opt = h.options().dtype(torch::kFloat32);
+  auto part = torch::empty({T, P, 2}, opt);
+  auto ztgt = torch::empty({T}, opt);
+  auto lse = torch::empty({T}, opt);
+  auto loss = torch::zeros({}, opt);
+  dlb_lmloss_fwd(h.data_ptr(), w.data_ptr(), bias.data_ptr<float>(),
+                 tgt.data_ptr<int>(), part.data_ptr<float>(),
+                 ztgt.data_ptr<float>(), lse.data_ptr<float>(),
+                 loss.data_ptr<float>(), T, d, V, P,
+                 at::hip::getCurrentHIPStream().stream());
+  return {loss, lse};
+}
+
+// Returns {dh [T,d] fp32, dw [V,d] fp32, db [V] fp32}.
+static std::vector<torch::Tensor> lmloss_bwd(torch::Tensor h, torch::Tensor w,
+                                             torch::Tensor bias,
+                                             torch::Tensor tgt,
+                                             torch::Tensor lse,
+                                             torch::Tensor go) {
+  const long T = h.size(0);
+  const int d = h.size(1), V = w.size(0);
+  TORCH_CHECK(go.is_cuda() && go.scalar_type() == torch::kFloat32 &&
+              go.numel() == 1);
+  const int P = lmloss_partitions(T);
+  auto opt = h.options().dtype(torch::kFloat32);
+  auto dh = torch::zeros({T, d}, opt);   // fp32 atomics accumulate
+  auto dw = torch::empty({V, d}, opt);   // direct stores
+  auto db = torch::empty({V}, opt);
+  dlb_lmloss_bwd(h.data_ptr(), w.data_ptr(), bias.data_ptr<float>(),
+                 tgt.data_ptr<int>(), lse.data_ptr<float>(),
+                 go.data_ptr<float>(), dh.data_ptr<float>(),
+                 dw.data_ptr<float>(), db.data_ptr<float>(), T, d, V, P,
+                 at::hip::getCurrentHIPStream().stream());
+  return {dh, dw, db};
+}
+
 extern "C" void dlb_gconv_fwd(const void* x, const void* w, void* y, int N,
                               int IH, int IW, int C, int GW, int stride,
                               hipStream_t stream);
@@ -696,4 +781,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("maxpool_bwd", &maxpool_bwd);
   m.def("logsoftmax_fwd", &logsoftmax_fwd);
   m.def("logsoftmax_bwd", &logsoftmax_bwd);
+  m.def("lmloss_fwd", &lmloss_fwd,
+        "Fused decoder GEMM -> log_softmax -> NLL forward (no logits)");
+  m.def("lmloss_bwd", &lmloss_bwd,
+        "Fused LM loss backward: dh, dW, db with recomputed logits tiles");
 }

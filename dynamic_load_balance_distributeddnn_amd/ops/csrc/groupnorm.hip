@@ -88,7 +88,27 @@ __device__ inline ChunkMap chunk_map(int TC, int chunk_oct) {
   return m;
 }
 
+// Same-address LDS atomics from the TP lanes of one octet serialize
+// (+6..+100 us per dispatch, tools/gn_probe).  When TCe is a pow2 < 64
+// (then the octet loop is single-trip and all 256 threads are active),
+// fold the tp lanes with wave shuffles first; only lanes < TCe touch
+// LDS afterwards (contention drops to the 4 waves).
+__device__ inline bool tp_shuffle_ok(const ChunkMap& m) {
+  return (m.TCe == m.o1 - m.o0) && m.TCe < 64 &&
+         ((m.TCe & (m.TCe - 1)) == 0);
+}
+
+__device__ inline float tp_fold(float v, int TCe) {
+  for (int off = TCe; off < 64; off <<= 1) v += __shfl_down(v, off, 64);
+  return v;
+}
+
 // ---------------------------------------------------------------- forward
+// Dynamic LDS: [2 * chunk-channels] staged gamma/beta.  Per-octet
+// parameter reads come from LDS — per-lane 4-byte GLOBAL gathers here
+// compiled to a serialized load->use->waitcnt chain (~10-25 us of fixed
+// per-block latency, the measured small-shape floor); staging them with
+// one coalesced sweep removes it.
 extern "C" __global__ void __launch_bounds__(GN_BLOCK)
 gn_fwd_kernel(const GnSegs segs, bf16* __restrict__ y,
               const float* __restrict__ gamma, const float* __restrict__ beta,
@@ -102,12 +122,21 @@ gn_fwd_kernel(const GnSegs segs, bf16* __restrict__ y,
   const int Cg = C / G;
   // groups owned by this chunk (chunk boundaries are group-aligned)
   const int g0 = (m.o0 << 3) / Cg, g1 = (m.o1 << 3) / Cg;
+  const int cbase = m.o0 << 3;
+  const int cspan = (m.o1 - m.o0) << 3;
 
   __shared__ float s_sum[GN_MAXG];
   __shared__ float s_ssq[GN_MAXG];
   __shared__ float s_mean[GN_MAXG];
   __shared__ float s_rstd[GN_MAXG];
+  extern __shared__ float s_par[];  // [cspan] gamma, [cspan] beta
+  float* s_ga = s_par;
+  float* s_be = s_par + cspan;
   for (int g = g0 + t; g < g1; g += GN_BLOCK) { s_sum[g] = 0.f; s_ssq[g] = 0.f; }
+  for (int c = t; c < cspan; c += GN_BLOCK) {
+    s_ga[c] = gamma[cbase + c];
+    s_be[c] = beta[cbase + c];
+  }
   __syncthreads();
 
   if (m.active) {
@@ -128,11 +157,17 @@ gn_fwd_kernel(const GnSegs segs, bf16* __restrict__ y,
           ss[j] += v * v;
         }
       }
+      const bool shf = tp_shuffle_ok(m);
+      const int lane = t & 63;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        int g = (c0 + j) / Cg;
-        atomicAdd(&s_sum[g], s[j]);
-        atomicAdd(&s_ssq[g], ss[j]);
+        float v1 = shf ? tp_fold(s[j], m.TCe) : s[j];
+        float v2 = shf ? tp_fold(ss[j], m.TCe) : ss[j];
+        if (!shf || lane < m.TCe) {
+          int g = (c0 + j) / Cg;
+          atomicAdd(&s_sum[g], v1);
+          atomicAdd(&s_ssq[g], v2);
+        }
       }
     }
   }
@@ -162,8 +197,8 @@ gn_fwd_kernel(const GnSegs segs, bf16* __restrict__ y,
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       int c = c0 + j, g = c / Cg;
-      ga[j] = gamma[c];
-      be[j] = beta[c];
+      ga[j] = s_ga[c - cbase];
+      be[j] = s_be[c - cbase];
       mu[j] = s_mean[g];
       rs[j] = s_rstd[g];
     }
@@ -221,11 +256,17 @@ gn_stats_kernel(const GnSegs segs, float* __restrict__ mean_out,
           ss[j] += v * v;
         }
       }
+      const bool shf = tp_shuffle_ok(m);
+      const int lane = t & 63;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        int g = (c0 + j) / Cg;
-        atomicAdd(&s_sum[g], s[j]);
-        atomicAdd(&s_ssq[g], ss[j]);
+        float v1 = shf ? tp_fold(s[j], m.TCe) : s[j];
+        float v2 = shf ? tp_fold(ss[j], m.TCe) : ss[j];
+        if (!shf || lane < m.TCe) {
+          int g = (c0 + j) / Cg;
+          atomicAdd(&s_sum[g], v1);
+          atomicAdd(&s_ssq[g], v2);
+        }
       }
     }
   }
@@ -251,7 +292,7 @@ gn_bwd_kernel(const GnSegs segs, const bf16* __restrict__ dz,
               const float* __restrict__ beta,
               const float* __restrict__ mean_in,
               const float* __restrict__ rstd_in,
-              float* __restrict__ dgamma, float* __restrict__ dbeta,
+              float* __restrict__ dgb_part,
               const int HW, const int C, const int G, const int relu,
               const int accumulate, const int chunk_oct) {
   const int n = blockIdx.x;
@@ -265,9 +306,23 @@ gn_bwd_kernel(const GnSegs segs, const bf16* __restrict__ dz,
 
   __shared__ float s_s1[GN_MAXG];
   __shared__ float s_s2[GN_MAXG];
-  extern __shared__ float s_dgb[];  // [2*cspan]: dgamma then dbeta partials
-  for (int g = g0 + t; g < g1; g += GN_BLOCK) { s_s1[g] = 0.f; s_s2[g] = 0.f; }
+  __shared__ float s_mu[GN_MAXG];
+  __shared__ float s_rs[GN_MAXG];
+  extern __shared__ float s_dgb[];  // [2*cspan] dgamma/dbeta partials,
+                                    // then [cspan] gamma, [cspan] beta
+  float* s_ga = s_dgb + 2 * cspan;
+  float* s_be = s_dgb + 3 * cspan;
+  for (int g = g0 + t; g < g1; g += GN_BLOCK) {
+    s_s1[g] = 0.f;
+    s_s2[g] = 0.f;
+    s_mu[g] = mean_in[(long)n * G + g];
+    s_rs[g] = rstd_in[(long)n * G + g];
+  }
   for (int c = t; c < 2 * cspan; c += GN_BLOCK) s_dgb[c] = 0.f;
+  for (int c = t; c < cspan; c += GN_BLOCK) {
+    s_ga[c] = gamma[cbase + c];
+    s_be[c] = beta[cbase + c];
+  }
   __syncthreads();
 
   const bf16* db = dz + (long)n * HW * C;
@@ -282,10 +337,10 @@ gn_bwd_kernel(const GnSegs segs, const bf16* __restrict__ dz,
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         int c = c0 + j, g = c / Cg;
-        ga[j] = gamma[c];
-        be[j] = beta[c];
-        mu[j] = mean_in[(long)n * G + g];
-        rs[j] = rstd_in[(long)n * G + g];
+        ga[j] = s_ga[c - cbase];
+        be[j] = s_be[c - cbase];
+        mu[j] = s_mu[g];
+        rs[j] = s_rs[g];
       }
       float a1[8] = {0}, a2[8] = {0}, adg[8] = {0}, adb[8] = {0};
       #pragma unroll 4
@@ -306,22 +361,34 @@ gn_bwd_kernel(const GnSegs segs, const bf16* __restrict__ dz,
           adb[j] += dy;
         }
       }
+      const bool shf = tp_shuffle_ok(m);
+      const int lane = t & 63;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        int g = (c0 + j) / Cg;
-        atomicAdd(&s_s1[g], a1[j]);
-        atomicAdd(&s_s2[g], a2[j]);
-        atomicAdd(&s_dgb[c0 - cbase + j], adg[j]);
-        atomicAdd(&s_dgb[cspan + c0 - cbase + j], adb[j]);
+        float v1 = shf ? tp_fold(a1[j], m.TCe) : a1[j];
+        float v2 = shf ? tp_fold(a2[j], m.TCe) : a2[j];
+        float vg = shf ? tp_fold(adg[j], m.TCe) : adg[j];
+        float vb = shf ? tp_fold(adb[j], m.TCe) : adb[j];
+        if (!shf || lane < m.TCe) {
+          const int g = (c0 + j) / Cg;
+          atomicAdd(&s_s1[g], v1);
+          atomicAdd(&s_s2[g], v2);
+          atomicAdd(&s_dgb[c0 - cbase + j], vg);
+          atomicAdd(&s_dgb[cspan + c0 - cbase + j], vb);
+        }
       }
     }
   }
   __syncthreads();
 
-  // publish per-channel param grads (one global atomic per channel)
+  // publish per-(sample, channel) partials with plain stores — the
+  // deterministic column reduction (dlb_gn_dgb_reduce) follows; global
+  // atomics here cost a +10..30 us serialization tail (tools/gn_probe)
+  // and made dgamma order-dependent across runs.
   for (int c = t; c < cspan; c += GN_BLOCK) {
-    atomicAdd(&dgamma[cbase + c], s_dgb[c]);
-    atomicAdd(&dbeta[cbase + c], s_dgb[cspan + c]);
+    float* row = dgb_part + (long)n * 2 * C;
+    row[cbase + c] = s_dgb[c];
+    row[C + cbase + c] = s_dgb[cspan + c];
   }
 
   if (!m.active) return;
@@ -339,10 +406,10 @@ gn_bwd_kernel(const GnSegs segs, const bf16* __restrict__ dz,
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       int c = c0 + j, g = c / Cg;
-      ga[j] = gamma[c];
-      be[j] = beta[c];
-      mu[j] = mean_in[(long)n * G + g];
-      rs[j] = rstd_in[(long)n * G + g];
+      ga[j] = s_ga[c - cbase];
+      be[j] = s_be[c - cbase];
+      mu[j] = s_mu[g];
+      rs[j] = s_rs[g];
       k1[j] = s_s1[g] * inv_m;
       k2[j] = s_s2[g] * inv_m;
     }
@@ -414,7 +481,8 @@ extern "C" void dlb_gn_fwd_segs(const void* const* xs, const int* starts,
   sg.start[nseg] = starts[nseg];
   int chunk_oct, nchunks;
   gn_chunking(N, C, G, 0, &chunk_oct, &nchunks);
-  hipLaunchKernelGGL(gn_fwd_kernel, dim3(N, nchunks), dim3(GN_BLOCK), 0,
+  const size_t shmem = 2 * (size_t)(chunk_oct * 8) * sizeof(float);
+  hipLaunchKernelGGL(gn_fwd_kernel, dim3(N, nchunks), dim3(GN_BLOCK), shmem,
                      stream, sg, (bf16*)y, gamma, beta, mean, rstd, HW, C,
                      G, eps, relu, chunk_oct);
 }
@@ -423,7 +491,7 @@ extern "C" void dlb_gn_bwd_segs(const void* const* xs, const int* starts,
                                 int nseg, const void* dz, void* const* dxs,
                                 const float* gamma, const float* beta,
                                 const float* mean, const float* rstd,
-                                float* dgamma, float* dbeta,
+                                float* dgb_part,
                                 int N, int HW, int C, int G, int relu,
                                 int accumulate, hipStream_t stream) {
   GnSegs sg{};
@@ -437,10 +505,10 @@ extern "C" void dlb_gn_bwd_segs(const void* const* xs, const int* starts,
   sg.start[nseg] = starts[nseg];
   int chunk_oct, nchunks;
   gn_chunking(N, C, G, 1, &chunk_oct, &nchunks);
-  size_t shmem = 2 * (size_t)(chunk_oct * 8) * sizeof(float);
+  size_t shmem = 4 * (size_t)(chunk_oct * 8) * sizeof(float);
   hipLaunchKernelGGL(gn_bwd_kernel, dim3(N, nchunks), dim3(GN_BLOCK), shmem,
                      stream, sg, (const bf16*)dz, dsg, gamma, beta,
-                     mean, rstd, dgamma, dbeta, HW, C, G, relu, accumulate,
+                     mean, rstd, dgb_part, HW, C, G, relu, accumulate,
                      chunk_oct);
 }
 

@@ -36,3 +36,48 @@ extern "C" void dlb_slab_sum(const float* part, float* out, int splits,
   hipLaunchKernelGGL(slab_sum_kernel, dim3((unsigned)grid), dim3(block), 0,
                      stream, part, out, splits, len);
 }
+
+// --------- GroupNorm dgamma/dbeta deterministic column reduction --------
+// gn_bwd publishes per-sample partials part[n][2C] with plain stores
+// (global atomicAdd on the [2C] words measured a +10..30 us per-dispatch
+// tail under 512-way contention — tools/gn_probe — and made dgamma
+// nondeterministic).  Level 1 sums sample groups; level 2 adds the group
+// sums into the dgamma/dbeta buffers.  Fixed split order -> bitwise
+// deterministic across runs.
+extern "C" __global__ void __launch_bounds__(256)
+gn_dgb_l1_kernel(const float* __restrict__ part, float* __restrict__ mid,
+                 const int N, const int gs, const long len2) {
+  const long c = (long)blockIdx.x * 256 + threadIdx.x;
+  if (c >= len2) return;
+  const int n0 = blockIdx.y * gs;
+  const int n1 = min(n0 + gs, N);
+  float s = 0.f;
+  for (int n = n0; n < n1; ++n) s += part[(long)n * len2 + c];
+  mid[(long)blockIdx.y * len2 + c] = s;
+}
+
+extern "C" __global__ void __launch_bounds__(256)
+gn_dgb_l2_kernel(const float* __restrict__ mid, const int groups,
+                 const long len2, const int C, float* __restrict__ dgamma,
+                 float* __restrict__ dbeta) {
+  const long c = (long)blockIdx.x * 256 + threadIdx.x;
+  if (c >= len2) return;
+  float s = 0.f;
+  for (int g = 0; g < groups; ++g) s += mid[(long)g * len2 + c];
+  if (c < C)
+    dgamma[c] += s;
+  else
+    dbeta[c - C] += s;
+}
+
+extern "C" void dlb_gn_dgb_reduce(const float* part, float* mid, int N,
+                                  int groups, int C, float* dgamma,
+                                  float* dbeta, hipStream_t stream) {
+  const long len2 = 2L * C;
+  const int gs = (N + groups - 1) / groups;
+  const unsigned cb = (unsigned)((len2 + 255) / 256);
+  hipLaunchKernelGGL(gn_dgb_l1_kernel, dim3(cb, groups), dim3(256), 0,
+                     stream, part, mid, N, gs, len2);
+  hipLaunchKernelGGL(gn_dgb_l2_kernel, dim3(cb), dim3(256), 0, stream, mid,
+                     groups, len2, C, dgamma, dbeta);
+}

@@ -22,7 +22,7 @@ from . import available, ext
 # ops with a hand-written gfx950 kernel wired in (grown as kernels land)
 NATIVE_OPS: set[str] = {"group_norm_act", "conv2d", "layer_norm",
                         "causal_attention", "avg_pool2d", "log_softmax",
-                        "max_pool2d"}
+                        "max_pool2d", "lm_loss"}
 
 
 def _use_native(name: str, x: torch.Tensor) -> bool:
@@ -128,6 +128,26 @@ def log_softmax(x, dim=-1):
         from . import native
         return native.log_softmax(x)
     return F.log_softmax(x, dim=dim)
+
+
+def lm_loss(h, weight, bias, targets):
+    """Decoder GEMM -> log_softmax -> NLL(mean), fused.
+
+    ``h`` is the encoder output [S, B, d] (or [T, d]); weight/bias the
+    decoder Linear parameters; targets int64 [T].  On GPU the hand-
+    written gfx950 kernels stream [64, 64] logit MFMA tiles with an
+    online softmax — the [T, V] logits (~120 MB bf16 at the flagship
+    shape, reference dbs.py:371-374) are never materialized, and
+    backward recomputes the tiles for dh/dW/db.  CPU path is the plain
+    composition in fp32.
+    """
+    if (_use_native("lm_loss", h) and h.dtype == torch.bfloat16
+            and h.shape[-1] % 8 == 0 and h.shape[-1] <= 224):
+        from . import native
+        return native.lm_loss(h, weight, bias, targets)
+    logits = F.linear(h, weight, bias).reshape(-1, weight.shape[0])
+    return F.nll_loss(F.log_softmax(logits.float(), dim=-1),
+                      targets.reshape(-1))
 
 
 def max_pool2d(x, k, stride=None, padding=0):

@@ -26,8 +26,10 @@ def weight_bf16(weight: torch.Tensor) -> torch.Tensor:
     m = getattr(weight, "_dlb_bf16", None)
     if m is not None:
         return m
-    return weight.detach().to(torch.bfloat16) \
-        .contiguous(memory_format=torch.channels_last)
+    w = weight.detach().to(torch.bfloat16)
+    if w.dim() == 4:
+        return w.contiguous(memory_format=torch.channels_last)
+    return w.contiguous()
 
 
 class _GroupNormAct(torch.autograd.Function):
@@ -322,3 +324,35 @@ class _MaxPool2d(torch.autograd.Function):
 
 def max_pool2d(x, k, stride, pad):
     return _MaxPool2d.apply(x, k, stride, pad)
+
+
+class _LMLoss(torch.autograd.Function):
+    """Fused decoder GEMM -> log_softmax -> NLL (mean) — the [T, V]
+    logits are never materialized (reference criterion site
+    dbs.py:371-374 over Net/Transformer.py:95).
+    Backward recomputes logit tiles to produce dh, dW, db."""
+
+    @staticmethod
+    def forward(ctx, h2, weight, bias, targets):
+        wb = weight_bf16(weight)  # free view of the FlatSGD bf16 mirror
+        if not wb.is_contiguous():
+            wb = wb.contiguous()
+        tgt = targets.to(torch.int32)
+        bias32 = bias.detach().float().contiguous()
+        loss, lse = ext().lmloss_fwd(h2, wb, bias32, tgt)
+        ctx.save_for_backward(h2, wb, bias32, tgt, lse)
+        return loss
+
+    @staticmethod
+    def backward(ctx, go):
+        h2, wb, bias32, tgt, lse = ctx.saved_tensors
+        go32 = go.detach().to(torch.float32).reshape(1).contiguous()
+        dh, dw, db = ext().lmloss_bwd(h2, wb, bias32, tgt, lse, go32)
+        return (dh.to(h2.dtype), dw, db, None)
+
+
+def lm_loss(h, weight, bias, targets):
+    h2 = h.reshape(-1, h.shape[-1])
+    if not h2.is_contiguous():
+        h2 = h2.contiguous()
+    return _LMLoss.apply(h2, weight, bias, targets.reshape(-1))

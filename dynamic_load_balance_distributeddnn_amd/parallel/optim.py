@@ -46,7 +46,9 @@ class FlatSGD:
                 view = _strided_view(self.param_arena, off, p)
                 view.copy_(p.data)
                 p.data = view  # re-home the parameter into the arena
-                if self.bf16_mirror is not None and p.dim() == 4:
+                # conv (4D) and matmul (2D) weights get bf16 mirror
+                # views — the compute kernels read weights as bf16
+                if self.bf16_mirror is not None and p.dim() in (2, 4):
                     p._dlb_bf16 = _strided_view(self.bf16_mirror, off, p)
             self.refresh_mirror()
 
