@@ -20,6 +20,34 @@ typedef __hip_bfloat16 bf16;
 #define MAX_S 40
 #define MAX_D 104
 
+// philox2x32-10 counter hash for the attention-probability dropout mask
+// (reference regularization: nn.MultiheadAttention drops attn PROBS at
+// p=0.2, Net/Transformer.py:63-64).  Deterministic in (seed, element
+// index), so the backward recomputes the mask instead of storing it.
+__device__ inline unsigned philox_u32(unsigned long long seed,
+                                      unsigned idx) {
+  unsigned c0 = idx, c1 = (unsigned)(seed >> 32);
+  unsigned key = (unsigned)seed;
+#pragma unroll
+  for (int r = 0; r < 10; ++r) {
+    const unsigned long long prod = 0xD256D193ull * c0;
+    const unsigned hi = (unsigned)(prod >> 32), lo = (unsigned)prod;
+    c0 = hi ^ c1 ^ key;
+    c1 = lo;
+    key += 0x9E3779B9u;
+  }
+  return c0;
+}
+
+// keep-mask scale for probability element `idx`: 0 when dropped, else
+// 1/(1-pd).  pd == 0 short-circuits (uniform branch).
+__device__ inline float drop_scale(float pd, unsigned long long seed,
+                                   unsigned idx) {
+  if (pd <= 0.f) return 1.f;
+  const unsigned thresh = (unsigned)(pd * 4294967296.0);
+  return philox_u32(seed, idx) >= thresh ? 1.f / (1.f - pd) : 0.f;
+}
+
 struct AttnParams {
   const bf16 *q, *k, *v;  // base + per-tensor offset already applied
   bf16* o;                // [S, B, E] output slice (rowstride ld_o)
@@ -27,6 +55,8 @@ struct AttnParams {
   int S, B, H, DH;
   int ld_qkv, ld_o;       // row strides (elements)
   float scale;
+  float pd;               // attention-probability dropout (train mode)
+  unsigned long long seed;
 };
 
 __global__ void __launch_bounds__(AT_BLOCK)
@@ -83,11 +113,16 @@ attn_fwd_kernel(const AttnParams p) {
   }
   __syncthreads();
 
-  // save P for backward
+  // save PRE-dropout P for backward (the mask is philox-recomputed)
   float* pg = p.p_save + (long)bh * p.S * p.S;
   for (int idx = t; idx < p.S * p.S; idx += AT_BLOCK) pg[idx] = p_s[idx];
+  if (p.pd > 0.f) {
+    for (int idx = t; idx < p.S * p.S; idx += AT_BLOCK)
+      p_s[idx] *= drop_scale(p.pd, p.seed, (unsigned)(bh * p.S * p.S + idx));
+  }
+  __syncthreads();
 
-  // O = P V
+  // O = A V  (A = dropout(P))
   for (int idx = t; idx < SD; idx += AT_BLOCK) {
     const int i = idx / p.DH, e = idx % p.DH;
     float acc = 0.f;
@@ -100,11 +135,13 @@ attn_fwd_kernel(const AttnParams p) {
 struct AttnBwdParams {
   const bf16 *q, *k, *v;   // fwd inputs (slices, rowstride ld_qkv)
   const bf16* dout;        // [S, B, E] grad of O (rowstride ld_o)
-  const float* p_save;     // [B*H, S, S]
+  const float* p_save;     // [B*H, S, S] PRE-dropout probs
   bf16 *dq, *dk, *dv;      // grads: CONTIGUOUS [S,B,E] (row stride ld_g)
   int S, B, H, DH;
   int ld_qkv, ld_o, ld_g;
   float scale;
+  float pd;                // must match the forward's (mask recompute)
+  unsigned long long seed;
 };
 
 __global__ void __launch_bounds__(AT_BLOCK)
@@ -113,6 +150,7 @@ attn_bwd_kernel(const AttnBwdParams p) {
   __shared__ bf16 b_s[MAX_S * MAX_D];   // stage V/Q (reused)
   __shared__ float p_s[MAX_S * MAX_S];
   __shared__ float dp_s[MAX_S * MAX_S];
+  __shared__ float m_s[MAX_S * MAX_S];  // dropout keep-scales
   __shared__ float drow[MAX_S];
 
   const int bh = blockIdx.x;
@@ -122,7 +160,12 @@ attn_bwd_kernel(const AttnBwdParams p) {
   const int SD = p.S * p.DH;
 
   const float* pg = p.p_save + (long)bh * p.S * p.S;
-  for (int idx = t; idx < p.S * p.S; idx += AT_BLOCK) p_s[idx] = pg[idx];
+  for (int idx = t; idx < p.S * p.S; idx += AT_BLOCK) {
+    p_s[idx] = pg[idx];
+    m_s[idx] = p.pd > 0.f
+                   ? drop_scale(p.pd, p.seed, (unsigned)(bh * p.S * p.S + idx))
+                   : 1.f;
+  }
 
   // stage dO into a_s, V into b_s
   for (int i = t; i < SD; i += AT_BLOCK) {
@@ -132,15 +175,16 @@ attn_bwd_kernel(const AttnBwdParams p) {
   }
   __syncthreads();
 
-  // dV[j][e] = sum_{i>=j} P[i][j] * dO[i][e]
+  // dV[j][e] = sum_{i>=j} A[i][j] * dO[i][e]   (A = dropout(P))
   for (int idx = t; idx < SD; idx += AT_BLOCK) {
     const int j = idx / p.DH, e = idx % p.DH;
     float acc = 0.f;
     for (int i = j; i < p.S; ++i)
-      acc += p_s[i * p.S + j] * __bfloat162float(a_s[i * p.DH + e]);
+      acc += p_s[i * p.S + j] * m_s[i * p.S + j] *
+             __bfloat162float(a_s[i * p.DH + e]);
     p.dv[((long)j * p.B + b) * p.ld_g + hoff + e] = __float2bfloat16(acc);
   }
-  // dP[i][j] = dot(dO[i], V[j])
+  // dP[i][j] = dot(dO[i], V[j]) * keep-scale  (grad wrt PRE-dropout P)
   for (int idx = t; idx < p.S * p.S; idx += AT_BLOCK) {
     const int i = idx / p.S, j = idx % p.S;
     float acc = 0.f;
@@ -148,7 +192,7 @@ attn_bwd_kernel(const AttnBwdParams p) {
       for (int e = 0; e < p.DH; ++e)
         acc += __bfloat162float(a_s[i * p.DH + e]) *
                __bfloat162float(b_s[j * p.DH + e]);
-    dp_s[idx] = acc;
+    dp_s[idx] = acc * m_s[idx];
   }
   __syncthreads();
   // dS = P o (dP - rowsum(dP o P)) ; rowsum per row i
@@ -192,11 +236,11 @@ attn_bwd_kernel(const AttnBwdParams p) {
 
 extern "C" void dlb_attn_fwd(const void* q, const void* k, const void* v,
                              void* o, float* p_save, int S, int B, int H,
-                             int DH, int ld_qkv, int ld_o,
-                             hipStream_t stream) {
+                             int DH, int ld_qkv, int ld_o, float pd,
+                             unsigned long long seed, hipStream_t stream) {
   AttnParams p{(const bf16*)q, (const bf16*)k, (const bf16*)v, (bf16*)o,
                p_save, S, B, H, DH, ld_qkv, ld_o,
-               1.0f / sqrtf((float)DH)};
+               1.0f / sqrtf((float)DH), pd, seed};
   hipLaunchKernelGGL(attn_fwd_kernel, dim3(B * H), dim3(AT_BLOCK), 0, stream,
                      p);
 }
@@ -204,11 +248,12 @@ extern "C" void dlb_attn_fwd(const void* q, const void* k, const void* v,
 extern "C" void dlb_attn_bwd(const void* q, const void* k, const void* v,
                              const void* dout, const float* p_save, void* dq,
                              void* dk, void* dv, int S, int B, int H, int DH,
-                             int ld_qkv, int ld_o, int ld_g,
-                             hipStream_t stream) {
+                             int ld_qkv, int ld_o, int ld_g, float pd,
+                             unsigned long long seed, hipStream_t stream) {
   AttnBwdParams p{(const bf16*)q, (const bf16*)k, (const bf16*)v,
                   (const bf16*)dout, p_save, (bf16*)dq, (bf16*)dk, (bf16*)dv,
-                  S, B, H, DH, ld_qkv, ld_o, ld_g, 1.0f / sqrtf((float)DH)};
+                  S, B, H, DH, ld_qkv, ld_o, ld_g, 1.0f / sqrtf((float)DH),
+                  pd, seed};
   hipLaunchKernelGGL(attn_bwd_kernel, dim3(B * H), dim3(AT_BLOCK), 0, stream,
                      p);
 }

@@ -444,12 +444,13 @@ extern "C" void dlb_ln_bwd(const void* x, const void* dz, void* dx,
                            int R, int D, hipStream_t stream);
 extern "C" void dlb_attn_fwd(const void* q, const void* k, const void* v,
                              void* o, float* p_save, int S, int B, int H,
-                             int DH, int ld_qkv, int ld_o, hipStream_t stream);
+                             int DH, int ld_qkv, int ld_o, float pd,
+                             unsigned long long seed, hipStream_t stream);
 extern "C" void dlb_attn_bwd(const void* q, const void* k, const void* v,
                              const void* dout, const float* p_save, void* dq,
                              void* dk, void* dv, int S, int B, int H, int DH,
-                             int ld_qkv, int ld_o, int ld_g,
-                             hipStream_t stream);
+                             int ld_qkv, int ld_o, int ld_g, float pd,
+                             unsigned long long seed, hipStream_t stream);
 extern "C" void dlb_logsoftmax_fwd(const void* x, void* y, long R, int D,
                                    hipStream_t stream);
 extern "C" void dlb_logsoftmax_bwd(const void* y, const void* dy, void* dx,
@@ -535,8 +536,11 @@ static std::vector<torch::Tensor> ln_bwd(torch::Tensor x, torch::Tensor dz,
 
 // ---- causal attention ------------------------------------------------
 static std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
-                                           torch::Tensor v, int64_t nhead) {
-  // q/k/v: [S, B, E] bf16 slices sharing a storage row stride
+                                           torch::Tensor v, int64_t nhead,
+                                           double pd, int64_t seed) {
+  // q/k/v: [S, B, E] bf16 slices sharing a storage row stride.
+  // pd > 0 applies philox dropout to the attention PROBABILITIES
+  // (train mode; reference MHA p=0.2) keyed by (seed, element).
   TORCH_CHECK(q.is_cuda() && q.scalar_type() == torch::kBFloat16);
   const int S = q.size(0), B = q.size(1), E = q.size(2);
   const int DH = E / (int)nhead;
@@ -549,13 +553,15 @@ static std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
                              q.options().dtype(torch::kFloat32));
   dlb_attn_fwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
                p_save.data_ptr<float>(), S, B, (int)nhead, DH, ld, E,
+               (float)pd, (unsigned long long)seed,
                at::hip::getCurrentHIPStream().stream());
   return {o, p_save};
 }
 static std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
                                            torch::Tensor v, torch::Tensor do_,
                                            torch::Tensor p_save,
-                                           int64_t nhead) {
+                                           int64_t nhead, double pd,
+                                           int64_t seed) {
   const int S = q.size(0), B = q.size(1), E = q.size(2);
   const int DH = E / (int)nhead;
   const int ld = q.stride(0) / B;
@@ -565,7 +571,8 @@ static std::vector<torch::Tensor> attn_bwd(torch::Tensor q, torch::Tensor k,
   auto doc = do_.contiguous();
   dlb_attn_bwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), doc.data_ptr(),
                p_save.data_ptr<float>(), dq.data_ptr(), dk.data_ptr(),
-               dv.data_ptr(), S, B, (int)nhead, DH, ld, E, E,
+               dv.data_ptr(), S, B, (int)nhead, DH, ld, E, E, (float)pd,
+               (unsigned long long)seed,
                at::hip::getCurrentHIPStream().stream());
   return {dq, dk, dv};
 }

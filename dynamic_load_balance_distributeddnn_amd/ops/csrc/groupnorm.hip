@@ -439,11 +439,17 @@ gn_bwd_kernel(const GnSegs segs, const bf16* __restrict__ dz,
 }
 
 // ---------------------------------------------------------------- launchers
+// Block target for the (N, chunks) grid.  512 measured best at the
+// flagship N=512 (nchunks=1): every grid size beyond ~512 blocks pays a
+// fixed ~10-25 us per 512-block round that chunk-splitting cannot buy
+// back (tools/gn_probe; the LDS-atomic contention part of it is folded
+// away by tp_fold, the remainder is per-block lifetime).  Chunking still
+// engages for small per-rank batches (N < 512), where it fills CUs.
 static int gn_target_blocks(int bwd) {
   static int cached[2] = {-1, -1};
   if (cached[bwd] < 0) {
     const char* e = getenv(bwd ? "DLB_GN_TARGET_BWD" : "DLB_GN_TARGET");
-    cached[bwd] = e ? atoi(e) : 2048;
+    cached[bwd] = e ? atoi(e) : 512;
     if (cached[bwd] < 1) cached[bwd] = 1;
   }
   return cached[bwd];

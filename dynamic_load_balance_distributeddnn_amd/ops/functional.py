@@ -78,16 +78,17 @@ def layer_norm(x, normalized_shape, weight, bias, eps=1e-5):
 def causal_attention(q, k, v, nhead, dropout_p=0.0, training=False):
     """Causal multi-head self-attention on [S, B, E] qkv slices.
 
-    Note on dropout: the reference's nn.MultiheadAttention drops
-    attention PROBABILITIES (p=0.2).  The fused kernel applies no
-    internal dropout; regularization comes from the existing dropout on
-    the attention output (engine parity note in the module docs).  The
-    CPU path mirrors that (dropout_p ignored) so both paths match.
+    Matches the reference's nn.MultiheadAttention regularization:
+    attention PROBABILITIES are dropped at ``dropout_p`` in train mode
+    (Net/Transformer.py:63-64).  The fused gfx950 kernel applies a
+    philox mask in-kernel (recomputed for backward); the CPU path uses
+    SDPA's own probability dropout.
     """
+    pd = float(dropout_p) if training else 0.0
     if (_use_native("causal_attention", q) and q.dtype == torch.bfloat16
             and q.shape[0] <= 40 and q.shape[2] // nhead <= 104):
         from . import native
-        return native.causal_attention(q, k, v, nhead)
+        return native.causal_attention(q, k, v, nhead, pd)
     S, B, E = q.shape
     d = E // nhead
 
@@ -95,7 +96,7 @@ def causal_attention(q, k, v, nhead, dropout_p=0.0, training=False):
         return t.reshape(S, B * nhead, d).transpose(0, 1)
 
     out = F.scaled_dot_product_attention(
-        split(q), split(k), split(v), dropout_p=0.0, is_causal=True)
+        split(q), split(k), split(v), dropout_p=pd, is_causal=True)
     return out.transpose(0, 1).reshape(S, B, E)
 
 

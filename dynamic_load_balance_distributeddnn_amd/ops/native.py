@@ -227,22 +227,35 @@ def layer_norm(x, weight, bias, eps=1e-5):
 
 # -------------------------------------------------------------- attention
 class _CausalAttention(torch.autograd.Function):
+    """Fused causal MHA with philox dropout on attention PROBABILITIES
+    (reference regularization: nn.TransformerEncoderLayer drops attn
+    weights at p=0.2 — Net/Transformer.py:63-64).  p_save holds the
+    PRE-dropout probs; the mask is philox-recomputed in backward from
+    the same (seed, element) key, so nothing extra is stored."""
+
     @staticmethod
-    def forward(ctx, q, k, v, nhead):
-        o, p_save = ext().attn_fwd(q, k, v, nhead)
+    def forward(ctx, q, k, v, nhead, dropout_p):
+        # draw the per-call seed from torch's CPU generator so runs are
+        # reproducible under torch.manual_seed
+        seed = int(torch.randint(0, 2**62, (1,)).item()) if dropout_p > 0 \
+            else 0
+        o, p_save = ext().attn_fwd(q, k, v, nhead, float(dropout_p), seed)
         ctx.save_for_backward(q, k, v, p_save)
         ctx.nhead = nhead
+        ctx.pd = float(dropout_p)
+        ctx.seed = seed
         return o
 
     @staticmethod
     def backward(ctx, do):
         q, k, v, p_save = ctx.saved_tensors
-        dq, dk, dv = ext().attn_bwd(q, k, v, do, p_save, ctx.nhead)
-        return dq, dk, dv, None
+        dq, dk, dv = ext().attn_bwd(q, k, v, do, p_save, ctx.nhead,
+                                    ctx.pd, ctx.seed)
+        return dq, dk, dv, None, None
 
 
-def causal_attention(q, k, v, nhead):
-    return _CausalAttention.apply(q, k, v, nhead)
+def causal_attention(q, k, v, nhead, dropout_p=0.0):
+    return _CausalAttention.apply(q, k, v, nhead, dropout_p)
 
 
 # ------------------------------------------------------------ log_softmax

@@ -23,10 +23,23 @@ MI355X-native execution (same math):
 
 from __future__ import annotations
 
+import os
+
 import torch
 import torch.distributed as dist
 
 __all__ = ["GradientSynchronizer"]
+
+
+def _default_bucket_bytes() -> int:
+    """Tunable bucket size (env DLB_BUCKET_BYTES).
+
+    Default 4 MiB: DenseNet-121's ~28 MB of fp32 grads then form 7+
+    concurrent reductions — one per xGMI link class (each MI355X has 7
+    p2p links; a single ring all-reduce is per-link bound, so several
+    buckets in flight are what spreads traffic across links)."""
+    v = os.environ.get("DLB_BUCKET_BYTES")
+    return int(v) if v else (4 << 20)
 
 
 def _strided_view(arena: torch.Tensor, offset: int, p: torch.Tensor):
@@ -50,7 +63,7 @@ class GradientSynchronizer:
     def __init__(
         self,
         model: torch.nn.Module,
-        bucket_bytes: int = 4 << 20,
+        bucket_bytes: int | None = None,
         grad_dtype: torch.dtype = torch.float32,
         defer: bool = False,
     ):
@@ -59,6 +72,8 @@ class GradientSynchronizer:
         reduce (the LM path clips gradients first, reference dbs.py:274).
         Buckets still go out as concurrent async collectives."""
         self.defer = defer
+        if bucket_bytes is None:
+            bucket_bytes = _default_bucket_bytes()
         self.params = [p for p in model.parameters() if p.requires_grad]
         device = self.params[0].device
         total = sum((p.numel() + 7) & ~7 for p in self.params)
