@@ -17,6 +17,7 @@ Rebuilds reference dbs.py run()/train()/transformer_train()/validate()
 
 from __future__ import annotations
 
+import os
 import time
 from contextlib import nullcontext
 
@@ -90,6 +91,11 @@ class Trainer:
         if self.is_lm:
             self.dbs_interval = 0
         self._ema_iter_s: float | None = None
+        # hipGraph capture of the whole training step (zoo steps are
+        # hundreds of small kernels; replay removes launch overhead).
+        # world==1 only: per-rank shapes are then epoch-invariant and
+        # there are no collectives to capture.  DLB_NO_GRAPHS disables.
+        self._graph = None  # None=not tried, False=unavailable, else graph
 
         # datasets built once; partitioned fresh each epoch
         if self.is_lm:
@@ -193,6 +199,65 @@ class Trainer:
         t.step_done()
         return loss
 
+    # ---------------------------------------------- hipGraph step path
+    def _graph_body(self):
+        self.sync.zero()
+        with self._autocast():
+            out = self.model(self._static_x)
+            loss = self.criterion(out, self._static_y)
+        loss.backward()
+        self.sync.finish()
+        self.optimizer.step()
+        return loss
+
+    def _capture_graph(self, inputs, target) -> None:
+        """Capture one full training step into a hipGraph.  The 3 warmup
+        iterations update weights; optimizer state is snapshotted and
+        restored so capture has no training side effects."""
+        self._static_x = inputs.clone()
+        self._static_y = target.clone()
+        p0 = self.optimizer.param_arena.clone()
+        m0 = self.optimizer.momentum_buf.clone()
+        try:
+            side = torch.cuda.Stream()
+            side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(side):
+                for _ in range(3):
+                    self._graph_body()
+            torch.cuda.current_stream().wait_stream(side)
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                self._static_loss = self._graph_body()
+            self._graph = g
+        except Exception:
+            self._graph = False  # capture unsupported -> eager forever
+        finally:
+            with torch.no_grad():
+                self.optimizer.param_arena.copy_(p0)
+                self.optimizer.momentum_buf.copy_(m0)
+                self.optimizer.refresh_mirror()
+
+    def _graph_step(self, inputs, target, epoch, steps_per_epoch):
+        if self._graph is None:
+            self._capture_graph(inputs, target)
+        if self._graph is False:
+            return self._step(inputs, target, epoch, steps_per_epoch)
+        t = self.timer
+        t.iter_start()
+        self._static_x.copy_(inputs, non_blocking=True)
+        self._static_y.copy_(target, non_blocking=True)
+        self._graph.replay()
+        t.backward_done()   # whole replay counts as compute; no comm at
+        t.add_compute(self.fault.maybe_wait(epoch, steps_per_epoch))
+        t.comm_done()       # world==1 so the sync interval is ~0
+        t.step_done()
+        return self._static_loss
+
+    def _use_graphs(self) -> bool:
+        return (self.device.type == "cuda" and self.world_size == 1
+                and not self.is_lm
+                and not os.environ.get("DLB_NO_GRAPHS"))
+
     def train_epoch(self, epoch: int):
         """One epoch. Returns (compute_s, sync_s, mean loss)."""
         args = self.args
@@ -230,12 +295,13 @@ class Trainer:
                 epoch_loss += loss.detach()
                 since += 1
         else:
+            step_fn = self._graph_step if self._use_graphs() else self._step
             for inputs, target in source:
                 inputs = inputs.to(self.device, non_blocking=True)
                 if self.device.type == "cuda":
                     inputs = inputs.contiguous(memory_format=torch.channels_last)
                 target = target.to(self.device, non_blocking=True)
-                loss = self._step(inputs, target, epoch, steps)
+                loss = step_fn(inputs, target, epoch, steps)
                 epoch_loss += loss.detach()
 
         compute_s, sync_s = self.timer.epoch_totals()

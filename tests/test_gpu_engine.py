@@ -122,3 +122,28 @@ def test_densenet_grads_match_cpu_fp32_reference():
         assert cos > 0.85, (n1, cos)
     mean_cos = sum(c for c, _ in coses) / len(coses)
     assert mean_cos > 0.97, sorted(coses)[:5]
+
+
+@needs_gpu
+def test_engine_hipgraph_step_matches_eager(tmp_path, monkeypatch):
+    """world==1 CV epochs run as hipGraph replays (capture once, replay
+    per step) and must produce the same training result as eager: the
+    capture snapshots/restores optimizer state, so trajectories align."""
+    monkeypatch.setenv("DLB_SYNTH_SCALE", "0.01")
+    from dynamic_load_balance_distributeddnn_amd.engine import Trainer
+
+    def run(no_graphs):
+        if no_graphs:
+            monkeypatch.setenv("DLB_NO_GRAPHS", "1")
+        else:
+            monkeypatch.delenv("DLB_NO_GRAPHS", raising=False)
+        args = _args(["-d", "false", "-ws", "1", "-b", "64", "-e", "1",
+                      "-ds", "cifar10", "-m", "densenet"])
+        tr = Trainer(args, 0, 1, torch.device("cuda:0"), logger=None)
+        _, _, loss = tr.train_epoch(0)
+        return tr, loss
+
+    tr_g, loss_g = run(False)
+    assert tr_g._graph not in (None, False), "graph capture did not engage"
+    tr_e, loss_e = run(True)
+    assert abs(loss_g - loss_e) < 2e-2 * max(1.0, abs(loss_e))
