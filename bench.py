@@ -41,6 +41,11 @@ def get_args():
     p.add_argument("--device", default=None, help="override (e.g. cpu for debug)")
     p.add_argument("--no-graphs", action="store_true",
                    help="disable hipGraph capture of the training step")
+    p.add_argument("--engine", action="store_true",
+                   help="measure the ENGINE-level step (Trainer epoch: "
+                        "dataloader, H2D, DBS bookkeeping, hipEvent "
+                        "sensor) instead of the pre-staged kernel-"
+                        "throughput step — the trainer's steady state")
     p.add_argument("--no-channels-last", action="store_true",
                    help="disable NHWC layout (NHWC is the GPU default; the "
                         "gfx950 kernels are NHWC-native)")
@@ -63,6 +68,76 @@ def build(model_name, num_classes=10):
     return table[model_name]()
 
 
+def engine_bench(args, rank, world, device):
+    """Trainer-in-the-loop measurement: epochs with the real data path.
+
+    Complements the default pre-staged step (which isolates kernel
+    throughput): this includes the synthetic-dataset loader, H2D, the
+    per-epoch repartition, and the hipEvent DBS sensor."""
+    from dynamic_load_balance_distributeddnn_amd.cli import get_parser
+    from dynamic_load_balance_distributeddnn_amd.engine import Trainer
+    from dynamic_load_balance_distributeddnn_amd.scheduler import \
+        straggler_idle_pct
+
+    # size the synthetic dataset so one epoch is exactly --steps steps
+    os.environ["DLB_SYNTH_SCALE"] = str(
+        args.steps * args.global_batch / 50_000.0)
+    model_flag = {"densenet": "densenet", "resnet": "resnet",
+                  "regnet": "regnet", "googlenet": "googlenet",
+                  "mnistnet": "mnistnet",
+                  "transformer": "transformer"}.get(args.model, "densenet")
+    ds = ("wikitext2" if model_flag == "transformer"
+          else ("mnist" if model_flag == "mnistnet" else "cifar10"))
+    targs = get_parser().parse_args(
+        ["-d", "false" if device.type == "cuda" else "true",
+         "-ws", str(world), "-b", str(args.global_batch),
+         "-e", "2", "-ds", ds, "-m", model_flag])
+    tr = Trainer(targs, rank, world, device, logger=None)
+    tr.train_epoch(0)  # warmup epoch (hipGraph capture happens here)
+    steps = tr.steps_per_epoch
+    if world > 1:
+        dist.barrier()
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    compute_s, _, _ = tr.train_epoch(1)
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    if world > 1:
+        dist.barrier()
+    elapsed = time.perf_counter() - t0
+
+    idle = 0.0
+    if world > 1:
+        t = torch.tensor([elapsed],
+                         device=device if device.type == "cuda" else "cpu")
+        allt = torch.empty(world, dtype=t.dtype, device=t.device)
+        dist.all_gather_into_tensor(allt, t)
+        per = allt.cpu().numpy()
+        elapsed = float(per.max())
+        idle = straggler_idle_pct(per)
+    if rank == 0:
+        items = (args.global_batch * 35 if model_flag == "transformer"
+                 else args.global_batch)
+        print(json.dumps({
+            "metric": ("tokens_per_sec" if model_flag == "transformer"
+                       else "images_per_sec"),
+            "value": round(items * steps / elapsed, 2),
+            "unit": ("tokens/s" if model_flag == "transformer"
+                     else "images/s"),
+            "n_gpus": world, "steps": steps, "warmup": steps,
+            "ms_per_step": round(elapsed / steps * 1e3, 3),
+            "higher_is_better": True, "scaling": "strong",
+            "vs_baseline": None, "dtype": args.dtype, "data": "synthetic",
+            "straggler_idle_pct": round(idle, 3), "mode": "engine",
+            "config": {"model": model_flag,
+                       "global_batch": args.global_batch,
+                       "parallelism": f"dbs-dp{world}"},
+        }))
+    if world > 1:
+        dist.destroy_process_group()
+
+
 def main():
     args = get_args()
     rank = int(os.environ.get("RANK", 0))
@@ -82,6 +157,9 @@ def main():
         os.environ.setdefault("MASTER_PORT", "29531")
         backend = "nccl" if device.type == "cuda" else "gloo"
         dist.init_process_group(backend, rank=rank, world_size=world)
+
+    if args.engine:
+        return engine_bench(args, rank, world, device)
 
     from dynamic_load_balance_distributeddnn_amd.models import LM_CONFIG
     from dynamic_load_balance_distributeddnn_amd.parallel import \
