@@ -37,9 +37,9 @@ extern "C" void dlb_gn_bwd_segs(const void* const* xs, const int* starts,
                                 float* dgb_part,
                                 int N, int HW, int C, int G, int relu,
                                 int accumulate, hipStream_t stream);
-extern "C" void dlb_gn_dgb_reduce(const float* part, float* mid, int N,
-                                  int groups, int C, float* dgamma,
-                                  float* dbeta, hipStream_t stream);
+extern "C" void dlb_gn_dgb_reduce(const float* part, int N, int C,
+                                  float* dgamma, float* dbeta,
+                                  hipStream_t stream);
 extern "C" void dlb_gn_stats_segs(const void* const* xs, const int* starts,
                                   int nseg, float* mean, float* rstd, int N,
                                   int HW, int C, int G, float eps,
@@ -148,12 +148,9 @@ static std::vector<torch::Tensor> gn_bwd(std::vector<torch::Tensor> xs,
   }
   auto stream = at::hip::getCurrentHIPStream();
   // dgamma/dbeta path: per-sample partial rows [N, 2C] (plain stores in
-  // the kernel) + a deterministic two-level column reduction — replaces
-  // the contended global-atomic publish (see reduce.hip).
+  // the kernel) + one deterministic column-sum kernel — replaces the
+  // contended global-atomic publish (see reduce.hip).
   auto part = torch::empty({N, 2 * C},
-                           xs[0].options().dtype(torch::kFloat32));
-  const int rgroups = N >= 64 ? 16 : (N >= 8 ? 4 : 1);
-  auto midb = torch::empty({rgroups, 2 * C},
                            xs[0].options().dtype(torch::kFloat32));
   dlb_gn_bwd_segs(ptrs, starts, (int)xs.size(), dz.data_ptr(), dptrs,
                   gamma.data_ptr<float>(), beta.data_ptr<float>(),
@@ -161,9 +158,9 @@ static std::vector<torch::Tensor> gn_bwd(std::vector<torch::Tensor> xs,
                   part.data_ptr<float>(),
                   N, HW, C, (int)groups, relu ? 1 : 0, acc ? 1 : 0,
                   stream.stream());
-  dlb_gn_dgb_reduce(part.data_ptr<float>(), midb.data_ptr<float>(), N,
-                    rgroups, C, dgamma.data_ptr<float>(),
-                    dbeta.data_ptr<float>(), stream.stream());
+  dlb_gn_dgb_reduce(part.data_ptr<float>(), N, C,
+                    dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),
+                    stream.stream());
   out.push_back(dgamma);
   out.push_back(dbeta);
   return out;  // [dx_0..dx_{k-1}, dgamma, dbeta]
@@ -219,18 +216,13 @@ extern "C" int dlb_conv_wrw_nsplits(int N, int OH, int OW, int Ci, int Co,
 extern "C" void dlb_slab_sum(const float* part, float* out, int splits,
                              long len, hipStream_t stream);
 
-// Reduce [splits, len] fp32 slabs into `out` (len): one float4 streaming
-// kernel, summed in split order (deterministic) — replaces the generic
+// Reduce [splits, len] fp32 slabs into `out` (len) with the
+// deterministic column-sum kernel (reduce.hip) — replaces the generic
 // torch reducer that measured 11 µs/call at these shapes.
 static void slab_reduce_into(torch::Tensor part, torch::Tensor out,
                              int splits, long len) {
-  if (len % 4 == 0) {
-    dlb_slab_sum(part.data_ptr<float>(), out.data_ptr<float>(), splits, len,
-                 at::hip::getCurrentHIPStream().stream());
-  } else {
-    auto o = out.view({len});
-    torch::sum_out(o, part.view({splits, len}), 0);
-  }
+  dlb_slab_sum(part.data_ptr<float>(), out.data_ptr<float>(), splits, len,
+               at::hip::getCurrentHIPStream().stream());
 }
 
 static inline bool is_cl(const torch::Tensor& t) {

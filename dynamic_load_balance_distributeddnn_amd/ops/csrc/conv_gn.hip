@@ -102,13 +102,15 @@ __device__ inline void gnc_stage_ab(const GnCParams& p, float* sa, float* sb,
   }
 }
 
-// Normalized 8-channel load at (row m, channel k): one LDS segment
-// lookup + 8 FMAs against the staged tables.
-__device__ inline bf16x8_t gn_load8(const GnCParams& p, const short* soct,
-                                    const float* sa, const float* sb, int W,
-                                    int koff, int n_base, int m, int k,
-                                    int mbound) {
+// RAW 8-channel segment load at (row m, channel k) — no normalization:
+// the affine tables are applied at LDS-WRITE time (gnc_apply8), which
+// keeps the global-load loop free of dependent FMA chains (the fused
+// wrw measured ~15% behind the plain wrw with the math on the load
+// path — round-1 ROADMAP #5).
+__device__ inline bf16x8_t gn_raw8(const GnCParams& p, const short* soct,
+                                   int m, int k, int mbound, bool* okp) {
   const bool ok = (m < mbound) & (k < p.C);
+  *okp = ok;
   if (!ok) return gnc_zero8();
   const unsigned n = p.fd_hw.div((unsigned)m);
   const unsigned pix = (unsigned)m - n * (unsigned)p.HW;
@@ -116,8 +118,15 @@ __device__ inline bf16x8_t gn_load8(const GnCParams& p, const short* soct,
   const int cs = p.segs.start[si + 1] - p.segs.start[si];
   const bf16* ptr = p.segs.p[si] +
                     ((long)n * p.HW + pix) * cs + (k - p.segs.start[si]);
-  bf16x8_t x8 = *reinterpret_cast<const bf16x8_t*>(ptr);
-  const int base = ((int)n - n_base) * W + koff;
+  return *reinterpret_cast<const bf16x8_t*>(ptr);
+}
+
+// gn(x) = x*a + b (+ReLU) against the staged tables; zero for padding
+// lanes (the GEMM's K/M padding must stay zero, not b).
+__device__ inline bf16x8_t gnc_apply8(const GnCParams& p, bf16x8_t x8,
+                                      const float* sa, const float* sb,
+                                      int base, bool ok) {
+  if (!ok) return gnc_zero8();
   union { bf16x8_t v; bf16 h[8]; } in, out;
   in.v = x8;
 #pragma unroll
@@ -194,6 +203,7 @@ gnconv1x1_fwd_kernel(const GnCParams p) {
     for (int j = 0; j < FB; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
   bf16x8_t areg[APT], breg[BPT];
+  bool aok[APT];
 
   auto load_tile = [&](int kt) {
 #pragma unroll
@@ -201,10 +211,8 @@ gnconv1x1_fwd_kernel(const GnCParams p) {
       const int c = t + u * CONV_BLOCK;
       const int row = c / (BK / 8);
       const int k8 = (c % (BK / 8)) * 8;
-      const int par = (kt / BK) & 1;
-      areg[u] = gn_load8(p, s_oct, s_ga[par], s_gb[par], BK, k8, nb0,
-                         m0 + row, kt + k8, p.M);
-      if (c >= ACH) areg[u] = gnc_zero8();
+      areg[u] = gn_raw8(p, s_oct, m0 + row, kt + k8, p.M, &aok[u]);
+      if (c >= ACH) { areg[u] = gnc_zero8(); aok[u] = false; }
     }
 #pragma unroll
     for (int u = 0; u < BPT; ++u) {
@@ -219,14 +227,18 @@ gnconv1x1_fwd_kernel(const GnCParams p) {
     }
   };
 
-  auto write_tile = [&](int buf) {
+  auto write_tile = [&](int buf, int kt) {
+    const int par = (kt / BK) & 1;
 #pragma unroll
     for (int u = 0; u < APT; ++u) {
       const int c = t + u * CONV_BLOCK;
       if (c < ACH) {
         const int row = c / (BK / 8);
         const int k8 = (c % (BK / 8)) * 8;
-        *reinterpret_cast<bf16x8_t*>(&a_lds[buf][row * LDA + k8]) = areg[u];
+        const int n = (int)p.fd_hw.div((unsigned)(m0 + row));
+        const int base = (n - nb0) * BK + k8;
+        *reinterpret_cast<bf16x8_t*>(&a_lds[buf][row * LDA + k8]) =
+            gnc_apply8(p, areg[u], s_ga[par], s_gb[par], base, aok[u]);
       }
     }
 #pragma unroll
@@ -248,7 +260,7 @@ gnconv1x1_fwd_kernel(const GnCParams p) {
   stage_ab(0);
   __syncthreads();
   load_tile(0);
-  write_tile(0);
+  write_tile(0, 0);
   stage_ab(BK);
   __syncthreads();
 
@@ -276,7 +288,7 @@ gnconv1x1_fwd_kernel(const GnCParams p) {
       for (int j = 0; j < FB; ++j)
         acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
-    if (more) write_tile(buf ^ 1);
+    if (more) write_tile(buf ^ 1, kt + BK);
     if (kt + 2 * BK < p.C) stage_ab(kt + 2 * BK);
     __syncthreads();
     buf ^= 1;
@@ -356,6 +368,7 @@ gnconv1x1_wrw_kernel(const GnCParams p) {
     for (int j = 0; j < FB; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
   bf16x8_t dreg[DPT], xreg[XPT];
+  bool xok[XPT];
 
   auto load_chunk = [&](int mt) {
 #pragma unroll
@@ -380,15 +393,12 @@ gnconv1x1_wrw_kernel(const GnCParams p) {
       const int c = t + u * CONV_BLOCK;
       const int mm = c / (BKN / 8);
       const int k8 = (c % (BKN / 8)) * 8;
-      const int par = (mt / BM) & 1;
-      xreg[u] = gn_load8(p, s_oct, s_ga[par], s_gb[par], BKN, k8,
-                         (int)p.fd_hw.div((unsigned)mt), mt + mm, k0 + k8,
-                         mend);
-      if (c >= XCH) xreg[u] = gnc_zero8();
+      xreg[u] = gn_raw8(p, s_oct, mt + mm, k0 + k8, mend, &xok[u]);
+      if (c >= XCH) { xreg[u] = gnc_zero8(); xok[u] = false; }
     }
   };
 
-  auto write_chunk = [&](int buf) {
+  auto write_chunk = [&](int buf, int mt) {
 #pragma unroll
     for (int u = 0; u < DPT; ++u) {
       const int c = t + u * CONV_BLOCK;
@@ -398,13 +408,18 @@ gnconv1x1_wrw_kernel(const GnCParams p) {
         *reinterpret_cast<bf16x8_t*>(&dy_t[buf][mm * LD + c8]) = dreg[u];
       }
     }
+    const int par = (mt / BM) & 1;
+    const int nb = (int)p.fd_hw.div((unsigned)mt);
 #pragma unroll
     for (int u = 0; u < XPT; ++u) {
       const int c = t + u * CONV_BLOCK;
       if (c < XCH) {
         const int mm = c / (BKN / 8);
         const int k8 = (c % (BKN / 8)) * 8;
-        *reinterpret_cast<bf16x8_t*>(&x_t[buf][mm * LX + k8]) = xreg[u];
+        const int n = (int)p.fd_hw.div((unsigned)(mt + mm));
+        const int base = (n - nb) * BKN + k8;
+        *reinterpret_cast<bf16x8_t*>(&x_t[buf][mm * LX + k8]) =
+            gnc_apply8(p, xreg[u], s_ga[par], s_gb[par], base, xok[u]);
       }
     }
   };
@@ -419,7 +434,7 @@ gnconv1x1_wrw_kernel(const GnCParams p) {
   stage_ab(mstart);
   __syncthreads();
   load_chunk(mstart);
-  write_chunk(0);
+  write_chunk(0, mstart);
   stage_ab(mstart + BM);
   __syncthreads();
 
@@ -445,7 +460,7 @@ gnconv1x1_wrw_kernel(const GnCParams p) {
           acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
     }
-    if (more) write_chunk(buf ^ 1);
+    if (more) write_chunk(buf ^ 1, mt + BM);
     if (mt + 2 * BM < mend) stage_ab(mt + 2 * BM);
     __syncthreads();
     buf ^= 1;

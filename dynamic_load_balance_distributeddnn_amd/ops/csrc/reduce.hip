@@ -1,83 +1,82 @@
-// Split-K slab reduction for the conv weight-grad kernels (gfx950).
+// Deterministic column reductions for split-K weight-grad slabs and the
+// GroupNorm dgamma/dbeta partials (gfx950).
 //
-// The wrw kernels write per-split fp32 partial slabs [splits, Co*K]
-// (contention-free, deterministic).  Round 1 reduced them with
-// torch::sum_out — at::native::reduce_kernel measured 4.4% of the
-// DenseNet step (profiles/SUMMARY.md: 24.3 ms / 2160 calls ≈ 11 µs per
-// call for reductions whose traffic is < 1 µs at HBM speed; the generic
-// reducer's config is launch/occupancy-bound at these shapes).  This
-// kernel is a flat float4 streaming sum: out[i] = Σ_s part[s*len + i],
-// summed in split order (bitwise deterministic across runs).
+//   out[c] (+)= sum_{s<S} part[s*len + c]
+//
+// Profiled history: torch::sum_out ran these shapes at ~11 us/call
+// (generic-reducer config overhead, 4.4% of the DenseNet step); a flat
+// one-thread-per-column kernel was WORSE (~42 us: ceil(len/256) blocks
+// is 32 blocks at typical wrw sizes, with a serial split loop —
+// latency-bound).  This version parallelizes both dimensions inside a
+// block: 256 threads as 32 columns x 8 row-groups, each thread sums a
+// fixed row range with a 4-deep unrolled (ILP) loop, then an LDS tree
+// folds the 8 row-group partials in a FIXED order — bitwise
+// deterministic across runs, unlike fp32 atomics.
 #include "common.h"
 
-extern "C" __global__ void __launch_bounds__(256)
-slab_sum_kernel(const float* __restrict__ part, float* __restrict__ out,
-                const int splits, const long len) {
-  const long q = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 4;
-  if (q >= len) return;
-  float4 acc = *reinterpret_cast<const float4*>(part + q);
-  for (int s = 1; s < splits; ++s) {
-    float4 v = *reinterpret_cast<const float4*>(part + (long)s * len + q);
-    acc.x += v.x;
-    acc.y += v.y;
-    acc.z += v.z;
-    acc.w += v.w;
+#define CS_COLS 32
+#define CS_ROWG 8
+
+// Csplit >= 0 routes columns: c < Csplit -> outA[c], else outB[c-Csplit]
+// (the dgamma/dbeta pair); Csplit < 0 -> everything to outA.
+// add != 0 accumulates into pre-zeroed outputs, else overwrites.
+extern "C" __global__ void __launch_bounds__(CS_COLS* CS_ROWG)
+colsum_kernel(const float* __restrict__ part, const int S, const long len,
+              float* __restrict__ outA, float* __restrict__ outB,
+              const long Csplit, const int add) {
+  const int tc = threadIdx.x % CS_COLS;
+  const int tr = threadIdx.x / CS_COLS;
+  const long c = (long)blockIdx.x * CS_COLS + tc;
+  __shared__ float fold[CS_ROWG][CS_COLS];
+
+  float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
+  if (c < len) {
+    const int gs = (S + CS_ROWG - 1) / CS_ROWG;
+    const int s0 = tr * gs;
+    const int s1 = min(S, s0 + gs);
+    const float* p = part + c;
+    int s = s0;
+    for (; s + 4 <= s1; s += 4) {
+      a0 += p[(long)s * len];
+      a1 += p[(long)(s + 1) * len];
+      a2 += p[(long)(s + 2) * len];
+      a3 += p[(long)(s + 3) * len];
+    }
+    for (; s < s1; ++s) a0 += p[(long)s * len];
   }
-  *reinterpret_cast<float4*>(out + q) = acc;
+  fold[tr][tc] = (a0 + a1) + (a2 + a3);
+  __syncthreads();
+#pragma unroll
+  for (int off = CS_ROWG / 2; off > 0; off >>= 1) {
+    if (tr < off) fold[tr][tc] += fold[tr + off][tc];
+    __syncthreads();
+  }
+  if (tr == 0 && c < len) {
+    float v = fold[0][tc];
+    float* dst = (Csplit >= 0 && c >= Csplit) ? outB + (c - Csplit)
+                                              : outA + c;
+    if (add)
+      *dst += v;
+    else
+      *dst = v;
+  }
 }
 
-// len must be a multiple of 4 (every wrw slab is: K and C are multiples
-// of 8); the binding falls back to torch otherwise.
+// Split-K slab reduce: out[0:len] = sum over slabs (overwrite).
 extern "C" void dlb_slab_sum(const float* part, float* out, int splits,
                              long len, hipStream_t stream) {
-  const long quads = len / 4;
-  const int block = 256;
-  const long grid = (quads + block - 1) / block;
-  hipLaunchKernelGGL(slab_sum_kernel, dim3((unsigned)grid), dim3(block), 0,
-                     stream, part, out, splits, len);
+  const unsigned grid = (unsigned)((len + CS_COLS - 1) / CS_COLS);
+  hipLaunchKernelGGL(colsum_kernel, dim3(grid), dim3(CS_COLS * CS_ROWG), 0,
+                     stream, part, splits, len, out, nullptr, -1L, 0);
 }
 
-// --------- GroupNorm dgamma/dbeta deterministic column reduction --------
-// gn_bwd publishes per-sample partials part[n][2C] with plain stores
-// (global atomicAdd on the [2C] words measured a +10..30 us per-dispatch
-// tail under 512-way contention — tools/gn_probe — and made dgamma
-// nondeterministic).  Level 1 sums sample groups; level 2 adds the group
-// sums into the dgamma/dbeta buffers.  Fixed split order -> bitwise
-// deterministic across runs.
-extern "C" __global__ void __launch_bounds__(256)
-gn_dgb_l1_kernel(const float* __restrict__ part, float* __restrict__ mid,
-                 const int N, const int gs, const long len2) {
-  const long c = (long)blockIdx.x * 256 + threadIdx.x;
-  if (c >= len2) return;
-  const int n0 = blockIdx.y * gs;
-  const int n1 = min(n0 + gs, N);
-  float s = 0.f;
-  for (int n = n0; n < n1; ++n) s += part[(long)n * len2 + c];
-  mid[(long)blockIdx.y * len2 + c] = s;
-}
-
-extern "C" __global__ void __launch_bounds__(256)
-gn_dgb_l2_kernel(const float* __restrict__ mid, const int groups,
-                 const long len2, const int C, float* __restrict__ dgamma,
-                 float* __restrict__ dbeta) {
-  const long c = (long)blockIdx.x * 256 + threadIdx.x;
-  if (c >= len2) return;
-  float s = 0.f;
-  for (int g = 0; g < groups; ++g) s += mid[(long)g * len2 + c];
-  if (c < C)
-    dgamma[c] += s;
-  else
-    dbeta[c - C] += s;
-}
-
-extern "C" void dlb_gn_dgb_reduce(const float* part, float* mid, int N,
-                                  int groups, int C, float* dgamma,
-                                  float* dbeta, hipStream_t stream) {
+// GroupNorm dgamma/dbeta: part is [N, 2C] per-sample partial rows;
+// accumulates into the (pre-zeroed / accumulating) dgamma and dbeta.
+extern "C" void dlb_gn_dgb_reduce(const float* part, int N, int C,
+                                  float* dgamma, float* dbeta,
+                                  hipStream_t stream) {
   const long len2 = 2L * C;
-  const int gs = (N + groups - 1) / groups;
-  const unsigned cb = (unsigned)((len2 + 255) / 256);
-  hipLaunchKernelGGL(gn_dgb_l1_kernel, dim3(cb, groups), dim3(256), 0,
-                     stream, part, mid, N, gs, len2);
-  hipLaunchKernelGGL(gn_dgb_l2_kernel, dim3(cb), dim3(256), 0, stream, mid,
-                     groups, len2, C, dgamma, dbeta);
+  const unsigned grid = (unsigned)((len2 + CS_COLS - 1) / CS_COLS);
+  hipLaunchKernelGGL(colsum_kernel, dim3(grid), dim3(CS_COLS * CS_ROWG), 0,
+                     stream, part, N, len2, dgamma, dbeta, (long)C, 1);
 }
