@@ -13,24 +13,30 @@
 // path runs in (channels_last).  Grid = (N, channel-chunks): GroupNorm's
 // reductions are per-(sample, group), so a chunk of WHOLE groups is
 // fully independent of every other chunk — slicing C (at group-aligned
-// octet granularity) multiplies blocks without scratch buffers, extra
-// passes, or any cross-block coordination.  Round-1 ran one block per
-// sample (N=512 -> 2 blocks/CU) and the small-HW DenseNet layers were
-// latency-bound at 0.4-1.6 TB/s (profiles/, gn_bwd_bench); round-1's
-// alternative (HW slicing + scratch reduce) measured slower at every
-// target and is gone.  Threads within a block are (pixel, channel-octet)
-// as before: a thread owns a FIXED run of 8 channels across pixels
-// strided by TP, so partial sums live in 8 statically-indexed registers
-// (no scratch — CDNA guide rule 20) and the group merge is 8 LDS atomics
-// per thread per sweep.  All loads are 16-byte bf16x8, fully coalesced.
-// Requires C%8==0 (every channel count in the zoo satisfies this).
+// octet granularity, <= 128 octets so every thread owns exactly ONE
+// channel octet) multiplies blocks without any cross-block coordination.
+// Threads are (pixel, channel-octet): a thread owns a FIXED run of 8
+// channels across pixels strided by TP, so partial sums live in 8
+// statically-indexed registers (no scratch — CDNA guide rule 20).
+//
+// Reductions are DETERMINISTIC by construction: per-thread partials go
+// to a [TP][cspan] LDS image with plain stores (TP*cspan <= 2048
+// floats), then fixed-order tree/serial merges produce the group sums
+// and per-channel dgamma/dbeta.  The earlier LDS atomicAdd merge both
+// serialized under tp contention (+6..+100 us/dispatch, tools/gn_probe)
+// and made results order-dependent across runs; dgamma/dbeta global
+// atomics had the same two problems and are now per-sample partial rows
+// reduced by reduce.hip's colsum kernel.
+// All loads are 16-byte bf16x8, fully coalesced.  Requires C%8==0
+// (every channel count in the zoo satisfies this).
 
 #include "common.h"
 
 #ifndef GN_BLOCK
 #define GN_BLOCK 256
 #endif
-#define GN_MAXG 64  // num_groups <= 64 covers the zoo (8/16/32)
+#define GN_MAXG 64      // num_groups <= 64 covers the zoo (8/16/32)
+#define GN_MAXCHUNK 128  // octets per chunk (=> single octet per thread)
 
 typedef __hip_bfloat16 bf16;
 
@@ -65,8 +71,9 @@ __device__ inline const bf16* seg_locate(const GnSegs& sg, int c0, int& cloc,
   return sg.p[si];
 }
 
-// This block's octet range [o0, o1) and thread mapping for it.
-// chunk_oct is group-aligned, so [o0*8, o1*8) covers whole groups.
+// This block's octet range [o0, o1) and thread mapping.  chunk_oct is
+// group-aligned and <= GN_MAXCHUNK, so TCe == span and every active
+// thread owns exactly one octet (single-trip).
 struct ChunkMap {
   int o0, o1;    // octet range
   int TCe, TP;   // channel-octet threads, pixel stride
@@ -78,8 +85,7 @@ __device__ inline ChunkMap chunk_map(int TC, int chunk_oct) {
   ChunkMap m;
   m.o0 = blockIdx.y * chunk_oct;
   m.o1 = min(m.o0 + chunk_oct, TC);
-  const int span = m.o1 - m.o0;
-  m.TCe = span < GN_BLOCK ? span : GN_BLOCK;
+  m.TCe = m.o1 - m.o0;       // <= GN_MAXCHUNK <= GN_BLOCK
   m.TP = GN_BLOCK / m.TCe;
   const int t = threadIdx.x;
   m.tc = t % m.TCe;
@@ -88,27 +94,9 @@ __device__ inline ChunkMap chunk_map(int TC, int chunk_oct) {
   return m;
 }
 
-// Same-address LDS atomics from the TP lanes of one octet serialize
-// (+6..+100 us per dispatch, tools/gn_probe).  When TCe is a pow2 < 64
-// (then the octet loop is single-trip and all 256 threads are active),
-// fold the tp lanes with wave shuffles first; only lanes < TCe touch
-// LDS afterwards (contention drops to the 4 waves).
-__device__ inline bool tp_shuffle_ok(const ChunkMap& m) {
-  return (m.TCe == m.o1 - m.o0) && m.TCe < 64 &&
-         ((m.TCe & (m.TCe - 1)) == 0);
-}
-
-__device__ inline float tp_fold(float v, int TCe) {
-  for (int off = TCe; off < 64; off <<= 1) v += __shfl_down(v, off, 64);
-  return v;
-}
-
 // ---------------------------------------------------------------- forward
-// Dynamic LDS: [2 * chunk-channels] staged gamma/beta.  Per-octet
-// parameter reads come from LDS — per-lane 4-byte GLOBAL gathers here
-// compiled to a serialized load->use->waitcnt chain (~10-25 us of fixed
-// per-block latency, the measured small-shape floor); staging them with
-// one coalesced sweep removes it.
+// Dynamic LDS layout: [cspan] gamma | [cspan] beta | [TP*cspan] P0 |
+// [TP*cspan] P1  (P* hold per-(tp, channel) partials; TP*cspan <= 2048).
 extern "C" __global__ void __launch_bounds__(GN_BLOCK)
 gn_fwd_kernel(const GnSegs segs, bf16* __restrict__ y,
               const float* __restrict__ gamma, const float* __restrict__ beta,
@@ -120,63 +108,62 @@ gn_fwd_kernel(const GnSegs segs, bf16* __restrict__ y,
   const ChunkMap m = chunk_map(TC, chunk_oct);
   const int t = threadIdx.x;
   const int Cg = C / G;
-  // groups owned by this chunk (chunk boundaries are group-aligned)
   const int g0 = (m.o0 << 3) / Cg, g1 = (m.o1 << 3) / Cg;
   const int cbase = m.o0 << 3;
   const int cspan = (m.o1 - m.o0) << 3;
 
-  __shared__ float s_sum[GN_MAXG];
-  __shared__ float s_ssq[GN_MAXG];
   __shared__ float s_mean[GN_MAXG];
   __shared__ float s_rstd[GN_MAXG];
-  extern __shared__ float s_par[];  // [cspan] gamma, [cspan] beta
-  float* s_ga = s_par;
-  float* s_be = s_par + cspan;
-  for (int g = g0 + t; g < g1; g += GN_BLOCK) { s_sum[g] = 0.f; s_ssq[g] = 0.f; }
+  extern __shared__ float s_dyn[];
+  float* s_ga = s_dyn;
+  float* s_be = s_dyn + cspan;
+  float* P0 = s_dyn + 2 * cspan;
+  float* P1 = P0 + m.TP * cspan;
   for (int c = t; c < cspan; c += GN_BLOCK) {
     s_ga[c] = gamma[cbase + c];
     s_be[c] = beta[cbase + c];
   }
-  __syncthreads();
+
+  const int c0 = (m.o0 + m.tc) << 3;
+  int cloc, cs;
+  const bf16* sb = seg_locate(segs, c0, cloc, cs);
+  const bf16* xb = sb + (long)n * HW * cs + cloc;
 
   if (m.active) {
-    for (int oct = m.o0 + m.tc; oct < m.o1; oct += m.TCe) {
-      const int c0 = oct << 3;
-      int cloc, cs;
-      const bf16* sb = seg_locate(segs, c0, cloc, cs);
-      const bf16* xb = sb + (long)n * HW * cs + cloc;
-      float s[8] = {0, 0, 0, 0, 0, 0, 0, 0};
-      float ss[8] = {0, 0, 0, 0, 0, 0, 0, 0};
-      #pragma unroll 4
-      for (int p = m.tp; p < HW; p += m.TP) {
-        Bf16x8 chunk = *reinterpret_cast<const Bf16x8*>(xb + (long)p * cs);
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          float v = bf2f(chunk.v[j]);
-          s[j] += v;
-          ss[j] += v * v;
-        }
-      }
-      const bool shf = tp_shuffle_ok(m);
-      const int lane = t & 63;
+    float s[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+    float ss[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+    #pragma unroll 4
+    for (int p = m.tp; p < HW; p += m.TP) {
+      Bf16x8 chunk = *reinterpret_cast<const Bf16x8*>(xb + (long)p * cs);
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        float v1 = shf ? tp_fold(s[j], m.TCe) : s[j];
-        float v2 = shf ? tp_fold(ss[j], m.TCe) : ss[j];
-        if (!shf || lane < m.TCe) {
-          int g = (c0 + j) / Cg;
-          atomicAdd(&s_sum[g], v1);
-          atomicAdd(&s_ssq[g], v2);
-        }
+        float v = bf2f(chunk.v[j]);
+        s[j] += v;
+        ss[j] += v * v;
       }
+    }
+    const int pb = m.tp * cspan + (m.tc << 3);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      P0[pb + j] = s[j];
+      P1[pb + j] = ss[j];
     }
   }
   __syncthreads();
 
+  // fixed-order group merge (deterministic): one thread per group sums
+  // its channels across the TP rows
   const float inv_m = 1.0f / ((float)HW * Cg);
   for (int g = g0 + t; g < g1; g += GN_BLOCK) {
-    float mu = s_sum[g] * inv_m;
-    float var = s_ssq[g] * inv_m - mu * mu;
+    float su = 0.f, sq = 0.f;
+    const int cl0 = g * Cg - cbase;
+    for (int tp = 0; tp < m.TP; ++tp)
+      for (int c = cl0; c < cl0 + Cg; ++c) {
+        su += P0[tp * cspan + c];
+        sq += P1[tp * cspan + c];
+      }
+    float mu = su * inv_m;
+    float var = sq * inv_m - mu * mu;
     float r = rsqrtf(var + eps);
     s_mean[g] = mu;
     s_rstd[g] = r;
@@ -188,32 +175,26 @@ gn_fwd_kernel(const GnSegs segs, bf16* __restrict__ y,
   if (!m.active) return;
 
   bf16* yb = y + (long)n * HW * C;
-  for (int oct = m.o0 + m.tc; oct < m.o1; oct += m.TCe) {
-    const int c0 = oct << 3;
-    int cloc, cs;
-    const bf16* sb = seg_locate(segs, c0, cloc, cs);
-    const bf16* xb = sb + (long)n * HW * cs + cloc;
-    float ga[8], be[8], mu[8], rs[8];
+  float ga[8], be[8], mu[8], rs[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    int c = c0 + j, g = c / Cg;
+    ga[j] = s_ga[c - cbase];
+    be[j] = s_be[c - cbase];
+    mu[j] = s_mean[g];
+    rs[j] = s_rstd[g];
+  }
+  #pragma unroll 4
+  for (int p = m.tp; p < HW; p += m.TP) {
+    Bf16x8 chunk = *reinterpret_cast<const Bf16x8*>(xb + (long)p * cs);
+    Bf16x8 out;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      int c = c0 + j, g = c / Cg;
-      ga[j] = s_ga[c - cbase];
-      be[j] = s_be[c - cbase];
-      mu[j] = s_mean[g];
-      rs[j] = s_rstd[g];
+      float v = (bf2f(chunk.v[j]) - mu[j]) * rs[j] * ga[j] + be[j];
+      if (relu) v = fmaxf(v, 0.f);
+      out.v[j] = f2bf(v);
     }
-    #pragma unroll 4
-      for (int p = m.tp; p < HW; p += m.TP) {
-      Bf16x8 chunk = *reinterpret_cast<const Bf16x8*>(xb + (long)p * cs);
-      Bf16x8 out;
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        float v = (bf2f(chunk.v[j]) - mu[j]) * rs[j] * ga[j] + be[j];
-        if (relu) v = fmaxf(v, 0.f);
-        out.v[j] = f2bf(v);
-      }
-      *reinterpret_cast<Bf16x8*>(yb + (long)p * C + c0) = out;
-    }
+    *reinterpret_cast<Bf16x8*>(yb + (long)p * C + c0) = out;
   }
 }
 
@@ -222,6 +203,7 @@ gn_fwd_kernel(const GnSegs segs, bf16* __restrict__ y,
 // virtual concat, with NO normalize pass.  Used by the fused
 // GN->1x1-conv kernels (conv_gn.hip), which normalize at operand-load
 // time — the packed y activation is never materialized.
+// Dynamic LDS: [TP*cspan] P0 | [TP*cspan] P1.
 extern "C" __global__ void __launch_bounds__(GN_BLOCK)
 gn_stats_kernel(const GnSegs segs, float* __restrict__ mean_out,
                 float* __restrict__ rstd_out, const int HW, const int C,
@@ -232,49 +214,50 @@ gn_stats_kernel(const GnSegs segs, float* __restrict__ mean_out,
   const int t = threadIdx.x;
   const int Cg = C / G;
   const int g0 = (m.o0 << 3) / Cg, g1 = (m.o1 << 3) / Cg;
+  const int cbase = m.o0 << 3;
+  const int cspan = (m.o1 - m.o0) << 3;
 
-  __shared__ float s_sum[GN_MAXG];
-  __shared__ float s_ssq[GN_MAXG];
-  for (int g = g0 + t; g < g1; g += GN_BLOCK) { s_sum[g] = 0.f; s_ssq[g] = 0.f; }
-  __syncthreads();
+  extern __shared__ float s_dyn[];
+  float* P0 = s_dyn;
+  float* P1 = s_dyn + m.TP * cspan;
 
   if (m.active) {
-    for (int oct = m.o0 + m.tc; oct < m.o1; oct += m.TCe) {
-      const int c0 = oct << 3;
-      int cloc, cs;
-      const bf16* sb = seg_locate(segs, c0, cloc, cs);
-      const bf16* xb = sb + (long)n * HW * cs + cloc;
-      float s[8] = {0, 0, 0, 0, 0, 0, 0, 0};
-      float ss[8] = {0, 0, 0, 0, 0, 0, 0, 0};
-      #pragma unroll 4
-      for (int p = m.tp; p < HW; p += m.TP) {
-        Bf16x8 chunk = *reinterpret_cast<const Bf16x8*>(xb + (long)p * cs);
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          float v = bf2f(chunk.v[j]);
-          s[j] += v;
-          ss[j] += v * v;
-        }
-      }
-      const bool shf = tp_shuffle_ok(m);
-      const int lane = t & 63;
+    const int c0 = (m.o0 + m.tc) << 3;
+    int cloc, cs;
+    const bf16* sb = seg_locate(segs, c0, cloc, cs);
+    const bf16* xb = sb + (long)n * HW * cs + cloc;
+    float s[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+    float ss[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+    #pragma unroll 4
+    for (int p = m.tp; p < HW; p += m.TP) {
+      Bf16x8 chunk = *reinterpret_cast<const Bf16x8*>(xb + (long)p * cs);
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        float v1 = shf ? tp_fold(s[j], m.TCe) : s[j];
-        float v2 = shf ? tp_fold(ss[j], m.TCe) : ss[j];
-        if (!shf || lane < m.TCe) {
-          int g = (c0 + j) / Cg;
-          atomicAdd(&s_sum[g], v1);
-          atomicAdd(&s_ssq[g], v2);
-        }
+        float v = bf2f(chunk.v[j]);
+        s[j] += v;
+        ss[j] += v * v;
       }
+    }
+    const int pb = m.tp * cspan + (m.tc << 3);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      P0[pb + j] = s[j];
+      P1[pb + j] = ss[j];
     }
   }
   __syncthreads();
+
   const float inv_m = 1.0f / ((float)HW * Cg);
   for (int g = g0 + t; g < g1; g += GN_BLOCK) {
-    float mu = s_sum[g] * inv_m;
-    float var = s_ssq[g] * inv_m - mu * mu;
+    float su = 0.f, sq = 0.f;
+    const int cl0 = g * Cg - cbase;
+    for (int tp = 0; tp < m.TP; ++tp)
+      for (int c = cl0; c < cl0 + Cg; ++c) {
+        su += P0[tp * cspan + c];
+        sq += P1[tp * cspan + c];
+      }
+    float mu = su * inv_m;
+    float var = sq * inv_m - mu * mu;
     mean_out[(long)n * G + g] = mu;
     rstd_out[(long)n * G + g] = rsqrtf(var + eps);
   }
@@ -283,9 +266,11 @@ gn_stats_kernel(const GnSegs segs, float* __restrict__ mean_out,
 // ---------------------------------------------------------------- backward
 // dx_i = r * (g_c*dy_i - (s1 + xhat_i*s2) / m)   with per-group sums
 //   s1 = sum(g_c * dy),  s2 = sum(g_c * dy * xhat)
-// dgamma_c = sum_{n,p} dy*xhat ; dbeta_c = sum_{n,p} dy  (global atomics,
-// caller zero-fills).  ReLU mask recomputed as (xhat*g+b) > 0.
-// Dynamic LDS: [2 * chunk_oct*8] dgamma/dbeta partials, local channels.
+// dgamma_c/dbeta_c partials go to dgb_part[n][2C] rows (plain stores);
+// reduce.hip's colsum kernel folds them deterministically.
+// ReLU mask recomputed as (xhat*g+b) > 0.
+// Dynamic LDS: [2*cspan] dgb | [cspan] gamma | [cspan] beta |
+// [TP*cspan] P0 | [TP*cspan] P1.
 extern "C" __global__ void __launch_bounds__(GN_BLOCK)
 gn_bwd_kernel(const GnSegs segs, const bf16* __restrict__ dz,
               const GnSegsMut dxs, const float* __restrict__ gamma,
@@ -301,24 +286,23 @@ gn_bwd_kernel(const GnSegs segs, const bf16* __restrict__ dz,
   const int t = threadIdx.x;
   const int Cg = C / G;
   const int g0 = (m.o0 << 3) / Cg, g1 = (m.o1 << 3) / Cg;
-  const int cbase = m.o0 << 3;                  // first channel of chunk
-  const int cspan = (m.o1 - m.o0) << 3;         // channels in chunk
+  const int cbase = m.o0 << 3;
+  const int cspan = (m.o1 - m.o0) << 3;
 
   __shared__ float s_s1[GN_MAXG];
   __shared__ float s_s2[GN_MAXG];
   __shared__ float s_mu[GN_MAXG];
   __shared__ float s_rs[GN_MAXG];
-  extern __shared__ float s_dgb[];  // [2*cspan] dgamma/dbeta partials,
-                                    // then [cspan] gamma, [cspan] beta
-  float* s_ga = s_dgb + 2 * cspan;
-  float* s_be = s_dgb + 3 * cspan;
+  extern __shared__ float s_dyn[];
+  float* s_dgb = s_dyn;              // [2*cspan] merged dgamma/dbeta
+  float* s_ga = s_dyn + 2 * cspan;
+  float* s_be = s_ga + cspan;
+  float* P0 = s_be + cspan;          // [TP*cspan]
+  float* P1 = P0 + m.TP * cspan;
   for (int g = g0 + t; g < g1; g += GN_BLOCK) {
-    s_s1[g] = 0.f;
-    s_s2[g] = 0.f;
     s_mu[g] = mean_in[(long)n * G + g];
     s_rs[g] = rstd_in[(long)n * G + g];
   }
-  for (int c = t; c < 2 * cspan; c += GN_BLOCK) s_dgb[c] = 0.f;
   for (int c = t; c < cspan; c += GN_BLOCK) {
     s_ga[c] = gamma[cbase + c];
     s_be[c] = beta[cbase + c];
@@ -326,65 +310,86 @@ gn_bwd_kernel(const GnSegs segs, const bf16* __restrict__ dz,
   __syncthreads();
 
   const bf16* db = dz + (long)n * HW * C;
+  const int c0 = (m.o0 + m.tc) << 3;
+  int cloc, cs;
+  const bf16* sb = seg_locate(segs, c0, cloc, cs);
+  const bf16* xb = sb + (long)n * HW * cs + cloc;
 
+  float ga[8], be[8], mu[8], rs[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    int c = c0 + j, g = c / Cg;
+    ga[j] = s_ga[c - cbase];
+    be[j] = s_be[c - cbase];
+    mu[j] = s_mu[g];
+    rs[j] = s_rs[g];
+  }
+
+  float a1[8] = {0}, a2[8] = {0}, adg[8] = {0}, adb[8] = {0};
   if (m.active) {
-    for (int oct = m.o0 + m.tc; oct < m.o1; oct += m.TCe) {
-      const int c0 = oct << 3;
-      int cloc, cs;
-      const bf16* sb = seg_locate(segs, c0, cloc, cs);
-      const bf16* xb = sb + (long)n * HW * cs + cloc;
-      float ga[8], be[8], mu[8], rs[8];
+    #pragma unroll 4
+    for (int p = m.tp; p < HW; p += m.TP) {
+      Bf16x8 xc = *reinterpret_cast<const Bf16x8*>(xb + (long)p * cs);
+      Bf16x8 dc = *reinterpret_cast<const Bf16x8*>(db + (long)p * C + c0);
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        int c = c0 + j, g = c / Cg;
-        ga[j] = s_ga[c - cbase];
-        be[j] = s_be[c - cbase];
-        mu[j] = s_mu[g];
-        rs[j] = s_rs[g];
-      }
-      float a1[8] = {0}, a2[8] = {0}, adg[8] = {0}, adb[8] = {0};
-      #pragma unroll 4
-      for (int p = m.tp; p < HW; p += m.TP) {
-        Bf16x8 xc = *reinterpret_cast<const Bf16x8*>(xb + (long)p * cs);
-        Bf16x8 dc = *reinterpret_cast<const Bf16x8*>(db + (long)p * C + c0);
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          float xhat = (bf2f(xc.v[j]) - mu[j]) * rs[j];
-          float dy = bf2f(dc.v[j]);
-          if (relu) {
-            float yv = xhat * ga[j] + be[j];
-            dy = yv > 0.f ? dy : 0.f;
-          }
-          a1[j] += ga[j] * dy;
-          a2[j] += ga[j] * dy * xhat;
-          adg[j] += dy * xhat;
-          adb[j] += dy;
+        float xhat = (bf2f(xc.v[j]) - mu[j]) * rs[j];
+        float dy = bf2f(dc.v[j]);
+        if (relu) {
+          float yv = xhat * ga[j] + be[j];
+          dy = yv > 0.f ? dy : 0.f;
         }
-      }
-      const bool shf = tp_shuffle_ok(m);
-      const int lane = t & 63;
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        float v1 = shf ? tp_fold(a1[j], m.TCe) : a1[j];
-        float v2 = shf ? tp_fold(a2[j], m.TCe) : a2[j];
-        float vg = shf ? tp_fold(adg[j], m.TCe) : adg[j];
-        float vb = shf ? tp_fold(adb[j], m.TCe) : adb[j];
-        if (!shf || lane < m.TCe) {
-          const int g = (c0 + j) / Cg;
-          atomicAdd(&s_s1[g], v1);
-          atomicAdd(&s_s2[g], v2);
-          atomicAdd(&s_dgb[c0 - cbase + j], vg);
-          atomicAdd(&s_dgb[cspan + c0 - cbase + j], vb);
-        }
+        a1[j] += ga[j] * dy;
+        a2[j] += ga[j] * dy * xhat;
+        adg[j] += dy * xhat;
+        adb[j] += dy;
       }
     }
+    // phase 1: group sums
+    const int pb = m.tp * cspan + (m.tc << 3);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      P0[pb + j] = a1[j];
+      P1[pb + j] = a2[j];
+    }
+  }
+  __syncthreads();
+  for (int g = g0 + t; g < g1; g += GN_BLOCK) {
+    float v1 = 0.f, v2 = 0.f;
+    const int cl0 = g * Cg - cbase;
+    for (int tp = 0; tp < m.TP; ++tp)
+      for (int c = cl0; c < cl0 + Cg; ++c) {
+        v1 += P0[tp * cspan + c];
+        v2 += P1[tp * cspan + c];
+      }
+    s_s1[g] = v1;
+    s_s2[g] = v2;
+  }
+  __syncthreads();
+
+  // phase 2: per-channel dgamma/dbeta partials (reuse P0/P1)
+  if (m.active) {
+    const int pb = m.tp * cspan + (m.tc << 3);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      P0[pb + j] = adg[j];
+      P1[pb + j] = adb[j];
+    }
+  }
+  __syncthreads();
+  for (int c = t; c < cspan; c += GN_BLOCK) {
+    float dg = 0.f, db2 = 0.f;
+    for (int tp = 0; tp < m.TP; ++tp) {
+      dg += P0[tp * cspan + c];
+      db2 += P1[tp * cspan + c];
+    }
+    s_dgb[c] = dg;
+    s_dgb[cspan + c] = db2;
   }
   __syncthreads();
 
   // publish per-(sample, channel) partials with plain stores — the
-  // deterministic column reduction (dlb_gn_dgb_reduce) follows; global
-  // atomics here cost a +10..30 us serialization tail (tools/gn_probe)
-  // and made dgamma order-dependent across runs.
+  // deterministic column reduction (reduce.hip colsum) follows
   for (int c = t; c < cspan; c += GN_BLOCK) {
     float* row = dgb_part + (long)n * 2 * C;
     row[cbase + c] = s_dgb[c];
@@ -394,47 +399,40 @@ gn_bwd_kernel(const GnSegs segs, const bf16* __restrict__ dz,
   if (!m.active) return;
 
   const float inv_m = 1.0f / ((float)HW * Cg);
-  for (int oct = m.o0 + m.tc; oct < m.o1; oct += m.TCe) {
-    const int c0 = oct << 3;
-    int cloc, cs;
-    const bf16* sb = seg_locate(segs, c0, cloc, cs);
+  bf16* dxb;
+  {
     int si = 0;
     while (si + 1 < segs.nseg && c0 >= segs.start[si + 1]) ++si;
-    const bf16* xb = sb + (long)n * HW * cs + cloc;
-    bf16* dxb = dxs.p[si] + (long)n * HW * cs + cloc;
-    float ga[8], be[8], mu[8], rs[8], k1[8], k2[8];
+    dxb = dxs.p[si] + (long)n * HW * cs + cloc;
+  }
+  float k1[8], k2[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    int g = (c0 + j) / Cg;
+    k1[j] = s_s1[g] * inv_m;
+    k2[j] = s_s2[g] * inv_m;
+  }
+  #pragma unroll 4
+  for (int p = m.tp; p < HW; p += m.TP) {
+    Bf16x8 xc = *reinterpret_cast<const Bf16x8*>(xb + (long)p * cs);
+    Bf16x8 dc = *reinterpret_cast<const Bf16x8*>(db + (long)p * C + c0);
+    Bf16x8 out;
+    Bf16x8 prev;
+    if (accumulate)
+      prev = *reinterpret_cast<const Bf16x8*>(dxb + (long)p * cs);
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      int c = c0 + j, g = c / Cg;
-      ga[j] = s_ga[c - cbase];
-      be[j] = s_be[c - cbase];
-      mu[j] = s_mu[g];
-      rs[j] = s_rs[g];
-      k1[j] = s_s1[g] * inv_m;
-      k2[j] = s_s2[g] * inv_m;
-    }
-    #pragma unroll 4
-      for (int p = m.tp; p < HW; p += m.TP) {
-      Bf16x8 xc = *reinterpret_cast<const Bf16x8*>(xb + (long)p * cs);
-      Bf16x8 dc = *reinterpret_cast<const Bf16x8*>(db + (long)p * C + c0);
-      Bf16x8 out;
-      Bf16x8 prev;
-      if (accumulate)
-        prev = *reinterpret_cast<const Bf16x8*>(dxb + (long)p * cs);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        float xhat = (bf2f(xc.v[j]) - mu[j]) * rs[j];
-        float dy = bf2f(dc.v[j]);
-        if (relu) {
-          float yv = xhat * ga[j] + be[j];
-          dy = yv > 0.f ? dy : 0.f;
-        }
-        float v = rs[j] * (ga[j] * dy - (k1[j] + xhat * k2[j]));
-        if (accumulate) v += bf2f(prev.v[j]);
-        out.v[j] = f2bf(v);
+      float xhat = (bf2f(xc.v[j]) - mu[j]) * rs[j];
+      float dy = bf2f(dc.v[j]);
+      if (relu) {
+        float yv = xhat * ga[j] + be[j];
+        dy = yv > 0.f ? dy : 0.f;
       }
-      *reinterpret_cast<Bf16x8*>(dxb + (long)p * cs) = out;
+      float v = rs[j] * (ga[j] * dy - (k1[j] + xhat * k2[j]));
+      if (accumulate) v += bf2f(prev.v[j]);
+      out.v[j] = f2bf(v);
     }
+    *reinterpret_cast<Bf16x8*>(dxb + (long)p * cs) = out;
   }
 }
 
@@ -442,9 +440,8 @@ gn_bwd_kernel(const GnSegs segs, const bf16* __restrict__ dz,
 // Block target for the (N, chunks) grid.  512 measured best at the
 // flagship N=512 (nchunks=1): every grid size beyond ~512 blocks pays a
 // fixed ~10-25 us per 512-block round that chunk-splitting cannot buy
-// back (tools/gn_probe; the LDS-atomic contention part of it is folded
-// away by tp_fold, the remainder is per-block lifetime).  Chunking still
-// engages for small per-rank batches (N < 512), where it fills CUs.
+// back (tools/gn_probe).  Chunking still engages for small per-rank
+// batches (N < 512), where it fills CUs.
 static int gn_target_blocks(int bwd) {
   static int cached[2] = {-1, -1};
   if (cached[bwd] < 0) {
@@ -457,8 +454,9 @@ static int gn_target_blocks(int bwd) {
 
 static int gcd_i(int a, int b) { while (b) { int t = a % b; a = b; b = t; } return a; }
 
-// Chunk size in octets: group-aligned (a chunk covers whole groups) and
-// sized so N * nchunks ~ the block target.  Returns (chunk_oct, nchunks).
+// Chunk size in octets: group-aligned (a chunk covers whole groups),
+// <= GN_MAXCHUNK (single octet per thread), sized so N * nchunks ~ the
+// block target.  Returns (chunk_oct, nchunks).
 static void gn_chunking(int N, int C, int G, int bwd, int* chunk_oct,
                         int* nchunks) {
   const int TC = C >> 3;
@@ -469,8 +467,17 @@ static void gn_chunking(int N, int C, int G, int bwd, int* chunk_oct,
   if (want < 1) want = 1;
   if (want > units) want = units;
   int cu = (units + want - 1) / want;    // units per chunk
+  while (cu * u > GN_MAXCHUNK && cu > 1) --cu;
   *chunk_oct = cu * u;
   *nchunks = (TC + *chunk_oct - 1) / *chunk_oct;
+}
+
+static size_t gn_dyn_shmem(int chunk_oct, int fixed_cspans) {
+  // P0/P1 hold [TP][cspan] partials; TP*cspan = (GN_BLOCK/TCe)*TCe*8
+  // <= GN_BLOCK*8 = 2048 floats each — and a SMALLER last chunk has a
+  // LARGER TP, so size them at the bound, not at this chunk's TP.
+  const int cspan = chunk_oct * 8;
+  return (size_t)(fixed_cspans * cspan + 2 * (GN_BLOCK * 8)) * sizeof(float);
 }
 
 extern "C" void dlb_gn_fwd_segs(const void* const* xs, const int* starts,
@@ -487,10 +494,10 @@ extern "C" void dlb_gn_fwd_segs(const void* const* xs, const int* starts,
   sg.start[nseg] = starts[nseg];
   int chunk_oct, nchunks;
   gn_chunking(N, C, G, 0, &chunk_oct, &nchunks);
-  const size_t shmem = 2 * (size_t)(chunk_oct * 8) * sizeof(float);
-  hipLaunchKernelGGL(gn_fwd_kernel, dim3(N, nchunks), dim3(GN_BLOCK), shmem,
-                     stream, sg, (bf16*)y, gamma, beta, mean, rstd, HW, C,
-                     G, eps, relu, chunk_oct);
+  hipLaunchKernelGGL(gn_fwd_kernel, dim3(N, nchunks), dim3(GN_BLOCK),
+                     gn_dyn_shmem(chunk_oct, 2), stream, sg, (bf16*)y,
+                     gamma, beta, mean, rstd, HW, C, G, eps, relu,
+                     chunk_oct);
 }
 
 extern "C" void dlb_gn_bwd_segs(const void* const* xs, const int* starts,
@@ -511,11 +518,10 @@ extern "C" void dlb_gn_bwd_segs(const void* const* xs, const int* starts,
   sg.start[nseg] = starts[nseg];
   int chunk_oct, nchunks;
   gn_chunking(N, C, G, 1, &chunk_oct, &nchunks);
-  size_t shmem = 4 * (size_t)(chunk_oct * 8) * sizeof(float);
-  hipLaunchKernelGGL(gn_bwd_kernel, dim3(N, nchunks), dim3(GN_BLOCK), shmem,
-                     stream, sg, (const bf16*)dz, dsg, gamma, beta,
-                     mean, rstd, dgb_part, HW, C, G, relu, accumulate,
-                     chunk_oct);
+  hipLaunchKernelGGL(gn_bwd_kernel, dim3(N, nchunks), dim3(GN_BLOCK),
+                     gn_dyn_shmem(chunk_oct, 4), stream, sg,
+                     (const bf16*)dz, dsg, gamma, beta, mean, rstd,
+                     dgb_part, HW, C, G, relu, accumulate, chunk_oct);
 }
 
 extern "C" void dlb_gn_stats_segs(const void* const* xs, const int* starts,
@@ -531,6 +537,7 @@ extern "C" void dlb_gn_stats_segs(const void* const* xs, const int* starts,
   sg.start[nseg] = starts[nseg];
   int chunk_oct, nchunks;
   gn_chunking(N, C, G, 0, &chunk_oct, &nchunks);
-  hipLaunchKernelGGL(gn_stats_kernel, dim3(N, nchunks), dim3(GN_BLOCK), 0,
-                     stream, sg, mean, rstd, HW, C, G, eps, chunk_oct);
+  hipLaunchKernelGGL(gn_stats_kernel, dim3(N, nchunks), dim3(GN_BLOCK),
+                     gn_dyn_shmem(chunk_oct, 0), stream, sg, mean, rstd,
+                     HW, C, G, eps, chunk_oct);
 }
