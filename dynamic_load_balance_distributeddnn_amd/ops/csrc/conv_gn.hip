@@ -295,8 +295,10 @@ gnconv1x1_fwd_kernel(const GnCParams p) {
   }
 
   constexpr int LDO = BN + 8;
-  static_assert(BM * LDA >= 32 * LDO, "epilogue chunk must fit a_lds[0]");
-  bf16* o_lds = a_lds[0];
+  // bounce buffer: a_lds when it fits 32 output rows, else b_lds
+  static_assert(BM * LDA >= 32 * LDO || BN * LDB >= 32 * LDO,
+                "epilogue chunk must fit an LDS buffer");
+  bf16* o_lds = (BM * LDA >= 32 * LDO) ? a_lds[0] : b_lds[0];
 #pragma unroll
   for (int ch = 0; ch < BM / 32; ++ch) {
     __syncthreads();
@@ -519,6 +521,14 @@ extern "C" void dlb_gnconv1x1_fwd(const void* const* xs, const int* starts,
            relu);
   p.w = (const bf16*)w;
   p.y = (bf16*)y;
+  // Co=128 layers give grid.y=1; when the 128-row tiling can't fill the
+  // 256 CUs (small-HW DenseNet stages), halve BM to double the blocks.
+  if ((long)cdiv(p.M, 128) * cdiv(Co, 128) < 400) {
+    dim3 grid(cdiv(p.M, 64), cdiv(Co, 128));
+    hipLaunchKernelGGL((gnconv1x1_fwd_kernel<64, 128, 1, 4>), grid,
+                       dim3(CONV_BLOCK), 0, stream, p);
+    return;
+  }
   dim3 grid(cdiv(p.M, 128), cdiv(Co, 128));
   hipLaunchKernelGGL((gnconv1x1_fwd_kernel<128, 128, 2, 2>), grid,
                      dim3(CONV_BLOCK), 0, stream, p);
