@@ -142,8 +142,10 @@ def lm_loss(h, weight, bias, targets):
     backward recomputes the tiles for dh/dW/db.  CPU path is the plain
     composition in fp32.
     """
+    import os
     if (_use_native("lm_loss", h) and h.dtype == torch.bfloat16
-            and h.shape[-1] % 8 == 0 and h.shape[-1] <= 224):
+            and h.shape[-1] % 8 == 0 and h.shape[-1] <= 224
+            and not os.environ.get("DLB_NO_FUSED_LMLOSS")):
         from . import native
         return native.lm_loss(h, weight, bias, targets)
     logits = F.linear(h, weight, bias).reshape(-1, weight.shape[0])
