@@ -79,6 +79,41 @@ __device__ inline void lm_stage(const bf16* __restrict__ src, long row0,
   }
 }
 
+// Register-staged variant: the serialized stage->barrier->MFMA loop
+// measured ~3 ms per dispatch (5x the traffic bound) — prefetching the
+// NEXT tile's 7 bf16x8 loads per thread while the current tile's MFMAs
+// run hides the L2/L3 latency (conv.hip's pipeline idiom).
+// 64 rows * (LM_DPAD/8) chunks = 7 * LM_BLOCK exactly.
+#define LM_SREG (LM_BM * (LM_DPAD / 8) / LM_BLOCK)
+struct LmStage {
+  bf16x8_t v[LM_SREG];
+};
+
+__device__ inline void lm_sload(const bf16* __restrict__ src, long row0,
+                                long rmax, int d, LmStage& s) {
+  const int cpr = LM_DPAD / 8;
+#pragma unroll
+  for (int u = 0; u < LM_SREG; ++u) {
+    const int c = threadIdx.x + u * LM_BLOCK;
+    const int r = c / cpr;
+    const int k8 = (c % cpr) * 8;
+    s.v[u] = lm_zero8();
+    if (row0 + r < rmax && k8 < d)
+      s.v[u] = *reinterpret_cast<const bf16x8_t*>(src + (row0 + r) * d + k8);
+  }
+}
+
+__device__ inline void lm_swrite(const LmStage& s, bf16* dst) {
+  const int cpr = LM_DPAD / 8;
+#pragma unroll
+  for (int u = 0; u < LM_SREG; ++u) {
+    const int c = threadIdx.x + u * LM_BLOCK;
+    const int r = c / cpr;
+    const int k8 = (c % cpr) * 8;
+    *reinterpret_cast<bf16x8_t*>(dst + r * LM_LDH + k8) = s.v[u];
+  }
+}
+
 // z tile [64 t-rows x 64 v-cols] via MFMA from staged h and W images.
 // Wave w computes rows [w*16, w*16+16).  acc[j] covers cols [j*16,+16).
 __device__ inline void lm_ztile(const bf16* h_lds, const bf16* w_lds,
@@ -133,11 +168,16 @@ lmloss_fwd_kernel(const bf16* __restrict__ h, const bf16* __restrict__ w,
   for (int r = 0; r < 4; ++r)
     my_tgt[r] = (m0 + rbase + r < T) ? tgt[m0 + rbase + r] : -1;
 
+  LmStage ws;
+  if (v_tile0 < v_tile1) lm_sload(w, (long)v_tile0 * LM_BN, V, d, ws);
+  __syncthreads();
+  if (v_tile0 < v_tile1) lm_swrite(ws, w_lds);
+  __syncthreads();
+
   for (int vt = v_tile0; vt < v_tile1; ++vt) {
     const int n0 = vt * LM_BN;
-    __syncthreads();
-    lm_stage(w, n0, V, d, w_lds, LM_BN);
-    __syncthreads();
+    if (vt + 1 < v_tile1)  // prefetch next W tile during the MFMAs
+      lm_sload(w, (long)(vt + 1) * LM_BN, V, d, ws);
     f32x4 acc[4];
     lm_ztile(h_lds, w_lds, acc, lane, wave);
 #pragma unroll
@@ -160,6 +200,9 @@ lmloss_fwd_kernel(const bf16* __restrict__ h, const bf16* __restrict__ w,
         s_run[r] = s;
       }
     }
+    __syncthreads();          // all MFMA reads of w_lds complete
+    if (vt + 1 < v_tile1) lm_swrite(ws, w_lds);
+    __syncthreads();
   }
 
   // merge the 16 lanes of each (lane>>4) group: cols -> one (m, s)
@@ -273,11 +316,16 @@ lmloss_bwd_dh_kernel(const bf16* __restrict__ h, const bf16* __restrict__ w,
   for (int j = 0; j < LM_DPAD / 16; ++j) dacc[j] = {0.f, 0.f, 0.f, 0.f};
   float db_dummy[4] = {0.f, 0.f, 0.f, 0.f};
 
+  LmStage ws;
+  if (v_tile0 < v_tile1) lm_sload(w, (long)v_tile0 * LM_BN, V, d, ws);
+  __syncthreads();
+  if (v_tile0 < v_tile1) lm_swrite(ws, w_lds);
+  __syncthreads();
+
   for (int vt = v_tile0; vt < v_tile1; ++vt) {
     const int n0 = vt * LM_BN;
-    __syncthreads();
-    lm_stage(w, n0, V, d, w_lds, LM_BN);
-    __syncthreads();
+    if (vt + 1 < v_tile1)  // prefetch next W tile during this tile's math
+      lm_sload(w, (long)(vt + 1) * LM_BN, V, d, ws);
     f32x4 acc[4];
     lm_ztile(h_lds, w_lds, acc, lane, wave);
     lm_dp_tile(acc, bias, tgt, lse, m0, n0, T, V, scale, dp_lds, lane, wave,
@@ -296,6 +344,9 @@ lmloss_bwd_dh_kernel(const bf16* __restrict__ h, const bf16* __restrict__ w,
                                                           dacc[j], 0, 0, 0);
       }
     }
+    __syncthreads();          // w_lds reads complete before overwrite
+    if (vt + 1 < v_tile1) lm_swrite(ws, w_lds);
+    __syncthreads();
   }
 
   const int rbase = wave * 16 + (lane >> 4) * 4;
@@ -343,10 +394,15 @@ lmloss_bwd_dw_kernel(const bf16* __restrict__ h, const bf16* __restrict__ w,
   for (int j = 0; j < LM_DPAD / 16; ++j) wacc[j] = {0.f, 0.f, 0.f, 0.f};
   float db_part[4] = {0.f, 0.f, 0.f, 0.f};
 
+  LmStage hs;
+  lm_sload(h, 0, T, d, hs);
+  __syncthreads();
+  lm_swrite(hs, h_lds);
+  __syncthreads();
+
   for (long m0 = 0; m0 < T; m0 += LM_BM) {
-    __syncthreads();
-    lm_stage(h, m0, T, d, h_lds, LM_BM);
-    __syncthreads();
+    if (m0 + LM_BM < T)  // prefetch next h chunk during this chunk's math
+      lm_sload(h, m0 + LM_BM, T, d, hs);
     f32x4 acc[4];
     lm_ztile(h_lds, w_lds, acc, lane, wave);
     lm_dp_tile(acc, bias, tgt, lse, m0, n0, T, V, scale, dp_lds, lane, wave,
@@ -364,6 +420,9 @@ lmloss_bwd_dw_kernel(const bf16* __restrict__ h, const bf16* __restrict__ w,
                                                           wacc[j], 0, 0, 0);
       }
     }
+    __syncthreads();          // h_lds reads complete before overwrite
+    if (m0 + LM_BM < T) lm_swrite(hs, h_lds);
+    __syncthreads();
   }
 
   // db: this lane's partials cover cols j*16+(lane&15) summed over its
