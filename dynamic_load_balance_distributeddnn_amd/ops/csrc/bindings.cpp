@@ -105,7 +105,8 @@ static std::vector<torch::Tensor> gn_bwd(std::vector<torch::Tensor> xs,
                                          c10::optional<torch::Tensor>
                                              dgamma_out = c10::nullopt,
                                          c10::optional<torch::Tensor>
-                                             dbeta_out = c10::nullopt) {
+                                             dbeta_out = c10::nullopt,
+                                         bool dgb_defer = false) {
   // dx_accum: preallocated per-segment grad buffers — the kernel ADDS
   // into them (the dense-stream manual backward), instead of allocating
   // fresh outputs for autograd to sum pairwise.
@@ -132,6 +133,24 @@ static std::vector<torch::Tensor> gn_bwd(std::vector<torch::Tensor> xs,
   starts[xs.size()] = C;
   const int N = xs[0].size(0), HW = xs[0].size(1);
   TORCH_CHECK(dz.is_contiguous() && dz.size(2) == C);
+  auto stream = at::hip::getCurrentHIPStream();
+  // dgamma/dbeta path: per-sample partial rows [N, 2C] (plain stores in
+  // the kernel) + one deterministic column-sum kernel — replaces the
+  // contended global-atomic publish (see reduce.hip).  With dgb_defer
+  // the caller batches many layers' reductions into one launch
+  // (gn_dgb_reduce_multi) and gets the raw partial rows back instead.
+  auto part = torch::empty({N, 2 * C},
+                           xs[0].options().dtype(torch::kFloat32));
+  dlb_gn_bwd_segs(ptrs, starts, (int)xs.size(), dz.data_ptr(), dptrs,
+                  gamma.data_ptr<float>(), beta.data_ptr<float>(),
+                  mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                  part.data_ptr<float>(),
+                  N, HW, C, (int)groups, relu ? 1 : 0, acc ? 1 : 0,
+                  stream.stream());
+  if (dgb_defer) {
+    out.push_back(part);
+    return out;  // [dx_0..dx_{k-1}, part]
+  }
   torch::Tensor dgamma, dbeta;
   if (dgamma_out.has_value()) {
     TORCH_CHECK(dgamma_out->is_contiguous() && dgamma_out->numel() == C &&
@@ -146,24 +165,44 @@ static std::vector<torch::Tensor> gn_bwd(std::vector<torch::Tensor> xs,
     dgamma = gbuf[0];
     dbeta = gbuf[1];
   }
-  auto stream = at::hip::getCurrentHIPStream();
-  // dgamma/dbeta path: per-sample partial rows [N, 2C] (plain stores in
-  // the kernel) + one deterministic column-sum kernel — replaces the
-  // contended global-atomic publish (see reduce.hip).
-  auto part = torch::empty({N, 2 * C},
-                           xs[0].options().dtype(torch::kFloat32));
-  dlb_gn_bwd_segs(ptrs, starts, (int)xs.size(), dz.data_ptr(), dptrs,
-                  gamma.data_ptr<float>(), beta.data_ptr<float>(),
-                  mean.data_ptr<float>(), rstd.data_ptr<float>(),
-                  part.data_ptr<float>(),
-                  N, HW, C, (int)groups, relu ? 1 : 0, acc ? 1 : 0,
-                  stream.stream());
   dlb_gn_dgb_reduce(part.data_ptr<float>(), N, C,
                     dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),
                     stream.stream());
   out.push_back(dgamma);
   out.push_back(dbeta);
   return out;  // [dx_0..dx_{k-1}, dgamma, dbeta]
+}
+
+extern "C" void dlb_gn_dgb_reduce_multi(const void* const* parts,
+                                        void* const* dgs, void* const* dbs,
+                                        const int* Cs, int nl, int N,
+                                        hipStream_t stream);
+
+// Batched dgamma/dbeta reduction over many layers' deferred partials
+// (one launch; += into the pre-zeroed arena grad views).
+static void gn_dgb_reduce_multi(std::vector<torch::Tensor> parts,
+                                std::vector<torch::Tensor> dgs,
+                                std::vector<torch::Tensor> dbs) {
+  const int nl = (int)parts.size();
+  TORCH_CHECK(nl >= 1 && nl <= 52 && dgs.size() == parts.size() &&
+              dbs.size() == parts.size());
+  const void* pp[52];
+  void* pg[52];
+  void* pb[52];
+  int cs[52];
+  const int N = (int)parts[0].size(0);
+  for (int i = 0; i < nl; ++i) {
+    TORCH_CHECK(parts[i].is_contiguous() && parts[i].size(0) == N);
+    const int C = (int)(parts[i].size(1) / 2);
+    TORCH_CHECK(dgs[i].is_contiguous() && dgs[i].numel() == C &&
+                dbs[i].is_contiguous() && dbs[i].numel() == C);
+    pp[i] = parts[i].data_ptr();
+    pg[i] = dgs[i].data_ptr();
+    pb[i] = dbs[i].data_ptr();
+    cs[i] = C;
+  }
+  dlb_gn_dgb_reduce_multi(pp, pg, pb, cs, nl, N,
+                          at::hip::getCurrentHIPStream().stream());
 }
 
 // Stats-only GroupNorm over the virtual concat (mean/rstd for the fused
@@ -756,7 +795,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("xs"), py::arg("dz"), py::arg("gamma"), py::arg("beta"),
         py::arg("mean"), py::arg("rstd"), py::arg("groups"), py::arg("relu"),
         py::arg("dx_accum") = py::none(),
-        py::arg("dgamma_out") = py::none(), py::arg("dbeta_out") = py::none());
+        py::arg("dgamma_out") = py::none(), py::arg("dbeta_out") = py::none(),
+        py::arg("dgb_defer") = false);
+  m.def("gn_dgb_reduce_multi", &gn_dgb_reduce_multi,
+        "Batched deterministic dgamma/dbeta reduction (one launch)");
   m.def("gn_stats", &gn_stats, "Stats-only GroupNorm over virtual concat");
   m.def("gn_conv1x1_fwd", &gn_conv1x1_fwd,
         "Fused GroupNorm(+ReLU) -> 1x1 conv forward (stream never packed)");

@@ -80,3 +80,83 @@ extern "C" void dlb_gn_dgb_reduce(const float* part, int N, int C,
   hipLaunchKernelGGL(colsum_kernel, dim3(grid), dim3(CS_COLS * CS_ROWG), 0,
                      stream, part, N, len2, dgamma, dbeta, (long)C, 1);
 }
+
+// --------------------- batched multi-layer variant ----------------------
+// A dense-block backward produces up to ~50 per-layer dgamma/dbeta
+// partial buffers; reducing each with its own launch costs ~4-6 us of
+// latency+boundary apiece.  This folds a whole block's reductions into
+// ONE launch: a descriptor table maps block ranges to layers.
+#define CSM_MAX 52
+struct CsmDesc {
+  const float* part[CSM_MAX];
+  float* dg[CSM_MAX];
+  float* db[CSM_MAX];
+  int C[CSM_MAX];
+  int blk0[CSM_MAX + 1];
+  int nl;
+  int N;
+};
+
+extern "C" __global__ void __launch_bounds__(CS_COLS* CS_ROWG)
+colsum_multi_kernel(const CsmDesc d) {
+  int li = 0;
+  while (li + 1 < d.nl && (int)blockIdx.x >= d.blk0[li + 1]) ++li;
+  const int lb = blockIdx.x - d.blk0[li];
+  const long Cl = d.C[li];
+  const long len2 = 2 * Cl;
+  const int tc = threadIdx.x % CS_COLS;
+  const int tr = threadIdx.x / CS_COLS;
+  const long c = (long)lb * CS_COLS + tc;
+  __shared__ float fold[CS_ROWG][CS_COLS];
+
+  float a0 = 0.f, a1 = 0.f, a2 = 0.f, a3 = 0.f;
+  if (c < len2) {
+    const int gs = (d.N + CS_ROWG - 1) / CS_ROWG;
+    const int s0 = tr * gs;
+    const int s1 = min(d.N, s0 + gs);
+    const float* p = d.part[li] + c;
+    int s = s0;
+    for (; s + 4 <= s1; s += 4) {
+      a0 += p[(long)s * len2];
+      a1 += p[(long)(s + 1) * len2];
+      a2 += p[(long)(s + 2) * len2];
+      a3 += p[(long)(s + 3) * len2];
+    }
+    for (; s < s1; ++s) a0 += p[(long)s * len2];
+  }
+  fold[tr][tc] = (a0 + a1) + (a2 + a3);
+  __syncthreads();
+#pragma unroll
+  for (int off = CS_ROWG / 2; off > 0; off >>= 1) {
+    if (tr < off) fold[tr][tc] += fold[tr + off][tc];
+    __syncthreads();
+  }
+  if (tr == 0 && c < len2) {
+    float v = fold[0][tc];
+    if (c < Cl)
+      d.dg[li][c] += v;
+    else
+      d.db[li][c - Cl] += v;
+  }
+}
+
+extern "C" void dlb_gn_dgb_reduce_multi(const void* const* parts,
+                                        void* const* dgs, void* const* dbs,
+                                        const int* Cs, int nl, int N,
+                                        hipStream_t stream) {
+  CsmDesc d{};
+  d.nl = nl;
+  d.N = N;
+  int b = 0;
+  for (int i = 0; i < nl; ++i) {
+    d.part[i] = (const float*)parts[i];
+    d.dg[i] = (float*)dgs[i];
+    d.db[i] = (float*)dbs[i];
+    d.C[i] = Cs[i];
+    d.blk0[i] = b;
+    b += (int)((2L * Cs[i] + CS_COLS - 1) / CS_COLS);
+  }
+  d.blk0[nl] = b;
+  hipLaunchKernelGGL(colsum_multi_kernel, dim3(b), dim3(CS_COLS * CS_ROWG),
+                     0, stream, d);
+}

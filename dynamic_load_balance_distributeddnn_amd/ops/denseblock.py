@@ -117,10 +117,26 @@ class _DenseBlockFn(torch.autograd.Function):
         sink = getattr(params[0], "_dlb_sink", None)
         direct = sink is not None and params[0].grad is not None
 
-        def norm_outs(p_g, p_b):
+        # Direct mode defers every norm's dgamma/dbeta reduction: gn_bwd
+        # returns the raw [N, 2C] partial rows, and ONE batched colsum
+        # launch at the end of the block folds them all into the arena
+        # grad views (each per-layer launch measured ~4-6 us of
+        # latency/boundary; a DenseNet block has up to 49 of them).
+        dgb_batch = []  # (part, gamma_idx, beta_idx)
+
+        def norm_bwd(xs, dz3, gi, bi, mean, rstd, dx_accum=None):
+            """gn_bwd wrapper: returns the per-segment dx list."""
+            kw = dict(dx_accum=dx_accum) if dx_accum is not None else {}
             if direct:
-                return dict(dgamma_out=p_g.grad, dbeta_out=p_b.grad)
-            return {}
+                outs = ext().gn_bwd(xs, dz3, params[gi], params[bi], mean,
+                                    rstd, groups, True, dgb_defer=True, **kw)
+                dgb_batch.append((outs[-1], gi, bi))
+                return outs[:-1]
+            outs = ext().gn_bwd(xs, dz3, params[gi], params[bi], mean,
+                                rstd, groups, True, **kw)
+            pgrads[gi] = outs[-2]
+            pgrads[bi] = outs[-1]
+            return outs[:-2]
 
         def put(idx, grad):
             # grad already landed in the arena on the direct path
@@ -154,13 +170,8 @@ class _DenseBlockFn(torch.autograd.Function):
                 dwt.view(co, 1, 1, ci).permute(0, 3, 1, 2))
             # the transition norm is every segment's LAST consumer: its
             # backward writes the per-segment grad buffers fresh
-            outs = ext().gn_bwd(segs3, _to_nhwc3(dyt), gt, bt, mt, rt,
-                                groups, True,
-                                **norm_outs(params[6 * nlayers],
-                                            params[6 * nlayers + 1]))
-            dsegs = list(outs[:-2])
-            put(6 * nlayers, outs[-2])
-            put(6 * nlayers + 1, outs[-1])
+            dsegs = list(norm_bwd(segs3, _to_nhwc3(dyt), 6 * nlayers,
+                                  6 * nlayers + 1, mt, rt))
         else:
             # external grads arrive per segment (one consumer each);
             # clone into owned buffers the kernels then accumulate into
@@ -174,7 +185,7 @@ class _DenseBlockFn(torch.autograd.Function):
                 y1, m1, r1, w1c, h13, m2, r2, y2, w2c = \
                     saves[LW * li:LW * li + LW]
             g1, b1 = params[6 * li], params[6 * li + 1]
-            g2, b2 = params[6 * li + 3], params[6 * li + 4]
+
             in_segs = segs3[nlayers - li:]
             dfresh4 = _as4(dsegs[nlayers - 1 - li], n, h, w)
             y24 = _as4(y2, n, h, w)
@@ -186,13 +197,8 @@ class _DenseBlockFn(torch.autograd.Function):
             dw2 = ext().conv_wrw(y24, dfresh4, 3, 3, 1, 1, **w2_out)
             put(6 * li + 5, None if direct else
                 dw2.view(co2, 3, 3, ci2).permute(0, 3, 1, 2))
-            outs2 = ext().gn_bwd([h13], _to_nhwc3(dy2), g2, b2,
-                                 m2, r2, groups, True,
-                                 **norm_outs(params[6 * li + 3],
-                                             params[6 * li + 4]))
-            dh1, dg2, db2 = outs2
-            put(6 * li + 3, dg2)
-            put(6 * li + 4, db2)
+            (dh1,) = norm_bwd([h13], _to_nhwc3(dy2), 6 * li + 3,
+                              6 * li + 4, m2, r2)
             dh14 = _as4(dh1, n, h, w)
             dy1 = _conv_bwd_data(dh14, w1c, h, w, 1, 0)
             co1, ci1 = w1c.shape[0], w1c.shape[1]
@@ -209,13 +215,21 @@ class _DenseBlockFn(torch.autograd.Function):
                                      **w1_out)
             put(6 * li + 2, None if direct else
                 dw1.view(co1, 1, 1, ci1).permute(0, 3, 1, 2))
-            outs = ext().gn_bwd(in_segs, _to_nhwc3(dy1), g1, b1, m1, r1,
-                                groups, True,
-                                dx_accum=dsegs[nlayers - li:],
-                                **norm_outs(params[6 * li],
-                                            params[6 * li + 1]))
-            put(6 * li, outs[-2])
-            put(6 * li + 1, outs[-1])
+            norm_bwd(in_segs, _to_nhwc3(dy1), 6 * li, 6 * li + 1, m1, r1,
+                     dx_accum=dsegs[nlayers - li:])
+
+        if direct and dgb_batch:
+            # one launch reduces every deferred dgamma/dbeta; only then
+            # are the norm params bucket-ready
+            for s in range(0, len(dgb_batch), 52):
+                chunk = dgb_batch[s:s + 52]
+                ext().gn_dgb_reduce_multi(
+                    [p for p, _, _ in chunk],
+                    [params[gi].grad for _, gi, _ in chunk],
+                    [params[bi].grad for _, _, bi in chunk])
+            for _, gi, bi in dgb_batch:
+                sink.mark_ready(params[gi])
+                sink.mark_ready(params[bi])
 
         dseg0 = _as4(dsegs[-1], n, h, w)
         return (None, dseg0, *pgrads)

@@ -15,6 +15,9 @@ import torch
 
 pytestmark = pytest.mark.gpu
 
+needs_gpu = pytest.mark.skipif(not torch.cuda.is_available(),
+                               reason="needs ROCm GPU")
+
 
 def _grads(model):
     return {n: p.grad.clone() for n, p in model.named_parameters()}
@@ -149,3 +152,49 @@ def test_block_fn_engaged_on_flagship():
     finally:
         db.dense_block_forward = orig
     assert calls == [6, 12, 24, 16]
+
+
+@needs_gpu
+def test_direct_arena_grads_match_module_path():
+    """FlatSGD direct mode (arena grad views + DEFERRED batched
+    dgamma/dbeta reduction) must produce the same gradients as the
+    plain autograd/module path."""
+    import torch.nn.functional as F
+
+    from dynamic_load_balance_distributeddnn_amd.models import DenseNet121
+    from dynamic_load_balance_distributeddnn_amd.parallel import \
+        GradientSynchronizer
+    from dynamic_load_balance_distributeddnn_amd.parallel.optim import FlatSGD
+
+    torch.manual_seed(4)
+    x = torch.randn(16, 3, 32, 32, device="cuda") \
+        .to(memory_format=torch.channels_last)
+    y = torch.randint(0, 10, (16,), device="cuda")
+
+    def grads(direct):
+        torch.manual_seed(11)
+        model = DenseNet121(10).cuda().to(memory_format=torch.channels_last)
+        if direct:
+            sync = GradientSynchronizer(model)
+            opt = FlatSGD(sync, lr=0.0)
+            sync.zero()
+        else:
+            for p in model.parameters():
+                p.grad = None
+        with torch.autocast("cuda", dtype=torch.bfloat16):
+            loss = F.cross_entropy(model(x), y)
+        loss.backward()
+        return [p.grad.detach().float().clone()
+                for p in model.parameters()], loss.item()
+
+    gd, ld = grads(True)
+    gm, lm = grads(False)
+    assert abs(ld - lm) < 1e-2 * max(1.0, abs(lm))
+    bad = 0
+    for a, b in zip(gd, gm):
+        ref = b.norm().item()
+        if ref < 1e-8:
+            continue
+        if (a - b).norm().item() / ref > 2e-2:
+            bad += 1
+    assert bad == 0
