@@ -174,6 +174,7 @@ def test_direct_arena_grads_match_module_path():
     def grads(direct):
         torch.manual_seed(11)
         model = DenseNet121(10).cuda().to(memory_format=torch.channels_last)
+        sync = None
         if direct:
             sync = GradientSynchronizer(model)
             opt = FlatSGD(sync, lr=0.0)
@@ -184,8 +185,11 @@ def test_direct_arena_grads_match_module_path():
         with torch.autocast("cuda", dtype=torch.bfloat16):
             loss = F.cross_entropy(model(x), y)
         loss.backward()
-        return [p.grad.detach().float().clone()
-                for p in model.parameters()], loss.item()
+        out = [p.grad.detach().float().clone()
+               for p in model.parameters()], loss.item()
+        if sync is not None:
+            sync.detach()  # break the param->sink reference cycle
+        return out
 
     gd, ld = grads(True)
     gm, lm = grads(False)

@@ -68,6 +68,8 @@ def test_fusion_cuts_peak_memory():
         else:
             os.environ.pop("DLB_NO_BLOCK_FN", None)
         try:
+            import gc
+            gc.collect()  # free other tests' leftovers (arena cycles)
             torch.cuda.empty_cache()
             torch.cuda.reset_peak_memory_stats()
             torch.manual_seed(0)
