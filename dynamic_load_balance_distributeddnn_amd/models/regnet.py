@@ -39,7 +39,7 @@ class _SqueezeExcite(nn.Module):
     def forward(self, x):
         s = FD.adaptive_avg_pool1(x)
         s = self.expand(F.relu(self.reduce(s)))
-        return x * s.sigmoid()
+        return FD.se_mul(x, s)  # fused sigmoid-mul
 
 
 class _Block(nn.Module):

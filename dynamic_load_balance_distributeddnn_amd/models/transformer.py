@@ -58,10 +58,10 @@ class _EncoderLayer(nn.Module):
         attn = FD.causal_attention(q, k, v, self.nhead,
                                    self.dropout_p, self.training)
         attn = self.out_proj(attn)
-        x = self.norm1(x + F.dropout(attn, self.dropout_p, self.training))
-        ff = self.ffn2(F.dropout(F.relu(self.ffn1(x)),
-                                 self.dropout_p, self.training))
-        return self.norm2(x + F.dropout(ff, self.dropout_p, self.training))
+        x = self.norm1(x + FD.dropout(attn, self.dropout_p, self.training))
+        ff = self.ffn2(FD.dropout(F.relu(self.ffn1(x)),
+                                  self.dropout_p, self.training))
+        return self.norm2(x + FD.dropout(ff, self.dropout_p, self.training))
 
 
 class TransformerModel(nn.Module):
@@ -82,7 +82,8 @@ class TransformerModel(nn.Module):
     def forward_features(self, src):  # [S, B] int64 -> [S, B, d]
         """Encoder output BEFORE the decoder — the input of the fused
         LM loss head (ops.functional.lm_loss)."""
-        x = self.embed(src) * math.sqrt(self.d_model)
+        x = FD.embedding_scaled(src, self.embed.weight,
+                                math.sqrt(self.d_model))
         x = self.pos(x)
         if x.is_cuda and torch.is_autocast_enabled("cuda"):
             # keep the whole encoder in bf16 so residual adds don't

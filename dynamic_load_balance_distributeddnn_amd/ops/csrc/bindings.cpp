@@ -779,6 +779,81 @@ static torch::Tensor maxpool_bwd(torch::Tensor dy, torch::Tensor idx,
   return dx;
 }
 
+extern "C" void dlb_dropout(const void* x, void* y, long n, float pd,
+                            unsigned long long seed, hipStream_t stream);
+extern "C" void dlb_embed_fwd(const void* table, const int* idx, void* out,
+                              long T, int d, float scale,
+                              hipStream_t stream);
+extern "C" void dlb_embed_bwd(const void* dy, const int* idx, float* dtable,
+                              long T, int d, float scale,
+                              hipStream_t stream);
+extern "C" void dlb_se_fwd(const void* x, const void* g, void* y, long NHW,
+                           int HW, int C, hipStream_t stream);
+extern "C" void dlb_se_bwd(const void* x, const void* g, const void* dy,
+                           void* dx, float* dg, int N, int HW, int C,
+                           hipStream_t stream);
+
+// philox dropout; calling twice with the same seed applies the same
+// mask (the backward IS a forward on dy)
+static torch::Tensor dropout_op(torch::Tensor x, double pd, int64_t seed) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kBFloat16);
+  auto xc = x.contiguous();
+  auto y = torch::empty_like(xc);
+  dlb_dropout(xc.data_ptr(), y.data_ptr(), xc.numel(), (float)pd,
+              (unsigned long long)seed,
+              at::hip::getCurrentHIPStream().stream());
+  return y.view_as(x);
+}
+
+static torch::Tensor embed_fwd(torch::Tensor table, torch::Tensor idx,
+                               double scale) {
+  TORCH_CHECK(table.is_cuda() && table.is_contiguous() &&
+              table.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(idx.is_contiguous() && idx.scalar_type() == torch::kInt32);
+  const int d = (int)table.size(1);
+  TORCH_CHECK(d % 8 == 0);
+  const long T = idx.numel();
+  auto out = torch::empty({T, d}, table.options());
+  dlb_embed_fwd(table.data_ptr(), idx.data_ptr<int>(), out.data_ptr(), T, d,
+                (float)scale, at::hip::getCurrentHIPStream().stream());
+  return out;
+}
+
+static torch::Tensor embed_bwd(torch::Tensor dy, torch::Tensor idx,
+                               int64_t V, double scale) {
+  const int d = (int)dy.size(-1);
+  const long T = idx.numel();
+  auto dyc = dy.contiguous();
+  auto dt = torch::zeros({V, d}, dy.options().dtype(torch::kFloat32));
+  dlb_embed_bwd(dyc.data_ptr(), idx.data_ptr<int>(), dt.data_ptr<float>(),
+                T, d, (float)scale,
+                at::hip::getCurrentHIPStream().stream());
+  return dt;
+}
+
+// x [N,HW,C] bf16; gate [N,C] bf16 (pre-sigmoid)
+static torch::Tensor se_fwd(torch::Tensor x, torch::Tensor g) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && g.is_contiguous());
+  const int N = (int)x.size(0), HW = (int)x.size(1), C = (int)x.size(2);
+  TORCH_CHECK(C % 8 == 0 && g.numel() == (long)N * C);
+  auto y = torch::empty_like(x);
+  dlb_se_fwd(x.data_ptr(), g.data_ptr(), y.data_ptr(), (long)N * HW, HW, C,
+             at::hip::getCurrentHIPStream().stream());
+  return y;
+}
+
+static std::vector<torch::Tensor> se_bwd(torch::Tensor x, torch::Tensor g,
+                                         torch::Tensor dy) {
+  const int N = (int)x.size(0), HW = (int)x.size(1), C = (int)x.size(2);
+  auto dyc = dy.contiguous();
+  auto dx = torch::empty_like(x);
+  auto dg = torch::empty({N, C}, x.options().dtype(torch::kFloat32));
+  dlb_se_bwd(x.data_ptr(), g.data_ptr(), dyc.data_ptr(), dx.data_ptr(),
+             dg.data_ptr<float>(), N, HW, C,
+             at::hip::getCurrentHIPStream().stream());
+  return {dx, dg};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv_fwd", &conv_fwd, "Implicit-GEMM NHWC bf16 conv forward");
   m.def("conv_bwd_data", &conv_bwd_data, "NHWC bf16 conv input-grad");
@@ -826,4 +901,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Fused decoder GEMM -> log_softmax -> NLL forward (no logits)");
   m.def("lmloss_bwd", &lmloss_bwd,
         "Fused LM loss backward: dh, dW, db with recomputed logits tiles");
+  m.def("dropout", &dropout_op, "philox dropout (same seed = same mask)");
+  m.def("embed_fwd", &embed_fwd, "embedding lookup x scale (bf16)");
+  m.def("embed_bwd", &embed_bwd, "embedding grad scatter (fp32)");
+  m.def("se_fwd", &se_fwd, "SE sigmoid-gate broadcast multiply");
+  m.def("se_bwd", &se_bwd, "SE backward (dx, dgate)");
 }

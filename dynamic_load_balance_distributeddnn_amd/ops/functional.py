@@ -22,7 +22,8 @@ from . import available, ext
 # ops with a hand-written gfx950 kernel wired in (grown as kernels land)
 NATIVE_OPS: set[str] = {"group_norm_act", "conv2d", "layer_norm",
                         "causal_attention", "avg_pool2d", "log_softmax",
-                        "max_pool2d", "lm_loss"}
+                        "max_pool2d", "lm_loss", "dropout", "embedding",
+                        "se_mul"}
 
 
 def _use_native(name: str, x: torch.Tensor) -> bool:
@@ -151,6 +152,38 @@ def lm_loss(h, weight, bias, targets):
     logits = F.linear(h, weight, bias).reshape(-1, weight.shape[0])
     return F.nll_loss(F.log_softmax(logits.float(), dim=-1),
                       targets.reshape(-1))
+
+
+def dropout(x, p, training=True):
+    """Elementwise dropout: philox in-kernel mask on GPU
+    (recomputed for backward), torch RNG on CPU."""
+    if (training and p > 0 and _use_native("dropout", x)
+            and x.dtype == torch.bfloat16):
+        from . import native
+        return native.dropout(x, p)
+    return F.dropout(x, p, training)
+
+
+def embedding_scaled(idx, weight, scale):
+    """Embedding lookup x scale (Net/Transformer.py:91)."""
+    if (idx.is_cuda and _use_native("embedding", idx)
+            and weight.shape[1] % 8 == 0
+            and torch.is_autocast_enabled("cuda")):  # bf16 regime only
+        from . import native
+        return native.embedding_scaled(idx, weight, scale)
+    return F.embedding(idx, weight) * scale
+
+
+def se_mul(x, gate):
+    """x * sigmoid(gate) with gate [N,C,1,1] (RegNet SE,
+    Net/RegNet.py:21-22).  Fused sigmoid+broadcast-mul on GPU."""
+    if (_use_native("se_mul", x) and x.dtype == torch.bfloat16
+            and gate.dtype == torch.bfloat16
+            and x.is_contiguous(memory_format=torch.channels_last)
+            and x.shape[1] % 8 == 0):
+        from . import native
+        return native.se_mul(x, gate)
+    return x * gate.sigmoid()
 
 
 def max_pool2d(x, k, stride=None, padding=0):

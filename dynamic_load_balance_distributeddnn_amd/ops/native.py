@@ -369,3 +369,77 @@ def lm_loss(h, weight, bias, targets):
     if not h2.is_contiguous():
         h2 = h2.contiguous()
     return _LMLoss.apply(h2, weight, bias, targets.reshape(-1))
+
+
+# ----------------------------------------------------- dropout (K15)
+class _Dropout(torch.autograd.Function):
+    """philox dropout: the mask is a pure function of (seed, index), so
+    backward re-applies the same kernel to the upstream grad."""
+
+    @staticmethod
+    def forward(ctx, x, pd):
+        seed = int(torch.randint(0, 2**62, (1,)).item())
+        ctx.pd, ctx.seed = float(pd), seed
+        return ext().dropout(x, float(pd), seed)
+
+    @staticmethod
+    def backward(ctx, dy):
+        return ext().dropout(dy.contiguous(), ctx.pd, ctx.seed), None
+
+
+def dropout(x, pd):
+    return _Dropout.apply(x, pd)
+
+
+# ------------------------------------------- embedding x sqrt(d) (K11)
+class _EmbeddingScaled(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, idx2, weight, scale):
+        wb = weight_bf16(weight)
+        if not wb.is_contiguous():
+            wb = wb.contiguous()
+        idx32 = idx2.reshape(-1).to(torch.int32)
+        out = ext().embed_fwd(wb, idx32, float(scale))
+        ctx.save_for_backward(idx32)
+        ctx.meta = (weight.shape[0], float(scale), idx2.shape)
+        return out.view(*idx2.shape, -1)
+
+    @staticmethod
+    def backward(ctx, dy):
+        (idx32,) = ctx.saved_tensors
+        V, scale, _ = ctx.meta
+        dt = ext().embed_bwd(dy.reshape(idx32.numel(), -1), idx32, V, scale)
+        return None, dt, None
+
+
+def embedding_scaled(idx, weight, scale):
+    """table[idx] * scale — reference Net/Transformer.py:91 (K11)."""
+    return _EmbeddingScaled.apply(idx, weight, scale)
+
+
+# ------------------------------------------------- SE sigmoid-mul (K10)
+class _SEMul(torch.autograd.Function):
+    """y = x * sigmoid(gate) with NHWC x [N,HW,C] and gate [N,C]."""
+
+    @staticmethod
+    def forward(ctx, x3, g2):
+        y = ext().se_fwd(x3, g2)
+        ctx.save_for_backward(x3, g2)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x3, g2 = ctx.saved_tensors
+        dx, dg = ext().se_bwd(x3, g2, dy)
+        return dx, dg.to(g2.dtype)
+
+
+def se_mul(x, gate):
+    """x [N,C,H,W] channels_last * sigmoid(gate [N,C,1,1])."""
+    n, c, h, w = x.shape
+    x3 = _to_nhwc3(x)
+    g2 = gate.reshape(n, c)
+    if not g2.is_contiguous():
+        g2 = g2.contiguous()
+    y3 = _SEMul.apply(x3, g2)
+    return y3.view(n, h, w, c).permute(0, 3, 1, 2)
