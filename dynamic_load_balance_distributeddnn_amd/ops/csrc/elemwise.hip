@@ -116,6 +116,14 @@ se_fwd_kernel(const bf16* __restrict__ x, const bf16* __restrict__ g,
   *reinterpret_cast<Bf16x8e*>(y + e) = o;
 }
 
+extern "C" void dlb_se_fwd(const void* x, const void* g, void* y, long NHW,
+                           int HW, int C, hipStream_t stream) {
+  const long grid = (NHW * (long)C / 8 + 255) / 256;
+  hipLaunchKernelGGL(se_fwd_kernel, dim3((unsigned)grid), dim3(256), 0,
+                     stream, (const bf16*)x, (const bf16*)g, (bf16*)y, NHW,
+                     HW, C);
+}
+
 // dx = dy * sig(g);  dg[n,c] = sum_p dy*x * sig*(1-sig).
 // One thread per (n, c), serial over the HW pixels: dy/x/dx accesses at
 // channel stride stay coalesced across adjacent-c threads and the gate
