@@ -87,10 +87,12 @@ __device__ inline void lm_stage(const bf16* __restrict__ src, long row0,
 #define LM_SREG (LM_BM * (LM_DPAD / 8) / LM_BLOCK)
 struct LmStage {
   bf16x8_t v[LM_SREG];
+  float b;  // bias[n0 + tid] for tid < LM_BN (vocab-tile bias slice)
 };
 
 __device__ inline void lm_sload(const bf16* __restrict__ src, long row0,
-                                long rmax, int d, LmStage& s) {
+                                long rmax, int d, LmStage& s,
+                                const float* __restrict__ bias) {
   const int cpr = LM_DPAD / 8;
 #pragma unroll
   for (int u = 0; u < LM_SREG; ++u) {
@@ -101,9 +103,15 @@ __device__ inline void lm_sload(const bf16* __restrict__ src, long row0,
     if (row0 + r < rmax && k8 < d)
       s.v[u] = *reinterpret_cast<const bf16x8_t*>(src + (row0 + r) * d + k8);
   }
+  if (bias) {
+    s.b = 0.f;
+    if (threadIdx.x < LM_BN && row0 + threadIdx.x < rmax)
+      s.b = bias[row0 + threadIdx.x];
+  }
 }
 
-__device__ inline void lm_swrite(const LmStage& s, bf16* dst) {
+__device__ inline void lm_swrite(const LmStage& s, bf16* dst,
+                                 float* s_bias) {
   const int cpr = LM_DPAD / 8;
 #pragma unroll
   for (int u = 0; u < LM_SREG; ++u) {
@@ -112,6 +120,7 @@ __device__ inline void lm_swrite(const LmStage& s, bf16* dst) {
     const int k8 = (c % cpr) * 8;
     *reinterpret_cast<bf16x8_t*>(dst + r * LM_LDH + k8) = s.v[u];
   }
+  if (s_bias && threadIdx.x < LM_BN) s_bias[threadIdx.x] = s.b;
 }
 
 // z tile [64 t-rows x 64 v-cols] via MFMA from staged h and W images.
@@ -147,6 +156,7 @@ lmloss_fwd_kernel(const bf16* __restrict__ h, const bf16* __restrict__ w,
                   const int V, const int P, const int tiles_per_p) {
   __shared__ bf16 h_lds[LM_BM * LM_LDH];
   __shared__ bf16 w_lds[LM_BN * LM_LDH];
+  __shared__ float s_bias[LM_BN];
 
   const int t = threadIdx.x;
   const int lane = t & 63;
@@ -169,15 +179,16 @@ lmloss_fwd_kernel(const bf16* __restrict__ h, const bf16* __restrict__ w,
     my_tgt[r] = (m0 + rbase + r < T) ? tgt[m0 + rbase + r] : -1;
 
   LmStage ws;
-  if (v_tile0 < v_tile1) lm_sload(w, (long)v_tile0 * LM_BN, V, d, ws);
+  if (v_tile0 < v_tile1)
+    lm_sload(w, (long)v_tile0 * LM_BN, V, d, ws, bias);
   __syncthreads();
-  if (v_tile0 < v_tile1) lm_swrite(ws, w_lds);
+  if (v_tile0 < v_tile1) lm_swrite(ws, w_lds, s_bias);
   __syncthreads();
 
   for (int vt = v_tile0; vt < v_tile1; ++vt) {
     const int n0 = vt * LM_BN;
     if (vt + 1 < v_tile1)  // prefetch next W tile during the MFMAs
-      lm_sload(w, (long)(vt + 1) * LM_BN, V, d, ws);
+      lm_sload(w, (long)(vt + 1) * LM_BN, V, d, ws, bias);
     f32x4 acc[4];
     lm_ztile(h_lds, w_lds, acc, lane, wave);
 #pragma unroll
@@ -186,8 +197,9 @@ lmloss_fwd_kernel(const bf16* __restrict__ h, const bf16* __restrict__ w,
       float tile_max = -3.4e38f;
 #pragma unroll
       for (int j = 0; j < 4; ++j) {
-        const int col = n0 + j * 16 + (lane & 15);
-        z[j] = (col < V) ? acc[j][r] + bias[col] : -3.4e38f;
+        const int cl = j * 16 + (lane & 15);
+        const int col = n0 + cl;
+        z[j] = (col < V) ? acc[j][r] + s_bias[cl] : -3.4e38f;
         tile_max = fmaxf(tile_max, z[j]);
         if (col == my_tgt[r]) ztgt[m0 + rbase + r] = z[j];
       }
@@ -201,7 +213,7 @@ lmloss_fwd_kernel(const bf16* __restrict__ h, const bf16* __restrict__ w,
       }
     }
     __syncthreads();          // all MFMA reads of w_lds complete
-    if (vt + 1 < v_tile1) lm_swrite(ws, w_lds);
+    if (vt + 1 < v_tile1) lm_swrite(ws, w_lds, s_bias);
     __syncthreads();
   }
 
@@ -256,10 +268,13 @@ lmloss_combine_kernel(const float2* __restrict__ part,
 }
 
 // dP tile epilogue: write go/T * (softmax - onehot) to a [64][LM_LDP]
-// bf16 LDS image from the z accumulators.
-__device__ inline void lm_dp_tile(f32x4 acc[4], const float* __restrict__
-                                  bias, const int* __restrict__ tgt,
-                                  const float* __restrict__ lse, long m0,
+// bf16 LDS image from the z accumulators.  bias comes from the staged
+// LDS slice and lse/tgt from per-lane registers — per-tile 4-byte
+// global gathers here re-serialized every tile on waitcnt chains.
+__device__ inline void lm_dp_tile(f32x4 acc[4],
+                                  const float* __restrict__ s_bias,
+                                  const int my_tgt[4],
+                                  const float my_lse[4], long m0,
                                   int n0, int T, int V, float scale,
                                   bf16* dp_lds, int lane, int wave,
                                   float db_part[4]) {
@@ -267,20 +282,18 @@ __device__ inline void lm_dp_tile(f32x4 acc[4], const float* __restrict__
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
     const long row = m0 + rbase + r;
-    const float l = (row < T) ? lse[row] : 0.f;
-    const int ty = (row < T) ? tgt[row] : -1;
 #pragma unroll
     for (int j = 0; j < 4; ++j) {
-      const int col = n0 + j * 16 + (lane & 15);
+      const int cl = j * 16 + (lane & 15);
+      const int col = n0 + cl;
       float v = 0.f;
       if (row < T && col < V) {
-        v = __expf(acc[j][r] + bias[col] - l);
-        if (col == ty) v -= 1.f;
+        v = __expf(acc[j][r] + s_bias[cl] - my_lse[r]);
+        if (col == my_tgt[r]) v -= 1.f;
         v *= scale;
       }
       db_part[j] += v;
-      dp_lds[(rbase + r) * LM_LDP + j * 16 + (lane & 15)] =
-          __float2bfloat16(v);
+      dp_lds[(rbase + r) * LM_LDP + cl] = __float2bfloat16(v);
     }
   }
 }
@@ -299,6 +312,7 @@ lmloss_bwd_dh_kernel(const bf16* __restrict__ h, const bf16* __restrict__ w,
   __shared__ bf16 h_lds[LM_BM * LM_LDH];
   __shared__ bf16 w_lds[LM_BN * LM_LDH];
   __shared__ bf16 dp_lds[LM_BM * LM_LDP];
+  __shared__ float s_bias[LM_BN];
 
   const int t = threadIdx.x;
   const int lane = t & 63;
@@ -315,21 +329,31 @@ lmloss_bwd_dh_kernel(const bf16* __restrict__ h, const bf16* __restrict__ w,
 #pragma unroll
   for (int j = 0; j < LM_DPAD / 16; ++j) dacc[j] = {0.f, 0.f, 0.f, 0.f};
   float db_dummy[4] = {0.f, 0.f, 0.f, 0.f};
+  const int rbase = wave * 16 + (lane >> 4) * 4;
+  float my_lse[4];
+  int my_tgt[4];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const long row = m0 + rbase + r;
+    my_lse[r] = (row < T) ? lse[row] : 0.f;
+    my_tgt[r] = (row < T) ? tgt[row] : -1;
+  }
 
   LmStage ws;
-  if (v_tile0 < v_tile1) lm_sload(w, (long)v_tile0 * LM_BN, V, d, ws);
+  if (v_tile0 < v_tile1)
+    lm_sload(w, (long)v_tile0 * LM_BN, V, d, ws, bias);
   __syncthreads();
-  if (v_tile0 < v_tile1) lm_swrite(ws, w_lds);
+  if (v_tile0 < v_tile1) lm_swrite(ws, w_lds, s_bias);
   __syncthreads();
 
   for (int vt = v_tile0; vt < v_tile1; ++vt) {
     const int n0 = vt * LM_BN;
     if (vt + 1 < v_tile1)  // prefetch next W tile during this tile's math
-      lm_sload(w, (long)(vt + 1) * LM_BN, V, d, ws);
+      lm_sload(w, (long)(vt + 1) * LM_BN, V, d, ws, bias);
     f32x4 acc[4];
     lm_ztile(h_lds, w_lds, acc, lane, wave);
-    lm_dp_tile(acc, bias, tgt, lse, m0, n0, T, V, scale, dp_lds, lane, wave,
-               db_dummy);
+    lm_dp_tile(acc, s_bias, my_tgt, my_lse, m0, n0, T, V, scale, dp_lds,
+               lane, wave, db_dummy);
     __syncthreads();
     // dh[64, d] += dP[64, 64] @ W_tile[64, d]
 #pragma unroll
@@ -345,11 +369,10 @@ lmloss_bwd_dh_kernel(const bf16* __restrict__ h, const bf16* __restrict__ w,
       }
     }
     __syncthreads();          // w_lds reads complete before overwrite
-    if (vt + 1 < v_tile1) lm_swrite(ws, w_lds);
+    if (vt + 1 < v_tile1) lm_swrite(ws, w_lds, s_bias);
     __syncthreads();
   }
 
-  const int rbase = wave * 16 + (lane >> 4) * 4;
 #pragma unroll
   for (int j = 0; j < LM_DPAD / 16; ++j) {
     const int col = j * 16 + (lane & 15);
@@ -379,6 +402,7 @@ lmloss_bwd_dw_kernel(const bf16* __restrict__ h, const bf16* __restrict__ w,
   __shared__ bf16 w_lds[LM_BN * LM_LDH];
   __shared__ bf16 dp_lds[LM_BM * LM_LDP];
   __shared__ float db_lds[LM_BN];
+  __shared__ float s_bias[LM_BN];
 
   const int t = threadIdx.x;
   const int lane = t & 63;
@@ -387,26 +411,38 @@ lmloss_bwd_dw_kernel(const bf16* __restrict__ h, const bf16* __restrict__ w,
   const float scale = go[0] / (float)T;
 
   lm_stage(w, n0, V, d, w_lds, LM_BN);
-  if (t < LM_BN) db_lds[t] = 0.f;
+  if (t < LM_BN) {
+    db_lds[t] = 0.f;
+    s_bias[t] = (n0 + t < V) ? bias[n0 + t] : 0.f;
+  }
 
   f32x4 wacc[LM_DPAD / 16];
 #pragma unroll
   for (int j = 0; j < LM_DPAD / 16; ++j) wacc[j] = {0.f, 0.f, 0.f, 0.f};
   float db_part[4] = {0.f, 0.f, 0.f, 0.f};
+  const int rbase = wave * 16 + (lane >> 4) * 4;
 
   LmStage hs;
-  lm_sload(h, 0, T, d, hs);
+  lm_sload(h, 0, T, d, hs, nullptr);
   __syncthreads();
-  lm_swrite(hs, h_lds);
+  lm_swrite(hs, h_lds, nullptr);
   __syncthreads();
 
   for (long m0 = 0; m0 < T; m0 += LM_BM) {
     if (m0 + LM_BM < T)  // prefetch next h chunk during this chunk's math
-      lm_sload(h, m0 + LM_BM, T, d, hs);
+      lm_sload(h, m0 + LM_BM, T, d, hs, nullptr);
+    float my_lse[4];
+    int my_tgt[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const long row = m0 + rbase + r;
+      my_lse[r] = (row < T) ? lse[row] : 0.f;
+      my_tgt[r] = (row < T) ? tgt[row] : -1;
+    }
     f32x4 acc[4];
     lm_ztile(h_lds, w_lds, acc, lane, wave);
-    lm_dp_tile(acc, bias, tgt, lse, m0, n0, T, V, scale, dp_lds, lane, wave,
-               db_part);
+    lm_dp_tile(acc, s_bias, my_tgt, my_lse, m0, n0, T, V, scale, dp_lds,
+               lane, wave, db_part);
     __syncthreads();
     // dW[64 v, d] += dP^T[64 v, 64 t] @ h[64 t, d]
 #pragma unroll
@@ -421,7 +457,7 @@ lmloss_bwd_dw_kernel(const bf16* __restrict__ h, const bf16* __restrict__ w,
       }
     }
     __syncthreads();          // h_lds reads complete before overwrite
-    if (m0 + LM_BM < T) lm_swrite(hs, h_lds);
+    if (m0 + LM_BM < T) lm_swrite(hs, h_lds, nullptr);
     __syncthreads();
   }
 
