@@ -147,3 +147,19 @@ def test_engine_hipgraph_step_matches_eager(tmp_path, monkeypatch):
     assert tr_g._graph not in (None, False), "graph capture did not engage"
     tr_e, loss_e = run(True)
     assert abs(loss_g - loss_e) < 2e-2 * max(1.0, abs(loss_e))
+
+
+@needs_gpu
+def test_dbs_interval_mode_on_gpu(monkeypatch):
+    """`-dbsi` on GPU: the mid-epoch hipEvent drain + EMA + solver path
+    must run (world==1 makes the split trivial but exercises the timer
+    mechanics end-to-end on the device timeline)."""
+    monkeypatch.setenv("DLB_SYNTH_SCALE", "0.01")
+    from dynamic_load_balance_distributeddnn_amd.engine import Trainer
+
+    args = _args(["-d", "false", "-ws", "1", "-b", "64", "-e", "1",
+                  "-ds", "cifar10", "-m", "densenet", "-dbsi", "3"])
+    tr = Trainer(args, 0, 1, torch.device("cuda:0"), logger=None)
+    compute, sync, loss = tr.train_epoch(0)
+    assert compute > 0 and np.isfinite(loss)
+    assert tr._ema_iter_s is not None and tr._ema_iter_s > 0
