@@ -72,13 +72,14 @@ def test_fusion_cuts_peak_memory():
             gc.collect()  # free other tests' leftovers (arena cycles)
             torch.cuda.empty_cache()
             torch.cuda.reset_peak_memory_stats()
+            base = torch.cuda.memory_allocated()  # lingering tensors
             torch.manual_seed(0)
             model = DenseNet121().cuda().to(memory_format=torch.channels_last)
             x = torch.randn(64, 3, 32, 32, device="cuda") \
                 .to(memory_format=torch.channels_last)
             with torch.autocast("cuda", dtype=torch.bfloat16):
                 model(x).float().sum().backward()
-            return torch.cuda.max_memory_allocated()
+            return torch.cuda.max_memory_allocated() - base
         finally:
             os.environ.pop("DLB_NO_BLOCK_FN", None)
     fused, unfused = peak(False), peak(True)
