@@ -337,12 +337,13 @@ static torch::Tensor gn_conv1x1_wrw(std::vector<torch::Tensor> xs,
   TORCH_CHECK(HW >= 32 || (HW >= 16 && 64 % HW == 0),
               "row-chunk sample span exceeds the staged table");
   int splits = dlb_conv_wrw_nsplits(N, 1, HW, C, Co, 1, 1);
-  // Each split writes a full [Co, C] fp32 slab; at the deep-stream
-  // shapes (C >= 512, few tiles) the slab traffic exceeds the operand
-  // streams, so cap the split count there (sweepable).
+  // Sweepable split cap.  MEASURED (r2c22): capping LOSES — 128 -> -5%,
+  // 64 -> -15% flagship: the fp32 slab traffic is cheaper than the
+  // occupancy the extra splits buy on the few-tile deep-stream GEMMs.
+  // Default 0 = uncapped; kept as a knob for other batch shapes.
   static const int gnc_cap = [] {
     const char* e = getenv("DLB_GNC_SPLITS_MAX");
-    return e ? atoi(e) : 64;
+    return e ? atoi(e) : 0;
   }();
   if (gnc_cap > 0 && splits > gnc_cap) splits = gnc_cap;
   const int mps = ((N * HW + splits - 1) / splits + 63) / 64 * 64;
