@@ -57,8 +57,8 @@ class _Block(nn.Module):
                    if se_ratio > 0 else None)
         self.b = nn.Sequential(
             Conv2d(w_b, w_out, 1),
-            GroupNormAct(_gn_groups(w_out), w_out),
         )
+        self.norm_out = GroupNormAct(_gn_groups(w_out), w_out)
         self.proj = None
         if stride != 1 or w_in != w_out:
             self.proj = nn.Sequential(
@@ -71,8 +71,11 @@ class _Block(nn.Module):
         if self.se is not None:
             out = self.se(out)
         out = self.b(out)
-        out = out + (self.proj(x) if self.proj is not None else x)
-        return F.relu(out, inplace=True)
+        res = self.proj(x) if self.proj is not None else x
+        # relu(GN(out) + res) fused (ops.functional.group_norm_add_act)
+        return FD.group_norm_add_act(out, res, self.norm_out.num_groups,
+                                     self.norm_out.weight,
+                                     self.norm_out.bias, self.norm_out.eps)
 
 
 class RegNet(nn.Module):

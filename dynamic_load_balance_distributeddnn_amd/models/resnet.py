@@ -36,14 +36,16 @@ class _Basic(nn.Module):
             Conv2d(cin, width, 3, stride=stride, padding=1),
             GroupNormAct(_GN, width, relu=True),
             Conv2d(width, width, 3, padding=1),
-            GroupNormAct(_GN, width),
         )
+        self.norm_out = GroupNormAct(_GN, cout)  # params for the fused tail
         self.proj = _proj(cin, cout, stride) if (stride != 1 or cin != cout) else None
 
     def forward(self, x):
         out = self.a(x)
-        out = out + (self.proj(x) if self.proj is not None else x)
-        return F.relu(out, inplace=True)
+        res = self.proj(x) if self.proj is not None else x
+        # relu(GN(out) + res) in one kernel (ops.functional)
+        return FD.group_norm_add_act(out, res, _GN, self.norm_out.weight,
+                                     self.norm_out.bias, self.norm_out.eps)
 
 
 class _Bottleneck(nn.Module):
@@ -58,14 +60,15 @@ class _Bottleneck(nn.Module):
             Conv2d(width, width, 3, stride=stride, padding=1),
             GroupNormAct(_GN, width, relu=True),
             Conv2d(width, cout, 1),
-            GroupNormAct(_GN, cout),
         )
+        self.norm_out = GroupNormAct(_GN, cout)  # params for the fused tail
         self.proj = _proj(cin, cout, stride) if (stride != 1 or cin != cout) else None
 
     def forward(self, x):
         out = self.a(x)
-        out = out + (self.proj(x) if self.proj is not None else x)
-        return F.relu(out, inplace=True)
+        res = self.proj(x) if self.proj is not None else x
+        return FD.group_norm_add_act(out, res, _GN, self.norm_out.weight,
+                                     self.norm_out.bias, self.norm_out.eps)
 
 
 class ResNet(nn.Module):

@@ -28,6 +28,7 @@ static void sgd_momentum(torch::Tensor p, torch::Tensor g, torch::Tensor m,
 extern "C" void dlb_gn_fwd_segs(const void* const* xs, const int* starts,
                                 int nseg, void* y, const float* gamma,
                                 const float* beta, float* mean, float* rstd,
+                                const void* res,
                                 int N, int HW, int C, int G,
                                 float eps, int relu, hipStream_t stream);
 extern "C" void dlb_gn_bwd_segs(const void* const* xs, const int* starts,
@@ -35,6 +36,7 @@ extern "C" void dlb_gn_bwd_segs(const void* const* xs, const int* starts,
                                 const float* gamma, const float* beta,
                                 const float* mean, const float* rstd,
                                 float* dgb_part,
+                                const void* res, void* dres,
                                 int N, int HW, int C, int G, int relu,
                                 int accumulate, hipStream_t stream);
 extern "C" void dlb_gn_dgb_reduce(const float* part, int N, int C,
@@ -62,7 +64,9 @@ extern "C" int dlb_gnconv1x1_wrw(const void* const* xs, const int* starts,
 static std::vector<torch::Tensor> gn_fwd(std::vector<torch::Tensor> xs,
                                          torch::Tensor gamma,
                                          torch::Tensor beta, int64_t groups,
-                                         double eps, bool relu) {
+                                         double eps, bool relu,
+                                         c10::optional<torch::Tensor>
+                                             res = c10::nullopt) {
   TORCH_CHECK(!xs.empty() && xs.size() <= 56);
   int C = 0;
   const void* ptrs[56];
@@ -85,9 +89,16 @@ static std::vector<torch::Tensor> gn_fwd(std::vector<torch::Tensor> xs,
                            xs[0].options().dtype(torch::kFloat32));
   auto rstd = torch::empty_like(mean);
   auto stream = at::hip::getCurrentHIPStream();
+  const void* resp = nullptr;
+  if (res.has_value()) {
+    TORCH_CHECK(res->is_contiguous() && res->scalar_type() == torch::kBFloat16
+                && res->numel() == (long)N * HW * C,
+                "residual must be a packed bf16 [N,HW,C] tensor");
+    resp = res->data_ptr();
+  }
   dlb_gn_fwd_segs(ptrs, starts, (int)xs.size(), y.data_ptr(),
                   gamma.data_ptr<float>(), beta.data_ptr<float>(),
-                  mean.data_ptr<float>(), rstd.data_ptr<float>(), N,
+                  mean.data_ptr<float>(), rstd.data_ptr<float>(), resp, N,
                   HW, C, (int)groups, (float)eps, relu ? 1 : 0,
                   stream.stream());
   return {y, mean, rstd};
@@ -106,7 +117,9 @@ static std::vector<torch::Tensor> gn_bwd(std::vector<torch::Tensor> xs,
                                              dgamma_out = c10::nullopt,
                                          c10::optional<torch::Tensor>
                                              dbeta_out = c10::nullopt,
-                                         bool dgb_defer = false) {
+                                         bool dgb_defer = false,
+                                         c10::optional<torch::Tensor>
+                                             res = c10::nullopt) {
   // dx_accum: preallocated per-segment grad buffers — the kernel ADDS
   // into them (the dense-stream manual backward), instead of allocating
   // fresh outputs for autograd to sum pairwise.
@@ -141,15 +154,24 @@ static std::vector<torch::Tensor> gn_bwd(std::vector<torch::Tensor> xs,
   // (gn_dgb_reduce_multi) and gets the raw partial rows back instead.
   auto part = torch::empty({N, 2 * C},
                            xs[0].options().dtype(torch::kFloat32));
+  const void* resp = nullptr;
+  void* dresp = nullptr;
+  torch::Tensor dres;
+  if (res.has_value()) {
+    resp = res->data_ptr();
+    dres = torch::empty({N, HW, C}, xs[0].options());
+    dresp = dres.data_ptr();
+  }
   dlb_gn_bwd_segs(ptrs, starts, (int)xs.size(), dz.data_ptr(), dptrs,
                   gamma.data_ptr<float>(), beta.data_ptr<float>(),
                   mean.data_ptr<float>(), rstd.data_ptr<float>(),
-                  part.data_ptr<float>(),
+                  part.data_ptr<float>(), resp, dresp,
                   N, HW, C, (int)groups, relu ? 1 : 0, acc ? 1 : 0,
                   stream.stream());
+  if (res.has_value()) out.push_back(dres);
   if (dgb_defer) {
     out.push_back(part);
-    return out;  // [dx_0..dx_{k-1}, part]
+    return out;  // [dx_0..dx_{k-1}, (dres,) part]
   }
   torch::Tensor dgamma, dbeta;
   if (dgamma_out.has_value()) {
@@ -874,13 +896,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("p"), py::arg("g"), py::arg("m"), py::arg("lr"),
         py::arg("mu"), py::arg("mirror") = py::none(),
         "Fused SGD momentum step over flat arenas (gfx950)");
-  m.def("gn_fwd", &gn_fwd, "Fused GroupNorm(+ReLU) forward, NHWC bf16");
+  m.def("gn_fwd", &gn_fwd, "Fused GroupNorm(+residual-add)(+ReLU) forward",
+        py::arg("xs"), py::arg("gamma"), py::arg("beta"), py::arg("groups"),
+        py::arg("eps"), py::arg("relu"), py::arg("res") = py::none());
   m.def("gn_bwd", &gn_bwd, "Fused GroupNorm(+ReLU) backward, NHWC bf16",
         py::arg("xs"), py::arg("dz"), py::arg("gamma"), py::arg("beta"),
         py::arg("mean"), py::arg("rstd"), py::arg("groups"), py::arg("relu"),
         py::arg("dx_accum") = py::none(),
         py::arg("dgamma_out") = py::none(), py::arg("dbeta_out") = py::none(),
-        py::arg("dgb_defer") = false);
+        py::arg("dgb_defer") = false, py::arg("res") = py::none());
   m.def("gn_dgb_reduce_multi", &gn_dgb_reduce_multi,
         "Batched deterministic dgamma/dbeta reduction (one launch)");
   m.def("gn_stats", &gn_stats, "Stats-only GroupNorm over virtual concat");

@@ -101,6 +101,7 @@ extern "C" __global__ void __launch_bounds__(GN_BLOCK)
 gn_fwd_kernel(const GnSegs segs, bf16* __restrict__ y,
               const float* __restrict__ gamma, const float* __restrict__ beta,
               float* __restrict__ mean_out, float* __restrict__ rstd_out,
+              const bf16* __restrict__ res,
               const int HW, const int C, const int G, const float eps,
               const int relu, const int chunk_oct) {
   const int n = blockIdx.x;
@@ -175,6 +176,7 @@ gn_fwd_kernel(const GnSegs segs, bf16* __restrict__ y,
   if (!m.active) return;
 
   bf16* yb = y + (long)n * HW * C;
+  const bf16* resb = res ? res + (long)n * HW * C : nullptr;
   float ga[8], be[8], mu[8], rs[8];
 #pragma unroll
   for (int j = 0; j < 8; ++j) {
@@ -187,10 +189,13 @@ gn_fwd_kernel(const GnSegs segs, bf16* __restrict__ y,
   #pragma unroll 4
   for (int p = m.tp; p < HW; p += m.TP) {
     Bf16x8 chunk = *reinterpret_cast<const Bf16x8*>(xb + (long)p * cs);
+    Bf16x8 rv;
+    if (resb) rv = *reinterpret_cast<const Bf16x8*>(resb + (long)p * C + c0);
     Bf16x8 out;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       float v = (bf2f(chunk.v[j]) - mu[j]) * rs[j] * ga[j] + be[j];
+      if (resb) v += bf2f(rv.v[j]);
       if (relu) v = fmaxf(v, 0.f);
       out.v[j] = f2bf(v);
     }
@@ -278,6 +283,7 @@ gn_bwd_kernel(const GnSegs segs, const bf16* __restrict__ dz,
               const float* __restrict__ mean_in,
               const float* __restrict__ rstd_in,
               float* __restrict__ dgb_part,
+              const bf16* __restrict__ res, bf16* __restrict__ dres,
               const int HW, const int C, const int G, const int relu,
               const int accumulate, const int chunk_oct) {
   const int n = blockIdx.x;
@@ -310,6 +316,8 @@ gn_bwd_kernel(const GnSegs segs, const bf16* __restrict__ dz,
   __syncthreads();
 
   const bf16* db = dz + (long)n * HW * C;
+  const bf16* resb = res ? res + (long)n * HW * C : nullptr;
+  bf16* dresb = dres ? dres + (long)n * HW * C : nullptr;
   const int c0 = (m.o0 + m.tc) << 3;
   int cloc, cs;
   const bf16* sb = seg_locate(segs, c0, cloc, cs);
@@ -331,12 +339,16 @@ gn_bwd_kernel(const GnSegs segs, const bf16* __restrict__ dz,
     for (int p = m.tp; p < HW; p += m.TP) {
       Bf16x8 xc = *reinterpret_cast<const Bf16x8*>(xb + (long)p * cs);
       Bf16x8 dc = *reinterpret_cast<const Bf16x8*>(db + (long)p * C + c0);
+      Bf16x8 rv;
+      if (resb)
+        rv = *reinterpret_cast<const Bf16x8*>(resb + (long)p * C + c0);
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         float xhat = (bf2f(xc.v[j]) - mu[j]) * rs[j];
         float dy = bf2f(dc.v[j]);
         if (relu) {
           float yv = xhat * ga[j] + be[j];
+          if (resb) yv += bf2f(rv.v[j]);
           dy = yv > 0.f ? dy : 0.f;
         }
         a1[j] += ga[j] * dy;
@@ -416,7 +428,11 @@ gn_bwd_kernel(const GnSegs segs, const bf16* __restrict__ dz,
   for (int p = m.tp; p < HW; p += m.TP) {
     Bf16x8 xc = *reinterpret_cast<const Bf16x8*>(xb + (long)p * cs);
     Bf16x8 dc = *reinterpret_cast<const Bf16x8*>(db + (long)p * C + c0);
+    Bf16x8 rv;
+    if (resb)
+      rv = *reinterpret_cast<const Bf16x8*>(resb + (long)p * C + c0);
     Bf16x8 out;
+    Bf16x8 drv;
     Bf16x8 prev;
     if (accumulate)
       prev = *reinterpret_cast<const Bf16x8*>(dxb + (long)p * cs);
@@ -426,13 +442,17 @@ gn_bwd_kernel(const GnSegs segs, const bf16* __restrict__ dz,
       float dy = bf2f(dc.v[j]);
       if (relu) {
         float yv = xhat * ga[j] + be[j];
+        if (resb) yv += bf2f(rv.v[j]);
         dy = yv > 0.f ? dy : 0.f;
       }
+      if (dresb) drv.v[j] = f2bf(dy);  // residual grad = masked dy
       float v = rs[j] * (ga[j] * dy - (k1[j] + xhat * k2[j]));
       if (accumulate) v += bf2f(prev.v[j]);
       out.v[j] = f2bf(v);
     }
     *reinterpret_cast<Bf16x8*>(dxb + (long)p * cs) = out;
+    if (dresb)
+      *reinterpret_cast<Bf16x8*>(dresb + (long)p * C + c0) = drv;
   }
 }
 
@@ -483,6 +503,7 @@ static size_t gn_dyn_shmem(int chunk_oct, int fixed_cspans) {
 extern "C" void dlb_gn_fwd_segs(const void* const* xs, const int* starts,
                                 int nseg, void* y, const float* gamma,
                                 const float* beta, float* mean, float* rstd,
+                                const void* res,
                                 int N, int HW, int C, int G,
                                 float eps, int relu, hipStream_t stream) {
   GnSegs sg{};
@@ -496,8 +517,8 @@ extern "C" void dlb_gn_fwd_segs(const void* const* xs, const int* starts,
   gn_chunking(N, C, G, 0, &chunk_oct, &nchunks);
   hipLaunchKernelGGL(gn_fwd_kernel, dim3(N, nchunks), dim3(GN_BLOCK),
                      gn_dyn_shmem(chunk_oct, 2), stream, sg, (bf16*)y,
-                     gamma, beta, mean, rstd, HW, C, G, eps, relu,
-                     chunk_oct);
+                     gamma, beta, mean, rstd, (const bf16*)res,
+                     HW, C, G, eps, relu, chunk_oct);
 }
 
 extern "C" void dlb_gn_bwd_segs(const void* const* xs, const int* starts,
@@ -505,6 +526,7 @@ extern "C" void dlb_gn_bwd_segs(const void* const* xs, const int* starts,
                                 const float* gamma, const float* beta,
                                 const float* mean, const float* rstd,
                                 float* dgb_part,
+                                const void* res, void* dres,
                                 int N, int HW, int C, int G, int relu,
                                 int accumulate, hipStream_t stream) {
   GnSegs sg{};
@@ -521,7 +543,8 @@ extern "C" void dlb_gn_bwd_segs(const void* const* xs, const int* starts,
   hipLaunchKernelGGL(gn_bwd_kernel, dim3(N, nchunks), dim3(GN_BLOCK),
                      gn_dyn_shmem(chunk_oct, 4), stream, sg,
                      (const bf16*)dz, dsg, gamma, beta, mean, rstd,
-                     dgb_part, HW, C, G, relu, accumulate, chunk_oct);
+                     dgb_part, (const bf16*)res, (bf16*)dres,
+                     HW, C, G, relu, accumulate, chunk_oct);
 }
 
 extern "C" void dlb_gn_stats_segs(const void* const* xs, const int* starts,

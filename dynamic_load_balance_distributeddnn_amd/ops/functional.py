@@ -61,6 +61,22 @@ def group_norm_act(x, num_groups, weight, bias, eps=1e-5, relu=False):
     return F.relu(out, inplace=True) if relu else out
 
 
+def group_norm_add_act(x, res, num_groups, weight, bias, eps=1e-5):
+    """relu(GroupNorm(x) + residual) — the ResNet/RegNet junction,
+    fused into the GN kernel on GPU (residual rides the normalize and
+    backward sweeps; the torch composition costs ~13% of the ResNet-101
+    step in elementwise kernels)."""
+    if (_use_native("group_norm_act", x)
+            and res.dtype == x.dtype and res.shape == x.shape
+            and res.is_contiguous(memory_format=torch.channels_last)):
+        from . import native
+        if native.gn_native_ok(x, num_groups, weight):
+            return native.group_norm_add_act(x, res, num_groups, weight,
+                                             bias, eps)
+    out = F.group_norm(x, num_groups, weight, bias, eps) + res
+    return F.relu(out, inplace=True)
+
+
 def linear(x, weight, bias=None):
     # Plain library GEMM: hipBLASLt via F.linear (the north star allows
     # vendor GEMM libraries for unfused matmuls; fused hot ops are ours).

@@ -141,3 +141,40 @@ def test_densenet_virtual_concat_trains():
     loss.backward()
     assert torch.isfinite(loss)
     assert all(torch.isfinite(p.grad).all() for p in m.parameters())
+
+
+def test_gn_add_relu_fused_matches_fp32():
+    """relu(GN(x) + residual) fused (ResNet/RegNet junction) vs the
+    fp32 torch composition, forward and all gradients."""
+    import torch.nn.functional as F
+
+    from dynamic_load_balance_distributeddnn_amd.ops import native
+
+    torch.manual_seed(6)
+    N, C, H, W = 16, 128, 8, 8
+    x = torch.randn(N, C, H, W, device="cuda").bfloat16() \
+        .to(memory_format=torch.channels_last).requires_grad_()
+    r = torch.randn(N, C, H, W, device="cuda").bfloat16() \
+        .to(memory_format=torch.channels_last).requires_grad_()
+    g = torch.randn(C, device="cuda").requires_grad_()
+    b = torch.randn(C, device="cuda").requires_grad_()
+
+    y = native.group_norm_add_act(x, r, 32, g, b, 1e-5)
+    dz = torch.randn_like(y)
+    y.backward(dz)
+
+    x32 = x.detach().float().requires_grad_()
+    r32 = r.detach().float().requires_grad_()
+    g32 = g.detach().clone().requires_grad_()
+    b32 = b.detach().clone().requires_grad_()
+    ref = F.relu(F.group_norm(x32, 32, g32, b32, 1e-5) + r32)
+    ref.backward(dz.float())
+
+    def rel(a, bb):
+        return (a.float() - bb).norm().item() / max(bb.norm().item(), 1e-9)
+
+    assert rel(y, ref) < 1.5e-2
+    assert rel(x.grad, x32.grad) < 2e-2
+    assert rel(r.grad, r32.grad) < 1.5e-2
+    assert rel(g.grad, g32.grad) < 1.5e-2
+    assert rel(b.grad, b32.grad) < 1.5e-2
