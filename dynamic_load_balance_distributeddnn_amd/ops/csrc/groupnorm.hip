@@ -97,7 +97,8 @@ __device__ inline ChunkMap chunk_map(int TC, int chunk_oct) {
 // ---------------------------------------------------------------- forward
 // Dynamic LDS layout: [cspan] gamma | [cspan] beta | [TP*cspan] P0 |
 // [TP*cspan] P1  (P* hold per-(tp, channel) partials; TP*cspan <= 2048).
-extern "C" __global__ void __launch_bounds__(GN_BLOCK)
+template <bool HASRES>
+__global__ void __launch_bounds__(GN_BLOCK)
 gn_fwd_kernel(const GnSegs segs, bf16* __restrict__ y,
               const float* __restrict__ gamma, const float* __restrict__ beta,
               float* __restrict__ mean_out, float* __restrict__ rstd_out,
@@ -176,7 +177,7 @@ gn_fwd_kernel(const GnSegs segs, bf16* __restrict__ y,
   if (!m.active) return;
 
   bf16* yb = y + (long)n * HW * C;
-  const bf16* resb = res ? res + (long)n * HW * C : nullptr;
+  const bf16* resb = HASRES ? res + (long)n * HW * C : nullptr;
   float ga[8], be[8], mu[8], rs[8];
 #pragma unroll
   for (int j = 0; j < 8; ++j) {
@@ -190,12 +191,13 @@ gn_fwd_kernel(const GnSegs segs, bf16* __restrict__ y,
   for (int p = m.tp; p < HW; p += m.TP) {
     Bf16x8 chunk = *reinterpret_cast<const Bf16x8*>(xb + (long)p * cs);
     Bf16x8 rv;
-    if (resb) rv = *reinterpret_cast<const Bf16x8*>(resb + (long)p * C + c0);
+    if (HASRES)
+      rv = *reinterpret_cast<const Bf16x8*>(resb + (long)p * C + c0);
     Bf16x8 out;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       float v = (bf2f(chunk.v[j]) - mu[j]) * rs[j] * ga[j] + be[j];
-      if (resb) v += bf2f(rv.v[j]);
+      if (HASRES) v += bf2f(rv.v[j]);
       if (relu) v = fmaxf(v, 0.f);
       out.v[j] = f2bf(v);
     }
@@ -276,7 +278,8 @@ gn_stats_kernel(const GnSegs segs, float* __restrict__ mean_out,
 // ReLU mask recomputed as (xhat*g+b) > 0.
 // Dynamic LDS: [2*cspan] dgb | [cspan] gamma | [cspan] beta |
 // [TP*cspan] P0 | [TP*cspan] P1.
-extern "C" __global__ void __launch_bounds__(GN_BLOCK)
+template <bool HASRES>
+__global__ void __launch_bounds__(GN_BLOCK)
 gn_bwd_kernel(const GnSegs segs, const bf16* __restrict__ dz,
               const GnSegsMut dxs, const float* __restrict__ gamma,
               const float* __restrict__ beta,
@@ -316,8 +319,8 @@ gn_bwd_kernel(const GnSegs segs, const bf16* __restrict__ dz,
   __syncthreads();
 
   const bf16* db = dz + (long)n * HW * C;
-  const bf16* resb = res ? res + (long)n * HW * C : nullptr;
-  bf16* dresb = dres ? dres + (long)n * HW * C : nullptr;
+  const bf16* resb = HASRES ? res + (long)n * HW * C : nullptr;
+  bf16* dresb = HASRES ? dres + (long)n * HW * C : nullptr;
   const int c0 = (m.o0 + m.tc) << 3;
   int cloc, cs;
   const bf16* sb = seg_locate(segs, c0, cloc, cs);
@@ -340,7 +343,7 @@ gn_bwd_kernel(const GnSegs segs, const bf16* __restrict__ dz,
       Bf16x8 xc = *reinterpret_cast<const Bf16x8*>(xb + (long)p * cs);
       Bf16x8 dc = *reinterpret_cast<const Bf16x8*>(db + (long)p * C + c0);
       Bf16x8 rv;
-      if (resb)
+      if (HASRES)
         rv = *reinterpret_cast<const Bf16x8*>(resb + (long)p * C + c0);
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
@@ -348,7 +351,7 @@ gn_bwd_kernel(const GnSegs segs, const bf16* __restrict__ dz,
         float dy = bf2f(dc.v[j]);
         if (relu) {
           float yv = xhat * ga[j] + be[j];
-          if (resb) yv += bf2f(rv.v[j]);
+          if (HASRES) yv += bf2f(rv.v[j]);
           dy = yv > 0.f ? dy : 0.f;
         }
         a1[j] += ga[j] * dy;
@@ -429,7 +432,7 @@ gn_bwd_kernel(const GnSegs segs, const bf16* __restrict__ dz,
     Bf16x8 xc = *reinterpret_cast<const Bf16x8*>(xb + (long)p * cs);
     Bf16x8 dc = *reinterpret_cast<const Bf16x8*>(db + (long)p * C + c0);
     Bf16x8 rv;
-    if (resb)
+    if (HASRES)
       rv = *reinterpret_cast<const Bf16x8*>(resb + (long)p * C + c0);
     Bf16x8 out;
     Bf16x8 drv;
@@ -442,16 +445,16 @@ gn_bwd_kernel(const GnSegs segs, const bf16* __restrict__ dz,
       float dy = bf2f(dc.v[j]);
       if (relu) {
         float yv = xhat * ga[j] + be[j];
-        if (resb) yv += bf2f(rv.v[j]);
+        if (HASRES) yv += bf2f(rv.v[j]);
         dy = yv > 0.f ? dy : 0.f;
       }
-      if (dresb) drv.v[j] = f2bf(dy);  // residual grad = masked dy
+      if (HASRES) drv.v[j] = f2bf(dy);  // residual grad = masked dy
       float v = rs[j] * (ga[j] * dy - (k1[j] + xhat * k2[j]));
       if (accumulate) v += bf2f(prev.v[j]);
       out.v[j] = f2bf(v);
     }
     *reinterpret_cast<Bf16x8*>(dxb + (long)p * cs) = out;
-    if (dresb)
+    if (HASRES)
       *reinterpret_cast<Bf16x8*>(dresb + (long)p * C + c0) = drv;
   }
 }
@@ -515,10 +518,16 @@ extern "C" void dlb_gn_fwd_segs(const void* const* xs, const int* starts,
   sg.start[nseg] = starts[nseg];
   int chunk_oct, nchunks;
   gn_chunking(N, C, G, 0, &chunk_oct, &nchunks);
-  hipLaunchKernelGGL(gn_fwd_kernel, dim3(N, nchunks), dim3(GN_BLOCK),
-                     gn_dyn_shmem(chunk_oct, 2), stream, sg, (bf16*)y,
-                     gamma, beta, mean, rstd, (const bf16*)res,
-                     HW, C, G, eps, relu, chunk_oct);
+  if (res)
+    hipLaunchKernelGGL((gn_fwd_kernel<true>), dim3(N, nchunks),
+                       dim3(GN_BLOCK), gn_dyn_shmem(chunk_oct, 2), stream,
+                       sg, (bf16*)y, gamma, beta, mean, rstd,
+                       (const bf16*)res, HW, C, G, eps, relu, chunk_oct);
+  else
+    hipLaunchKernelGGL((gn_fwd_kernel<false>), dim3(N, nchunks),
+                       dim3(GN_BLOCK), gn_dyn_shmem(chunk_oct, 2), stream,
+                       sg, (bf16*)y, gamma, beta, mean, rstd, nullptr,
+                       HW, C, G, eps, relu, chunk_oct);
 }
 
 extern "C" void dlb_gn_bwd_segs(const void* const* xs, const int* starts,
@@ -540,11 +549,18 @@ extern "C" void dlb_gn_bwd_segs(const void* const* xs, const int* starts,
   sg.start[nseg] = starts[nseg];
   int chunk_oct, nchunks;
   gn_chunking(N, C, G, 1, &chunk_oct, &nchunks);
-  hipLaunchKernelGGL(gn_bwd_kernel, dim3(N, nchunks), dim3(GN_BLOCK),
-                     gn_dyn_shmem(chunk_oct, 4), stream, sg,
-                     (const bf16*)dz, dsg, gamma, beta, mean, rstd,
-                     dgb_part, (const bf16*)res, (bf16*)dres,
-                     HW, C, G, relu, accumulate, chunk_oct);
+  if (res)
+    hipLaunchKernelGGL((gn_bwd_kernel<true>), dim3(N, nchunks),
+                       dim3(GN_BLOCK), gn_dyn_shmem(chunk_oct, 4), stream,
+                       sg, (const bf16*)dz, dsg, gamma, beta, mean, rstd,
+                       dgb_part, (const bf16*)res, (bf16*)dres,
+                       HW, C, G, relu, accumulate, chunk_oct);
+  else
+    hipLaunchKernelGGL((gn_bwd_kernel<false>), dim3(N, nchunks),
+                       dim3(GN_BLOCK), gn_dyn_shmem(chunk_oct, 4), stream,
+                       sg, (const bf16*)dz, dsg, gamma, beta, mean, rstd,
+                       dgb_part, nullptr, nullptr,
+                       HW, C, G, relu, accumulate, chunk_oct);
 }
 
 extern "C" void dlb_gn_stats_segs(const void* const* xs, const int* starts,
