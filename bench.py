@@ -155,7 +155,12 @@ def main():
     if world > 1:
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         os.environ.setdefault("MASTER_PORT", "29531")
-        backend = "nccl" if device.type == "cuda" else "gloo"
+        # DLB_BENCH_BACKEND=gloo lets a 1-GPU box exercise the
+        # multi-rank bench path (RCCL refuses 2 ranks on one GPU);
+        # the driver's real SCALE runs use the default (RCCL).
+        backend = os.environ.get(
+            "DLB_BENCH_BACKEND",
+            "nccl" if device.type == "cuda" else "gloo")
         dist.init_process_group(backend, rank=rank, world_size=world)
 
     if args.engine:
