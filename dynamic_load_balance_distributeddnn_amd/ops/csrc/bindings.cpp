@@ -199,6 +199,56 @@ extern "C" void dlb_gn_dgb_reduce_multi(const void* const* parts,
                                         void* const* dgs, void* const* dbs,
                                         const int* Cs, int nl, int N,
                                         hipStream_t stream);
+extern "C" void dlb_chansum(const void* x, float* sum, float* ssq, int N,
+                            int HW, int Cs, hipStream_t stream);
+extern "C" void dlb_gn_stats_sums(const void* const* sums,
+                                  const void* const* ssqs,
+                                  const int* starts, int nseg, float* mean,
+                                  float* rstd, int N, int HW, int C, int G,
+                                  float eps, hipStream_t stream);
+
+// Per-channel (sum, ssq) of one [N, HW, Cs] bf16 segment.
+static std::vector<torch::Tensor> chan_sums(torch::Tensor x) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 3 && x.is_contiguous() &&
+              x.scalar_type() == torch::kBFloat16);
+  const int N = (int)x.size(0), HW = (int)x.size(1), Cs = (int)x.size(2);
+  auto sum = torch::empty({N, Cs}, x.options().dtype(torch::kFloat32));
+  auto ssq = torch::empty_like(sum);
+  dlb_chansum(x.data_ptr(), sum.data_ptr<float>(), ssq.data_ptr<float>(),
+              N, HW, Cs, at::hip::getCurrentHIPStream().stream());
+  return {sum, ssq};
+}
+
+// mean/rstd [N, G] from accumulated per-channel sums (newest-first
+// segment lists, mirroring the activation segment order).
+static std::vector<torch::Tensor> gn_stats_from_sums(
+    std::vector<torch::Tensor> sums, std::vector<torch::Tensor> ssqs,
+    int64_t groups, int64_t HW, double eps) {
+  const int nseg = (int)sums.size();
+  TORCH_CHECK(nseg >= 1 && nseg <= 56 && ssqs.size() == sums.size());
+  const void* ps[56];
+  const void* qs[56];
+  int starts[57];
+  int C = 0;
+  const int N = (int)sums[0].size(0);
+  for (int i = 0; i < nseg; ++i) {
+    TORCH_CHECK(sums[i].is_contiguous() && ssqs[i].is_contiguous() &&
+                sums[i].size(0) == N);
+    ps[i] = sums[i].data_ptr();
+    qs[i] = ssqs[i].data_ptr();
+    starts[i] = C;
+    C += (int)sums[i].size(1);
+  }
+  starts[nseg] = C;
+  TORCH_CHECK(C % groups == 0);
+  auto mean = torch::empty({N, groups},
+                           sums[0].options().dtype(torch::kFloat32));
+  auto rstd = torch::empty_like(mean);
+  dlb_gn_stats_sums(ps, qs, starts, nseg, mean.data_ptr<float>(),
+                    rstd.data_ptr<float>(), N, (int)HW, C, (int)groups,
+                    (float)eps, at::hip::getCurrentHIPStream().stream());
+  return {mean, rstd};
+}
 
 // Batched dgamma/dbeta reduction over many layers' deferred partials
 // (one launch; += into the pre-zeroed arena grad views).
@@ -907,6 +957,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("dgb_defer") = false, py::arg("res") = py::none());
   m.def("gn_dgb_reduce_multi", &gn_dgb_reduce_multi,
         "Batched deterministic dgamma/dbeta reduction (one launch)");
+  m.def("chan_sums", &chan_sums, "per-channel (sum, ssq) of a segment");
+  m.def("gn_stats_from_sums", &gn_stats_from_sums,
+        "GroupNorm mean/rstd from accumulated per-channel sums");
   m.def("gn_stats", &gn_stats, "Stats-only GroupNorm over virtual concat");
   m.def("gn_conv1x1_fwd", &gn_conv1x1_fwd,
         "Fused GroupNorm(+ReLU) -> 1x1 conv forward (stream never packed)");
