@@ -64,11 +64,14 @@ class _DenseBlockFn(torch.autograd.Function):
         fused = hw >= 32 or (hw >= 16 and 128 % hw == 0)
         segs3 = [_to_nhwc3(seg0)]
         saves = []
-        # Incremental norm1 statistics: the stream's per-channel
-        # (sum, ssq) accumulate as segments append, so each layer's
-        # mean/rstd derive from O(C) sums instead of re-reading the
-        # O(N*HW*C) stream (gn_stats was 1.26 ms/step on the flagship).
-        inc = fused and not os.environ.get("DLB_NO_INCSTATS")
+        # Incremental norm1 statistics (OPT-IN, experimental): derive
+        # each layer's mean/rstd from accumulated per-channel (sum, ssq)
+        # instead of re-reading the stream.  MEASURED SLOWER at the
+        # flagship (r2c26: 18.7k vs 19.0k img/s in the same session —
+        # the gn_stats stream re-reads are already Infinity-Cache-served,
+        # while chansum's thin per-(n,c) grid is latency-bound), so the
+        # default stays on gn_stats.
+        inc = fused and bool(os.environ.get("DLB_INCSTATS"))
         if inc:
             s0, q0 = ext().chan_sums(segs3[0])
             sums, ssqs = [s0], [q0]
