@@ -155,8 +155,8 @@ lmloss_fwd_kernel(const bf16* __restrict__ h, const bf16* __restrict__ w,
                   float* __restrict__ ztgt, const int T, const int d,
                   const int V, const int P, const int tiles_per_p) {
   __shared__ bf16 h_lds[LM_BM * LM_LDH];
-  __shared__ bf16 w_lds[2][LM_BN * LM_LDH];
-  __shared__ float s_bias[2][LM_BN];
+  __shared__ bf16 w_lds[LM_BN * LM_LDH];
+  __shared__ float s_bias[LM_BN];
 
   const int t = threadIdx.x;
   const int lane = t & 63;
@@ -179,11 +179,10 @@ lmloss_fwd_kernel(const bf16* __restrict__ h, const bf16* __restrict__ w,
     my_tgt[r] = (m0 + rbase + r < T) ? tgt[m0 + rbase + r] : -1;
 
   LmStage ws;
-  int buf = 0;
   if (v_tile0 < v_tile1)
     lm_sload(w, (long)v_tile0 * LM_BN, V, d, ws, bias);
   __syncthreads();
-  if (v_tile0 < v_tile1) lm_swrite(ws, w_lds[0], s_bias[0]);
+  if (v_tile0 < v_tile1) lm_swrite(ws, w_lds, s_bias);
   __syncthreads();
 
   for (int vt = v_tile0; vt < v_tile1; ++vt) {
@@ -191,10 +190,7 @@ lmloss_fwd_kernel(const bf16* __restrict__ h, const bf16* __restrict__ w,
     if (vt + 1 < v_tile1)  // prefetch next W tile during the MFMAs
       lm_sload(w, (long)(vt + 1) * LM_BN, V, d, ws, bias);
     f32x4 acc[4];
-    lm_ztile(h_lds, w_lds[buf], acc, lane, wave);
-    // write the prefetched tile into the OTHER buffer — no read clash,
-    // so ONE barrier per tile publishes it and joins this tile's reads
-    if (vt + 1 < v_tile1) lm_swrite(ws, w_lds[buf ^ 1], s_bias[buf ^ 1]);
+    lm_ztile(h_lds, w_lds, acc, lane, wave);
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       float z[4];
@@ -203,7 +199,7 @@ lmloss_fwd_kernel(const bf16* __restrict__ h, const bf16* __restrict__ w,
       for (int j = 0; j < 4; ++j) {
         const int cl = j * 16 + (lane & 15);
         const int col = n0 + cl;
-        z[j] = (col < V) ? acc[j][r] + s_bias[buf][cl] : -3.4e38f;
+        z[j] = (col < V) ? acc[j][r] + s_bias[cl] : -3.4e38f;
         tile_max = fmaxf(tile_max, z[j]);
         if (col == my_tgt[r]) ztgt[m0 + rbase + r] = z[j];
       }
@@ -216,8 +212,9 @@ lmloss_fwd_kernel(const bf16* __restrict__ h, const bf16* __restrict__ w,
         s_run[r] = s;
       }
     }
+    __syncthreads();          // all MFMA reads of w_lds complete
+    if (vt + 1 < v_tile1) lm_swrite(ws, w_lds, s_bias);
     __syncthreads();
-    buf ^= 1;
   }
 
   // merge the 16 lanes of each (lane>>4) group: cols -> one (m, s)
@@ -313,9 +310,9 @@ lmloss_bwd_dh_kernel(const bf16* __restrict__ h, const bf16* __restrict__ w,
                      const int T, const int d, const int V, const int P,
                      const int tiles_per_p) {
   __shared__ bf16 h_lds[LM_BM * LM_LDH];
-  __shared__ bf16 w_lds[2][LM_BN * LM_LDH];
+  __shared__ bf16 w_lds[LM_BN * LM_LDH];
   __shared__ bf16 dp_lds[LM_BM * LM_LDP];
-  __shared__ float s_bias[2][LM_BN];
+  __shared__ float s_bias[LM_BN];
 
   const int t = threadIdx.x;
   const int lane = t & 63;
@@ -343,11 +340,10 @@ lmloss_bwd_dh_kernel(const bf16* __restrict__ h, const bf16* __restrict__ w,
   }
 
   LmStage ws;
-  int buf = 0;
   if (v_tile0 < v_tile1)
     lm_sload(w, (long)v_tile0 * LM_BN, V, d, ws, bias);
   __syncthreads();
-  if (v_tile0 < v_tile1) lm_swrite(ws, w_lds[0], s_bias[0]);
+  if (v_tile0 < v_tile1) lm_swrite(ws, w_lds, s_bias);
   __syncthreads();
 
   for (int vt = v_tile0; vt < v_tile1; ++vt) {
@@ -355,13 +351,10 @@ lmloss_bwd_dh_kernel(const bf16* __restrict__ h, const bf16* __restrict__ w,
     if (vt + 1 < v_tile1)  // prefetch next W tile during this tile's math
       lm_sload(w, (long)(vt + 1) * LM_BN, V, d, ws, bias);
     f32x4 acc[4];
-    lm_ztile(h_lds, w_lds[buf], acc, lane, wave);
-    lm_dp_tile(acc, s_bias[buf], my_tgt, my_lse, m0, n0, T, V, scale,
-               dp_lds, lane, wave, db_dummy);
-    // publish the prefetched tile into the OTHER buffer (reads of it
-    // last happened before the previous iteration's closing barrier)
-    if (vt + 1 < v_tile1) lm_swrite(ws, w_lds[buf ^ 1], s_bias[buf ^ 1]);
-    __syncthreads();          // dp_lds ready; ztile reads of buf joined
+    lm_ztile(h_lds, w_lds, acc, lane, wave);
+    lm_dp_tile(acc, s_bias, my_tgt, my_lse, m0, n0, T, V, scale, dp_lds,
+               lane, wave, db_dummy);
+    __syncthreads();
     // dh[64, d] += dP[64, 64] @ W_tile[64, d]
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {
@@ -370,14 +363,14 @@ lmloss_bwd_dh_kernel(const bf16* __restrict__ h, const bf16* __restrict__ w,
                   (lane >> 4) * 8]);
 #pragma unroll
       for (int j = 0; j < LM_DPAD / 16; ++j) {
-        bf16x8_t bfrag = lm_tr_frag<LM_LDH>(w_lds[buf], ks * 32, j * 16,
-                                            lane);
+        bf16x8_t bfrag = lm_tr_frag<LM_LDH>(w_lds, ks * 32, j * 16, lane);
         dacc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag,
                                                           dacc[j], 0, 0, 0);
       }
     }
-    __syncthreads();          // joins dh reads of buf; publishes buf^1
-    buf ^= 1;
+    __syncthreads();          // w_lds reads complete before overwrite
+    if (vt + 1 < v_tile1) lm_swrite(ws, w_lds, s_bias);
+    __syncthreads();
   }
 
 #pragma unroll
@@ -405,7 +398,7 @@ lmloss_bwd_dw_kernel(const bf16* __restrict__ h, const bf16* __restrict__ w,
                      const float* __restrict__ go, float* __restrict__ dw,
                      float* __restrict__ db, const int T, const int d,
                      const int V) {
-  __shared__ bf16 h_lds[2][LM_BM * LM_LDH];
+  __shared__ bf16 h_lds[LM_BM * LM_LDH];
   __shared__ bf16 w_lds[LM_BN * LM_LDH];
   __shared__ bf16 dp_lds[LM_BM * LM_LDP];
   __shared__ float db_lds[LM_BN];
@@ -430,10 +423,9 @@ lmloss_bwd_dw_kernel(const bf16* __restrict__ h, const bf16* __restrict__ w,
   const int rbase = wave * 16 + (lane >> 4) * 4;
 
   LmStage hs;
-  int buf = 0;
   lm_sload(h, 0, T, d, hs, nullptr);
   __syncthreads();
-  lm_swrite(hs, h_lds[0], nullptr);
+  lm_swrite(hs, h_lds, nullptr);
   __syncthreads();
 
   for (long m0 = 0; m0 < T; m0 += LM_BM) {
@@ -448,11 +440,10 @@ lmloss_bwd_dw_kernel(const bf16* __restrict__ h, const bf16* __restrict__ w,
       my_tgt[r] = (row < T) ? tgt[row] : -1;
     }
     f32x4 acc[4];
-    lm_ztile(h_lds[buf], w_lds, acc, lane, wave);
+    lm_ztile(h_lds, w_lds, acc, lane, wave);
     lm_dp_tile(acc, s_bias, my_tgt, my_lse, m0, n0, T, V, scale, dp_lds,
                lane, wave, db_part);
-    if (m0 + LM_BM < T) lm_swrite(hs, h_lds[buf ^ 1], nullptr);
-    __syncthreads();          // dp_lds ready; ztile reads of buf joined
+    __syncthreads();
     // dW[64 v, d] += dP^T[64 v, 64 t] @ h[64 t, d]
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {
@@ -460,14 +451,14 @@ lmloss_bwd_dw_kernel(const bf16* __restrict__ h, const bf16* __restrict__ w,
                                           wave * 16, lane);
 #pragma unroll
       for (int j = 0; j < LM_DPAD / 16; ++j) {
-        bf16x8_t bfrag = lm_tr_frag<LM_LDH>(h_lds[buf], ks * 32, j * 16,
-                                            lane);
+        bf16x8_t bfrag = lm_tr_frag<LM_LDH>(h_lds, ks * 32, j * 16, lane);
         wacc[j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag,
                                                           wacc[j], 0, 0, 0);
       }
     }
-    __syncthreads();          // joins dW reads of buf; publishes buf^1
-    buf ^= 1;
+    __syncthreads();          // h_lds reads complete before overwrite
+    if (m0 + LM_BM < T) lm_swrite(hs, h_lds, nullptr);
+    __syncthreads();
   }
 
   // db: this lane's partials cover cols j*16+(lane&15) summed over its
