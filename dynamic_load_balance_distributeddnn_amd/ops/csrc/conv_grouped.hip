@@ -144,64 +144,114 @@ gconv_bwd_kernel(const GConvParams p) {
       p.out + (((long)n * p.IH + ih) * p.IW + iw) * p.C + ci8) = o;
 }
 
-// wrw: grid (G, 9 taps, splits); 128-m stages with vector loads; thread
-// = (co_l, ci_l, m-replica); LDS tree-reduce over replicas at the end.
+// wrw v3: the per-(group, tap) weight grad is a [GW x GW] output GEMM
+// with K = M (dW[co,ci] = sum_m dy[m,co] * x_tap[m,ci]) — exactly one
+// v_mfma_f32_16x16x32_bf16 tile per 32 pixels.  v2 ran it as scalar
+// LDS FMAs with dy re-staged per tap across 9 blocks and measured 28%
+// of the RegNetY step (183 ms, profiles r2c30).  Here: grid (G,
+// splits); one block stages dy ONCE and the 9 shifted x images per
+// 128-m chunk, and the 4 waves split the 9 taps (wave w owns taps
+// {w, w+4, w+8}), each accumulating its [16,16] tiles via
+// transpose-read fragments.  GW=8 groups zero-pad to the 16-wide tile.
+#define GWP 16  // padded tile width (GW is 8 or 16 in the zoo)
+
+typedef __attribute__((__vector_size__(4 * sizeof(float)))) float gf32x4;
+typedef __attribute__((__vector_size__(4 * sizeof(__bf16)))) __bf16 gtrvec;
+#define GLDS3 __attribute__((address_space(3)))
+
+// ds_read_b64_tr_b16 fragment from a row-major [m][GWP] LDS image:
+// element j = img[mbase + (lane>>4)*8 + j][lane & 15] — the [n=col]
+// [k=m] operand both A (cols=co) and B (cols=ci) need here.
+__device__ inline bf16x8_t gtr_frag(const bf16* img, int mbase, int lane) {
+  const int j15 = lane & 15, q = lane >> 4;
+  const int row = mbase + q * 8 + (j15 >> 2);
+  const int col = 4 * (j15 & 3);
+  auto p0 = (GLDS3 gtrvec*)((GLDS3 bf16*)img + (long)row * GWP + col);
+  auto p1 = (GLDS3 gtrvec*)((GLDS3 bf16*)img + (long)(row + 4) * GWP + col);
+  gtrvec lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p0);
+  gtrvec hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p1);
+  union { struct { gtrvec a, b; } t; bf16x8_t v; } u;
+  u.t.a = lo;
+  u.t.b = hi;
+  return u.v;
+}
+
 __global__ void __launch_bounds__(GBLOCK)
 gconv_wrw_kernel(const GConvParams p) {
-  __shared__ bf16 dy_s[128 * 16];  // [mm][co_l]
-  __shared__ bf16 x_s[128 * 16];   // [mm][ci_l]
-  __shared__ float red[GBLOCK];
+  __shared__ bf16 dy_s[128 * GWP];      // [mm][co_l]
+  __shared__ bf16 x_s[9][128 * GWP];    // per-tap shifted [mm][ci_l]
 
   const int g = blockIdx.x;
-  const int tap = blockIdx.y;
-  const int r = tap / 3, s = tap % 3;
   const int co0 = g * p.GW, ci0 = g * p.GW;
   const int t = threadIdx.x;
-  const int pairs = p.GW * p.GW;          // 64 or 256
-  const int rep_n = GBLOCK / pairs;       // 4 or 1
-  const int pair = t % pairs;
-  const int rep = t / pairs;
-  const int col = pair % p.GW;            // ci_l
-  const int row = pair / p.GW;            // co_l
+  const int lane = t & 63;
+  const int wave = t >> 6;
   const int M = p.N * p.OH * p.OW;
-  const int mstart = blockIdx.z * p.m_per_split;
+  const int mstart = blockIdx.y * p.m_per_split;
   const int mend = min(M, mstart + p.m_per_split);
+  const int ntaps = (wave < 1) ? 3 : 2;  // taps {w, w+4, w+8<9}
 
-  float acc = 0.f;
+  gf32x4 acc[3];
+#pragma unroll
+  for (int i = 0; i < 3; ++i) acc[i] = {0.f, 0.f, 0.f, 0.f};
+
   for (int mt = mstart; mt < mend; mt += 128) {
-    for (int c = t; c < 128 * (p.GW / 8); c += GBLOCK) {
-      const int mm = c / (p.GW / 8);
-      const int c8 = (c % (p.GW / 8)) * 8;
+    // stage dy once and each tap's shifted x (zero-padded cols >= GW)
+    for (int c = t; c < 128 * (GWP / 8); c += GBLOCK) {
+      const int mm = c / (GWP / 8);
+      const int c8 = (c % (GWP / 8)) * 8;
       const int m = mt + mm;
       bf16x8_t dv = {0, 0, 0, 0, 0, 0, 0, 0};
-      bf16x8_t xv = {0, 0, 0, 0, 0, 0, 0, 0};
-      if (m < mend) {
-        const int n = m / (p.OH * p.OW);
+      int n = 0, oh = 0, ow = 0;
+      const bool mok = m < mend && c8 < p.GW;
+      if (mok) {
+        n = m / (p.OH * p.OW);
         const int rem = m % (p.OH * p.OW);
-        const int oh = rem / p.OW, ow = rem % p.OW;
+        oh = rem / p.OW;
+        ow = rem % p.OW;
         dv = *reinterpret_cast<const bf16x8_t*>(
             p.dy + (((long)n * p.OH + oh) * p.OW + ow) * p.C + co0 + c8);
-        const int ih = oh * p.stride - 1 + r;
-        const int iw = ow * p.stride - 1 + s;
-        if (ih >= 0 && ih < p.IH && iw >= 0 && iw < p.IW)
-          xv = *reinterpret_cast<const bf16x8_t*>(
-              p.x + (((long)n * p.IH + ih) * p.IW + iw) * p.C + ci0 + c8);
       }
-      *reinterpret_cast<bf16x8_t*>(&dy_s[mm * 16 + c8]) = dv;
-      *reinterpret_cast<bf16x8_t*>(&x_s[mm * 16 + c8]) = xv;
+      *reinterpret_cast<bf16x8_t*>(&dy_s[mm * GWP + c8]) = dv;
+#pragma unroll
+      for (int tap = 0; tap < 9; ++tap) {
+        bf16x8_t xv = {0, 0, 0, 0, 0, 0, 0, 0};
+        if (mok) {
+          const int ih = oh * p.stride - 1 + tap / 3;
+          const int iw = ow * p.stride - 1 + tap % 3;
+          if (ih >= 0 && ih < p.IH && iw >= 0 && iw < p.IW)
+            xv = *reinterpret_cast<const bf16x8_t*>(
+                p.x + (((long)n * p.IH + ih) * p.IW + iw) * p.C + ci0 + c8);
+        }
+        *reinterpret_cast<bf16x8_t*>(&x_s[tap][mm * GWP + c8]) = xv;
+      }
     }
     __syncthreads();
-    const int mlim = min(128, mend - mt);
-    for (int mm = rep; mm < mlim; mm += rep_n)
-      acc += __bfloat162float(dy_s[mm * 16 + row]) *
-             __bfloat162float(x_s[mm * 16 + col]);
+#pragma unroll
+    for (int ks = 0; ks < 4; ++ks) {
+      bf16x8_t afrag = gtr_frag(dy_s, ks * 32, lane);
+      for (int ti = 0; ti < ntaps; ++ti) {
+        bf16x8_t bfrag = gtr_frag(x_s[wave + 4 * ti], ks * 32, lane);
+        acc[ti] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag,
+                                                          acc[ti], 0, 0, 0);
+      }
+    }
     __syncthreads();
   }
-  red[t] = acc;
-  __syncthreads();
-  if (rep == 0) {
-    for (int rr = 1; rr < rep_n; ++rr) acc += red[pair + rr * pairs];
-    atomicAdd(&p.dw[(((long)(co0 + row) * 3 + r) * 3 + s) * p.GW + col], acc);
+
+  // C/D layout: col = lane&15 (ci_l), row = (lane>>4)*4 + rr (co_l)
+  const int col = lane & 15;
+  for (int ti = 0; ti < ntaps; ++ti) {
+    const int tap = wave + 4 * ti;
+    if (col >= p.GW) continue;
+#pragma unroll
+    for (int rr = 0; rr < 4; ++rr) {
+      const int row = (lane >> 4) * 4 + rr;
+      if (row >= p.GW) continue;
+      atomicAdd(&p.dw[(((long)(co0 + row) * 3 + tap / 3) * 3 + tap % 3) *
+                          p.GW + col],
+                acc[ti][rr]);
+    }
   }
 }
 
@@ -235,11 +285,11 @@ extern "C" void dlb_gconv_wrw(const void* x, const void* dy, float* dw, int N,
   GConvParams p{(const bf16*)x, nullptr, (const bf16*)dy, nullptr, dw,
                 N, IH, IW, C, OH, OW, GW, stride, 0};
   const int M = N * OH * OW;
-  const long tiles = (long)(C / GW) * 9;
+  const long tiles = C / GW;  // one block covers all 9 taps of a group
   int splits = (int)std::min<long>(std::max<long>(1, 2048 / tiles),
                                    std::max<long>(1, M / (4 * 128)));
   p.m_per_split = cdiv(cdiv(M, splits), 128) * 128;
   splits = cdiv(M, p.m_per_split);
-  dim3 grid(C / GW, 9, splits);
+  dim3 grid(C / GW, splits);
   hipLaunchKernelGGL(gconv_wrw_kernel, grid, dim3(GBLOCK), 0, stream, p);
 }
