@@ -16,143 +16,6 @@ typedef __attribute__((ext_vector_type(8))) short bf16x8_t;
 
 #define GBLOCK 256
 
-struct GConvParams {
-  const bf16* x;   // [N, H, W, C]
-  const bf16* w;   // [C, 3, 3, GW]  (channels_last [Co,GW,3,3])
-  const bf16* dy;  // bwd/wrw
-  bf16* out;       // y or dx
-  float* dw;       // wrw
-  int N, IH, IW, C, OH, OW, GW, stride;
-  int m_per_split;
-};
-
-// fwd: block = (co-octet, 256-pixel tile); weights for the octet in LDS.
-__global__ void __launch_bounds__(GBLOCK)
-gconv_fwd_kernel(const GConvParams p) {
-  __shared__ bf16 wlds[8 * 9 * 16];  // [j][tap][ci], GW<=16
-
-  const int co8 = blockIdx.x * 8;
-  const int g = co8 / p.GW;
-  const int ci0 = g * p.GW;
-  const int t = threadIdx.x;
-
-  // stage the octet's weights: 8 rows of [9*GW] contiguous
-  for (int c = t; c < 8 * 9 * p.GW / 8; c += GBLOCK) {
-    const int j = c / (9 * p.GW / 8);
-    const int e8 = (c % (9 * p.GW / 8)) * 8;
-    *reinterpret_cast<bf16x8_t*>(&wlds[(j * 9 * p.GW) + e8]) =
-        *reinterpret_cast<const bf16x8_t*>(p.w + (long)(co8 + j) * 9 * p.GW +
-                                           e8);
-  }
-  __syncthreads();
-
-  const long M = (long)p.N * p.OH * p.OW;
-  const long m = (long)blockIdx.y * GBLOCK + t;
-  if (m >= M) return;
-  const int ow = (int)(m % p.OW);
-  const int oh = (int)((m / p.OW) % p.OH);
-  const int n = (int)(m / ((long)p.OH * p.OW));
-
-  float acc[8] = {0, 0, 0, 0, 0, 0, 0, 0};
-  for (int r = 0; r < 3; ++r) {
-    const int ih = oh * p.stride - 1 + r;
-    if (ih < 0 || ih >= p.IH) continue;
-    for (int s = 0; s < 3; ++s) {
-      const int iw = ow * p.stride - 1 + s;
-      if (iw < 0 || iw >= p.IW) continue;
-      const bf16* xp = p.x + (((long)n * p.IH + ih) * p.IW + iw) * p.C + ci0;
-      const bf16* wp = &wlds[(r * 3 + s) * p.GW];
-      for (int c8 = 0; c8 < p.GW; c8 += 8) {
-        bf16x8_t xv = *reinterpret_cast<const bf16x8_t*>(xp + c8);
-        const bf16* xe = reinterpret_cast<const bf16*>(&xv);
-#pragma unroll
-        for (int e = 0; e < 8; ++e) {
-          const float x1 = __bfloat162float(xe[e]);
-#pragma unroll
-          for (int j = 0; j < 8; ++j)
-            acc[j] += x1 * __bfloat162float(wp[j * 9 * p.GW + c8 + e]);
-        }
-      }
-    }
-  }
-  bf16x8_t o;
-  bf16* ov = reinterpret_cast<bf16*>(&o);
-#pragma unroll
-  for (int j = 0; j < 8; ++j) ov[j] = __float2bfloat16(acc[j]);
-  *reinterpret_cast<bf16x8_t*>(
-      p.out + (((long)n * p.OH + oh) * p.OW + ow) * p.C + co8) = o;
-}
-
-// bwd-data: block = (ci-octet, 256-pixel tile); the group's weight
-// columns for this ci-octet staged in LDS: [co_l][tap][8].
-__global__ void __launch_bounds__(GBLOCK)
-gconv_bwd_kernel(const GConvParams p) {
-  __shared__ bf16 wlds[16 * 9 * 8];  // [co_l][tap][jj]
-
-  const int ci8 = blockIdx.x * 8;
-  const int g = ci8 / p.GW;
-  const int co0 = g * p.GW;
-  const int cl0 = ci8 - g * p.GW;
-  const int t = threadIdx.x;
-
-  for (int c = t; c < p.GW * 9 * 8; c += GBLOCK) {
-    const int jj = c % 8;
-    const int tap = (c / 8) % 9;
-    const int col = c / 72;
-    wlds[c] = p.w[((long)(co0 + col) * 9 + tap) * p.GW + cl0 + jj];
-  }
-  __syncthreads();
-
-  const long M = (long)p.N * p.IH * p.IW;
-  const long m = (long)blockIdx.y * GBLOCK + t;
-  if (m >= M) return;
-  const int iw = (int)(m % p.IW);
-  const int ih = (int)((m / p.IW) % p.IH);
-  const int n = (int)(m / ((long)p.IH * p.IW));
-
-  float acc[8] = {0, 0, 0, 0, 0, 0, 0, 0};
-  for (int r = 0; r < 3; ++r) {
-    const int ohn = ih + 1 - r;
-    if (ohn < 0 || ohn % p.stride) continue;
-    const int oh = ohn / p.stride;
-    if (oh >= p.OH) continue;
-    for (int s = 0; s < 3; ++s) {
-      const int own = iw + 1 - s;
-      if (own < 0 || own % p.stride) continue;
-      const int ow = own / p.stride;
-      if (ow >= p.OW) continue;
-      const bf16* dp = p.dy + (((long)n * p.OH + oh) * p.OW + ow) * p.C + co0;
-      const int tap = r * 3 + s;
-      for (int c8 = 0; c8 < p.GW; c8 += 8) {
-        bf16x8_t dv = *reinterpret_cast<const bf16x8_t*>(dp + c8);
-        const bf16* de = reinterpret_cast<const bf16*>(&dv);
-#pragma unroll
-        for (int e = 0; e < 8; ++e) {
-          const float d1 = __bfloat162float(de[e]);
-#pragma unroll
-          for (int j = 0; j < 8; ++j)
-            acc[j] += d1 * __bfloat162float(wlds[((c8 + e) * 9 + tap) * 8 + j]);
-        }
-      }
-    }
-  }
-  bf16x8_t o;
-  bf16* ov = reinterpret_cast<bf16*>(&o);
-#pragma unroll
-  for (int j = 0; j < 8; ++j) ov[j] = __float2bfloat16(acc[j]);
-  *reinterpret_cast<bf16x8_t*>(
-      p.out + (((long)n * p.IH + ih) * p.IW + iw) * p.C + ci8) = o;
-}
-
-// wrw v3: the per-(group, tap) weight grad is a [GW x GW] output GEMM
-// with K = M (dW[co,ci] = sum_m dy[m,co] * x_tap[m,ci]) — exactly one
-// v_mfma_f32_16x16x32_bf16 tile per 32 pixels.  v2 ran it as scalar
-// LDS FMAs with dy re-staged per tap across 9 blocks and measured 28%
-// of the RegNetY step (183 ms, profiles r2c30).  Here: grid (G,
-// splits); one block stages dy ONCE and the 9 shifted x images per
-// 128-m chunk, and the 4 waves split the 9 taps (wave w owns taps
-// {w, w+4, w+8}), each accumulating its [16,16] tiles via
-// transpose-read fragments.  GW=8 groups zero-pad to the 16-wide tile.
 #define GWP 16  // padded tile width (GW is 8 or 16 in the zoo)
 
 typedef __attribute__((__vector_size__(4 * sizeof(float)))) float gf32x4;
@@ -175,6 +38,155 @@ __device__ inline bf16x8_t gtr_frag(const bf16* img, int mbase, int lane) {
   u.t.b = hi;
   return u.v;
 }
+
+struct GConvParams {
+  const bf16* x;   // [N, H, W, C]
+  const bf16* w;   // [C, 3, 3, GW]  (channels_last [Co,GW,3,3])
+  const bf16* dy;  // bwd/wrw
+  bf16* out;       // y or dx
+  float* dw;       // wrw
+  int N, IH, IW, C, OH, OW, GW, stride;
+  int m_per_split;
+};
+
+// ---- MFMA forward/backward ------------------------------------------
+// Per group, forward is y[M, GW] = A[M, 9*GWP] @ B[9*GWP, GW] with A the
+// tap-gathered input (zero-padded to GWP per tap) and B the group's
+// weight — small-N GEMM on v_mfma_f32_16x16x32_bf16 (the direct VALU
+// versions below measured ~20% of the RegNetY step).  K = 9*GWP = 144,
+// padded to 160 (5 k-steps).  Backward-data is the same shape with the
+// tap-shift on the dy gather and B[k=tap*GWP+co][ci] = w[co][tap][ci].
+#define GK 160
+#define GLDA (GK + 8)
+
+template <bool BWD>
+__global__ void __launch_bounds__(GBLOCK)
+gconv_mfma_kernel(const GConvParams p) {
+  __shared__ bf16 a_lds[128 * GLDA];
+  __shared__ bf16 b_lds[16 * GLDA];
+  __shared__ bf16 o_lds[128 * GWP];
+
+  const int g = blockIdx.x;
+  const int c0 = g * p.GW;            // fwd: co0 == ci0 == c0
+  const int t = threadIdx.x;
+  const int lane = t & 63;
+  const int wave = t >> 6;
+  // output pixel space: fwd = (OH, OW), bwd = (IH, IW)
+  const int PH = BWD ? p.IH : p.OH;
+  const int PW = BWD ? p.IW : p.OW;
+  const long M = (long)p.N * PH * PW;
+  const long m0 = (long)blockIdx.y * 128;
+
+  // stage B [16 rows = out-channel][k]: fwd B[n][tap*GWP+ci] =
+  // w[c0+n][tap][ci]; bwd B[n=ci][tap*GWP+co] = w[c0+co][tap][ci]
+  for (int c = t; c < 16 * (GK / 8); c += GBLOCK) {
+    const int nrow = c / (GK / 8);
+    const int k8 = (c % (GK / 8)) * 8;
+    union { bf16x8_t v; bf16 h[8]; } u;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int k = k8 + j;
+      const int tap = k / GWP;
+      const int cl = k - tap * GWP;
+      bf16 w = (bf16)__float2bfloat16(0.f);
+      if (tap < 9 && cl < p.GW && nrow < p.GW) {
+        if (BWD)
+          w = p.w[((long)(c0 + cl) * 9 + tap) * p.GW + nrow];
+        else
+          w = p.w[((long)(c0 + nrow) * 9 + tap) * p.GW + cl];
+      }
+      u.h[j] = w;
+    }
+    *reinterpret_cast<bf16x8_t*>(&b_lds[nrow * GLDA + k8]) = u.v;
+  }
+
+  // stage A: per output pixel row, the 9 tap-gathered GW-slices
+  for (int c = t; c < 128 * (GK / 8); c += GBLOCK) {
+    const int mm = c / (GK / 8);
+    const int k8 = (c % (GK / 8)) * 8;
+    const int tap = k8 / GWP;           // GWP=16, k8 multiple of 8
+    const int cl8 = k8 - tap * GWP;     // 0 or 8
+    bf16x8_t v = {0, 0, 0, 0, 0, 0, 0, 0};
+    const long m = m0 + mm;
+    if (m < M && tap < 9 && cl8 < p.GW) {
+      const int pw = (int)(m % PW);
+      const int ph = (int)((m / PW) % PH);
+      const int n = (int)(m / ((long)PH * PW));
+      const int r = tap / 3, s = tap % 3;
+      bool ok;
+      int sh, sw;
+      if (BWD) {
+        const int ohn = ph + 1 - r, own = pw + 1 - s;
+        ok = ohn >= 0 && own >= 0 && ohn % p.stride == 0 &&
+             own % p.stride == 0;
+        sh = ohn / p.stride;
+        sw = own / p.stride;
+        ok = ok && sh < p.OH && sw < p.OW;
+      } else {
+        sh = ph * p.stride - 1 + r;
+        sw = pw * p.stride - 1 + s;
+        ok = sh >= 0 && sh < p.IH && sw >= 0 && sw < p.IW;
+      }
+      if (ok) {
+        const bf16* src = BWD ? p.dy : p.x;
+        const int SH = BWD ? p.OH : p.IH;
+        const int SW = BWD ? p.OW : p.IW;
+        v = *reinterpret_cast<const bf16x8_t*>(
+            src + (((long)n * SH + sh) * SW + sw) * p.C + c0 + cl8);
+      }
+    }
+    *reinterpret_cast<bf16x8_t*>(&a_lds[mm * GLDA + k8]) = v;
+  }
+  __syncthreads();
+
+  gf32x4 acc[2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i) acc[i] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+  for (int ks = 0; ks < GK / 32; ++ks) {
+    const int k8 = ks * 32 + (lane >> 4) * 8;
+    bf16x8_t bfrag = *reinterpret_cast<const bf16x8_t*>(
+        &b_lds[(lane & 15) * GLDA + k8]);
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      bf16x8_t afrag = *reinterpret_cast<const bf16x8_t*>(
+          &a_lds[(wave * 32 + i * 16 + (lane & 15)) * GLDA + k8]);
+      acc[i] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc[i],
+                                                       0, 0, 0);
+    }
+  }
+
+  // bounce [128][GWP] through LDS for coalesced 16B row writes
+  __syncthreads();
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    const int col = lane & 15;
+#pragma unroll
+    for (int rr = 0; rr < 4; ++rr) {
+      const int row = wave * 32 + i * 16 + (lane >> 4) * 4 + rr;
+      o_lds[row * GWP + col] = __float2bfloat16(acc[i][rr]);
+    }
+  }
+  __syncthreads();
+  for (int c = t; c < 128 * (p.GW / 8); c += GBLOCK) {
+    const int mm = c / (p.GW / 8);
+    const int c8 = (c % (p.GW / 8)) * 8;
+    const long m = m0 + mm;
+    if (m >= M) continue;
+    *reinterpret_cast<bf16x8_t*>(p.out + m * p.C + c0 + c8) =
+        *reinterpret_cast<bf16x8_t*>(&o_lds[mm * GWP + c8]);
+  }
+}
+
+// wrw v3: the per-(group, tap) weight grad is a [GW x GW] output GEMM
+// with K = M (dW[co,ci] = sum_m dy[m,co] * x_tap[m,ci]) — exactly one
+// v_mfma_f32_16x16x32_bf16 tile per 32 pixels.  v2 ran it as scalar
+// LDS FMAs with dy re-staged per tap across 9 blocks and measured 28%
+// of the RegNetY step (183 ms, profiles r2c30).  Here: grid (G,
+// splits); one block stages dy ONCE and the 9 shifted x images per
+// 128-m chunk, and the 4 waves split the 9 taps (wave w owns taps
+// {w, w+4, w+8}), each accumulating its [16,16] tiles via
+// transpose-read fragments.  GW=8 groups zero-pad to the 16-wide tile.
 
 __global__ void __launch_bounds__(GBLOCK)
 gconv_wrw_kernel(const GConvParams p) {
@@ -262,8 +274,9 @@ extern "C" void dlb_gconv_fwd(const void* x, const void* w, void* y, int N,
   const int OW = (IW + 2 - 3) / stride + 1;
   GConvParams p{(const bf16*)x, (const bf16*)w, nullptr, (bf16*)y, nullptr,
                 N, IH, IW, C, OH, OW, GW, stride, 0};
-  dim3 grid(C / 8, cdiv((long)N * OH * OW, GBLOCK));
-  hipLaunchKernelGGL(gconv_fwd_kernel, grid, dim3(GBLOCK), 0, stream, p);
+  dim3 grid(C / GW, cdiv((long)N * OH * OW, 128));
+  hipLaunchKernelGGL((gconv_mfma_kernel<false>), grid, dim3(GBLOCK), 0,
+                     stream, p);
 }
 
 extern "C" void dlb_gconv_bwd(const void* dy, const void* w, void* dx, int N,
@@ -273,8 +286,9 @@ extern "C" void dlb_gconv_bwd(const void* dy, const void* w, void* dx, int N,
   const int OW = (IW + 2 - 3) / stride + 1;
   GConvParams p{nullptr, (const bf16*)w, (const bf16*)dy, (bf16*)dx, nullptr,
                 N, IH, IW, C, OH, OW, GW, stride, 0};
-  dim3 grid(C / 8, cdiv((long)N * IH * IW, GBLOCK));
-  hipLaunchKernelGGL(gconv_bwd_kernel, grid, dim3(GBLOCK), 0, stream, p);
+  dim3 grid(C / GW, cdiv((long)N * IH * IW, 128));
+  hipLaunchKernelGGL((gconv_mfma_kernel<true>), grid, dim3(GBLOCK), 0,
+                     stream, p);
 }
 
 extern "C" void dlb_gconv_wrw(const void* x, const void* dy, float* dw, int N,
