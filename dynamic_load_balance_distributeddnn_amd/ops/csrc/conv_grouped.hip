@@ -66,8 +66,9 @@ gconv_mfma_kernel(const GConvParams p) {
   __shared__ bf16 b_lds[16 * GLDA];
   __shared__ bf16 o_lds[128 * GWP];
 
-  const int g = blockIdx.x;
-  const int c0 = g * p.GW;            // fwd: co0 == ci0 == c0
+  // a tile covers GWP consecutive channels = GWP/GW whole groups
+  // (2 groups for GW=8 — the weight is block-diagonal across them)
+  const int c0 = blockIdx.x * GWP;
   const int t = threadIdx.x;
   const int lane = t & 63;
   const int wave = t >> 6;
@@ -87,13 +88,15 @@ gconv_mfma_kernel(const GConvParams p) {
     for (int j = 0; j < 8; ++j) {
       const int k = k8 + j;
       const int tap = k / GWP;
-      const int cl = k - tap * GWP;
+      const int cl = k - tap * GWP;      // in-tile channel (k side)
       bf16 w = (bf16)__float2bfloat16(0.f);
-      if (tap < 9 && cl < p.GW && nrow < p.GW) {
-        if (BWD)
-          w = p.w[((long)(c0 + cl) * 9 + tap) * p.GW + nrow];
-        else
-          w = p.w[((long)(c0 + nrow) * 9 + tap) * p.GW + cl];
+      const int ga = c0 + nrow, gb = c0 + cl;
+      if (tap < 9 && ga < p.C && gb < p.C &&
+          ga / p.GW == gb / p.GW) {      // block-diagonal across groups
+        if (BWD)  // ga = ci, gb = co
+          w = p.w[((long)gb * 9 + tap) * p.GW + (ga - (ga / p.GW) * p.GW)];
+        else      // ga = co, gb = ci
+          w = p.w[((long)ga * 9 + tap) * p.GW + (gb - (gb / p.GW) * p.GW)];
       }
       u.h[j] = w;
     }
@@ -108,7 +111,7 @@ gconv_mfma_kernel(const GConvParams p) {
     const int cl8 = k8 - tap * GWP;     // 0 or 8
     bf16x8_t v = {0, 0, 0, 0, 0, 0, 0, 0};
     const long m = m0 + mm;
-    if (m < M && tap < 9 && cl8 < p.GW) {
+    if (m < M && tap < 9 && c0 + cl8 < p.C) {
       const int pw = (int)(m % PW);
       const int ph = (int)((m / PW) % PH);
       const int n = (int)(m / ((long)PH * PW));
@@ -168,11 +171,11 @@ gconv_mfma_kernel(const GConvParams p) {
     }
   }
   __syncthreads();
-  for (int c = t; c < 128 * (p.GW / 8); c += GBLOCK) {
-    const int mm = c / (p.GW / 8);
-    const int c8 = (c % (p.GW / 8)) * 8;
+  for (int c = t; c < 128 * (GWP / 8); c += GBLOCK) {
+    const int mm = c / (GWP / 8);
+    const int c8 = (c % (GWP / 8)) * 8;
     const long m = m0 + mm;
-    if (m >= M) continue;
+    if (m >= M || c0 + c8 >= p.C) continue;
     *reinterpret_cast<bf16x8_t*>(p.out + m * p.C + c0 + c8) =
         *reinterpret_cast<bf16x8_t*>(&o_lds[mm * GWP + c8]);
   }
@@ -193,8 +196,7 @@ gconv_wrw_kernel(const GConvParams p) {
   __shared__ bf16 dy_s[128 * GWP];      // [mm][co_l]
   __shared__ bf16 x_s[9][128 * GWP];    // per-tap shifted [mm][ci_l]
 
-  const int g = blockIdx.x;
-  const int co0 = g * p.GW, ci0 = g * p.GW;
+  const int c0 = blockIdx.x * GWP;  // GWP channels = 1-2 whole groups
   const int t = threadIdx.x;
   const int lane = t & 63;
   const int wave = t >> 6;
@@ -215,14 +217,14 @@ gconv_wrw_kernel(const GConvParams p) {
       const int m = mt + mm;
       bf16x8_t dv = {0, 0, 0, 0, 0, 0, 0, 0};
       int n = 0, oh = 0, ow = 0;
-      const bool mok = m < mend && c8 < p.GW;
+      const bool mok = m < mend && c0 + c8 < p.C;
       if (mok) {
         n = m / (p.OH * p.OW);
         const int rem = m % (p.OH * p.OW);
         oh = rem / p.OW;
         ow = rem % p.OW;
         dv = *reinterpret_cast<const bf16x8_t*>(
-            p.dy + (((long)n * p.OH + oh) * p.OW + ow) * p.C + co0 + c8);
+            p.dy + (((long)n * p.OH + oh) * p.OW + ow) * p.C + c0 + c8);
       }
       *reinterpret_cast<bf16x8_t*>(&dy_s[mm * GWP + c8]) = dv;
 #pragma unroll
@@ -233,7 +235,7 @@ gconv_wrw_kernel(const GConvParams p) {
           const int iw = ow * p.stride - 1 + tap % 3;
           if (ih >= 0 && ih < p.IH && iw >= 0 && iw < p.IW)
             xv = *reinterpret_cast<const bf16x8_t*>(
-                p.x + (((long)n * p.IH + ih) * p.IW + iw) * p.C + ci0 + c8);
+                p.x + (((long)n * p.IH + ih) * p.IW + iw) * p.C + c0 + c8);
         }
         *reinterpret_cast<bf16x8_t*>(&x_s[tap][mm * GWP + c8]) = xv;
       }
@@ -251,17 +253,18 @@ gconv_wrw_kernel(const GConvParams p) {
     __syncthreads();
   }
 
-  // C/D layout: col = lane&15 (ci_l), row = (lane>>4)*4 + rr (co_l)
+  // C/D layout: col = lane&15 (in-tile ci), row = (lane>>4)*4 + rr (co);
+  // keep only same-group (block-diagonal) in-bounds entries
   const int col = lane & 15;
   for (int ti = 0; ti < ntaps; ++ti) {
     const int tap = wave + 4 * ti;
-    if (col >= p.GW) continue;
 #pragma unroll
     for (int rr = 0; rr < 4; ++rr) {
       const int row = (lane >> 4) * 4 + rr;
-      if (row >= p.GW) continue;
-      atomicAdd(&p.dw[(((long)(co0 + row) * 3 + tap / 3) * 3 + tap % 3) *
-                          p.GW + col],
+      const int co = c0 + row, ci = c0 + col;
+      if (co >= p.C || ci >= p.C || co / p.GW != ci / p.GW) continue;
+      atomicAdd(&p.dw[(((long)co * 3 + tap / 3) * 3 + tap % 3) * p.GW +
+                      (ci - (ci / p.GW) * p.GW)],
                 acc[ti][rr]);
     }
   }
@@ -274,7 +277,7 @@ extern "C" void dlb_gconv_fwd(const void* x, const void* w, void* y, int N,
   const int OW = (IW + 2 - 3) / stride + 1;
   GConvParams p{(const bf16*)x, (const bf16*)w, nullptr, (bf16*)y, nullptr,
                 N, IH, IW, C, OH, OW, GW, stride, 0};
-  dim3 grid(C / GW, cdiv((long)N * OH * OW, 128));
+  dim3 grid(cdiv(C, GWP), cdiv((long)N * OH * OW, 128));
   hipLaunchKernelGGL((gconv_mfma_kernel<false>), grid, dim3(GBLOCK), 0,
                      stream, p);
 }
@@ -286,7 +289,7 @@ extern "C" void dlb_gconv_bwd(const void* dy, const void* w, void* dx, int N,
   const int OW = (IW + 2 - 3) / stride + 1;
   GConvParams p{nullptr, (const bf16*)w, (const bf16*)dy, (bf16*)dx, nullptr,
                 N, IH, IW, C, OH, OW, GW, stride, 0};
-  dim3 grid(C / GW, cdiv((long)N * IH * IW, 128));
+  dim3 grid(cdiv(C, GWP), cdiv((long)N * IH * IW, 128));
   hipLaunchKernelGGL((gconv_mfma_kernel<true>), grid, dim3(GBLOCK), 0,
                      stream, p);
 }
@@ -299,11 +302,11 @@ extern "C" void dlb_gconv_wrw(const void* x, const void* dy, float* dw, int N,
   GConvParams p{(const bf16*)x, nullptr, (const bf16*)dy, nullptr, dw,
                 N, IH, IW, C, OH, OW, GW, stride, 0};
   const int M = N * OH * OW;
-  const long tiles = C / GW;  // one block covers all 9 taps of a group
+  const long tiles = cdiv(C, GWP);  // one block: all 9 taps, GWP chans
   int splits = (int)std::min<long>(std::max<long>(1, 2048 / tiles),
                                    std::max<long>(1, M / (4 * 128)));
   p.m_per_split = cdiv(cdiv(M, splits), 128) * 128;
   splits = cdiv(M, p.m_per_split);
-  dim3 grid(C / GW, splits);
+  dim3 grid(cdiv(C, GWP), splits);
   hipLaunchKernelGGL(gconv_wrw_kernel, grid, dim3(GBLOCK), 0, stream, p);
 }
