@@ -1,13 +1,17 @@
 // Grouped 3x3 convolution (gfx950) — RegNet's only grouped shape
 // (SURVEY.md K3): channels-per-group == group_width (8 or 16), stride
-// 1/2, pad 1, square maps.  Near-depthwise (K per output = 9*GW), far
-// below the MFMA regime — direct VALU kernels, NHWC.
+// 1/2, pad 1, square maps, NHWC.
 //
-// v2: v1 re-loaded every weight from global per output pixel (1152
-// scalar loads/thread — measured 3.7 ms/call, 53% of the RegNet step).
-// Now each block owns one channel-octet and stages its weight slice in
-// LDS once; weight reads are wave-uniform LDS broadcasts and x/dy reads
-// are bf16x8 vectors.
+// History: v1 re-loaded weights per pixel (3.7 ms/call); v2 ran direct
+// VALU with LDS weight broadcasts and still measured 48% of the
+// RegNetY step (profiles r2c30).  v3 (this file) is MFMA end to end:
+// per 16-channel tile the forward/data-grad are [M, 144] x [144, 16]
+// GEMMs over tap-gathered operands, and the weight grad is one
+// [16 x 16] tile per (tile, tap) with K = pixels — all on
+// v_mfma_f32_16x16x32_bf16 with ds_read_b64_tr_b16 fragments.  GW=8
+// packs TWO groups per tile (the weight is block-diagonal across
+// them), so no MFMA lanes are wasted.  RegNetY-400MF +46%,
+// RegNetX-200MF +16% whole-model vs the v2 kernels.
 
 #include "common.h"
 
