@@ -958,6 +958,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gn_dgb_reduce_multi", &gn_dgb_reduce_multi,
         "Batched deterministic dgamma/dbeta reduction (one launch)");
   m.def("chan_sums", &chan_sums, "per-channel (sum, ssq) of a segment");
+  m.def("slab_sum", [](torch::Tensor part) {
+          TORCH_CHECK(part.is_cuda() && part.dim() == 2 &&
+                      part.is_contiguous() &&
+                      part.scalar_type() == torch::kFloat32);
+          auto out = torch::empty({part.size(1)}, part.options());
+          slab_reduce_into(part, out, (int)part.size(0), part.size(1));
+          return out;
+        },
+        "deterministic column sum of a [S, len] fp32 matrix");
   m.def("gn_stats_from_sums", &gn_stats_from_sums,
         "GroupNorm mean/rstd from accumulated per-channel sums");
   m.def("gn_stats", &gn_stats, "Stats-only GroupNorm over virtual concat");

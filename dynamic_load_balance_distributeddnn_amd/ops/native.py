@@ -136,7 +136,14 @@ class _Conv2d(torch.autograd.Function):
 
         dbias = None
         if has_bias and ctx.needs_input_grad[2]:
-            dbias = dy.sum(dim=(0, 2, 3), dtype=torch.float32)
+            if dy.shape[1] % 8 == 0:
+                # per-(sample, channel) sums then a deterministic column
+                # fold — torch's generic reduce measured 2.4 ms/step on
+                # GoogLeNet's biased convs (profiles r2c30)
+                s3 = _to_nhwc3(dy)
+                dbias = ext().slab_sum(ext().chan_sums(s3)[0])
+            else:
+                dbias = dy.sum(dim=(0, 2, 3), dtype=torch.float32)
 
         return dx, dweight, dbias, None, None
 
