@@ -333,8 +333,12 @@ struct ConvBwdParams {
   ConvGeom g;  // fd_pix: /(IH*IW), fd_w: /IW, fd_c: /Co, fd_s: /S
 };
 
+// n/ih/iw are the m-row's pixel decomposition, hoisted by the caller —
+// they are loop-invariant across k-tiles and keeping them out of the
+// loader shortens the address chain ahead of the global load.
 __device__ inline bf16x8_t dcol_load8(const ConvBwdParams& p, int m, int k,
-                                      bool vec) {
+                                      bool vec, unsigned n, unsigned ih,
+                                      unsigned iw) {
   bool ok = (m < p.M) & (k < p.K);
   m = ok ? m : 0;
   k = ok ? k : 0;
@@ -342,18 +346,23 @@ __device__ inline bf16x8_t dcol_load8(const ConvBwdParams& p, int m, int k,
     bf16x8_t v = *reinterpret_cast<const bf16x8_t*>(p.dy + (long)m * p.Co + k);
     return mask8(v, ok);
   }
-  unsigned n, rem, ih, iw;
-  p.g.fd_pix.divmod(m, n, rem);
-  p.g.fd_w.divmod(rem, ih, iw);
   if (vec) {
     unsigned rs, co, r, sx;
     p.g.fd_c.divmod(k, rs, co);
     p.g.fd_s.divmod(rs, r, sx);
     const int ohn = (int)ih + p.pad - (int)r;
     const int own = (int)iw + p.pad - (int)sx;
-    const int oh = ohn / p.stride, ow = own / p.stride;
-    ok &= (ohn >= 0) & (own >= 0) & (oh * p.stride == ohn) &
-          (ow * p.stride == own) & (oh < p.OH) & (ow < p.OW);
+    int oh, ow;
+    if (p.stride == 1) {  // uniform branch; avoids two runtime idivs
+      oh = ohn;
+      ow = own;
+      ok &= (ohn >= 0) & (own >= 0) & (oh < p.OH) & (ow < p.OW);
+    } else {
+      oh = ohn / p.stride;
+      ow = own / p.stride;
+      ok &= (ohn >= 0) & (own >= 0) & (oh * p.stride == ohn) &
+            (ow * p.stride == own) & (oh < p.OH) & (ow < p.OW);
+    }
     const long off = ok ? (((long)n * p.OH + oh) * p.OW + ow) * p.Co + co : 0;
     bf16x8_t v = *reinterpret_cast<const bf16x8_t*>(p.dy + off);
     return mask8(v, ok);
@@ -368,9 +377,18 @@ __device__ inline bf16x8_t dcol_load8(const ConvBwdParams& p, int m, int k,
     p.g.fd_s.divmod(rs, r, sx);
     const int ohn = (int)ih + p.pad - (int)r;
     const int own = (int)iw + p.pad - (int)sx;
-    const int oh = ohn / p.stride, ow = own / p.stride;
-    const bool e = ok & (kk < p.K) & (ohn >= 0) & (own >= 0) &
-                   (oh * p.stride == ohn) & (ow * p.stride == own) &
+    int oh, ow;
+    bool dv;
+    if (p.stride == 1) {
+      oh = ohn;
+      ow = own;
+      dv = true;
+    } else {
+      oh = ohn / p.stride;
+      ow = own / p.stride;
+      dv = (oh * p.stride == ohn) & (ow * p.stride == own);
+    }
+    const bool e = ok & (kk < p.K) & (ohn >= 0) & (own >= 0) & dv &
                    (oh < p.OH) & (ow < p.OW);
     const long off = e ? (((long)n * p.OH + oh) * p.OW + ow) * p.Co + co : 0;
     bf16 t = p.dy[off];
@@ -414,13 +432,24 @@ conv_bwd_data_kernel(const ConvBwdParams p) {
 
   bf16x8_t areg[APT], breg[BPT];
 
+  // Pixel decomposition of each owned m-row: invariant across k-tiles.
+  unsigned mn[APT], mih[APT], miw[APT];
+#pragma unroll
+  for (int u = 0; u < APT; ++u) {
+    const int c = t + u * CONV_BLOCK;
+    const int m = m0 + c / (BK / 8);
+    unsigned rem;
+    p.g.fd_pix.divmod(m < p.M ? m : 0, mn[u], rem);
+    p.g.fd_w.divmod(rem, mih[u], miw[u]);
+  }
+
   auto load_tile = [&](int kt) {
 #pragma unroll
     for (int u = 0; u < APT; ++u) {
       const int c = t + u * CONV_BLOCK;
       const int row = c / (BK / 8);
       const int k8 = (c % (BK / 8)) * 8;
-      areg[u] = dcol_load8(p, m0 + row, kt + k8, vec);
+      areg[u] = dcol_load8(p, m0 + row, kt + k8, vec, mn[u], mih[u], miw[u]);
       if (c >= ACH) areg[u] = zero8();
     }
 #pragma unroll
