@@ -95,15 +95,16 @@ maxpool_bwd_kernel(const bf16* __restrict__ dy,
     float acc[VEC];
 #pragma unroll
     for (int j = 0; j < VEC; ++j) acc[j] = 0.f;
+    const bool s1 = (stride == 1);  // uniform; skips runtime idivs
     for (int r = 0; r < k; ++r) {
       const int ohn = ih + pad - r;
-      if (ohn < 0 || ohn % stride) continue;
-      const int oh = ohn / stride;
+      if (ohn < 0 || (!s1 && ohn % stride)) continue;
+      const int oh = s1 ? ohn : ohn / stride;
       if (oh >= OH) continue;
       for (int s = 0; s < k; ++s) {
         const int own = iw + pad - s;
-        if (own < 0 || own % stride) continue;
-        const int ow = own / stride;
+        if (own < 0 || (!s1 && own % stride)) continue;
+        const int ow = s1 ? own : own / stride;
         if (ow >= OW) continue;
         const long o = (((long)n * OH + oh) * OW + ow) * C + c;
         const unsigned char want = (unsigned char)(r * k + s);
