@@ -146,3 +146,29 @@ def test_conv_and_gn_bitwise_deterministic():
     assert torch.equal(z1, z2)
     for a, b2 in zip(g1, g2):
         assert torch.equal(a, b2)
+
+
+@needs_gpu
+def test_conv_bias_grad_deterministic_and_correct():
+    """The biased-conv dbias path (chansum -> deterministic colsum) must
+    be bit-identical across runs and match the fp32 reduction."""
+    from dynamic_load_balance_distributeddnn_amd.ops import native
+
+    def run():
+        torch.manual_seed(3)
+        x = torch.randn(32, 192, 16, 16, device="cuda").bfloat16() \
+            .to(memory_format=torch.channels_last).requires_grad_()
+        w = (torch.randn(64, 192, 1, 1, device="cuda") * 0.05).requires_grad_()
+        b = torch.zeros(64, device="cuda").requires_grad_()
+        y = native.conv2d(x, w, b, 1, 0)
+        (y.float() ** 2).mean().backward()
+        return b.grad.clone(), y.detach().clone()
+
+    (db1, y1), (db2, y2) = run(), run()
+    assert torch.equal(y1, y2)
+    assert torch.equal(db1, db2)
+    # reference reduction of the actual upstream grad
+    y1 = y1.detach().requires_grad_()
+    (y1.float() ** 2).mean().backward()
+    ref = y1.grad.float().sum(dim=(0, 2, 3))
+    assert torch.allclose(db1, ref, rtol=1e-4, atol=1e-5)
