@@ -318,6 +318,11 @@ extern "C" void dlb_conv_bwd_data(const void* dy, const void* wt, void* dx,
                                   int N, int IH, int IW, int Ci, int OH,
                                   int OW, int Co, int R, int S, int stride,
                                   int pad, hipStream_t stream);
+extern "C" void dlb_conv_bwd_data_acc(const void* dy, const void* wt,
+                                      void* dx, int N, int IH, int IW,
+                                      int Ci, int OH, int OW, int Co, int R,
+                                      int S, int stride, int pad,
+                                      hipStream_t stream);
 extern "C" void dlb_conv_wrw(const void* x, const void* dy, float* dw, int N,
                              int IH, int IW, int Ci, int OH, int OW, int Co,
                              int R, int S, int stride, int pad, int splits,
@@ -467,13 +472,28 @@ static torch::Tensor conv_fwd(torch::Tensor x, torch::Tensor w,
 // (the natural layout — the kernel gathers it directly, no host reshape).
 static torch::Tensor conv_bwd_data(torch::Tensor dy, torch::Tensor w,
                                    int64_t IH, int64_t IW, int64_t stride,
-                                   int64_t pad) {
+                                   int64_t pad,
+                                   c10::optional<torch::Tensor> accum_into
+                                       = c10::nullopt) {
+  // accum_into: an existing [N, Ci, IH, IW] channels_last bf16 grad —
+  // the generic kernel ADDS the data-grad into it (residual junctions),
+  // replacing a separate elementwise add pass over the tensor.
   TORCH_CHECK(dy.is_cuda() && is_cl(dy) && is_cl(w));
   const int N = dy.size(0), Co = dy.size(1), OH = dy.size(2), OW = dy.size(3);
   const int Ci = w.size(1), R = w.size(2), S = w.size(3);
+  auto stream = at::hip::getCurrentHIPStream();
+  if (accum_into.has_value()) {
+    auto& dx0 = *accum_into;
+    TORCH_CHECK(is_cl(dx0) && dx0.size(0) == N && dx0.size(1) == Ci &&
+                dx0.size(2) == IH && dx0.size(3) == IW &&
+                dx0.scalar_type() == dy.scalar_type());
+    dlb_conv_bwd_data_acc(dy.data_ptr(), w.data_ptr(), dx0.data_ptr(), N,
+                          (int)IH, (int)IW, Ci, OH, OW, Co, R, S,
+                          (int)stride, (int)pad, stream.stream());
+    return dx0;
+  }
   auto dx = torch::empty({N, Ci, IH, IW},
                          dy.options().memory_format(torch::MemoryFormat::ChannelsLast));
-  auto stream = at::hip::getCurrentHIPStream();
   // 3x3/s1/p1: the LDS-halo kernel in transpose-read mode reads the
   // weight unmodified (no flip+copy transform, ~2x the generic kernel)
   // measured: the halo-WTR route wins for shallow reductions (DenseNet
@@ -937,7 +957,10 @@ static std::vector<torch::Tensor> se_bwd(torch::Tensor x, torch::Tensor g,
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv_fwd", &conv_fwd, "Implicit-GEMM NHWC bf16 conv forward");
-  m.def("conv_bwd_data", &conv_bwd_data, "NHWC bf16 conv input-grad");
+  m.def("conv_bwd_data", &conv_bwd_data, "NHWC bf16 conv input-grad",
+        py::arg("dy"), py::arg("w"), py::arg("IH"), py::arg("IW"),
+        py::arg("stride"), py::arg("pad"),
+        py::arg("accum_into") = py::none());
   m.def("conv_wrw", &conv_wrw,
         py::arg("x"), py::arg("dy"), py::arg("R"), py::arg("S"),
         py::arg("stride"), py::arg("pad"), py::arg("out") = py::none(),

@@ -68,6 +68,20 @@ __device__ inline bf16x8_t mask8(bf16x8_t v, bool ok) {
   return u.h;
 }
 
+// elementwise bf16 add of two octets (fp32 intermediate) — the
+// accumulate epilogue of the data-grad kernel (residual junctions).
+__device__ inline bf16x8_t add8(bf16x8_t a, bf16x8_t b) {
+  union U { bf16x8_t v; bf16 h[8]; };
+  U ua, ub, uo;
+  ua.v = a;
+  ub.v = b;
+#pragma unroll
+  for (int j = 0; j < 8; ++j)
+    uo.h[j] = __float2bfloat16(__bfloat162float(ua.h[j]) +
+                               __bfloat162float(ub.h[j]));
+  return uo.v;
+}
+
 __device__ inline bf16x8_t im2col_load8(const bf16* __restrict__ x, int m,
                                         int k, int IH, int IW, int Ci, int OH,
                                         int OW, int S, int stride, int pad,
@@ -397,7 +411,7 @@ __device__ inline bf16x8_t dcol_load8(const ConvBwdParams& p, int m, int k,
   return v;
 }
 
-template <int BM, int BN, int WM, int WN>
+template <int BM, int BN, int WM, int WN, bool ACCUM = false>
 __global__ void __launch_bounds__(CONV_BLOCK)
 conv_bwd_data_kernel(const ConvBwdParams p) {
   constexpr int WTM = BM / WM;
@@ -592,11 +606,20 @@ conv_bwd_data_kernel(const ConvBwdParams p) {
       const int m = m0 + ch * 32 + row;
       if (m >= p.M) continue;
       if (n0 + c8 + 7 < p.Ci) {
-        *reinterpret_cast<bf16x8_t*>(p.dx + (long)m * p.Ci + n0 + c8) =
-            *reinterpret_cast<const bf16x8_t*>(&o_lds[row * LDO + c8]);
+        bf16x8_t nv = *reinterpret_cast<const bf16x8_t*>(
+            &o_lds[row * LDO + c8]);
+        bf16x8_t* dst =
+            reinterpret_cast<bf16x8_t*>(p.dx + (long)m * p.Ci + n0 + c8);
+        if (ACCUM) nv = add8(nv, *dst);
+        *dst = nv;
       } else {
-        for (int j = 0; j < 8 && n0 + c8 + j < p.Ci; ++j)
-          p.dx[(long)m * p.Ci + n0 + c8 + j] = o_lds[row * LDO + c8 + j];
+        for (int j = 0; j < 8 && n0 + c8 + j < p.Ci; ++j) {
+          bf16* dst = p.dx + (long)m * p.Ci + n0 + c8 + j;
+          const bf16 nv = o_lds[row * LDO + c8 + j];
+          *dst = ACCUM ? (bf16)__float2bfloat16(__bfloat162float(nv) +
+                                                __bfloat162float(*dst))
+                       : nv;
+        }
       }
     }
   }
@@ -799,10 +822,11 @@ extern "C" void dlb_conv_fwd(const void* x, const void* w, void* y,
                        dim3(CONV_BLOCK), 0, stream, p);
 }
 
-extern "C" void dlb_conv_bwd_data(const void* dy, const void* w, void* dx,
-                                  int N, int IH, int IW, int Ci, int OH,
-                                  int OW, int Co, int R, int S, int stride,
-                                  int pad, hipStream_t stream) {
+template <bool ACCUM>
+static void launch_bwd_data(const void* dy, const void* w, void* dx, int N,
+                            int IH, int IW, int Ci, int OH, int OW, int Co,
+                            int R, int S, int stride, int pad,
+                            hipStream_t stream) {
   ConvBwdParams p{(const bf16*)dy, (const bf16*)w, (bf16*)dx, N, IH, IW, Ci,
                   OH, OW, Co, R, S, stride, pad, N * IH * IW, R * S * Co,
                   R * S * Ci, {}};
@@ -815,17 +839,37 @@ extern "C" void dlb_conv_bwd_data(const void* dy, const void* w, void* dx,
   pick_tile(Ci, bm, bn);
   dim3 grid(cdiv(p.M, bm), cdiv(Ci, bn));
   if (bm == 128 && bn == 128)
-    hipLaunchKernelGGL((conv_bwd_data_kernel<128, 128, 2, 2>), grid,
+    hipLaunchKernelGGL((conv_bwd_data_kernel<128, 128, 2, 2, ACCUM>), grid,
                        dim3(CONV_BLOCK), 0, stream, p);
   else if (bm == 128 && bn == 64)
-    hipLaunchKernelGGL((conv_bwd_data_kernel<128, 64, 2, 2>), grid,
+    hipLaunchKernelGGL((conv_bwd_data_kernel<128, 64, 2, 2, ACCUM>), grid,
                        dim3(CONV_BLOCK), 0, stream, p);
   else if (bm == 128 && bn == 32)
-    hipLaunchKernelGGL((conv_bwd_data_kernel<128, 32, 4, 1>), grid,
+    hipLaunchKernelGGL((conv_bwd_data_kernel<128, 32, 4, 1, ACCUM>), grid,
                        dim3(CONV_BLOCK), 0, stream, p);
   else
-    hipLaunchKernelGGL((conv_bwd_data_kernel<256, 16, 4, 1>), grid,
+    hipLaunchKernelGGL((conv_bwd_data_kernel<256, 16, 4, 1, ACCUM>), grid,
                        dim3(CONV_BLOCK), 0, stream, p);
+}
+
+extern "C" void dlb_conv_bwd_data(const void* dy, const void* w, void* dx,
+                                  int N, int IH, int IW, int Ci, int OH,
+                                  int OW, int Co, int R, int S, int stride,
+                                  int pad, hipStream_t stream) {
+  launch_bwd_data<false>(dy, w, dx, N, IH, IW, Ci, OH, OW, Co, R, S, stride,
+                         pad, stream);
+}
+
+// dx += data-grad (residual-junction fusion: the skip grad already lives
+// in dx and the junction add rides the epilogue instead of a separate
+// elementwise kernel).
+extern "C" void dlb_conv_bwd_data_acc(const void* dy, const void* w,
+                                      void* dx, int N, int IH, int IW,
+                                      int Ci, int OH, int OW, int Co, int R,
+                                      int S, int stride, int pad,
+                                      hipStream_t stream) {
+  launch_bwd_data<true>(dy, w, dx, N, IH, IW, Ci, OH, OW, Co, R, S, stride,
+                        pad, stream);
 }
 
 // split-count query: how many per-split slabs the wrw launch will write.
