@@ -65,10 +65,10 @@ class _Bottleneck(nn.Module):
         self.proj = _proj(cin, cout, stride) if (stride != 1 or cin != cout) else None
 
     def forward(self, x):
-        out = self.a(x)
-        res = self.proj(x) if self.proj is not None else x
-        return FD.group_norm_add_act(out, res, _GN, self.norm_out.weight,
-                                     self.norm_out.bias, self.norm_out.eps)
+        # one Function per block on GPU: junction add fused into conv1's
+        # data-grad epilogue, direct-arena weight grads, batched
+        # dgamma/dbeta (ops/resblock.py); composes per-layer elsewhere
+        return FD.res_bottleneck(self, x)
 
 
 class ResNet(nn.Module):

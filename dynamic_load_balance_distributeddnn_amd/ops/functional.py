@@ -77,6 +77,30 @@ def group_norm_add_act(x, res, num_groups, weight, bias, eps=1e-5):
     return F.relu(out, inplace=True)
 
 
+def res_bottleneck(mod, x):
+    """One ResNet bottleneck block (models.resnet._Bottleneck).
+
+    GPU path: a single autograd Function with a manual reverse walk
+    (ops/resblock.py): conv1's data grad accumulates into the skip grad
+    inside the kernel epilogue, weight grads land directly in the flat
+    arena, and the block's four dgamma/dbeta reductions batch into one
+    launch.  Fallback: the per-layer module composition.
+    """
+    import os
+    if (_use_native("group_norm_act", x)
+            and not os.environ.get("DLB_NO_BLOCK_FN")):
+        from . import native, resblock
+        if (native.gn_native_ok(x, mod.norm_out.num_groups,
+                                mod.norm_out.weight)
+                and resblock.bottleneck_fn_ok(mod, x)):
+            return resblock.bottleneck_forward(mod, x)
+    out = mod.a(x)
+    res = mod.proj(x) if mod.proj is not None else x
+    return group_norm_add_act(out, res, mod.norm_out.num_groups,
+                              mod.norm_out.weight, mod.norm_out.bias,
+                              mod.norm_out.eps)
+
+
 def linear(x, weight, bias=None):
     # Plain library GEMM: hipBLASLt via F.linear (the north star allows
     # vendor GEMM libraries for unfused matmuls; fused hot ops are ours).
