@@ -67,15 +67,10 @@ class _Block(nn.Module):
             )
 
     def forward(self, x):
-        out = self.a(x)
-        if self.se is not None:
-            out = self.se(out)
-        out = self.b(out)
-        res = self.proj(x) if self.proj is not None else x
-        # relu(GN(out) + res) fused (ops.functional.group_norm_add_act)
-        return FD.group_norm_add_act(out, res, self.norm_out.num_groups,
-                                     self.norm_out.weight,
-                                     self.norm_out.bias, self.norm_out.eps)
+        # one Function per block on GPU (ops/regblock.py): junction add
+        # fused into conv1's data-grad epilogue, direct-arena weight
+        # grads, batched dgamma/dbeta; composes per-layer elsewhere
+        return FD.reg_block(self, x)
 
 
 class RegNet(nn.Module):

@@ -837,9 +837,20 @@ static torch::Tensor gconv_bwd(torch::Tensor dy, torch::Tensor w, int64_t IH,
   return dx;
 }
 static torch::Tensor gconv_wrw(torch::Tensor x, torch::Tensor dy, int64_t GW,
-                               int64_t stride) {
+                               int64_t stride,
+                               c10::optional<torch::Tensor> out
+                                   = c10::nullopt) {
+  // out: a pre-zeroed [C, 9*GW] fp32 buffer (flat-arena grad view) the
+  // kernel's atomics accumulate into directly.
   const int N = x.size(0), C = x.size(1), IH = x.size(2), IW = x.size(3);
-  auto dw = torch::zeros({C, 9 * GW}, x.options().dtype(torch::kFloat32));
+  torch::Tensor dw;
+  if (out.has_value()) {
+    dw = *out;
+    TORCH_CHECK(dw.is_contiguous() && dw.numel() == (long)C * 9 * GW &&
+                dw.scalar_type() == torch::kFloat32);
+  } else {
+    dw = torch::zeros({C, 9 * GW}, x.options().dtype(torch::kFloat32));
+  }
   dlb_gconv_wrw(x.data_ptr(), dy.data_ptr(), dw.data_ptr<float>(), N, IH, IW,
                 C, (int)GW, (int)stride,
                 at::hip::getCurrentHIPStream().stream());
@@ -1010,7 +1021,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_bwd", &attn_bwd);
   m.def("gconv_fwd", &gconv_fwd);
   m.def("gconv_bwd", &gconv_bwd);
-  m.def("gconv_wrw", &gconv_wrw);
+  m.def("gconv_wrw", &gconv_wrw, "grouped 3x3 weight-grad",
+        py::arg("x"), py::arg("dy"), py::arg("GW"),
+        py::arg("stride"), py::arg("out") = py::none());
   m.def("maxpool_fwd", &maxpool_fwd);
   m.def("maxpool_bwd", &maxpool_bwd);
   m.def("logsoftmax_fwd", &logsoftmax_fwd);

@@ -101,6 +101,29 @@ def res_bottleneck(mod, x):
                               mod.norm_out.eps)
 
 
+def reg_block(mod, x):
+    """One RegNet block (models.regnet._Block) — the bottleneck
+    block-Function treatment (ops/regblock.py) with the SE middle's
+    backward chained manually; falls back to the per-layer composition.
+    """
+    import os
+    if (_use_native("group_norm_act", x)
+            and not os.environ.get("DLB_NO_BLOCK_FN")):
+        from . import native, regblock
+        if (native.gn_native_ok(x, mod.norm_out.num_groups,
+                                mod.norm_out.weight)
+                and regblock.regblock_fn_ok(mod, x)):
+            return regblock.regblock_forward(mod, x)
+    out = mod.a(x)
+    if mod.se is not None:
+        out = mod.se(out)
+    out = mod.b(out)
+    res = mod.proj(x) if mod.proj is not None else x
+    return group_norm_add_act(out, res, mod.norm_out.num_groups,
+                              mod.norm_out.weight, mod.norm_out.bias,
+                              mod.norm_out.eps)
+
+
 def linear(x, weight, bias=None):
     # Plain library GEMM: hipBLASLt via F.linear (the north star allows
     # vendor GEMM libraries for unfused matmuls; fused hot ops are ours).
