@@ -66,12 +66,15 @@ class _RegBlockFn(torch.autograd.Function):
         if has_se:
             wr, br, we, be = params[9:13]
             s0 = ext().gavg_fwd(_as4(u3, n, h2, w2)).reshape(n, -1)
-            s0f = s0.float()
-            wrf = wr.reshape(wr.shape[0], -1)
-            wef = we.reshape(we.shape[0], -1)
-            s1 = torch.addmm(br, s0f, wrf.t())
-            rr = torch.relu(s1)
-            gate = torch.addmm(be, rr, wef.t()).bfloat16().contiguous()
+            # gate math in fp32 regardless of the surrounding autocast:
+            # backward reuses these tensors in fp32 matrix products
+            with torch.autocast("cuda", enabled=False):
+                s0f = s0.float()
+                wrf = wr.reshape(wr.shape[0], -1)
+                wef = we.reshape(we.shape[0], -1)
+                s1 = torch.addmm(br, s0f, wrf.t())
+                rr = torch.relu(s1)
+                gate = torch.addmm(be, rr, wef.t()).bfloat16().contiguous()
             v3 = ext().se_fwd(u3, gate)
             se_saves = (s0f, s1, rr, gate)
         else:

@@ -52,10 +52,10 @@ def test_regblock_matches_per_layer(cin, cout, stride, gw, se, monkeypatch):
     ya, dxa, ga = _run(m, x0, True, monkeypatch)
     yb, dxb, gb = _run(m, x0, False, monkeypatch)
 
-    ytol = 0.0 if se == 0.0 else 3e-2
+    ytol = 0.0 if se == 0.0 else 5e-2
     assert (ya.float() - yb.float()).abs().max() <= ytol
-    assert (dxa.float() - dxb.float()).abs().max() <= 3e-2
-    rtol = 1e-3 if se == 0.0 else 3e-2
+    assert (dxa.float() - dxb.float()).abs().max() <= 5e-2
+    rtol = 1e-3 if se == 0.0 else 5e-2
     for n in ga:
         d = (ga[n].float() - gb[n].float()).abs().max().item()
         s = gb[n].float().abs().max().item() + 1e-6
