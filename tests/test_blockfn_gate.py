@@ -11,7 +11,7 @@ def test_bottleneck_gate_rejects_cpu():
         _Bottleneck
     from dynamic_load_balance_distributeddnn_amd.ops import resblock
 
-    m = _Bottleneck(64, 16, 1)
+    m = _Bottleneck(64, 32, 1)
     x = torch.randn(2, 64, 8, 8)
     assert not resblock.bottleneck_fn_ok(m, x)
     y = m(x)  # falls back to the per-layer composition
@@ -42,7 +42,7 @@ def test_blockfn_grads_match_reference_composition_cpu():
         _Bottleneck
 
     torch.manual_seed(0)
-    m = _Bottleneck(32, 8, 1)
+    m = _Bottleneck(32, 32, 1)
     x = torch.randn(2, 32, 8, 8, requires_grad=True)
     y = m(x)
 
