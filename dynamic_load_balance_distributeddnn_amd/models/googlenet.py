@@ -57,11 +57,10 @@ class _Inception(nn.Module):
         )
 
     def forward(self, x):
-        return torch.cat(
-            [self.branch1(x), self.branch2(x), self.branch3(x),
-             self.branch4(self.branch4_pool(x))],
-            dim=1,
-        )
+        # one Function per module on GPU (ops/googblock.py): branch
+        # data-grads accumulate in the conv epilogues instead of three
+        # autograd adds; composes per-branch elsewhere
+        return FD.inception(self, x)
 
 
 class GoogLeNet(nn.Module):

@@ -124,6 +124,34 @@ def reg_block(mod, x):
                               mod.norm_out.eps)
 
 
+def inception(mod, x):
+    """One GoogLeNet Inception module — the block-Function treatment
+    (ops/googblock.py): the pool branch's maxpool backward writes the
+    input grad fresh and the other branches accumulate into it, weight
+    and bias grads land directly in the flat arena, and the module's
+    seven GroupNorm reductions batch into one launch.  Falls back to
+    the per-branch composition."""
+    import os
+    if (_use_native("group_norm_act", x)
+            and not os.environ.get("DLB_NO_BLOCK_FN")):
+        from . import googblock
+        if inception_native_ok(mod, x):
+            return googblock.inception_forward(mod, x)
+    import torch
+    return torch.cat(
+        [mod.branch1(x), mod.branch2(x), mod.branch3(x),
+         mod.branch4(mod.branch4_pool(x))],
+        dim=1,
+    )
+
+
+def inception_native_ok(mod, x):
+    from . import googblock, native
+    return (native.gn_native_ok(x, mod.branch1[1].num_groups,
+                                mod.branch1[1].weight)
+            and googblock.inception_fn_ok(mod, x))
+
+
 def linear(x, weight, bias=None):
     # Plain library GEMM: hipBLASLt via F.linear (the north star allows
     # vendor GEMM libraries for unfused matmuls; fused hot ops are ours).
