@@ -105,9 +105,13 @@ class _InceptionFn(torch.autograd.Function):
             wp = params[wi]
             co, ci = wp.shape[0], wp.shape[1]
             if direct:
+                g4 = wp.grad.permute(0, 2, 3, 1)
+                # reshape of a non-viewable permute would be a silent
+                # copy (grad dropped) — the arena stores 4D params
+                # channels_last, assert it
+                assert g4.is_contiguous(), "arena grad not channels_last"
                 ext().conv_wrw(inp4, dh4, k, k, 1, k // 2,
-                               out=wp.grad.permute(0, 2, 3, 1)
-                               .reshape(co, k * k * ci))
+                               out=g4.reshape(co, k * k * ci))
                 sink.mark_ready(wp)
             else:
                 dw = ext().conv_wrw(inp4, dh4, k, k, 1, k // 2)

@@ -109,9 +109,13 @@ class _BottleneckFn(torch.autograd.Function):
             wp_ = params[wi]
             co, ci = wp_.shape[0], wp_.shape[1]
             if direct:
+                g4 = wp_.grad.permute(0, 2, 3, 1)
+                # a non-viewable permute would reshape into a COPY and
+                # silently drop the grad — the arena stores 4D params
+                # channels_last, assert it
+                assert g4.is_contiguous(), "arena grad not channels_last"
                 ext().conv_wrw(x4, dy4, R, S, st, pad,
-                               out=wp_.grad.permute(0, 2, 3, 1)
-                               .reshape(co, R * S * ci))
+                               out=g4.reshape(co, R * S * ci))
                 sink.mark_ready(wp_)
             else:
                 dw = ext().conv_wrw(x4, dy4, R, S, st, pad)
